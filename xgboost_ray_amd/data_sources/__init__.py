@@ -1,0 +1,28 @@
+"""Pluggable data sources (reference data_sources/__init__.py).
+
+Order matters: sources are probed in order and the first match wins.
+Distributed-dataframe sources (Modin/Dask/Petastorm/ray.data) from the
+reference are not applicable in this Ray-less single-node environment;
+their loading capability is covered by the file-based (CSV/Parquet
+multi-file) and ObjectStore sources.
+"""
+
+from xgboost_ray_amd.data_sources.data_source import DataSource, RayFileType
+from xgboost_ray_amd.data_sources.numpy import Numpy
+from xgboost_ray_amd.data_sources.pandas import Pandas
+from xgboost_ray_amd.data_sources.csv import CSV
+from xgboost_ray_amd.data_sources.parquet import Parquet
+from xgboost_ray_amd.data_sources.object_store import ObjectStore
+
+data_sources = [Numpy, Pandas, CSV, Parquet, ObjectStore]
+
+__all__ = [
+    "DataSource",
+    "RayFileType",
+    "Numpy",
+    "Pandas",
+    "CSV",
+    "Parquet",
+    "ObjectStore",
+    "data_sources",
+]

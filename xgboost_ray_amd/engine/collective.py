@@ -1,0 +1,105 @@
+"""Collective communication layer: RCCL over xGMI via torch.distributed.
+
+MI355X-native replacement for the reference's three communication
+mechanisms (SURVEY.md #5 'Distributed communication backend'):
+
+- Rabit tracker + worker ring (reference compat/tracker.py, main.py:225-324)
+  -> a torch.distributed process group bootstrapped over TCP on
+  127.0.0.1, rebuilt per training attempt with the currently-alive world
+  size (matching the reference's restart-tracker-per-attempt design,
+  reference main.py:256-283, 1207).
+- NCCL histogram AllReduce inside CUDA gpu_hist -> explicit RCCL AllReduce
+  of int64 fixed-point histograms (backend "nccl" IS RCCL on ROCm), issued
+  on a side HIP stream so the transfer overlaps the next feature block's
+  build.
+- CPU-path Rabit allreduce -> gloo backend.
+
+world_size == 1 degrades to a no-op local implementation (no process
+group), so single-actor training needs no rendezvous.
+"""
+
+import datetime
+import os
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+
+
+class Collective:
+    def __init__(
+        self,
+        rank: int = 0,
+        world_size: int = 1,
+        master_addr: str = "127.0.0.1",
+        master_port: Optional[int] = None,
+        backend: Optional[str] = None,
+        device: Optional[torch.device] = None,
+        timeout_s: float = 300.0,
+    ):
+        self.rank = rank
+        self.world_size = world_size
+        self.device = device or torch.device("cpu")
+        self._group = None
+        self._initialized_here = False
+        if world_size <= 1:
+            return
+        if backend is None:
+            backend = "nccl" if self.device.type == "cuda" else "gloo"
+        self.backend = backend
+        if dist.is_initialized():
+            # reuse the outer process group (e.g. launched via torchrun)
+            self._group = dist.group.WORLD
+            return
+        os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+        dist.init_process_group(
+            backend=backend,
+            init_method=f"tcp://{master_addr}:{master_port}",
+            rank=rank,
+            world_size=world_size,
+            timeout=datetime.timedelta(seconds=timeout_s),
+        )
+        self._group = dist.group.WORLD
+        self._initialized_here = True
+
+    @property
+    def is_distributed(self) -> bool:
+        return self.world_size > 1
+
+    def shutdown(self):
+        if self._initialized_here and dist.is_initialized():
+            dist.destroy_process_group()
+        self._group = None
+        self._initialized_here = False
+
+    # -- collectives -------------------------------------------------------
+    def allreduce_(self, tensor: torch.Tensor, op: str = "sum") -> torch.Tensor:
+        if not self.is_distributed:
+            return tensor
+        ops = {"sum": dist.ReduceOp.SUM, "max": dist.ReduceOp.MAX, "min": dist.ReduceOp.MIN}
+        dist.all_reduce(tensor, op=ops[op], group=self._group)
+        return tensor
+
+    def allreduce_async(self, tensor: torch.Tensor, op: str = "sum"):
+        """Launch an async AllReduce; returns a handle with .wait() (or None)."""
+        if not self.is_distributed:
+            return None
+        ops = {"sum": dist.ReduceOp.SUM, "max": dist.ReduceOp.MAX}
+        return dist.all_reduce(tensor, op=ops[op], group=self._group, async_op=True)
+
+    def broadcast_(self, tensor: torch.Tensor, src: int = 0) -> torch.Tensor:
+        if not self.is_distributed:
+            return tensor
+        dist.broadcast(tensor, src=src, group=self._group)
+        return tensor
+
+    def allgather_obj(self, obj) -> List:
+        if not self.is_distributed:
+            return [obj]
+        out = [None] * self.world_size
+        dist.all_gather_object(out, obj, group=self._group)
+        return out
+
+    def barrier(self):
+        if self.is_distributed:
+            dist.barrier(group=self._group)

@@ -1,0 +1,223 @@
+"""Evaluation metrics, designed for distributed evaluation.
+
+Each metric reduces its local shard to a small float64 stats vector that is
+summed across workers with one AllReduce (reference semantics: XGBoost's
+per-round metric AllReduce inside Rabit), then
+finalized identically on every rank. AUC is computed from a fixed
+score-histogram (16384 bins) so it needs no global sort; mAP/NDCG reduce
+per-query sums.
+"""
+
+from typing import Optional
+
+import torch
+
+_AUC_BINS = 16384
+
+
+class Metric:
+    name = "base"
+    higher_better = False
+
+    def local_stats(self, margin, label, weight, qid, obj) -> torch.Tensor:
+        raise NotImplementedError
+
+    def finalize(self, stats: torch.Tensor) -> float:
+        raise NotImplementedError
+
+
+def _w(label, weight):
+    if weight is None:
+        return torch.ones_like(label, dtype=torch.float64)
+    return weight.double()
+
+
+class RMSE(Metric):
+    name = "rmse"
+
+    def local_stats(self, margin, label, weight, qid, obj):
+        pred = margin.double()
+        w = _w(label, weight)
+        se = (w * (pred - label.double()) ** 2).sum()
+        return torch.stack([se, w.sum()])
+
+    def finalize(self, s):
+        return float(torch.sqrt(s[0] / s[1]))
+
+
+class MAE(Metric):
+    name = "mae"
+
+    def local_stats(self, margin, label, weight, qid, obj):
+        w = _w(label, weight)
+        return torch.stack(
+            [(w * (margin.double() - label.double()).abs()).sum(), w.sum()]
+        )
+
+    def finalize(self, s):
+        return float(s[0] / s[1])
+
+
+class LogLoss(Metric):
+    name = "logloss"
+
+    def local_stats(self, margin, label, weight, qid, obj):
+        p = torch.sigmoid(margin.double()).clamp(1e-16, 1 - 1e-16)
+        y = label.double()
+        w = _w(label, weight)
+        ll = -(y * torch.log(p) + (1 - y) * torch.log(1 - p))
+        return torch.stack([(w * ll).sum(), w.sum()])
+
+    def finalize(self, s):
+        return float(s[0] / s[1])
+
+
+class BinaryError(Metric):
+    name = "error"
+
+    def __init__(self, threshold: float = 0.5):
+        self.threshold = threshold
+        if threshold != 0.5:
+            self.name = f"error@{threshold}"
+
+    def local_stats(self, margin, label, weight, qid, obj):
+        p = torch.sigmoid(margin.double())
+        w = _w(label, weight)
+        wrong = (p > self.threshold).double() != label.double()
+        return torch.stack([(w * wrong).sum(), w.sum()])
+
+    def finalize(self, s):
+        return float(s[0] / s[1])
+
+
+class AUC(Metric):
+    """Binned ROC AUC: weighted pos/neg histograms over sigmoid(margin)."""
+
+    name = "auc"
+    higher_better = True
+
+    def local_stats(self, margin, label, weight, qid, obj):
+        p = torch.sigmoid(margin.double())
+        w = _w(label, weight)
+        b = torch.clamp((p * _AUC_BINS).long(), max=_AUC_BINS - 1)
+        pos = torch.zeros(_AUC_BINS, dtype=torch.float64, device=margin.device)
+        neg = torch.zeros(_AUC_BINS, dtype=torch.float64, device=margin.device)
+        y = label.double()
+        pos.scatter_add_(0, b, w * y)
+        neg.scatter_add_(0, b, w * (1 - y))
+        return torch.cat([pos, neg])
+
+    def finalize(self, s):
+        pos, neg = s[:_AUC_BINS], s[_AUC_BINS:]
+        # iterate bins ascending: AUC = sum over bins of
+        # neg_below * pos_here + 0.5 * neg_here * pos_here
+        cneg = torch.cumsum(neg, 0) - neg
+        auc = (cneg * pos).sum() + 0.5 * (neg * pos).sum()
+        denom = pos.sum() * neg.sum()
+        return float(auc / denom) if denom > 0 else 0.5
+
+
+class MLogLoss(Metric):
+    name = "mlogloss"
+
+    def local_stats(self, margin, label, weight, qid, obj):
+        logp = torch.log_softmax(margin.double(), dim=1)
+        w = _w(label, weight)
+        ll = -logp[torch.arange(len(label), device=margin.device), label.long()]
+        return torch.stack([(w * ll).sum(), w.sum()])
+
+    def finalize(self, s):
+        return float(s[0] / s[1])
+
+
+class MError(Metric):
+    name = "merror"
+
+    def local_stats(self, margin, label, weight, qid, obj):
+        pred = margin.argmax(dim=1)
+        w = _w(label, weight)
+        wrong = (pred != label.long()).double()
+        return torch.stack([(w * wrong).sum(), w.sum()])
+
+    def finalize(self, s):
+        return float(s[0] / s[1])
+
+
+class _PerGroupMetric(Metric):
+    higher_better = True
+
+    def group_score(self, m, y) -> float:
+        raise NotImplementedError
+
+    def local_stats(self, margin, label, weight, qid, obj):
+        if qid is None:
+            qid = torch.zeros(len(label), dtype=torch.int64, device=margin.device)
+        total = torch.zeros(2, dtype=torch.float64, device=margin.device)
+        change = torch.ones_like(qid, dtype=torch.bool)
+        change[1:] = qid[1:] != qid[:-1]
+        starts = torch.nonzero(change).flatten().tolist() + [len(qid)]
+        for gi in range(len(starts) - 1):
+            s, e = starts[gi], starts[gi + 1]
+            total[0] += self.group_score(margin[s:e], label[s:e])
+            total[1] += 1
+        return total
+
+    def finalize(self, s):
+        return float(s[0] / s[1]) if s[1] > 0 else 0.0
+
+
+class NDCG(_PerGroupMetric):
+    name = "ndcg"
+
+    def __init__(self, k: Optional[int] = None):
+        self.k = k
+        if k:
+            self.name = f"ndcg@{k}"
+
+    def group_score(self, m, y):
+        n = len(y)
+        k = min(self.k or n, n)
+        order = torch.argsort(m.double(), descending=True, stable=True)
+        gains = torch.pow(2.0, y.double()) - 1.0
+        disc = 1.0 / torch.log2(
+            torch.arange(n, device=m.device, dtype=torch.float64) + 2.0
+        )
+        dcg = (gains[order][:k] * disc[:k]).sum()
+        ideal, _ = torch.sort(gains, descending=True)
+        idcg = (ideal[:k] * disc[:k]).sum()
+        return float(dcg / idcg) if idcg > 0 else 1.0
+
+
+class MAP(_PerGroupMetric):
+    name = "map"
+
+    def group_score(self, m, y):
+        order = torch.argsort(m.double(), descending=True, stable=True)
+        rel = (y[order] > 0).double()
+        if rel.sum() == 0:
+            return 1.0
+        csum = torch.cumsum(rel, 0)
+        ranks = torch.arange(1, len(y) + 1, device=m.device, dtype=torch.float64)
+        ap = ((csum / ranks) * rel).sum() / rel.sum()
+        return float(ap)
+
+
+def get_metric(name: str) -> Metric:
+    if name.startswith("error@"):
+        return BinaryError(float(name.split("@")[1]))
+    if name.startswith("ndcg@"):
+        return NDCG(int(name.split("@")[1]))
+    table = {
+        "rmse": RMSE,
+        "mae": MAE,
+        "logloss": LogLoss,
+        "error": BinaryError,
+        "auc": AUC,
+        "mlogloss": MLogLoss,
+        "merror": MError,
+        "ndcg": NDCG,
+        "map": MAP,
+    }
+    if name not in table:
+        raise ValueError(f"Unsupported eval_metric: {name}")
+    return table[name]()

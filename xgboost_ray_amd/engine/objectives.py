@@ -1,0 +1,288 @@
+"""Objective functions: per-row gradient/hessian computation.
+
+MI355X-native equivalent of XGBoost's C++/CUDA objective registry
+(reference delegates these to libxgboost; SURVEY.md #2.3 'Objectives &
+metrics'). All math runs on torch tensors so the same code executes on CPU
+(tests) and on GPU (fused elementwise kernels take over for the hot ones).
+Custom Python objectives (callables) stay supported, matching the
+reference's passthrough (reference test_xgboost_api.py:77-152).
+"""
+
+import math
+from typing import Optional
+
+import torch
+
+_EPS = 1e-16
+
+
+class Objective:
+    name = "base"
+    n_class = 0
+    default_metric = "rmse"
+
+    def prob_to_margin(self, base_score: float) -> float:
+        return base_score
+
+    def gradients(
+        self,
+        margin: torch.Tensor,
+        label: torch.Tensor,
+        weight: Optional[torch.Tensor] = None,
+        qid: Optional[torch.Tensor] = None,
+    ) -> torch.Tensor:
+        """Return fp32 gpair [n, 2] (or [n, k, 2] for multiclass)."""
+        raise NotImplementedError
+
+    def transform_prediction(self, margin: torch.Tensor) -> torch.Tensor:
+        return margin
+
+    def validate_label(self, label: torch.Tensor):
+        pass
+
+    def _apply_weight(self, g, h, weight):
+        if weight is not None:
+            g = g * weight
+            h = h * weight
+        return torch.stack([g, h], dim=-1)
+
+
+class SquaredError(Objective):
+    name = "reg:squarederror"
+    default_metric = "rmse"
+
+    def gradients(self, margin, label, weight=None, qid=None):
+        g = margin - label
+        h = torch.ones_like(margin)
+        return self._apply_weight(g, h, weight)
+
+
+class AbsoluteError(Objective):
+    name = "reg:absoluteerror"
+    default_metric = "mae"
+
+    def gradients(self, margin, label, weight=None, qid=None):
+        g = torch.sign(margin - label)
+        h = torch.ones_like(margin)
+        return self._apply_weight(g, h, weight)
+
+
+class Logistic(Objective):
+    """binary:logistic - probability output."""
+
+    name = "binary:logistic"
+    default_metric = "logloss"
+
+    def prob_to_margin(self, base_score):
+        base_score = min(max(base_score, _EPS), 1.0 - _EPS)
+        return -math.log(1.0 / base_score - 1.0)
+
+    def gradients(self, margin, label, weight=None, qid=None):
+        p = torch.sigmoid(margin)
+        g = p - label
+        h = torch.clamp(p * (1.0 - p), min=_EPS)
+        return self._apply_weight(g, h, weight)
+
+    def transform_prediction(self, margin):
+        return torch.sigmoid(margin)
+
+    def validate_label(self, label):
+        if bool((label < 0).any()) or bool((label > 1).any()):
+            raise ValueError("label must be in [0,1] for binary:logistic")
+
+
+class LogisticRaw(Logistic):
+    """binary:logitraw - margin output."""
+
+    name = "binary:logitraw"
+    default_metric = "logloss"
+
+    def transform_prediction(self, margin):
+        return margin
+
+
+class RegLogistic(Logistic):
+    name = "reg:logistic"
+    default_metric = "rmse"
+
+
+class SoftmaxBase(Objective):
+    default_metric = "mlogloss"
+
+    def __init__(self, n_class: int):
+        if n_class < 2:
+            raise ValueError("multi objectives need num_class >= 2")
+        self.n_class = n_class
+
+    def gradients(self, margin, label, weight=None, qid=None):
+        # margin: [n, k]
+        p = torch.softmax(margin, dim=1)
+        y = torch.nn.functional.one_hot(label.long(), self.n_class).to(p.dtype)
+        g = p - y
+        h = torch.clamp(2.0 * p * (1.0 - p), min=_EPS)
+        if weight is not None:
+            g = g * weight.unsqueeze(1)
+            h = h * weight.unsqueeze(1)
+        return torch.stack([g, h], dim=-1)  # [n, k, 2]
+
+    def validate_label(self, label):
+        mx = int(label.max()) if label.numel() else 0
+        if mx >= self.n_class:
+            raise ValueError(
+                f"label {mx} >= num_class {self.n_class}"
+            )
+
+
+class SoftmaxClass(SoftmaxBase):
+    name = "multi:softmax"
+    default_metric = "merror"
+
+    def transform_prediction(self, margin):
+        return margin.argmax(dim=1).to(torch.float32)
+
+
+class SoftProb(SoftmaxBase):
+    name = "multi:softprob"
+    default_metric = "mlogloss"
+
+    def transform_prediction(self, margin):
+        return torch.softmax(margin, dim=1)
+
+
+class LambdaRankBase(Objective):
+    """Pairwise LambdaRank objectives over qid groups.
+
+    Deterministic all-pairs formulation (pairs of rows with different
+    relevance within a query group); NDCG variant weights each pair by its
+    |delta-NDCG|. Reference semantics: rank:pairwise / rank:ndcg passed
+    through to libxgboost (reference test_end_to_end.py:374-424).
+    """
+
+    ndcg_weighting = False
+    default_metric = "map"
+
+    def gradients(self, margin, label, weight=None, qid=None):
+        if qid is None:
+            raise ValueError("rank objectives require qid/group information")
+        device = margin.device
+        g = torch.zeros_like(margin)
+        h = torch.zeros_like(margin)
+        qid = qid.to(device)
+        # group boundaries (qid must be sorted)
+        change = torch.ones_like(qid, dtype=torch.bool)
+        change[1:] = qid[1:] != qid[:-1]
+        starts = torch.nonzero(change).flatten().tolist() + [len(qid)]
+        for gi in range(len(starts) - 1):
+            s, e = starts[gi], starts[gi + 1]
+            if e - s < 2:
+                continue
+            m = margin[s:e]
+            y = label[s:e]
+            diff_y = y.unsqueeze(1) - y.unsqueeze(0)  # [g, g]
+            pos = diff_y > 0  # i more relevant than j
+            if not bool(pos.any()):
+                continue
+            sij = m.unsqueeze(1) - m.unsqueeze(0)
+            rho = torch.sigmoid(-sij)  # d/ds_i of log-loss for pair (i,j)
+            lam = rho
+            hess = torch.clamp(rho * (1.0 - rho), min=_EPS)
+            if self.ndcg_weighting:
+                # |delta NDCG| for swapping i, j at current ranking
+                order = torch.argsort(m, descending=True)
+                rank = torch.empty_like(order)
+                rank[order] = torch.arange(e - s, device=device)
+                gain = torch.pow(2.0, y) - 1.0
+                disc = 1.0 / torch.log2(rank.to(m.dtype) + 2.0)
+                ideal_gain, _ = torch.sort(gain, descending=True)
+                ideal_disc = 1.0 / torch.log2(
+                    torch.arange(e - s, device=device, dtype=m.dtype) + 2.0
+                )
+                idcg = (ideal_gain * ideal_disc).sum()
+                if idcg <= 0:
+                    continue
+                dg = gain.unsqueeze(1) - gain.unsqueeze(0)
+                dd = disc.unsqueeze(1) - disc.unsqueeze(0)
+                delta = torch.abs(dg * dd) / idcg
+                lam = lam * delta
+                hess = hess * delta
+            lam = torch.where(pos, lam, torch.zeros_like(lam))
+            hess = torch.where(pos, hess, torch.zeros_like(hess))
+            # pair (i, j): i should rank above j -> push m_i up, m_j down
+            g[s:e] += -lam.sum(dim=1) + lam.sum(dim=0)
+            h[s:e] += hess.sum(dim=1) + hess.sum(dim=0)
+        return self._apply_weight(g, h, weight)
+
+
+class RankPairwise(LambdaRankBase):
+    name = "rank:pairwise"
+    default_metric = "map"
+
+
+class RankNDCG(LambdaRankBase):
+    name = "rank:ndcg"
+    ndcg_weighting = True
+    default_metric = "ndcg"
+
+
+class CustomObjective(Objective):
+    """Wraps a user callable obj(preds, dtrain) -> (grad, hess).
+
+    Matches the xgboost custom-objective convention the reference passes
+    through to every actor (reference test_xgboost_api.py:77-152).
+    """
+
+    name = "custom"
+    default_metric = "rmse"
+
+    def __init__(self, fn, n_class=0):
+        self.fn = fn
+        self.n_class = n_class
+
+    def gradients(self, margin, label, weight=None, qid=None):
+        import numpy as np
+
+        class _Shim:
+            def __init__(self, label, weight):
+                self._label = label
+                self._weight = weight
+
+            def get_label(self):
+                return self._label
+
+            def get_weight(self):
+                return self._weight
+
+        preds = margin.detach().cpu().numpy()
+        shim = _Shim(
+            label.detach().cpu().numpy(),
+            None if weight is None else weight.detach().cpu().numpy(),
+        )
+        grad, hess = self.fn(preds, shim)
+        g = torch.as_tensor(np.asarray(grad), dtype=torch.float32, device=margin.device)
+        h = torch.as_tensor(np.asarray(hess), dtype=torch.float32, device=margin.device)
+        return torch.stack([g.reshape(margin.shape), h.reshape(margin.shape)], dim=-1)
+
+
+_REGISTRY = {
+    "reg:squarederror": SquaredError,
+    "reg:linear": SquaredError,  # legacy alias
+    "reg:absoluteerror": AbsoluteError,
+    "reg:logistic": RegLogistic,
+    "binary:logistic": Logistic,
+    "binary:logitraw": LogisticRaw,
+    "rank:pairwise": RankPairwise,
+    "rank:ndcg": RankNDCG,
+}
+
+
+def get_objective(name_or_fn, num_class: int = 0) -> Objective:
+    if callable(name_or_fn):
+        return CustomObjective(name_or_fn, num_class)
+    name = name_or_fn or "reg:squarederror"
+    if name in ("multi:softmax",):
+        return SoftmaxClass(num_class)
+    if name in ("multi:softprob",):
+        return SoftProb(num_class)
+    if name not in _REGISTRY:
+        raise ValueError(f"Unsupported objective: {name}")
+    return _REGISTRY[name]()

@@ -1,0 +1,709 @@
+"""The boosting engine: depth-wise GPU histogram tree growing.
+
+MI355X-native replacement for XGBoost's ``gpu_hist`` updater + training
+loop (the reference delegates this entirely to ``xgb.train``, reference
+main.py:745; SURVEY.md #2.3 row 2 calls it the centerpiece). Per round:
+
+  grad/hess kernel -> int64 fixed-point quantization -> per-depth
+  LDS-tiled histogram build (smaller child only; sibling derived by the
+  histogram-subtraction trick) -> RCCL AllReduce of the built histograms
+  over xGMI -> best-split scan -> stable row partition -> next depth.
+
+Int64 fixed-point gradients make every reduction an integer sum:
+bitwise-deterministic regardless of GPU atomic ordering, feature-block
+schedule, or world size - which is what makes checkpoint-resume models
+exactly reproducible (reference test_fault_tolerance.py:401-449).
+"""
+
+import math
+from dataclasses import dataclass, field
+from typing import Callable, Dict, List, Optional, Sequence, Tuple, Union
+
+import numpy as np
+import torch
+
+from xgboost_ray_amd import ops
+from xgboost_ray_amd.booster import Booster, Tree
+from xgboost_ray_amd.engine.collective import Collective
+from xgboost_ray_amd.engine.metrics import get_metric
+from xgboost_ray_amd.engine.objectives import Objective, get_objective
+from xgboost_ray_amd.engine.quantile import BinnedMatrix
+
+_QUANT_BITS = 30
+
+
+@dataclass
+class TrainParams:
+    objective: Union[str, Callable] = "reg:squarederror"
+    eta: float = 0.3
+    max_depth: int = 6
+    max_bin: int = 256
+    reg_lambda: float = 1.0
+    reg_alpha: float = 0.0
+    gamma: float = 0.0
+    min_child_weight: float = 1.0
+    subsample: float = 1.0
+    colsample_bytree: float = 1.0
+    colsample_bylevel: float = 1.0
+    base_score: Optional[float] = None
+    num_class: int = 0
+    eval_metric: List[str] = field(default_factory=list)
+    seed: int = 0
+    tree_method: str = "hist"
+    verbosity: int = 1
+    nthread: int = 0
+    disable_default_eval_metric: bool = False
+
+    @classmethod
+    def from_dict(cls, params: Dict) -> "TrainParams":
+        p = cls()
+        alias = {
+            "learning_rate": "eta",
+            "lambda": "reg_lambda",
+            "alpha": "reg_alpha",
+            "min_split_loss": "gamma",
+            "random_state": "seed",
+            "n_jobs": "nthread",
+        }
+        for k, v in (params or {}).items():
+            k = alias.get(k, k)
+            if k == "eval_metric":
+                p.eval_metric = [v] if isinstance(v, str) else list(v)
+            elif hasattr(p, k):
+                setattr(p, k, v)
+            # unknown params are accepted silently (xgboost behavior)
+        p.eta = float(p.eta)
+        p.max_depth = int(p.max_depth)
+        p.max_bin = int(p.max_bin)
+        p.num_class = int(p.num_class or 0)
+        p.seed = int(p.seed or 0)
+        return p
+
+
+@dataclass
+class EvalPack:
+    """One eval set: raw features + incremental margin state."""
+
+    name: str
+    X: Optional[torch.Tensor]  # None => the training matrix itself
+    label: Optional[torch.Tensor] = None
+    weight: Optional[torch.Tensor] = None
+    qid: Optional[torch.Tensor] = None
+    base_margin: Optional[torch.Tensor] = None
+    margin: Optional[torch.Tensor] = None
+
+
+@dataclass
+class _Node:
+    nid: int
+    depth: int
+    start: int
+    count: int
+    sum_g: int  # quantized
+    sum_h: int
+    hist: Optional[torch.Tensor] = None  # [F, B, 2] int64 (this depth only)
+
+
+class CallbackList:
+    def __init__(self, callbacks):
+        self.callbacks = list(callbacks or [])
+
+    def before_iteration(self, booster, iteration, evals_log) -> bool:
+        stop = False
+        for cb in self.callbacks:
+            if hasattr(cb, "before_iteration"):
+                stop |= bool(cb.before_iteration(booster, iteration, evals_log))
+        return stop
+
+    def after_iteration(self, booster, iteration, evals_log) -> bool:
+        stop = False
+        for cb in self.callbacks:
+            if hasattr(cb, "after_iteration"):
+                stop |= bool(cb.after_iteration(booster, iteration, evals_log))
+            elif callable(cb):
+                stop |= bool(cb(booster, iteration, evals_log))
+        return stop
+
+
+class BoostingEngine:
+    """Single-rank engine; collectives make it data-parallel."""
+
+    def __init__(
+        self,
+        params: Dict,
+        dtrain: BinnedMatrix,
+        collective: Optional[Collective] = None,
+        rank: int = 0,
+        custom_objective: Optional[Callable] = None,
+    ):
+        self.p = TrainParams.from_dict(params)
+        self.raw_params = dict(params or {})
+        self.dtrain = dtrain
+        self.coll = collective or Collective()
+        self.rank = rank
+        self.device = dtrain.bins.device
+        obj_spec = custom_objective or self.p.objective
+        self.obj: Objective = get_objective(obj_spec, self.p.num_class)
+        self.n_class = max(1, self.p.num_class)
+        if self.p.base_score is None:
+            self.p.base_score = 0.5
+        self.base_margin_const = self.obj.prob_to_margin(self.p.base_score)
+
+        n = dtrain.n_rows
+        shape = (n,) if self.n_class == 1 else (n, self.n_class)
+        self.margin = torch.full(
+            shape, self.base_margin_const, dtype=torch.float32, device=self.device
+        )
+        if dtrain.base_margin is not None:
+            bm = dtrain.base_margin.reshape(shape)
+            self.margin = self.margin + bm
+        if dtrain.label is not None:
+            self.obj.validate_label(dtrain.label)
+        self.feat_bins = dtrain.cuts.feat_bins().to(self.device)
+        self.n_bins = dtrain.cuts.max_bins
+        self.iteration = 0
+        self.booster = Booster(
+            params={
+                "objective": getattr(self.obj, "name", "custom"),
+                "base_score": self.p.base_score,
+                "num_class": self.p.num_class,
+                "num_feature": dtrain.n_features,
+                "max_depth": self.p.max_depth,
+                "eta": self.p.eta,
+            }
+        )
+
+    # -- resume ------------------------------------------------------------
+    def load_model(self, model: Booster):
+        """Continue training from an existing model (xgb_model kwarg)."""
+        self.booster = model
+        self.iteration = model.num_boosted_rounds()
+        # recompute margins through the binned matrix (thresholds are exact
+        # cut values for models we trained, so binned replay is exact)
+        n = self.dtrain.n_rows
+        shape = (n,) if self.n_class == 1 else (n, self.n_class)
+        base = self.obj.prob_to_margin(model.base_score)
+        self.base_margin_const = base
+        self.margin = torch.full(
+            shape, base, dtype=torch.float32, device=self.device
+        )
+        if self.dtrain.base_margin is not None:
+            self.margin = self.margin + self.dtrain.base_margin.reshape(shape)
+        self._replay_trees_binned(model)
+
+    def _replay_trees_binned(self, model: Booster):
+        cuts = self.dtrain.cuts
+        cut_ptr = cuts.cut_ptr.cpu().numpy()
+        cuts_np = cuts.cuts_flat.cpu().numpy()
+        for i, t in enumerate(model.trees):
+            cls = model.tree_info[i] if i < len(model.tree_info) else 0
+            split_bin = np.zeros(t.num_nodes, dtype=np.int32)
+            for nid in range(t.num_nodes):
+                if t.feat[nid] >= 0:
+                    f = int(t.feat[nid])
+                    lo, hi = cut_ptr[f], cut_ptr[f + 1]
+                    b = np.searchsorted(cuts_np[lo:hi], t.thr[nid], side="left")
+                    split_bin[nid] = b - 1
+            margin_view = (
+                self.margin if self.n_class == 1 else self.margin[:, cls]
+            )
+            self._add_tree_margin_binned(t, split_bin, margin_view)
+
+    def _add_tree_margin_binned(self, t: Tree, split_bin, margin_view):
+        bins = self.dtrain.bins
+        n = bins.shape[0]
+        cur = torch.zeros(n, dtype=torch.int64, device=self.device)
+        feat = torch.from_numpy(t.feat.astype(np.int64)).to(self.device)
+        sbin = torch.as_tensor(split_bin, dtype=torch.int64, device=self.device)
+        left = torch.from_numpy(t.left.astype(np.int64)).to(self.device)
+        dl = torch.from_numpy(t.default_left.astype(np.uint8)).to(self.device)
+        val = torch.from_numpy(t.value).to(self.device)
+        while True:
+            f = feat[cur]
+            active = f >= 0
+            if not bool(active.any()):
+                break
+            rows = torch.nonzero(active).flatten()
+            fa = f[rows]
+            bv = bins[rows, fa].long()
+            miss = bv == ops.MISSING_BIN
+            go_left = bv <= sbin[cur[rows]]
+            go_left = torch.where(miss, dl[cur[rows]].bool(), go_left)
+            cur[rows] = torch.where(go_left, left[cur[rows]], left[cur[rows]] + 1)
+        margin_view += val[cur]
+
+    # -- one boosting round --------------------------------------------------
+    def update(self):
+        it = self.iteration
+        label = self.dtrain.label
+        gpair = self.obj.gradients(
+            self.margin, label, self.dtrain.weight, self.dtrain.qid
+        )
+        trees, classes = [], []
+        for cls in range(self.n_class):
+            gp = gpair if self.n_class == 1 else gpair[:, cls, :]
+            tree = self._grow_tree(gp.contiguous(), it, cls)
+            trees.append(tree)
+            classes.append(cls)
+        self.booster.append_round(trees, classes)
+        self.iteration += 1
+        return trees
+
+    def _quantize(self, gpair: torch.Tensor) -> Tuple[torch.Tensor, float, float]:
+        mx = torch.stack(
+            [gpair[:, 0].abs().max(), gpair[:, 1].abs().max()]
+        ).double()
+        if self.coll.is_distributed:
+            mx_d = mx.to(self.device) if self.device.type == "cuda" else mx
+            self.coll.allreduce_(mx_d, op="max")
+            mx = mx_d.cpu()
+        scale_g = float(2.0**_QUANT_BITS) / max(float(mx[0]), 1e-300)
+        scale_h = float(2.0**_QUANT_BITS) / max(float(mx[1]), 1e-300)
+        return ops.quantize_gpair(gpair, scale_g, scale_h), scale_g, scale_h
+
+    def _sample_rows(self, it: int, cls: int) -> torch.Tensor:
+        n = self.dtrain.n_rows
+        if self.p.subsample >= 1.0:
+            return torch.arange(n, dtype=torch.int32, device=self.device)
+        gen = torch.Generator(device="cpu")
+        gen.manual_seed(
+            (self.p.seed * 2654435761 + it * 97 + cls * 31 + self.rank) % (2**63)
+        )
+        mask = torch.rand(n, generator=gen) < self.p.subsample
+        return torch.nonzero(mask.to(self.device)).flatten().to(torch.int32)
+
+    def _sample_features(self, it: int, cls: int) -> Optional[torch.Tensor]:
+        F = self.dtrain.n_features
+        if self.p.colsample_bytree >= 1.0:
+            return None
+        k = max(1, int(round(F * self.p.colsample_bytree)))
+        gen = torch.Generator(device="cpu")
+        gen.manual_seed((self.p.seed * 2654435761 + it * 97 + cls * 31 + 7) % (2**63))
+        perm = torch.randperm(F, generator=gen)[:k]
+        mask = torch.zeros(F, dtype=torch.bool)
+        mask[perm] = True
+        return mask.to(self.device)
+
+    def _grow_tree(self, gpair: torch.Tensor, it: int, cls: int) -> Tree:
+        gq, scale_g, scale_h = self._quantize(gpair)
+        self._scale_g_cur = scale_g
+        self._leaf_segs = []
+        ridx = self._sample_rows(it, cls)
+        feat_mask = self._sample_features(it, cls)
+        n_local = int(ridx.numel())
+
+        root_sum = gq[ridx.long()].sum(dim=0) if n_local else torch.zeros(
+            2, dtype=torch.int64, device=self.device
+        )
+        if self.coll.is_distributed:
+            self.coll.allreduce_(root_sum)
+        root = _Node(
+            nid=0,
+            depth=0,
+            start=0,
+            count=n_local,
+            sum_g=int(root_sum[0]),
+            sum_h=int(root_sum[1]),
+        )
+
+        # growing tree arrays (python lists; tree sizes are tiny)
+        feat_l, thr_l, sbin_l = [-1], [0.0], [-1]
+        left_l, dl_l, val_l, gain_l, cover_l = [-1], [0], [0.0], [0.0], [0.0]
+        parent_l = [-1]
+
+        cuts_flat_cpu = self.dtrain.cuts.cuts_flat.cpu().numpy()
+        cut_ptr_cpu = self.dtrain.cuts.cut_ptr.cpu().numpy()
+
+        frontier: List[_Node] = [root]
+        parent_hist: Dict[int, torch.Tensor] = {}
+        # map nid -> node bookkeeping for building
+        for depth in range(self.p.max_depth):
+            if not frontier:
+                break
+            # ---- decide build list: root, or per sibling pair the smaller
+            build_nodes: List[_Node] = []
+            derive_nodes: List[Tuple[_Node, _Node]] = []  # (derive, built_sibling)
+            if depth == 0:
+                build_nodes = [root]
+            else:
+                i = 0
+                while i < len(frontier):
+                    a = frontier[i]
+                    b = frontier[i + 1] if i + 1 < len(frontier) else None
+                    if b is not None and parent_l[a.nid] == parent_l[b.nid]:
+                        # siblings: build the globally smaller one. Use the
+                        # quantized hessian sum as the size proxy - it is
+                        # identical on every rank (local counts are not).
+                        small, big = (a, b) if a.sum_h <= b.sum_h else (b, a)
+                        build_nodes.append(small)
+                        derive_nodes.append((big, small))
+                        i += 2
+                    else:
+                        build_nodes.append(a)
+                        i += 1
+
+            K = len(build_nodes)
+            starts = torch.tensor(
+                [nd.start for nd in build_nodes], dtype=torch.int64
+            )
+            counts = torch.tensor(
+                [nd.count for nd in build_nodes], dtype=torch.int64
+            )
+            hist = ops.build_histogram(
+                self.dtrain.bins, gq, ridx, starts, counts, self.n_bins
+            )
+            if self.coll.is_distributed:
+                self.coll.allreduce_(hist)
+            for k, nd in enumerate(build_nodes):
+                nd.hist = hist[k]
+            for big, small in derive_nodes:
+                pid = parent_l[big.nid]
+                big.hist = parent_hist[pid] - small.hist
+            parent_hist = {}
+
+            # ---- split scan over the whole frontier
+            all_hist = torch.stack([nd.hist for nd in frontier])
+            pg = torch.tensor(
+                [nd.sum_g for nd in frontier], dtype=torch.int64, device=self.device
+            )
+            ph = torch.tensor(
+                [nd.sum_h for nd in frontier], dtype=torch.int64, device=self.device
+            )
+            fb = self.feat_bins
+            if feat_mask is not None:
+                fb = torch.where(
+                    feat_mask, self.feat_bins, torch.zeros_like(self.feat_bins)
+                )
+            best = ops.find_splits(
+                all_hist,
+                pg,
+                ph,
+                fb,
+                scale_g,
+                scale_h,
+                self.p.reg_lambda,
+                self.p.reg_alpha,
+                self.p.gamma,
+                self.p.min_child_weight,
+            )
+            gain = best["gain"].cpu().numpy()
+            bfeat = best["feature"].cpu().numpy()
+            bbin = best["bin"].cpu().numpy()
+            bdl = best["default_left"].cpu().numpy()
+            blg = best["left_g"].cpu().numpy()
+            blh = best["left_h"].cpu().numpy()
+
+            split_nodes: List[_Node] = []
+            sf, sb, sdl = [], [], []
+            children_meta = []
+            for k, nd in enumerate(frontier):
+                if gain[k] <= 0 or bfeat[k] < 0 or not np.isfinite(gain[k]):
+                    self._finalize_leaf(nd, val_l, cover_l, scale_h)
+                    continue
+                f = int(bfeat[k])
+                b = int(bbin[k])
+                # split after bin b: left iff bin <= b; bin b covers
+                # [cut[b-1], cut[b]) so "bin <= b" <=> v < cut[b] = thr.
+                thr = float(cuts_flat_cpu[cut_ptr_cpu[f] + b])
+                lid = len(feat_l)
+                feat_l[nd.nid] = f
+                thr_l[nd.nid] = thr
+                sbin_l[nd.nid] = b
+                left_l[nd.nid] = lid
+                dl_l[nd.nid] = int(bdl[k])
+                gain_l[nd.nid] = float(gain[k])
+                cover_l[nd.nid] = float(nd.sum_h) / scale_h
+                for _ in range(2):
+                    feat_l.append(-1)
+                    thr_l.append(0.0)
+                    sbin_l.append(-1)
+                    left_l.append(-1)
+                    dl_l.append(0)
+                    val_l.append(0.0)
+                    gain_l.append(0.0)
+                    cover_l.append(0.0)
+                    parent_l.append(nd.nid)
+                split_nodes.append(nd)
+                sf.append(f)
+                sb.append(b)
+                sdl.append(int(bdl[k]))
+                children_meta.append(
+                    (lid, int(blg[k]), int(blh[k]))
+                )
+
+            if not split_nodes:
+                frontier = []
+                break
+
+            sstarts = torch.tensor(
+                [nd.start for nd in split_nodes], dtype=torch.int64
+            )
+            scounts = torch.tensor(
+                [nd.count for nd in split_nodes], dtype=torch.int64
+            )
+            ridx, left_counts = ops.partition_rows(
+                self.dtrain.bins,
+                ridx,
+                sstarts,
+                scounts,
+                torch.tensor(sf, dtype=torch.int32),
+                torch.tensor(sb, dtype=torch.int32),
+                torch.tensor(sdl, dtype=torch.uint8),
+            )
+
+            new_frontier: List[_Node] = []
+            for k, nd in enumerate(split_nodes):
+                lid, lg, lh = children_meta[k]
+                lcount = int(left_counts[k])
+                lnode = _Node(
+                    nid=lid,
+                    depth=depth + 1,
+                    start=nd.start,
+                    count=lcount,
+                    sum_g=lg,
+                    sum_h=lh,
+                )
+                rnode = _Node(
+                    nid=lid + 1,
+                    depth=depth + 1,
+                    start=nd.start + lcount,
+                    count=nd.count - lcount,
+                    sum_g=nd.sum_g - lg,
+                    sum_h=nd.sum_h - lh,
+                )
+                parent_hist[nd.nid] = nd.hist
+                nd.hist = None
+                new_frontier.extend([lnode, rnode])
+            frontier = new_frontier
+
+        # remaining frontier nodes (max depth reached) become leaves
+        final_segments = []
+        for nd in frontier:
+            self._finalize_leaf(nd, val_l, cover_l, scale_h)
+        # leaf margin update: walk all leaves via the segment structure -
+        # every row's final node is its segment's node. Collect leaf segs.
+        # (split nodes consumed their segments; leaves kept them)
+        # We track them via a second pass: recompute from the tree by
+        # replaying partitions is wasteful; instead we collected leaf info
+        # in _finalize_leaf.
+        leaf_starts = [s for (s, c, v) in self._leaf_segs]
+        leaf_counts = [c for (s, c, v) in self._leaf_segs]
+        leaf_vals = [v for (s, c, v) in self._leaf_segs]
+        if leaf_starts:
+            ops.update_margins(
+                self.margin if self.n_class == 1 else self.margin[:, cls],
+                ridx,
+                torch.tensor(leaf_starts, dtype=torch.int64),
+                torch.tensor(leaf_counts, dtype=torch.int64),
+                np.asarray(leaf_vals, dtype=np.float32),
+            )
+        self._leaf_segs = []
+
+        tree = Tree(
+            feat=np.asarray(feat_l, np.int32),
+            thr=np.asarray(thr_l, np.float32),
+            left=np.asarray(left_l, np.int32),
+            default_left=np.asarray(dl_l, np.uint8),
+            value=np.asarray(val_l, np.float32),
+            gain=np.asarray(gain_l, np.float32),
+            cover=np.asarray(cover_l, np.float32),
+            parent=np.asarray(
+                [2147483647 if p < 0 else p for p in parent_l], np.int32
+            ),
+        )
+        return tree
+
+    _leaf_segs: List[Tuple[int, int, float]] = []
+
+    def _finalize_leaf(self, nd: _Node, val_l, cover_l, scale_h):
+        G = nd.sum_g / self._scale_g_cur
+        H = nd.sum_h / scale_h
+        w = _calc_weight(G, H, self.p.reg_lambda, self.p.reg_alpha)
+        val_l[nd.nid] = self.p.eta * w
+        cover_l[nd.nid] = H
+        self._leaf_segs.append((nd.start, nd.count, self.p.eta * w))
+
+    # -- evaluation ----------------------------------------------------------
+    def eval_sets(self, evals: Sequence[EvalPack], feval=None) -> Dict[str, Dict[str, float]]:
+        metric_names = list(self.p.eval_metric)
+        if not metric_names and not self.p.disable_default_eval_metric:
+            metric_names = [self.obj.default_metric]
+        out: Dict[str, Dict[str, float]] = {}
+        local_stats = []
+        specs = []
+        for ev in evals:
+            margin = self.margin if ev.X is None else ev.margin
+            label = self.dtrain.label if ev.X is None else ev.label
+            weight = self.dtrain.weight if ev.X is None else ev.weight
+            qid = self.dtrain.qid if ev.X is None else ev.qid
+            for mname in metric_names:
+                m = get_metric(mname)
+                st = m.local_stats(margin, label, weight, qid, self.obj)
+                local_stats.append(st)
+                specs.append((ev.name, m))
+        if specs:
+            flat = torch.cat([s.flatten() for s in local_stats])
+            if self.coll.is_distributed:
+                flat_d = flat.to(self.device) if self.device.type == "cuda" else flat
+                self.coll.allreduce_(flat_d)
+                flat = flat_d.cpu()
+            pos = 0
+            for (ename, m), st in zip(specs, local_stats):
+                sz = st.numel()
+                val = m.finalize(flat[pos : pos + sz])
+                pos += sz
+                out.setdefault(ename, {})[m.name] = val
+        if feval is not None:
+            for ev in evals:
+                margin = self.margin if ev.X is None else ev.margin
+                label = self.dtrain.label if ev.X is None else ev.label
+
+                class _Shim:
+                    def __init__(self, l):
+                        self._l = l
+
+                    def get_label(self):
+                        return self._l
+
+                name, val = feval(
+                    margin.cpu().numpy(),
+                    _Shim(None if label is None else label.cpu().numpy()),
+                )
+                out.setdefault(ev.name, {})[name] = float(val)
+        return out
+
+    def update_eval_margins(self, evals: Sequence[EvalPack], trees, classes):
+        """Incrementally add the new round's trees to eval-set margins."""
+        for ev in evals:
+            if ev.X is None:
+                continue
+            for t, cls in zip(trees, classes):
+                mv = ev.margin if self.n_class == 1 else ev.margin[:, cls]
+                flat_feat = torch.from_numpy(t.feat).to(ev.X.device)
+                ops.predict_trees(
+                    ev.X,
+                    flat_feat,
+                    torch.from_numpy(t.thr).to(ev.X.device),
+                    torch.from_numpy(t.left).to(ev.X.device),
+                    torch.from_numpy(t.default_left).to(ev.X.device),
+                    torch.from_numpy(t.value).to(ev.X.device),
+                    torch.tensor([0, t.num_nodes], dtype=torch.int32),
+                    mv,
+                )
+
+    def init_eval_margins(self, evals: Sequence[EvalPack], model: Optional[Booster]):
+        for ev in evals:
+            if ev.X is None:
+                continue
+            n = ev.X.shape[0]
+            shape = (n,) if self.n_class == 1 else (n, self.n_class)
+            ev.margin = torch.full(
+                shape,
+                self.base_margin_const,
+                dtype=torch.float32,
+                device=ev.X.device,
+            )
+            if ev.base_margin is not None:
+                ev.margin += ev.base_margin.reshape(shape)
+            if model is not None and model.trees:
+                pm = model.predict_margin_tensor(ev.X)
+                ev.margin = pm.to(torch.float32) + (
+                    ev.base_margin.reshape(shape) if ev.base_margin is not None else 0
+                )
+
+    # quantization scale of the current tree (set in _grow_tree via _quantize)
+    _scale_g_cur: float = 1.0
+
+
+def _calc_weight(G: float, H: float, reg_lambda: float, reg_alpha: float) -> float:
+    if H + reg_lambda <= 0:
+        return 0.0
+    if reg_alpha > 0:
+        G = math.copysign(max(abs(G) - reg_alpha, 0.0), G)
+    return -G / (H + reg_lambda)
+
+
+def run_training(
+    params: Dict,
+    dtrain: BinnedMatrix,
+    num_boost_round: int,
+    evals: Sequence[EvalPack] = (),
+    collective: Optional[Collective] = None,
+    rank: int = 0,
+    xgb_model: Optional[Booster] = None,
+    callbacks=None,
+    early_stopping_rounds: Optional[int] = None,
+    verbose_eval: Union[bool, int] = False,
+    obj: Optional[Callable] = None,
+    feval: Optional[Callable] = None,
+    evals_result: Optional[Dict] = None,
+) -> Booster:
+    """The per-rank training loop (reference: xgb.train inside
+    RayXGBoostActor.train, reference main.py:722-752)."""
+    engine = BoostingEngine(
+        params, dtrain, collective, rank, custom_objective=obj
+    )
+    if xgb_model is not None:
+        engine.load_model(xgb_model)
+    engine.init_eval_margins(evals, xgb_model)
+    cb = CallbackList(callbacks)
+    log: Dict[str, Dict[str, List[float]]] = {}
+    best_score, best_iter, stall = None, None, 0
+    maximize = None
+    start_iter = engine.iteration
+
+    for rnd in range(num_boost_round):
+        it = engine.iteration
+        if cb.before_iteration(engine.booster, it, log):
+            break
+        trees = engine.update()
+        classes = engine.booster.tree_info[-len(trees):]
+        engine.update_eval_margins(evals, trees, classes)
+        if evals:
+            results = engine.eval_sets(evals, feval)
+            for ename, md in results.items():
+                for mname, v in md.items():
+                    log.setdefault(ename, {}).setdefault(mname, []).append(v)
+            if verbose_eval and rank == 0:
+                interval = 1 if verbose_eval is True else int(verbose_eval)
+                if it % interval == 0:
+                    msg = "\t".join(
+                        f"{e}-{m}:{v:.5f}"
+                        for e, md in results.items()
+                        for m, v in md.items()
+                    )
+                    print(f"[{it}]\t{msg}", flush=True)
+            if early_stopping_rounds:
+                last_eval = list(results.keys())[-1]
+                last_metric = list(results[last_eval].keys())[-1]
+                score = results[last_eval][last_metric]
+                if maximize is None:
+                    maximize = get_metric(last_metric.split("@")[0]).higher_better
+                improved = (
+                    best_score is None
+                    or (score > best_score if maximize else score < best_score)
+                )
+                if improved:
+                    best_score, best_iter, stall = score, it, 0
+                else:
+                    stall += 1
+                    if stall >= early_stopping_rounds:
+                        engine.booster.best_iteration = best_iter
+                        engine.booster.best_score = best_score
+                        engine.booster.set_attr(
+                            best_iteration=str(best_iter),
+                            best_score=str(best_score),
+                        )
+                        break
+        if cb.after_iteration(engine.booster, it, log):
+            break
+
+    if best_iter is not None and engine.booster.best_iteration is None:
+        engine.booster.best_iteration = best_iter
+        engine.booster.best_score = best_score
+        engine.booster.set_attr(
+            best_iteration=str(best_iter), best_score=str(best_score)
+        )
+    if evals_result is not None:
+        evals_result.update(log)
+    return engine.booster

@@ -1,0 +1,142 @@
+"""Device-dispatched compute primitives for the boosting engine.
+
+Each op has two implementations:
+
+- a pure-PyTorch CPU reference (``ops.cpu``) used for CPU training and as the
+  numerics oracle in tests, and
+- a hand-written CDNA4 HIP kernel (``ops.gpu`` -> ``xgboost_ray_amd/csrc``)
+  used whenever tensors live on a GPU.
+
+On a machine with a GPU the HIP extension is REQUIRED: if a CUDA tensor
+reaches an op and the extension is not importable, we raise instead of
+silently falling back to slow eager code (this is the MI355X-native compute
+path, reference-equivalent of XGBoost's CUDA gpu_hist updater, see
+reference main.py:745 / SURVEY.md #2.3).
+
+Histogram determinism: gradients are quantized to int64 fixed point before
+histogram accumulation, so per-bin sums are integer adds - associative and
+therefore bitwise-identical regardless of atomic ordering, row partition
+order, or world size. This is what makes checkpoint-resume and
+elastic-restart models deterministic (reference test_fault_tolerance.py
+testSameResultWithAndWithoutError semantics).
+"""
+
+import torch
+
+from xgboost_ray_amd.ops import cpu as _cpu
+
+MISSING_BIN = 255  # uint8 marker for "value missing" in the binned matrix
+
+_gpu_mod = None
+_gpu_checked = False
+
+
+def _gpu():
+    """Load the HIP extension, failing loudly if unavailable."""
+    global _gpu_mod, _gpu_checked
+    if not _gpu_checked:
+        _gpu_checked = True
+        from xgboost_ray_amd.ops import gpu as gpu_backend
+
+        _gpu_mod = gpu_backend
+    if _gpu_mod is None:  # pragma: no cover
+        raise RuntimeError("GPU op requested but HIP extension failed to load")
+    return _gpu_mod
+
+
+def _impl(t: torch.Tensor):
+    return _gpu() if t.is_cuda else _cpu
+
+
+def quantize_gpair(gpair: torch.Tensor, scale_g: float, scale_h: float) -> torch.Tensor:
+    """fp32 [n,2] gradient pairs -> int64 [n,2] fixed point."""
+    return _impl(gpair).quantize_gpair(gpair, scale_g, scale_h)
+
+
+def bin_matrix(values, cuts_flat, cut_ptr):
+    """fp32 [n, F] feature matrix -> uint8 [n, F] bin indices.
+
+    bin = number of cut points strictly below the value, clamped to the
+    feature's bin count - 1; NaN -> MISSING_BIN.
+    """
+    return _impl(values).bin_matrix(values, cuts_flat, cut_ptr)
+
+
+def build_histogram(bins, gpair_q, ridx, starts, counts, n_bins):
+    """Accumulate per-(node, feature, bin) int64 gradient-pair histograms.
+
+    bins: uint8 [n, F]; gpair_q: int64 [n, 2]; ridx: int32 [n] row index
+    array partitioned into contiguous per-node segments; starts/counts:
+    int64 [K] segment descriptors for the K nodes to build.
+    Returns int64 [K, F, n_bins, 2]. Missing values (bin==MISSING_BIN) are
+    skipped; their mass is recovered as node_total - feature_sum.
+    """
+    return _impl(bins).build_histogram(bins, gpair_q, ridx, starts, counts, n_bins)
+
+
+def find_splits(
+    hist,
+    parent_g,
+    parent_h,
+    feat_bins,
+    scale_g,
+    scale_h,
+    reg_lambda,
+    reg_alpha,
+    gamma,
+    min_child_weight,
+):
+    """Best-split scan over histograms.
+
+    hist: int64 [K, F, B, 2]; parent_g/parent_h: int64 [K] quantized node
+    sums; feat_bins: int32 [F] per-feature bin counts.
+    Returns dict of tensors (on hist.device) each [K]:
+      gain f32, feature i32, bin i32, default_left u8,
+      left_g/left_h int64 (quantized left-child sums incl. missing if
+      default_left).
+    """
+    return _impl(hist).find_splits(
+        hist,
+        parent_g,
+        parent_h,
+        feat_bins,
+        scale_g,
+        scale_h,
+        reg_lambda,
+        reg_alpha,
+        gamma,
+        min_child_weight,
+    )
+
+
+def partition_rows(bins, ridx, starts, counts, split_feat, split_bin, default_left):
+    """Stable in-place partition of each node's ridx segment by its split.
+
+    Rows with bin <= split_bin (or missing & default_left) go left.
+    Returns (ridx_out int32 [n], left_counts int64 [K]).
+    """
+    return _impl(bins).partition_rows(
+        bins, ridx, starts, counts, split_feat, split_bin, default_left
+    )
+
+
+def predict_trees(
+    X, feat, thr, left, default_left, value, tree_ptr, out, tree_weight=1.0
+):
+    """Accumulate tree-walk predictions for a forest slice into ``out``.
+
+    X: fp32 [n, F] raw features (NaN = missing). Tree arrays are flat SoA
+    concatenations with tree_ptr: int32 [T+1] node offsets.
+    feat: int32 (-1 => leaf), thr: f32, left: int32 (right = left + 1),
+    default_left: u8, value: f32 leaf weight. out: f32 [n] updated in place.
+    Decision rule (XGBoost semantics): go left iff x < thr; missing follows
+    default_left.
+    """
+    return _impl(X).predict_trees(
+        X, feat, thr, left, default_left, value, tree_ptr, out, tree_weight
+    )
+
+
+def update_margins(margin, ridx, starts, counts, leaf_values):
+    """margin[ridx[seg_k]] += leaf_values[k] for each final-leaf segment."""
+    return _impl(margin).update_margins(margin, ridx, starts, counts, leaf_values)

@@ -1,0 +1,224 @@
+"""Pure-PyTorch CPU reference implementations of the engine ops.
+
+These are the numerics oracle: the HIP kernels in ``csrc/`` are tested for
+bitwise (integer ops) / close (float ops) agreement against these functions.
+Integer fixed-point histogram accumulation makes the CPU and GPU training
+paths produce identical trees.
+"""
+
+import torch
+
+MISSING_BIN = 255
+
+
+def quantize_gpair(gpair: torch.Tensor, scale_g: float, scale_h: float) -> torch.Tensor:
+    out = torch.empty(gpair.shape, dtype=torch.int64, device=gpair.device)
+    out[:, 0] = torch.round(gpair[:, 0].double() * scale_g).to(torch.int64)
+    out[:, 1] = torch.round(gpair[:, 1].double() * scale_h).to(torch.int64)
+    return out
+
+
+def bin_matrix(values: torch.Tensor, cuts_flat: torch.Tensor, cut_ptr: torch.Tensor):
+    n, F = values.shape
+    out = torch.empty((n, F), dtype=torch.uint8, device=values.device)
+    for f in range(F):
+        lo, hi = int(cut_ptr[f]), int(cut_ptr[f + 1])
+        nb = hi - lo
+        col = values[:, f]
+        if nb == 0:
+            out[:, f] = torch.where(
+                torch.isnan(col),
+                torch.tensor(MISSING_BIN, dtype=torch.uint8),
+                torch.tensor(0, dtype=torch.uint8),
+            )
+            continue
+        cuts = cuts_flat[lo:hi].contiguous()
+        # XGBoost HistogramCuts::SearchBin semantics: bin = count of cuts
+        # <= value (upper_bound), clamped to nb - 1. Bin b covers
+        # [cut[b-1], cut[b]), so a split after bin b <=> "v < cut[b] goes
+        # left" - exactly the float threshold rule used at predict time.
+        b = torch.searchsorted(cuts, col.contiguous(), side="right")
+        b = torch.clamp(b, max=nb - 1).to(torch.uint8)
+        b = torch.where(torch.isnan(col), torch.full_like(b, MISSING_BIN), b)
+        out[:, f] = b
+    return out
+
+
+def build_histogram(bins, gpair_q, ridx, starts, counts, n_bins):
+    K = len(starts)
+    n, F = bins.shape
+    hist = torch.zeros((K, F, n_bins, 2), dtype=torch.int64, device=bins.device)
+    foff = torch.arange(F, dtype=torch.int64) * n_bins
+    for k in range(K):
+        s, c = int(starts[k]), int(counts[k])
+        if c == 0:
+            continue
+        idx = ridx[s : s + c].long()
+        rb = bins[idx].long()  # [c, F]
+        valid = rb != MISSING_BIN
+        flat = foff.unsqueeze(0) + rb  # [c, F]
+        g = gpair_q[idx, 0].unsqueeze(1).expand_as(flat)
+        h = gpair_q[idx, 1].unsqueeze(1).expand_as(flat)
+        hk = hist[k].view(F * n_bins, 2)
+        fidx = flat[valid]
+        hk[:, 0].scatter_add_(0, fidx, g[valid])
+        hk[:, 1].scatter_add_(0, fidx, h[valid])
+    return hist
+
+
+def _calc_score(G, H, reg_lambda, reg_alpha):
+    """XGBoost split score: ThresholdL1(G, alpha)^2 / (H + lambda)."""
+    if reg_alpha > 0:
+        G = torch.sign(G) * torch.clamp(G.abs() - reg_alpha, min=0.0)
+    denom = H + reg_lambda
+    return torch.where(denom > 0, G * G / denom, torch.zeros_like(G))
+
+
+def find_splits(
+    hist,
+    parent_g,
+    parent_h,
+    feat_bins,
+    scale_g,
+    scale_h,
+    reg_lambda,
+    reg_alpha,
+    gamma,
+    min_child_weight,
+):
+    K, F, B, _ = hist.shape
+    dev = hist.device
+    # Dequantize to double for the scan: scan math must be identical on CPU
+    # and GPU, and the quantized ints are exactly representable in f64.
+    g = hist[..., 0].double() / scale_g  # [K, F, B]
+    h = hist[..., 1].double() / scale_h
+    gq = hist[..., 0]
+    hq = hist[..., 1]
+
+    GL = torch.cumsum(g, dim=2)
+    HL = torch.cumsum(h, dim=2)
+    GLq = torch.cumsum(gq, dim=2)
+    HLq = torch.cumsum(hq, dim=2)
+    Gp = (parent_g.double() / scale_g).view(K, 1, 1)
+    Hp = (parent_h.double() / scale_h).view(K, 1, 1)
+    # Missing mass per (node, feature) = parent - feature total.
+    Gmiss = Gp - GL[:, :, -1:].clone()
+    Hmiss = Hp - HL[:, :, -1:].clone()
+
+    parent_score = _calc_score(Gp, Hp, reg_lambda, reg_alpha)
+
+    # Candidate split after bin b (left = bins <= b); last bin excluded.
+    bin_idx = torch.arange(B, device=dev).view(1, 1, B)
+    valid_bin = bin_idx < (feat_bins.to(dev).view(1, F, 1) - 1)
+
+    best = {
+        "gain": torch.full((K,), -1.0, dtype=torch.float64, device=dev),
+        "feature": torch.full((K,), -1, dtype=torch.int32, device=dev),
+        "bin": torch.zeros((K,), dtype=torch.int32, device=dev),
+        "default_left": torch.zeros((K,), dtype=torch.uint8, device=dev),
+        "left_g": torch.zeros((K,), dtype=torch.int64, device=dev),
+        "left_h": torch.zeros((K,), dtype=torch.int64, device=dev),
+    }
+
+    for default_left in (1, 0):
+        if default_left:
+            gl, hl = GL + Gmiss, HL + Hmiss
+        else:
+            gl, hl = GL, HL
+        gr, hr = Gp - gl, Hp - hl
+        ok = (
+            valid_bin
+            & (hl >= min_child_weight)
+            & (hr >= min_child_weight)
+        )
+        gain = (
+            0.5
+            * (
+                _calc_score(gl, hl, reg_lambda, reg_alpha)
+                + _calc_score(gr, hr, reg_lambda, reg_alpha)
+                - parent_score
+            )
+            - gamma
+        )
+        gain = torch.where(ok, gain, torch.full_like(gain, -float("inf")))
+        flat = gain.view(K, F * B)
+        mx, arg = flat.max(dim=1)
+        upd = mx > best["gain"]
+        f_sel = (arg // B).to(torch.int32)
+        b_sel = (arg % B).to(torch.int32)
+        best["gain"] = torch.where(upd, mx, best["gain"])
+        best["feature"] = torch.where(upd, f_sel, best["feature"])
+        best["bin"] = torch.where(upd, b_sel, best["bin"])
+        best["default_left"] = torch.where(
+            upd,
+            torch.full((K,), default_left, dtype=torch.uint8, device=dev),
+            best["default_left"],
+        )
+        karange = torch.arange(K, device=dev)
+        lgq = GLq[karange, f_sel.long(), b_sel.long()]
+        lhq = HLq[karange, f_sel.long(), b_sel.long()]
+        if default_left:
+            # left child also receives the missing mass (exact in quantized
+            # space: parent_q - feature_total_q).
+            lgq = lgq + (parent_g - GLq[:, :, -1][karange, f_sel.long()])
+            lhq = lhq + (parent_h - HLq[:, :, -1][karange, f_sel.long()])
+        best["left_g"] = torch.where(upd, lgq, best["left_g"])
+        best["left_h"] = torch.where(upd, lhq, best["left_h"])
+
+    best["gain"] = best["gain"].float()
+    return best
+
+
+def partition_rows(bins, ridx, starts, counts, split_feat, split_bin, default_left):
+    K = len(starts)
+    out = ridx.clone()
+    left_counts = torch.zeros(K, dtype=torch.int64)
+    for k in range(K):
+        s, c = int(starts[k]), int(counts[k])
+        if c == 0:
+            continue
+        idx = ridx[s : s + c]
+        fv = bins[idx.long(), int(split_feat[k])]
+        miss = fv == MISSING_BIN
+        go_left = fv <= int(split_bin[k])
+        if int(default_left[k]):
+            go_left = go_left | miss
+        else:
+            go_left = go_left & ~miss
+        out[s : s + c] = torch.cat([idx[go_left], idx[~go_left]])
+        left_counts[k] = int(go_left.sum())
+    return out, left_counts
+
+
+def predict_trees(X, feat, thr, left, default_left, value, tree_ptr, out, tree_weight=1.0):
+    n = X.shape[0]
+    T = len(tree_ptr) - 1
+    rows = torch.arange(n)
+    for t in range(T):
+        base = int(tree_ptr[t])
+        cur = torch.full((n,), base, dtype=torch.int64)
+        while True:
+            f = feat[cur].long()
+            is_leaf = f < 0
+            if bool(is_leaf.all()):
+                break
+            active = ~is_leaf
+            ca = cur[active]
+            fa = f[active]
+            x = X[rows[active], fa]
+            miss = torch.isnan(x)
+            go_left = x < thr[ca]
+            go_left = torch.where(miss, default_left[ca].bool(), go_left)
+            nxt = torch.where(go_left, left[ca].long(), left[ca].long() + 1) + base
+            cur = cur.clone()
+            cur[active] = nxt
+        out += value[cur] * tree_weight
+    return out
+
+
+def update_margins(margin, ridx, starts, counts, leaf_values):
+    for k in range(len(starts)):
+        s, c = int(starts[k]), int(counts[k])
+        if c:
+            margin[ridx[s : s + c].long()] += float(leaf_values[k])
+    return margin

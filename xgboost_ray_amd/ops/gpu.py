@@ -1,0 +1,132 @@
+"""GPU op implementations backed by the hand-written CDNA4 HIP extension.
+
+The extension (``xgboost_ray_amd/csrc``) is compiled in-tree for gfx950
+(``__graft_entry__.build()`` / ``python setup.py build_ext --inplace``).
+On a GPU machine it is REQUIRED - there is no silent eager fallback for
+CUDA tensors: if the .so is missing, ops raise immediately.
+"""
+
+import os
+
+import torch
+
+_ext = None
+_load_error = None
+
+
+def _load():
+    global _ext, _load_error
+    if _ext is not None:
+        return _ext
+    try:
+        import importlib
+
+        _ext = importlib.import_module("xgboost_ray_amd._hip_ops")
+    except ImportError as e:
+        _load_error = e
+        raise RuntimeError(
+            "xgboost_ray_amd HIP extension (_hip_ops) is not built. "
+            "Run `python -m xgboost_ray_amd.build` or "
+            "`python __graft_entry__.py build` to compile it for gfx950. "
+            f"Original error: {e}"
+        ) from e
+    return _ext
+
+
+def quantize_gpair(gpair, scale_g, scale_h):
+    return _load().quantize_gpair(gpair.contiguous(), scale_g, scale_h)
+
+
+def bin_matrix(values, cuts_flat, cut_ptr):
+    return _load().bin_matrix(
+        values.contiguous(),
+        cuts_flat.to(values.device),
+        cut_ptr.to(values.device),
+    )
+
+
+def build_histogram(bins, gpair_q, ridx, starts, counts, n_bins):
+    dev = bins.device
+    return _load().build_histogram(
+        bins,
+        gpair_q,
+        ridx,
+        starts.to(dev),
+        counts.to(dev),
+        int(n_bins),
+    )
+
+
+def find_splits(
+    hist,
+    parent_g,
+    parent_h,
+    feat_bins,
+    scale_g,
+    scale_h,
+    reg_lambda,
+    reg_alpha,
+    gamma,
+    min_child_weight,
+):
+    dev = hist.device
+    out = _load().find_splits(
+        hist,
+        parent_g.to(dev),
+        parent_h.to(dev),
+        feat_bins.to(dev),
+        float(scale_g),
+        float(scale_h),
+        float(reg_lambda),
+        float(reg_alpha),
+        float(gamma),
+        float(min_child_weight),
+    )
+    return {
+        "gain": out[0],
+        "feature": out[1],
+        "bin": out[2],
+        "default_left": out[3],
+        "left_g": out[4],
+        "left_h": out[5],
+    }
+
+
+def partition_rows(bins, ridx, starts, counts, split_feat, split_bin, default_left):
+    dev = bins.device
+    return _load().partition_rows(
+        bins,
+        ridx,
+        starts.to(dev),
+        counts.to(dev),
+        split_feat.to(dev),
+        split_bin.to(dev),
+        default_left.to(dev),
+    )
+
+
+def predict_trees(X, feat, thr, left, default_left, value, tree_ptr, out, tree_weight=1.0):
+    dev = X.device
+    _load().predict_trees(
+        X,
+        feat.to(dev),
+        thr.to(dev),
+        left.to(dev),
+        default_left.to(dev),
+        value.to(dev),
+        tree_ptr.to(dev),
+        out,
+        float(tree_weight),
+    )
+    return out
+
+
+def update_margins(margin, ridx, starts, counts, leaf_values):
+    import numpy as np
+
+    dev = margin.device
+    lv = torch.as_tensor(leaf_values, dtype=torch.float32, device=dev)
+    _load().update_margins(
+        margin, ridx, starts.to(dev), counts.to(dev), lv
+    )
+    return margin

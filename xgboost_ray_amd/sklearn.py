@@ -1,0 +1,496 @@
+"""scikit-learn estimator API (reference sklearn.py:1-1083).
+
+Drop-in equivalents of the reference's five estimators, built on this
+package's train()/predict() instead of xgboost.sklearn internals. RayDMatrix
+passthrough (reference sklearn.py:280-334) and the n_jobs -> num_actors
+mapping (reference sklearn.py:341-355) are preserved.
+"""
+
+import warnings
+from typing import Any, Dict, List, Optional, Tuple, Union
+
+import numpy as np
+
+from xgboost_ray_amd.booster import Booster
+from xgboost_ray_amd.main import RayParams, predict, train
+from xgboost_ray_amd.matrix import RayDMatrix
+
+_PARAM_NAMES = (
+    "max_depth",
+    "learning_rate",
+    "gamma",
+    "min_child_weight",
+    "max_delta_step",
+    "subsample",
+    "colsample_bytree",
+    "colsample_bylevel",
+    "colsample_bynode",
+    "reg_alpha",
+    "reg_lambda",
+    "scale_pos_weight",
+    "base_score",
+    "missing",
+    "num_parallel_tree",
+    "monotone_constraints",
+    "interaction_constraints",
+    "importance_type",
+    "max_bin",
+    "tree_method",
+    "booster",
+    "verbosity",
+    "random_state",
+    "seed",
+)
+
+
+class RayXGBMixin:
+    """Shared Ray-parameter plumbing (reference RayXGBMixin, sklearn.py:341+)."""
+
+    def _ray_set_ray_params_n_jobs(
+        self, ray_params: Optional[Union[RayParams, Dict]], n_jobs: Optional[int]
+    ) -> RayParams:
+        if ray_params is None:
+            if n_jobs is None or n_jobs in (-1, 0):
+                n_jobs = 1
+            ray_params = RayParams(num_actors=n_jobs)
+        elif n_jobs is not None:
+            warnings.warn(
+                "Both `ray_params` and `n_jobs` are set; ignoring `n_jobs`."
+            )
+        if isinstance(ray_params, dict):
+            ray_params = RayParams(**ray_params)
+        return ray_params
+
+    def _ray_predict(
+        self,
+        X,
+        output_margin=False,
+        validate_features=True,
+        base_margin=None,
+        iteration_range=None,
+        ray_params=None,
+        _remote=None,
+        ray_dmatrix_params=None,
+    ):
+        ray_params = self._ray_set_ray_params_n_jobs(
+            ray_params, getattr(self, "n_jobs", None)
+        )
+        if isinstance(X, RayDMatrix):
+            data = X
+        else:
+            data = RayDMatrix(
+                X, base_margin=base_margin, missing=getattr(self, "missing", None),
+                **(ray_dmatrix_params or {}),
+            )
+        return predict(
+            self.get_booster(),
+            data,
+            ray_params=ray_params,
+            _remote=_remote,
+            output_margin=output_margin,
+            iteration_range=iteration_range,
+        )
+
+
+def _check_if_params_are_ray_dmatrix(X, sample_weight, base_margin, eval_set,
+                                     sample_weight_eval_set, base_margin_eval_set):
+    """RayDMatrix passthrough detection (reference sklearn.py:280-334)."""
+    train_dmatrix = None
+    evals = ()
+    if isinstance(X, RayDMatrix):
+        train_dmatrix = X
+        if sample_weight is not None or base_margin is not None:
+            raise ValueError(
+                "When X is a RayDMatrix, pass label/weight/base_margin "
+                "through the RayDMatrix itself."
+            )
+        if eval_set:
+            if any(not isinstance(e[0], RayDMatrix) for e in eval_set):
+                raise ValueError(
+                    "When X is a RayDMatrix, eval_set entries must be "
+                    "(RayDMatrix, name) tuples."
+                )
+            evals = tuple(eval_set)
+    return train_dmatrix, evals
+
+
+class _RayXGBModel(RayXGBMixin):
+    _estimator_type = "regressor"
+
+    def __init__(
+        self,
+        max_depth: Optional[int] = None,
+        learning_rate: Optional[float] = None,
+        n_estimators: int = 100,
+        objective: Optional[str] = None,
+        booster: Optional[str] = None,
+        tree_method: Optional[str] = None,
+        n_jobs: Optional[int] = None,
+        gamma: Optional[float] = None,
+        min_child_weight: Optional[float] = None,
+        max_delta_step: Optional[float] = None,
+        subsample: Optional[float] = None,
+        colsample_bytree: Optional[float] = None,
+        colsample_bylevel: Optional[float] = None,
+        colsample_bynode: Optional[float] = None,
+        reg_alpha: Optional[float] = None,
+        reg_lambda: Optional[float] = None,
+        scale_pos_weight: Optional[float] = None,
+        base_score: Optional[float] = None,
+        random_state: Optional[int] = None,
+        missing: Optional[float] = None,
+        num_parallel_tree: Optional[int] = None,
+        monotone_constraints=None,
+        interaction_constraints=None,
+        importance_type: Optional[str] = None,
+        max_bin: Optional[int] = None,
+        verbosity: Optional[int] = None,
+        early_stopping_rounds: Optional[int] = None,
+        eval_metric=None,
+        **kwargs,
+    ):
+        self.max_depth = max_depth
+        self.learning_rate = learning_rate
+        self.n_estimators = n_estimators
+        self.objective = objective
+        self.booster = booster
+        self.tree_method = tree_method
+        self.n_jobs = n_jobs
+        self.gamma = gamma
+        self.min_child_weight = min_child_weight
+        self.max_delta_step = max_delta_step
+        self.subsample = subsample
+        self.colsample_bytree = colsample_bytree
+        self.colsample_bylevel = colsample_bylevel
+        self.colsample_bynode = colsample_bynode
+        self.reg_alpha = reg_alpha
+        self.reg_lambda = reg_lambda
+        self.scale_pos_weight = scale_pos_weight
+        self.base_score = base_score
+        self.random_state = random_state
+        self.missing = missing
+        self.num_parallel_tree = num_parallel_tree
+        self.monotone_constraints = monotone_constraints
+        self.interaction_constraints = interaction_constraints
+        self.importance_type = importance_type
+        self.max_bin = max_bin
+        self.verbosity = verbosity
+        self.early_stopping_rounds = early_stopping_rounds
+        self.eval_metric = eval_metric
+        self.kwargs = kwargs
+        self._Booster: Optional[Booster] = None
+
+    # sklearn plumbing ------------------------------------------------------
+    def get_params(self, deep=True) -> Dict:
+        import inspect
+
+        sig = inspect.signature(_RayXGBModel.__init__)
+        params = {
+            k: getattr(self, k)
+            for k in sig.parameters
+            if k not in ("self", "kwargs")
+        }
+        params.update(getattr(self, "kwargs", {}))
+        return params
+
+    def set_params(self, **params):
+        for k, v in params.items():
+            if hasattr(self, k):
+                setattr(self, k, v)
+            else:
+                self.kwargs[k] = v
+        return self
+
+    def get_booster(self) -> Booster:
+        if self._Booster is None:
+            raise ValueError("Model not fitted yet; call fit() first.")
+        return self._Booster
+
+    def load_model(self, fname):
+        self._Booster = Booster()
+        self._Booster.load_model(fname)
+        return self
+
+    def save_model(self, fname):
+        self.get_booster().save_model(fname)
+
+    def get_xgb_params(self) -> Dict:
+        params = {"objective": self.objective or "reg:squarederror"}
+        for name in _PARAM_NAMES:
+            val = getattr(self, name, None)
+            if val is not None and name not in ("missing",):
+                key = "eta" if name == "learning_rate" else name
+                key = "seed" if name == "random_state" else key
+                params[key] = val
+        if self.eval_metric is not None:
+            params["eval_metric"] = self.eval_metric
+        params.update(self.kwargs or {})
+        return params
+
+    def _build_dmatrices(
+        self, X, y, sample_weight, base_margin, eval_set,
+        sample_weight_eval_set, base_margin_eval_set, qid=None,
+        ray_dmatrix_params=None,
+    ):
+        train_dmatrix, evals = _check_if_params_are_ray_dmatrix(
+            X, sample_weight, base_margin, eval_set,
+            sample_weight_eval_set, base_margin_eval_set,
+        )
+        if train_dmatrix is None:
+            dm_params = dict(ray_dmatrix_params or {})
+            train_dmatrix = RayDMatrix(
+                X, label=y, weight=sample_weight, base_margin=base_margin,
+                missing=self.missing, qid=qid, **dm_params,
+            )
+            evals = []
+            for i, (ex, ey) in enumerate(eval_set or []):
+                ew = (sample_weight_eval_set or [None] * len(eval_set))[i]
+                ebm = (base_margin_eval_set or [None] * len(eval_set))[i]
+                evals.append(
+                    (
+                        RayDMatrix(
+                            ex, label=ey, weight=ew, base_margin=ebm,
+                            missing=self.missing,
+                        ),
+                        f"validation_{i}",
+                    )
+                )
+            evals = tuple(evals)
+        return train_dmatrix, evals
+
+    def _fit(
+        self, X, y, sample_weight=None, base_margin=None, eval_set=None,
+        sample_weight_eval_set=None, base_margin_eval_set=None,
+        early_stopping_rounds=None, verbose=True, xgb_model=None,
+        qid=None, ray_params=None, _remote=None, ray_dmatrix_params=None,
+        callbacks=None, extra_params=None,
+    ):
+        ray_params = self._ray_set_ray_params_n_jobs(ray_params, self.n_jobs)
+        train_dmatrix, evals = self._build_dmatrices(
+            X, y, sample_weight, base_margin, eval_set,
+            sample_weight_eval_set, base_margin_eval_set, qid=qid,
+            ray_dmatrix_params=ray_dmatrix_params,
+        )
+        params = self.get_xgb_params()
+        if extra_params:
+            params.update(extra_params)
+        evals_result: Dict = {}
+        if isinstance(xgb_model, _RayXGBModel):
+            xgb_model = xgb_model.get_booster()
+        self._Booster = train(
+            params,
+            train_dmatrix,
+            num_boost_round=self.n_estimators,
+            evals=evals,
+            evals_result=evals_result,
+            ray_params=ray_params,
+            _remote=_remote,
+            xgb_model=xgb_model,
+            early_stopping_rounds=(
+                early_stopping_rounds
+                if early_stopping_rounds is not None
+                else self.early_stopping_rounds
+            ),
+            verbose_eval=verbose,
+            callbacks=callbacks,
+        )
+        self.evals_result_ = evals_result
+        if self._Booster.best_iteration is not None:
+            self.best_iteration = self._Booster.best_iteration
+            self.best_score = self._Booster.best_score
+        return self
+
+    @property
+    def feature_importances_(self):
+        bst = self.get_booster()
+        n_feat = bst.num_features
+        gains = np.zeros(n_feat)
+        for t in bst.trees:
+            for nid in range(t.num_nodes):
+                if t.feat[nid] >= 0:
+                    gains[t.feat[nid]] += t.gain[nid]
+        total = gains.sum()
+        return gains / total if total > 0 else gains
+
+
+class RayXGBRegressor(_RayXGBModel):
+    """Distributed XGBoost-style regressor (reference sklearn.py:450-599)."""
+
+    def __init__(self, objective="reg:squarederror", **kwargs):
+        super().__init__(objective=objective, **kwargs)
+
+    def fit(
+        self, X, y, *, sample_weight=None, base_margin=None, eval_set=None,
+        sample_weight_eval_set=None, base_margin_eval_set=None,
+        early_stopping_rounds=None, verbose=True, xgb_model=None,
+        ray_params=None, _remote=None, ray_dmatrix_params=None,
+        callbacks=None, **kwargs,
+    ):
+        return self._fit(
+            X, y, sample_weight, base_margin, eval_set,
+            sample_weight_eval_set, base_margin_eval_set,
+            early_stopping_rounds, verbose, xgb_model,
+            ray_params=ray_params, _remote=_remote,
+            ray_dmatrix_params=ray_dmatrix_params, callbacks=callbacks,
+        )
+
+    def predict(
+        self, X, *, output_margin=False, base_margin=None,
+        iteration_range=None, ray_params=None, _remote=None,
+        ray_dmatrix_params=None, **kwargs,
+    ):
+        return self._ray_predict(
+            X, output_margin=output_margin, base_margin=base_margin,
+            iteration_range=iteration_range, ray_params=ray_params,
+            _remote=_remote, ray_dmatrix_params=ray_dmatrix_params,
+        )
+
+
+class RayXGBRFRegressor(RayXGBRegressor):
+    """Random-forest-style regressor (reference sklearn.py:602-640)."""
+
+    def __init__(
+        self, learning_rate=1.0, subsample=0.8, colsample_bytree=0.8,
+        reg_lambda=1e-5, **kwargs,
+    ):
+        super().__init__(
+            learning_rate=learning_rate, subsample=subsample,
+            colsample_bytree=colsample_bytree, reg_lambda=reg_lambda, **kwargs,
+        )
+
+
+class RayXGBClassifier(_RayXGBModel):
+    """Distributed classifier (reference sklearn.py:643-877)."""
+
+    _estimator_type = "classifier"
+
+    def __init__(self, objective="binary:logistic", **kwargs):
+        super().__init__(objective=objective, **kwargs)
+
+    def fit(
+        self, X, y, *, sample_weight=None, base_margin=None, eval_set=None,
+        sample_weight_eval_set=None, base_margin_eval_set=None,
+        early_stopping_rounds=None, verbose=True, xgb_model=None,
+        ray_params=None, _remote=None, ray_dmatrix_params=None,
+        callbacks=None, **kwargs,
+    ):
+        extra_params = {}
+        if isinstance(X, RayDMatrix):
+            # cannot inspect labels; trust user-provided objective/num_class
+            if self.objective and self.objective.startswith("multi"):
+                nc = (self.kwargs or {}).get("num_class")
+                if nc:
+                    extra_params["num_class"] = nc
+            self.classes_ = None
+            self.n_classes_ = (self.kwargs or {}).get("num_class", 2)
+        else:
+            y = np.asarray(y)
+            self.classes_ = np.unique(y)
+            self.n_classes_ = len(self.classes_)
+            if self.n_classes_ > 2:
+                extra_params["objective"] = "multi:softprob"
+                extra_params["num_class"] = self.n_classes_
+            else:
+                extra_params["objective"] = self.objective or "binary:logistic"
+            # encode labels to 0..k-1
+            if not np.array_equal(self.classes_, np.arange(self.n_classes_)):
+                lut = {c: i for i, c in enumerate(self.classes_)}
+                y = np.asarray([lut[v] for v in y], dtype=np.float32)
+        return self._fit(
+            X, y, sample_weight, base_margin, eval_set,
+            sample_weight_eval_set, base_margin_eval_set,
+            early_stopping_rounds, verbose, xgb_model,
+            ray_params=ray_params, _remote=_remote,
+            ray_dmatrix_params=ray_dmatrix_params, callbacks=callbacks,
+            extra_params=extra_params,
+        )
+
+    def predict_proba(
+        self, X, *, base_margin=None, iteration_range=None, ray_params=None,
+        _remote=None, ray_dmatrix_params=None, **kwargs,
+    ):
+        raw = self._ray_predict(
+            X, output_margin=False, base_margin=base_margin,
+            iteration_range=iteration_range, ray_params=ray_params,
+            _remote=_remote, ray_dmatrix_params=ray_dmatrix_params,
+        )
+        raw = np.asarray(raw)
+        if raw.ndim == 2:
+            return raw  # softprob output
+        return np.vstack([1.0 - raw, raw]).T
+
+    def predict(
+        self, X, *, output_margin=False, base_margin=None,
+        iteration_range=None, ray_params=None, _remote=None,
+        ray_dmatrix_params=None, **kwargs,
+    ):
+        if output_margin:
+            return self._ray_predict(
+                X, output_margin=True, base_margin=base_margin,
+                iteration_range=iteration_range, ray_params=ray_params,
+                _remote=_remote, ray_dmatrix_params=ray_dmatrix_params,
+            )
+        proba = self.predict_proba(
+            X, base_margin=base_margin, iteration_range=iteration_range,
+            ray_params=ray_params, _remote=_remote,
+            ray_dmatrix_params=ray_dmatrix_params,
+        )
+        idx = np.argmax(proba, axis=1)
+        if getattr(self, "classes_", None) is not None:
+            return self.classes_[idx]
+        return idx
+
+
+class RayXGBRFClassifier(RayXGBClassifier):
+    """Random-forest-style classifier (reference sklearn.py:880-917)."""
+
+    def __init__(
+        self, learning_rate=1.0, subsample=0.8, colsample_bytree=0.8,
+        reg_lambda=1e-5, **kwargs,
+    ):
+        super().__init__(
+            learning_rate=learning_rate, subsample=subsample,
+            colsample_bytree=colsample_bytree, reg_lambda=reg_lambda, **kwargs,
+        )
+
+
+class RayXGBRanker(_RayXGBModel):
+    """Learning-to-rank estimator, qid-based (reference sklearn.py:920-1083)."""
+
+    def __init__(self, objective="rank:pairwise", **kwargs):
+        if objective and not objective.startswith("rank:"):
+            raise ValueError("RayXGBRanker requires a rank:* objective")
+        super().__init__(objective=objective, **kwargs)
+
+    def fit(
+        self, X, y, *, qid=None, group=None, sample_weight=None,
+        base_margin=None, eval_set=None, eval_qid=None,
+        sample_weight_eval_set=None, base_margin_eval_set=None,
+        early_stopping_rounds=None, verbose=False, xgb_model=None,
+        ray_params=None, _remote=None, ray_dmatrix_params=None,
+        callbacks=None, **kwargs,
+    ):
+        if group is not None:
+            raise ValueError(
+                "`group` is not supported; pass per-row `qid` instead."
+            )
+        if qid is None and not isinstance(X, RayDMatrix):
+            raise ValueError("RayXGBRanker.fit requires `qid`")
+        return self._fit(
+            X, y, sample_weight, base_margin, eval_set,
+            sample_weight_eval_set, base_margin_eval_set,
+            early_stopping_rounds, verbose, xgb_model, qid=qid,
+            ray_params=ray_params, _remote=_remote,
+            ray_dmatrix_params=ray_dmatrix_params, callbacks=callbacks,
+        )
+
+    def predict(
+        self, X, *, output_margin=False, base_margin=None,
+        iteration_range=None, ray_params=None, _remote=None,
+        ray_dmatrix_params=None, **kwargs,
+    ):
+        return self._ray_predict(
+            X, output_margin=output_margin, base_margin=base_margin,
+            iteration_range=iteration_range, ray_params=ray_params,
+            _remote=_remote, ray_dmatrix_params=ray_dmatrix_params,
+        )
