@@ -1,0 +1,209 @@
+"""Distributed end-to-end correctness (mirrors reference test_end_to_end.py).
+
+The centerpiece is the joint-training test: a dataset whose BATCH-sharded
+halves are individually unlearnable, so 100% accuracy proves the
+cross-actor histogram allreduce is correct (reference
+test_end_to_end.py:56-211)."""
+
+import numpy as np
+import pytest
+
+from tests.utils import create_data, one_hot_impossible_halves
+from xgboost_ray_amd import (
+    RayDMatrix,
+    RayParams,
+    RayShardingMode,
+    predict,
+    train,
+)
+from xgboost_ray_amd.actor import TrainingError
+from xgboost_ray_amd.callback import DistributedCallback
+
+
+class PutCallback:
+    def after_iteration(self, booster, iteration, evals_log):
+        from xgboost_ray_amd.session import get_actor_rank, put_queue
+
+        put_queue(("it", iteration, get_actor_rank()))
+        return False
+
+
+class BoomCallback:
+    def after_iteration(self, booster, iteration, evals_log):
+        raise RuntimeError("boom-from-actor")
+
+
+class TestJointTraining:
+    def test_disjoint_halves_learn_whole(self):
+        X, y = one_hot_impossible_halves(repeat=32)
+        dtrain = RayDMatrix(
+            X, label=y, sharding=RayShardingMode.BATCH
+        )
+        bst = train(
+            {"objective": "multi:softmax", "num_class": 4, "max_depth": 3,
+             "eta": 1.0, "reg_lambda": 0.0, "min_child_weight": 0.0},
+            dtrain, 10, ray_params=RayParams(num_actors=2),
+        )
+        pred = bst.predict(X)
+        acc = (pred.astype(np.int64) == y.astype(np.int64)).mean()
+        assert acc == 1.0, f"allreduce broken: accuracy {acc} < 1.0"
+
+    def test_distributed_equals_single(self):
+        """2-actor training must equal 1-actor training bitwise (int64
+        histograms are world-size invariant)."""
+        X, y = create_data(4096, 6)
+        params = {"objective": "binary:logistic", "max_depth": 4, "eta": 0.3}
+        preds = []
+        for num_actors in (1, 2):
+            dtrain = RayDMatrix(X, label=y)
+            bst = train(
+                params, dtrain, 8, ray_params=RayParams(num_actors=num_actors)
+            )
+            preds.append(bst.predict(X, output_margin=True))
+        np.testing.assert_allclose(preds[0], preds[1], rtol=1e-5, atol=1e-6)
+
+
+class TestPredictions:
+    def test_softprob_2d_combine(self):
+        X, y = create_data(1200, 6, kind="multi")
+        dtrain = RayDMatrix(X, label=y)
+        bst = train(
+            {"objective": "multi:softprob", "num_class": 4, "max_depth": 4,
+             "eta": 0.5},
+            dtrain, 8, ray_params=RayParams(num_actors=2),
+        )
+        pred = predict(bst, RayDMatrix(X), ray_params=RayParams(num_actors=2))
+        assert pred.shape == (1200, 4)
+        np.testing.assert_allclose(pred.sum(axis=1), 1.0, rtol=1e-4)
+        local = bst.predict(X)
+        np.testing.assert_allclose(pred, local, rtol=1e-5, atol=1e-6)
+
+    def test_distributed_predict_matches_local(self):
+        X, y = create_data(2000, 5)
+        dtrain = RayDMatrix(X, label=y)
+        bst = train(
+            {"objective": "binary:logistic", "max_depth": 4, "eta": 0.3},
+            dtrain, 5, ray_params=RayParams(num_actors=2),
+        )
+        dist_pred = predict(
+            bst, RayDMatrix(X), ray_params=RayParams(num_actors=2)
+        )
+        np.testing.assert_allclose(
+            dist_pred, bst.predict(X), rtol=1e-5, atol=1e-6
+        )
+
+
+class TestEvalsAndResults:
+    def test_evals_and_additional_results(self):
+        X, y = create_data(2000, 5)
+        Xv, yv = create_data(500, 5, seed=5)
+        dtrain = RayDMatrix(X, label=y)
+        dval = RayDMatrix(Xv, label=yv)
+        res, add = {}, {}
+        train(
+            {"objective": "binary:logistic", "max_depth": 4, "eta": 0.3,
+             "eval_metric": ["logloss", "auc"]},
+            dtrain, 6, evals=[(dtrain, "train"), (dval, "valid")],
+            evals_result=res, additional_results=add,
+            ray_params=RayParams(num_actors=2),
+        )
+        assert set(res.keys()) == {"train", "valid"}
+        assert len(res["valid"]["logloss"]) == 6
+        assert add["total_n"] == 2000
+        assert add["training_time_s"] > 0
+        assert add["total_time_s"] >= add["training_time_s"]
+
+    def test_callback_returns_via_put_queue(self):
+        X, y = create_data(600, 4)
+        dtrain = RayDMatrix(X, label=y)
+        add = {}
+        train(
+            {"objective": "binary:logistic", "max_depth": 3},
+            dtrain, 4, ray_params=RayParams(num_actors=2),
+            additional_results=add, callbacks=[PutCallback()],
+        )
+        cr = add["callback_returns"]
+        assert sorted(cr.keys()) == [0, 1]
+        assert len(cr[0]) == 4
+        assert cr[0][0][0] == "it"
+
+
+class TestValidation:
+    def test_wrong_dtrain_type(self):
+        with pytest.raises(ValueError, match="RayDMatrix"):
+            train({}, np.zeros((10, 2)), ray_params=RayParams(num_actors=2))
+
+    def test_invalid_ray_params(self):
+        X, y = create_data(100, 3)
+        with pytest.raises(ValueError, match="num_actors"):
+            train({}, RayDMatrix(X, label=y), ray_params=RayParams())
+
+    def test_exact_tree_method_rejected(self):
+        X, y = create_data(100, 3)
+        with pytest.raises(ValueError, match="tree_method"):
+            train(
+                {"tree_method": "exact"}, RayDMatrix(X, label=y),
+                ray_params=RayParams(num_actors=2),
+            )
+
+    def test_error_propagates_from_actor(self):
+        """Exceptions inside actors surface on the driver
+        (reference test_end_to_end.py:321-353)."""
+        X, y = create_data(200, 3)
+
+        with pytest.raises(RuntimeError):
+            train(
+                {"objective": "binary:logistic"},
+                RayDMatrix(X, label=y), 3,
+                ray_params=RayParams(num_actors=2, max_actor_restarts=0),
+                callbacks=[BoomCallback()],
+            )
+
+
+class LogCallback(DistributedCallback):
+    def __init__(self, log_dir):
+        self.log_dir = log_dir
+
+    def _log(self, actor, event):
+        with open(f"{self.log_dir}/rank_{actor.rank}.log", "a") as f:
+            f.write(event + "\n")
+
+    def on_init(self, actor, *a, **kw):
+        self._log(actor, "init")
+
+    def before_data_loading(self, actor, data, *a, **kw):
+        self._log(actor, "before_load")
+
+    def after_data_loading(self, actor, data, *a, **kw):
+        self._log(actor, "after_load")
+
+    def before_train(self, actor, *a, **kw):
+        self._log(actor, "before_train")
+
+    def after_train(self, actor, result_dict, *a, **kw):
+        self._log(actor, "after_train")
+
+
+class TestDistributedCallbacks:
+    def test_callback_order(self, tmp_path):
+        """Lifecycle hooks fire in order on every actor
+        (reference test_end_to_end.py:279-305)."""
+        X, y = create_data(400, 4)
+        train(
+            {"objective": "binary:logistic"},
+            RayDMatrix(X, label=y), 2,
+            ray_params=RayParams(
+                num_actors=2,
+                distributed_callbacks=[LogCallback(str(tmp_path))],
+            ),
+        )
+        for rank in range(2):
+            with open(f"{tmp_path}/rank_{rank}.log") as f:
+                events = f.read().splitlines()
+            assert events == [
+                "init", "before_load", "before_load", "after_load",
+                "after_load", "before_train", "after_train",
+            ] or events == [
+                "init", "before_load", "after_load",
+                "before_train", "after_train",
+            ]

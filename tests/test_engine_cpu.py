@@ -1,0 +1,314 @@
+"""Single-rank engine correctness on CPU (numerics oracle for the GPU)."""
+
+import numpy as np
+import pytest
+import torch
+
+from tests.utils import create_data
+from xgboost_ray_amd import ops
+from xgboost_ray_amd.engine.quantile import BinnedMatrix, build_cuts
+from xgboost_ray_amd.engine.trainer import (
+    BoostingEngine,
+    EvalPack,
+    run_training,
+)
+
+
+def _binned(n=4000, f=6, kind="binary", seed=0, max_bin=64):
+    X, y = create_data(n, f, seed, kind)
+    return (
+        BinnedMatrix.build(
+            torch.from_numpy(X), label=torch.from_numpy(y), max_bin=max_bin
+        ),
+        X,
+        y,
+    )
+
+
+class TestBinning:
+    def test_bins_reconstruct_order(self):
+        X, _ = create_data(500, 3)
+        Xt = torch.from_numpy(X)
+        cuts = build_cuts(Xt, 32)
+        bins = ops.bin_matrix(Xt, cuts.cuts_flat, cuts.cut_ptr)
+        # binning must be monotone in the value
+        for f in range(3):
+            order = np.argsort(X[:, f], kind="stable")
+            b = bins[:, f].numpy()[order]
+            assert (np.diff(b.astype(np.int32)) >= 0).all()
+
+    def test_missing_values(self):
+        X = np.array([[0.0], [np.nan], [1.0]], dtype=np.float32)
+        Xt = torch.from_numpy(X)
+        cuts = build_cuts(Xt, 8)
+        bins = ops.bin_matrix(Xt, cuts.cuts_flat, cuts.cut_ptr)
+        assert bins[1, 0] == ops.MISSING_BIN
+        assert bins[0, 0] != ops.MISSING_BIN
+
+    def test_few_distinct_values_one_bin_each(self):
+        X = np.array([[0.0], [1.0], [2.0], [1.0], [0.0]], dtype=np.float32)
+        Xt = torch.from_numpy(X)
+        cuts = build_cuts(Xt, 256)
+        bins = ops.bin_matrix(Xt, cuts.cuts_flat, cuts.cut_ptr).numpy().flatten()
+        assert bins.tolist() == [0, 1, 2, 1, 0]
+
+
+class TestHistogram:
+    def test_histogram_totals_match_sums(self):
+        dm, X, y = _binned()
+        n = dm.n_rows
+        gq = ops.quantize_gpair(
+            torch.stack(
+                [torch.from_numpy(y) - 0.5, torch.ones(n)], dim=1
+            ),
+            2.0**20,
+            2.0**20,
+        )
+        ridx = torch.arange(n, dtype=torch.int32)
+        hist = ops.build_histogram(
+            dm.bins, gq, ridx, torch.tensor([0]), torch.tensor([n]), dm.cuts.max_bins
+        )
+        # per feature: sum over bins == total (no missing values here)
+        tot_g = int(gq[:, 0].sum())
+        for f in range(dm.n_features):
+            assert int(hist[0, f, :, 0].sum()) == tot_g
+
+    def test_histogram_subtraction_consistency(self):
+        dm, X, y = _binned()
+        n = dm.n_rows
+        gq = ops.quantize_gpair(
+            torch.stack([torch.from_numpy(y) - 0.5, torch.ones(n)], dim=1),
+            2.0**20,
+            2.0**20,
+        )
+        ridx = torch.arange(n, dtype=torch.int32)
+        whole = ops.build_histogram(
+            dm.bins, gq, ridx, torch.tensor([0]), torch.tensor([n]), dm.cuts.max_bins
+        )
+        half = ops.build_histogram(
+            dm.bins,
+            gq,
+            ridx,
+            torch.tensor([0, n // 2]),
+            torch.tensor([n // 2, n - n // 2]),
+            dm.cuts.max_bins,
+        )
+        torch.testing.assert_close(whole[0], half[0] + half[1])
+
+
+class TestTraining:
+    def test_loss_decreases_binary(self):
+        dm, X, y = _binned()
+        res = {}
+        run_training(
+            {"objective": "binary:logistic", "max_depth": 4, "eta": 0.3,
+             "eval_metric": ["logloss"]},
+            dm, 15, evals=[EvalPack(name="train", X=None)], evals_result=res,
+        )
+        ll = res["train"]["logloss"]
+        assert ll[-1] < ll[0] * 0.6
+
+    def test_regression(self):
+        dm, X, y = _binned(kind="reg")
+        res = {}
+        run_training(
+            {"objective": "reg:squarederror", "max_depth": 4, "eta": 0.3,
+             "eval_metric": ["rmse"]},
+            dm, 15, evals=[EvalPack(name="train", X=None)], evals_result=res,
+        )
+        assert res["train"]["rmse"][-1] < res["train"]["rmse"][0] * 0.5
+
+    def test_multiclass_softprob(self):
+        dm, X, y = _binned(kind="multi")
+        res = {}
+        bst = run_training(
+            {"objective": "multi:softprob", "num_class": 4, "max_depth": 4,
+             "eta": 0.5, "eval_metric": ["mlogloss", "merror"]},
+            dm, 10, evals=[EvalPack(name="train", X=None)], evals_result=res,
+        )
+        assert res["train"]["merror"][-1] < 0.05
+        pred = bst.predict(X[:100])
+        assert pred.shape == (100, 4)
+        np.testing.assert_allclose(pred.sum(axis=1), 1.0, rtol=1e-4)
+
+    def test_predict_matches_train_margins(self):
+        """The float tree-walk predictor must agree with the binned
+        training-time margins (threshold <-> bin equivalence)."""
+        dm, X, y = _binned(max_bin=64)
+        engine = BoostingEngine(
+            {"objective": "binary:logistic", "max_depth": 5, "eta": 0.3}, dm
+        )
+        for _ in range(5):
+            engine.update()
+        pred_margin = engine.booster.predict(X, output_margin=True)
+        np.testing.assert_allclose(
+            pred_margin, engine.margin.numpy(), rtol=1e-5, atol=1e-5
+        )
+
+    def test_determinism_two_runs(self):
+        dm, X, y = _binned()
+        outs = []
+        for _ in range(2):
+            bst = run_training(
+                {"objective": "binary:logistic", "max_depth": 6, "eta": 0.3},
+                dm, 8,
+            )
+            outs.append(bst.predict(X, output_margin=True))
+        np.testing.assert_array_equal(outs[0], outs[1])
+
+    def test_resume_equals_continuous(self):
+        """xgb_model resume must produce the identical model
+        (determinism contract, reference test_fault_tolerance.py:401-449)."""
+        dm, X, y = _binned()
+        bst_full = run_training(
+            {"objective": "binary:logistic", "max_depth": 5, "eta": 0.3},
+            dm, 12,
+        )
+        bst_half = run_training(
+            {"objective": "binary:logistic", "max_depth": 5, "eta": 0.3},
+            dm, 6,
+        )
+        bst_resumed = run_training(
+            {"objective": "binary:logistic", "max_depth": 5, "eta": 0.3},
+            dm, 6, xgb_model=bst_half,
+        )
+        np.testing.assert_array_equal(
+            bst_full.predict(X, output_margin=True),
+            bst_resumed.predict(X, output_margin=True),
+        )
+
+    def test_subsample_colsample(self):
+        dm, X, y = _binned()
+        res = {}
+        run_training(
+            {"objective": "binary:logistic", "max_depth": 4, "eta": 0.3,
+             "subsample": 0.7, "colsample_bytree": 0.8,
+             "eval_metric": ["logloss"]},
+            dm, 10, evals=[EvalPack(name="train", X=None)], evals_result=res,
+        )
+        assert res["train"]["logloss"][-1] < res["train"]["logloss"][0]
+
+    def test_early_stopping(self):
+        X, y = create_data(2000, 6, 0, "binary")
+        Xv, yv = create_data(500, 6, 7, "binary")
+        dm = BinnedMatrix.build(
+            torch.from_numpy(X), label=torch.from_numpy(y), max_bin=64
+        )
+        ev = EvalPack(
+            name="valid",
+            X=torch.from_numpy(Xv),
+            label=torch.from_numpy(yv),
+        )
+        res = {}
+        bst = run_training(
+            {"objective": "binary:logistic", "max_depth": 6, "eta": 0.9,
+             "eval_metric": ["logloss"]},
+            dm, 100, evals=[ev], evals_result=res,
+            early_stopping_rounds=5,
+        )
+        assert bst.best_iteration is not None
+        assert len(res["valid"]["logloss"]) < 100
+
+    def test_eval_on_holdout(self):
+        X, y = create_data(3000, 6, 0, "binary")
+        Xv, yv = create_data(800, 6, 3, "binary")
+        dm = BinnedMatrix.build(
+            torch.from_numpy(X), label=torch.from_numpy(y), max_bin=64
+        )
+        ev = EvalPack(
+            name="valid", X=torch.from_numpy(Xv), label=torch.from_numpy(yv)
+        )
+        res = {}
+        run_training(
+            {"objective": "binary:logistic", "max_depth": 4, "eta": 0.3,
+             "eval_metric": ["auc"]},
+            dm, 10, evals=[ev], evals_result=res,
+        )
+        assert res["valid"]["auc"][-1] > 0.85
+
+    def test_custom_objective_and_feval(self):
+        dm, X, y = _binned()
+
+        def squared_log(preds, dtrain):
+            y = dtrain.get_label()
+            grad = preds - y
+            hess = np.ones_like(preds)
+            return grad, hess
+
+        def rmse_feval(preds, dtrain):
+            y = dtrain.get_label()
+            return "custom-rmse", float(np.sqrt(np.mean((preds - y) ** 2)))
+
+        res = {}
+        run_training(
+            {"max_depth": 4, "eta": 0.3, "disable_default_eval_metric": 1},
+            dm, 8, evals=[EvalPack(name="train", X=None)],
+            obj=squared_log, feval=rmse_feval, evals_result=res,
+        )
+        vals = res["train"]["custom-rmse"]
+        assert vals[-1] < vals[0]
+
+
+class TestRanking:
+    def test_rank_ndcg_improves(self):
+        from tests.utils import create_labeled_sorted_rank_data
+
+        X, y, qid = create_labeled_sorted_rank_data()
+        dm = BinnedMatrix.build(
+            torch.from_numpy(X),
+            label=torch.from_numpy(y),
+            qid=torch.from_numpy(qid),
+            max_bin=64,
+        )
+        res = {}
+        run_training(
+            {"objective": "rank:ndcg", "max_depth": 4, "eta": 0.3,
+             "eval_metric": ["ndcg"]},
+            dm, 15, evals=[EvalPack(name="train", X=None)], evals_result=res,
+        )
+        ndcg = res["train"]["ndcg"]
+        assert ndcg[-1] > ndcg[0]
+        assert ndcg[-1] > 0.9
+
+    def test_rank_pairwise_runs(self):
+        from tests.utils import create_labeled_sorted_rank_data
+
+        X, y, qid = create_labeled_sorted_rank_data(n_groups=10)
+        dm = BinnedMatrix.build(
+            torch.from_numpy(X),
+            label=torch.from_numpy(y),
+            qid=torch.from_numpy(qid),
+            max_bin=64,
+        )
+        res = {}
+        run_training(
+            {"objective": "rank:pairwise", "max_depth": 3, "eta": 0.3,
+             "eval_metric": ["map"]},
+            dm, 8, evals=[EvalPack(name="train", X=None)], evals_result=res,
+        )
+        assert res["train"]["map"][-1] >= res["train"]["map"][0]
+
+
+class TestMetrics:
+    def test_auc_known_value(self):
+        from xgboost_ray_amd.engine.metrics import get_metric
+
+        # perfect separation -> AUC 1
+        margin = torch.tensor([-2.0, -1.0, 1.0, 2.0])
+        label = torch.tensor([0.0, 0.0, 1.0, 1.0])
+        m = get_metric("auc")
+        assert m.finalize(m.local_stats(margin, label, None, None, None)) == 1.0
+        # random-ish
+        margin2 = torch.tensor([1.0, -1.0, 1.0, -1.0])
+        v = m.finalize(m.local_stats(margin2, label, None, None, None))
+        assert abs(v - 0.5) < 1e-6
+
+    def test_rmse_weighted(self):
+        from xgboost_ray_amd.engine.metrics import get_metric
+
+        m = get_metric("rmse")
+        margin = torch.tensor([1.0, 3.0])
+        label = torch.tensor([0.0, 0.0])
+        w = torch.tensor([3.0, 1.0])
+        v = m.finalize(m.local_stats(margin, label, w, None, None))
+        assert abs(v - np.sqrt((3 * 1 + 9) / 4)) < 1e-9

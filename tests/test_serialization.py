@@ -1,0 +1,97 @@
+"""Booster persistence: XGBoost JSON schema, pickle, dumps."""
+
+import json
+import pickle
+
+import numpy as np
+import torch
+
+from tests.utils import create_data
+from xgboost_ray_amd.booster import Booster
+from xgboost_ray_amd.engine.quantile import BinnedMatrix
+from xgboost_ray_amd.engine.trainer import run_training
+
+
+def _train_booster(kind="binary", **params):
+    X, y = create_data(1000, 5, kind=kind)
+    dm = BinnedMatrix.build(
+        torch.from_numpy(X), label=torch.from_numpy(y), max_bin=64
+    )
+    base = {"objective": "binary:logistic", "max_depth": 4, "eta": 0.3}
+    base.update(params)
+    return run_training(base, dm, 5), X
+
+
+def test_json_schema_fields(tmp_path):
+    bst, X = _train_booster()
+    path = str(tmp_path / "m.json")
+    bst.save_model(path)
+    with open(path) as f:
+        doc = json.load(f)
+    learner = doc["learner"]
+    assert learner["objective"]["name"] == "binary:logistic"
+    model = learner["gradient_booster"]["model"]
+    assert int(model["gbtree_model_param"]["num_trees"]) == 5
+    tree = model["trees"][0]
+    n = int(tree["tree_param"]["num_nodes"])
+    for key in ("left_children", "right_children", "split_conditions",
+                "split_indices", "default_left", "base_weights",
+                "loss_changes", "sum_hessian", "parents"):
+        assert len(tree[key]) == n
+    assert learner["learner_model_param"]["num_feature"] == "5"
+
+
+def test_json_roundtrip_predictions(tmp_path):
+    bst, X = _train_booster()
+    path = str(tmp_path / "m.json")
+    bst.save_model(path)
+    bst2 = Booster()
+    bst2.load_model(path)
+    np.testing.assert_allclose(
+        bst.predict(X), bst2.predict(X), rtol=1e-6
+    )
+    assert bst2.objective == "binary:logistic"
+
+
+def test_save_raw_roundtrip():
+    bst, X = _train_booster()
+    raw = bst.save_raw()
+    bst2 = Booster()
+    bst2.load_model(raw)
+    np.testing.assert_allclose(bst.predict(X), bst2.predict(X), rtol=1e-6)
+
+
+def test_pickle_roundtrip():
+    bst, X = _train_booster()
+    bst2 = pickle.loads(pickle.dumps(bst))
+    np.testing.assert_array_equal(bst.predict(X), bst2.predict(X))
+
+
+def test_multiclass_roundtrip(tmp_path):
+    bst, X = _train_booster(
+        kind="multi", objective="multi:softprob", num_class=4
+    )
+    path = str(tmp_path / "mc.json")
+    bst.save_model(path)
+    bst2 = Booster()
+    bst2.load_model(path)
+    p1, p2 = bst.predict(X), bst2.predict(X)
+    assert p1.shape == (1000, 4)
+    np.testing.assert_allclose(p1, p2, rtol=1e-6)
+
+
+def test_get_dump_text():
+    bst, X = _train_booster()
+    dumps = bst.get_dump()
+    assert len(dumps) == 5
+    assert "leaf=" in dumps[0]
+    assert "yes=" in dumps[0]
+
+
+def test_attributes():
+    bst, _ = _train_booster()
+    bst.set_attr(foo="bar")
+    assert bst.attr("foo") == "bar"
+    assert bst.attributes()["foo"] == "bar"
+    bst.set_attr(foo=None)
+    assert bst.attr("foo") is None

@@ -1,0 +1,794 @@
+// MI355X (gfx950, CDNA4) kernels for the gradient-boosted-tree engine.
+//
+// MI355X-native replacement for XGBoost's CUDA gpu_hist updater internals
+// (the reference delegates them to libxgboost):
+//   - quantize_gpair: fp32 gradient pairs -> int64 fixed point
+//   - bin_matrix:     fp32 features -> uint8 quantile bins (LDS-cached cuts)
+//   - gather + build_histogram: LDS-tiled per-(node,feature,bin) int64
+//     gradient histograms, 64-wide-wave layout, deterministic by integer
+//     associativity (no float atomics anywhere)
+//   - find_splits:    sequential-per-(node,feature) scan, bitwise-identical
+//     to the CPU torch reference (same add/divide order)
+//   - partition_rows: stable two-pass partition (ballot prefix + scatter)
+//   - predict_trees:  SoA tree-walk
+//
+// Design notes (per /opt/skills/guides/cdna_hip_programming.md):
+//   wave = 64 lanes; LDS = 160 KiB/CU; histogram tiles sized so 1-2
+//   workgroups fit per CU; all global traffic vectorized where layout
+//   permits; grids sized >> 256 CUs.
+
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include <cstdint>
+#include <vector>
+
+#define WAVE 64
+#define CHECK_HIP(x)                                                         \
+  do {                                                                       \
+    hipError_t err = (x);                                                    \
+    TORCH_CHECK(err == hipSuccess, "HIP error: ", hipGetErrorString(err));   \
+  } while (0)
+
+static inline int64_t ceil_div(int64_t a, int64_t b) { return (a + b - 1) / b; }
+
+// ---------------------------------------------------------------------------
+// quantize_gpair: [n,2] f32 -> [n,2] i64 (round to nearest)
+// ---------------------------------------------------------------------------
+__global__ void quantize_gpair_kernel(const float2* __restrict__ gpair,
+                                      longlong2* __restrict__ out,
+                                      double scale_g, double scale_h,
+                                      int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    float2 gp = gpair[i];
+    longlong2 q;
+    q.x = llrint((double)gp.x * scale_g);
+    q.y = llrint((double)gp.y * scale_h);
+    out[i] = q;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// bin_matrix: values [n,F] -> bins u8 [n,F].
+// bin = count of cuts <= v (upper_bound), clamped to nb-1; NaN -> 255.
+// Cuts staged in LDS when they fit (F*max_bin floats; 200*255*4B = 204KB
+// does NOT fit -> fall back to L2-resident global reads, still fast since
+// cuts are tiny and re-read by every CU via L3).
+// ---------------------------------------------------------------------------
+template <bool LDS_CUTS>
+__global__ void bin_matrix_kernel(const float* __restrict__ values,
+                                  uint8_t* __restrict__ out,
+                                  const float* __restrict__ cuts,
+                                  const int64_t* __restrict__ cut_ptr,
+                                  int64_t n, int F, int total_cuts) {
+  extern __shared__ float lds_cuts[];
+  const float* C = cuts;
+  if (LDS_CUTS) {
+    for (int i = threadIdx.x; i < total_cuts; i += blockDim.x)
+      lds_cuts[i] = cuts[i];
+    __syncthreads();
+    C = lds_cuts;
+  }
+  int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t total = n * (int64_t)F;
+  for (; idx < total; idx += stride) {
+    int f = (int)(idx % F);
+    float v = values[idx];
+    int lo = (int)cut_ptr[f], hi = (int)cut_ptr[f + 1];
+    int nb = hi - lo;
+    uint8_t b;
+    if (isnan(v)) {
+      b = 255;
+    } else if (nb == 0) {
+      b = 0;
+    } else {
+      // upper_bound: first cut > v
+      int l = 0, r = nb;
+      while (l < r) {
+        int m = (l + r) >> 1;
+        if (C[lo + m] <= v) l = m + 1; else r = m;
+      }
+      if (l > nb - 1) l = nb - 1;
+      b = (uint8_t)l;
+    }
+    out[idx] = b;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// gather_gpair_seg: gpair_seg[i] = gpair_q[ridx[i]] for i in [0, n_seg)
+// (coalesced writes; the gather read hits L2/L3). Done once per depth so
+// the histogram kernel reads gradients coalesced per feature block.
+// ---------------------------------------------------------------------------
+__global__ void gather_gpair_kernel(const longlong2* __restrict__ gpair,
+                                    const int32_t* __restrict__ ridx,
+                                    longlong2* __restrict__ out, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) out[i] = gpair[(int64_t)(uint32_t)ridx[i]];
+}
+
+// ---------------------------------------------------------------------------
+// build_histogram: LDS-tiled int64 gradient-pair histograms.
+//
+// Grid: x = flattened (node, row-chunk), y = feature block.
+// Each workgroup accumulates a [FB][n_bins][2] int64 tile in LDS with
+// ds-atomics, then merges once into the global histogram with device
+// atomics. Integer accumulation => order-independent, bitwise
+// deterministic (the checkpoint-determinism contract).
+// ---------------------------------------------------------------------------
+#define HIST_THREADS 256
+#define HIST_ROWS_PER_WG 16384
+
+__global__ __launch_bounds__(HIST_THREADS) void build_histogram_kernel(
+    const uint8_t* __restrict__ bins,        // [n_rows_total, F]
+    const longlong2* __restrict__ gpair_seg, // [seg_total] segment order
+    const int32_t* __restrict__ ridx,        // [seg_total]
+    const int64_t* __restrict__ node_start,  // [K] segment starts
+    const int64_t* __restrict__ chunk_off,   // [K+1] cumulative chunks
+    long long* __restrict__ hist,            // [K, F, n_bins, 2]
+    int K, int F, int n_bins, int fb_size) {
+  // locate (node, chunk) from blockIdx.x via binary search on chunk_off
+  int wg = blockIdx.x;
+  int lo = 0, hi = K;
+  while (lo + 1 < hi) {
+    int m = (lo + hi) >> 1;
+    if (chunk_off[m] <= wg) lo = m; else hi = m;
+  }
+  const int node = lo;
+  const int64_t chunk_in_node = wg - chunk_off[node];
+  const int64_t seg_start = node_start[node];
+  // rows of this chunk within the node's segment
+  const int64_t row_lo = chunk_in_node * HIST_ROWS_PER_WG;
+
+  const int fb = blockIdx.y;
+  const int f0 = fb * fb_size;
+  const int fcount = fb_size < (F - f0) ? fb_size : (F - f0);
+
+  extern __shared__ unsigned long long lds_hist[];  // [fb_size][n_bins][2]
+  const int tile = fcount * n_bins * 2;
+  for (int i = threadIdx.x; i < tile; i += blockDim.x) lds_hist[i] = 0ull;
+  __syncthreads();
+
+  // row loop: lanes take consecutive rows -> gpair reads coalesced
+  // (segment order), bin reads are per-row contiguous byte runs.
+  // counts are stored at node_start[K..2K) (cat_start_count layout)
+  const int64_t node_count = node_start[K + node];
+  int64_t row_hi = row_lo + HIST_ROWS_PER_WG;
+  if (row_hi > node_count) row_hi = node_count;
+
+  for (int64_t i = row_lo + threadIdx.x; i < row_hi; i += blockDim.x) {
+    const int64_t seg_i = seg_start + i;
+    const longlong2 gp = gpair_seg[seg_i];
+    const uint64_t r = (uint32_t)ridx[seg_i];
+    const uint8_t* rowb = bins + r * (uint64_t)F + f0;
+    #pragma unroll 4
+    for (int f = 0; f < fcount; ++f) {
+      const int b = rowb[f];
+      if (b != 255) {
+        unsigned long long* cell = &lds_hist[((size_t)f * n_bins + b) * 2];
+        atomicAdd(cell, (unsigned long long)gp.x);
+        atomicAdd(cell + 1, (unsigned long long)gp.y);
+      }
+    }
+  }
+  __syncthreads();
+
+  // merge LDS tile into the global histogram
+  long long* ghist =
+      hist + (((size_t)node * F + f0) * n_bins) * 2;
+  for (int i = threadIdx.x; i < tile; i += blockDim.x) {
+    const unsigned long long v = lds_hist[i];
+    if (v != 0ull) {
+      atomicAdd((unsigned long long*)&ghist[i], v);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// find_splits: one thread per (node, feature); sequential 256-bin scan in
+// the exact same FP order as the CPU torch reference (int64 -> double per
+// bin, then left-to-right adds) so CPU and GPU grow identical trees.
+// ---------------------------------------------------------------------------
+struct SplitCand {
+  double gain;
+  int bin;
+  int default_left;
+  long long left_g;
+  long long left_h;
+};
+
+__device__ inline double calc_score(double G, double H, double lam, double alpha) {
+  if (alpha > 0.0) {
+    double t = fabs(G) - alpha;
+    G = t > 0.0 ? copysign(t, G) : 0.0;
+  }
+  double denom = H + lam;
+  return denom > 0.0 ? G * G / denom : 0.0;
+}
+
+__global__ void find_splits_kf_kernel(
+    const long long* __restrict__ hist,  // [K, F, B, 2]
+    const long long* __restrict__ parent_g, const long long* __restrict__ parent_h,
+    const int32_t* __restrict__ feat_bins, double scale_g, double scale_h,
+    double lam, double alpha, double gamma, double mcw,
+    double* __restrict__ out_gain,     // [K, F]
+    int32_t* __restrict__ out_bin,     // [K, F]
+    uint8_t* __restrict__ out_dl,      // [K, F]
+    long long* __restrict__ out_lg,    // [K, F]
+    long long* __restrict__ out_lh,    // [K, F]
+    int K, int F, int B) {
+  int64_t kf = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (kf >= (int64_t)K * F) return;
+  const int k = (int)(kf / F);
+  const int f = (int)(kf % F);
+  const long long* h = hist + ((size_t)k * F + f) * B * 2;
+  const int nb = feat_bins[f];
+
+  const double Gp = (double)parent_g[k] / scale_g;
+  const double Hp = (double)parent_h[k] / scale_h;
+  const double parent_score = calc_score(Gp, Hp, lam, alpha);
+
+  // feature totals (same order as cumsum's last element)
+  double Gtot = 0.0, Htot = 0.0;
+  long long Gtot_q = 0, Htot_q = 0;
+  for (int b = 0; b < B; ++b) {
+    Gtot += (double)h[b * 2] / scale_g;
+    Htot += (double)h[b * 2 + 1] / scale_h;
+    Gtot_q += h[b * 2];
+    Htot_q += h[b * 2 + 1];
+  }
+  const double Gmiss = Gp - Gtot;
+  const double Hmiss = Hp - Htot;
+  const long long Gmiss_q = parent_g[k] - Gtot_q;
+  const long long Hmiss_q = parent_h[k] - Htot_q;
+
+  SplitCand best = {-1.0, 0, 0, 0, 0};
+  // evaluate default_left = 1 first, then 0; strict > keeps CPU tie-break
+  for (int dl = 1; dl >= 0; --dl) {
+    double GL = 0.0, HL = 0.0;
+    long long GLq = 0, HLq = 0;
+    for (int b = 0; b < nb - 1 && b < B; ++b) {
+      GL += (double)h[b * 2] / scale_g;
+      HL += (double)h[b * 2 + 1] / scale_h;
+      GLq += h[b * 2];
+      HLq += h[b * 2 + 1];
+      double gl = dl ? GL + Gmiss : GL;
+      double hl = dl ? HL + Hmiss : HL;
+      double gr = Gp - gl, hr = Hp - hl;
+      if (hl < mcw || hr < mcw) continue;
+      double gain = 0.5 * (calc_score(gl, hl, lam, alpha) +
+                           calc_score(gr, hr, lam, alpha) - parent_score) -
+                    gamma;
+      if (gain > best.gain) {
+        best.gain = gain;
+        best.bin = b;
+        best.default_left = dl;
+        best.left_g = dl ? GLq + Gmiss_q : GLq;
+        best.left_h = dl ? HLq + Hmiss_q : HLq;
+      }
+    }
+  }
+  out_gain[kf] = best.gain;
+  out_bin[kf] = best.bin;
+  out_dl[kf] = (uint8_t)best.default_left;
+  out_lg[kf] = best.left_g;
+  out_lh[kf] = best.left_h;
+}
+
+__global__ void find_splits_reduce_kernel(
+    const double* __restrict__ kf_gain, const int32_t* __restrict__ kf_bin,
+    const uint8_t* __restrict__ kf_dl, const long long* __restrict__ kf_lg,
+    const long long* __restrict__ kf_lh,
+    float* __restrict__ out_gain, int32_t* __restrict__ out_feat,
+    int32_t* __restrict__ out_bin, uint8_t* __restrict__ out_dl,
+    long long* __restrict__ out_lg, long long* __restrict__ out_lh,
+    int K, int F) {
+  int k = blockIdx.x * blockDim.x + threadIdx.x;
+  if (k >= K) return;
+  double bg = -1.0;
+  int bf = -1, bb = 0, bdl = 0;
+  long long blg = 0, blh = 0;
+  // CPU semantics: dl=1 candidates evaluated first over all (f,b), dl=0
+  // overrides only with strictly greater gain. Within one dl, first
+  // occurrence (lowest feature) wins ties.
+  for (int want_dl = 1; want_dl >= 0; --want_dl) {
+    for (int f = 0; f < F; ++f) {
+      int64_t kf = (int64_t)k * F + f;
+      if ((int)kf_dl[kf] != want_dl) continue;
+      if (kf_gain[kf] > bg) {
+        bg = kf_gain[kf];
+        bf = f; bb = kf_bin[kf]; bdl = want_dl;
+        blg = kf_lg[kf]; blh = kf_lh[kf];
+      }
+    }
+  }
+  out_gain[k] = (float)bg;
+  out_feat[k] = bf;
+  out_bin[k] = bb;
+  out_dl[k] = (uint8_t)bdl;
+  out_lg[k] = blg;
+  out_lh[k] = blh;
+}
+
+// ---------------------------------------------------------------------------
+// partition_rows: stable two-pass partition of each node's ridx segment.
+// Pass 1 counts left-rows per 256-row block; host does a cumsum; pass 2
+// scatters with an in-block ballot prefix. Stability preserved because
+// blocks are processed in segment order and lanes in row order.
+// ---------------------------------------------------------------------------
+#define PART_THREADS 256
+
+__device__ inline bool go_left_pred(const uint8_t* bins, uint64_t r, int F,
+                                    int feat, int split_bin, int dl) {
+  const int b = bins[r * (uint64_t)F + feat];
+  if (b == 255) return dl != 0;
+  return b <= split_bin;
+}
+
+__global__ void partition_count_kernel(
+    const uint8_t* __restrict__ bins, const int32_t* __restrict__ ridx,
+    const int64_t* __restrict__ node_start,  // [K] then counts at [K..2K)
+    const int64_t* __restrict__ chunk_off,   // [K+1]
+    const int32_t* __restrict__ split_feat, const int32_t* __restrict__ split_bin,
+    const uint8_t* __restrict__ default_left,
+    int32_t* __restrict__ block_counts,  // [total_chunks]
+    int K, int F) {
+  int wg = blockIdx.x;
+  int lo = 0, hi = K;
+  while (lo + 1 < hi) {
+    int m = (lo + hi) >> 1;
+    if (chunk_off[m] <= wg) lo = m; else hi = m;
+  }
+  const int node = lo;
+  const int64_t chunk_in_node = wg - chunk_off[node];
+  const int64_t row_lo = chunk_in_node * PART_THREADS;
+  const int64_t count = node_start[K + node];
+  const int64_t seg_start = node_start[node];
+  const int feat = split_feat[node], sbin = split_bin[node];
+  const int dl = default_left[node];
+
+  const int64_t i = row_lo + threadIdx.x;
+  bool flag = false;
+  if (i < count) {
+    uint64_t r = (uint32_t)ridx[seg_start + i];
+    flag = go_left_pred(bins, r, F, feat, sbin, dl);
+  }
+  unsigned long long mask = __ballot(flag);
+  __shared__ int wave_sums[PART_THREADS / WAVE];
+  const int wave_id = threadIdx.x / WAVE;
+  if ((threadIdx.x & (WAVE - 1)) == 0) wave_sums[wave_id] = __popcll(mask);
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    int total = 0;
+    for (int w = 0; w < PART_THREADS / WAVE; ++w) total += wave_sums[w];
+    block_counts[wg] = total;
+  }
+}
+
+__global__ void partition_scatter_kernel(
+    const uint8_t* __restrict__ bins, const int32_t* __restrict__ ridx,
+    int32_t* __restrict__ ridx_out,
+    const int64_t* __restrict__ node_start,  // [K] + counts at [K..2K)
+    const int64_t* __restrict__ chunk_off,   // [K+1]
+    const int32_t* __restrict__ split_feat, const int32_t* __restrict__ split_bin,
+    const uint8_t* __restrict__ default_left,
+    const int64_t* __restrict__ left_before,   // [total_chunks] excl. prefix within node
+    const int64_t* __restrict__ node_left_total,  // [K]
+    int K, int F) {
+  int wg = blockIdx.x;
+  int lo = 0, hi = K;
+  while (lo + 1 < hi) {
+    int m = (lo + hi) >> 1;
+    if (chunk_off[m] <= wg) lo = m; else hi = m;
+  }
+  const int node = lo;
+  const int64_t chunk_in_node = wg - chunk_off[node];
+  const int64_t row_lo = chunk_in_node * PART_THREADS;
+  const int64_t count = node_start[K + node];
+  const int64_t seg_start = node_start[node];
+  const int feat = split_feat[node], sbin = split_bin[node];
+  const int dl = default_left[node];
+
+  const int64_t i = row_lo + threadIdx.x;
+  bool valid = i < count;
+  int32_t rv = 0;
+  bool flag = false;
+  if (valid) {
+    rv = ridx[seg_start + i];
+    flag = go_left_pred(bins, (uint32_t)rv, F, feat, sbin, dl);
+  }
+  unsigned long long mask = __ballot(flag);
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave_id = threadIdx.x / WAVE;
+  __shared__ int wave_sums[PART_THREADS / WAVE];
+  if (lane == 0) wave_sums[wave_id] = __popcll(mask);
+  __syncthreads();
+  int wave_left_before = 0;
+  for (int w = 0; w < wave_id; ++w) wave_left_before += wave_sums[w];
+
+  const int prefix_in_wave = __popcll(mask & ((lane == 0) ? 0ull : ((~0ull) >> (64 - lane))));
+  if (valid) {
+    const int64_t lbefore = left_before[wg];  // left rows in earlier chunks
+    if (flag) {
+      const int64_t pos = lbefore + wave_left_before + prefix_in_wave;
+      ridx_out[seg_start + pos] = rv;
+    } else {
+      // rights: position = node_left_total + (rows before me) - (lefts before me)
+      const int64_t rights_before =
+          (row_lo + threadIdx.x) - (lbefore + wave_left_before + prefix_in_wave);
+      ridx_out[seg_start + node_left_total[node] + rights_before] = rv;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// predict_trees: one thread per row, sequential over trees.
+// ---------------------------------------------------------------------------
+__global__ void predict_trees_kernel(
+    const float* __restrict__ X, const int32_t* __restrict__ feat,
+    const float* __restrict__ thr, const int32_t* __restrict__ left,
+    const uint8_t* __restrict__ default_left, const float* __restrict__ value,
+    const int32_t* __restrict__ tree_ptr, float* __restrict__ out,
+    float tree_weight, int64_t n, int F, int T) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    const float* row = X + i * (int64_t)F;
+    float acc = 0.0f;
+    for (int t = 0; t < T; ++t) {
+      int base = tree_ptr[t];
+      int nid = base;
+      int f = feat[nid];
+      while (f >= 0) {
+        float v = row[f];
+        bool goleft = isnan(v) ? (default_left[nid] != 0) : (v < thr[nid]);
+        nid = base + (goleft ? left[nid] : left[nid] + 1);
+        f = feat[nid];
+      }
+      acc += value[nid];
+    }
+    out[i] += acc * tree_weight;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// update_margins: margin[ridx[seg]] += leaf_val[node]
+// ---------------------------------------------------------------------------
+__global__ void update_margins_kernel(
+    float* __restrict__ margin, const int32_t* __restrict__ ridx,
+    const int64_t* __restrict__ node_start,  // [K] + counts at [K..2K)
+    const int64_t* __restrict__ chunk_off,   // [K+1]
+    const float* __restrict__ leaf_vals, int K) {
+  int wg = blockIdx.x;
+  int lo = 0, hi = K;
+  while (lo + 1 < hi) {
+    int m = (lo + hi) >> 1;
+    if (chunk_off[m] <= wg) lo = m; else hi = m;
+  }
+  const int node = lo;
+  const int64_t chunk_in_node = wg - chunk_off[node];
+  const int64_t row_lo = chunk_in_node * PART_THREADS;
+  const int64_t count = node_start[K + node];
+  const int64_t seg_start = node_start[node];
+  const float v = leaf_vals[node];
+  const int64_t i = row_lo + threadIdx.x;
+  if (i < count) {
+    margin[(uint32_t)ridx[seg_start + i]] += v;
+  }
+}
+
+// ===========================================================================
+// Host-side launchers / bindings
+// ===========================================================================
+
+static torch::Tensor make_chunk_off(const torch::Tensor& counts_cpu,
+                                    int64_t rows_per_chunk,
+                                    torch::Device dev, int64_t* total) {
+  const int64_t K = counts_cpu.size(0);
+  auto off = torch::zeros({K + 1}, torch::kInt64);
+  auto acc = off.accessor<int64_t, 1>();
+  auto cacc = counts_cpu.accessor<int64_t, 1>();
+  for (int64_t k = 0; k < K; ++k) {
+    acc[k + 1] = acc[k] + (cacc[k] + rows_per_chunk - 1) / rows_per_chunk;
+  }
+  *total = acc[K];
+  return off.to(dev);
+}
+
+static torch::Tensor cat_start_count(const torch::Tensor& starts_cpu,
+                                     const torch::Tensor& counts_cpu,
+                                     torch::Device dev) {
+  // layout used by kernels: [0..K) starts, [K..2K) counts
+  return torch::cat({starts_cpu, counts_cpu}).to(dev);
+}
+
+torch::Tensor quantize_gpair(torch::Tensor gpair, double scale_g, double scale_h) {
+  TORCH_CHECK(gpair.is_cuda() && gpair.dtype() == torch::kFloat32);
+  auto out = torch::empty_like(gpair, gpair.options().dtype(torch::kInt64));
+  int64_t n = gpair.size(0);
+  if (n == 0) return out;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  int64_t blocks = std::min<int64_t>(ceil_div(n, 256), 8192);
+  hipLaunchKernelGGL(quantize_gpair_kernel, dim3(blocks), dim3(256), 0,
+                     stream.stream(), (const float2*)gpair.data_ptr<float>(),
+                     (longlong2*)out.data_ptr<int64_t>(), scale_g, scale_h, n);
+  return out;
+}
+
+torch::Tensor bin_matrix(torch::Tensor values, torch::Tensor cuts_flat,
+                         torch::Tensor cut_ptr) {
+  TORCH_CHECK(values.is_cuda() && values.dim() == 2);
+  int64_t n = values.size(0);
+  int F = (int)values.size(1);
+  auto out = torch::empty({n, F}, values.options().dtype(torch::kUInt8));
+  if (n == 0) return out;
+  int total_cuts = (int)cuts_flat.size(0);
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  int64_t blocks = std::min<int64_t>(ceil_div(n * F, 256), 16384);
+  size_t lds = (size_t)total_cuts * sizeof(float);
+  if (lds <= 64 * 1024) {
+    hipLaunchKernelGGL((bin_matrix_kernel<true>), dim3(blocks), dim3(256), lds,
+                       stream.stream(), values.data_ptr<float>(),
+                       out.data_ptr<uint8_t>(), cuts_flat.data_ptr<float>(),
+                       cut_ptr.data_ptr<int64_t>(), n, F, total_cuts);
+  } else {
+    hipLaunchKernelGGL((bin_matrix_kernel<false>), dim3(blocks), dim3(256), 0,
+                       stream.stream(), values.data_ptr<float>(),
+                       out.data_ptr<uint8_t>(), cuts_flat.data_ptr<float>(),
+                       cut_ptr.data_ptr<int64_t>(), n, F, total_cuts);
+  }
+  return out;
+}
+
+torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
+                              torch::Tensor ridx, torch::Tensor starts,
+                              torch::Tensor counts, int64_t n_bins) {
+  TORCH_CHECK(bins.is_cuda() && bins.dtype() == torch::kUInt8);
+  const int K = (int)starts.size(0);
+  const int F = (int)bins.size(1);
+  auto dev = bins.device();
+  auto hist = torch::zeros({K, F, n_bins, 2},
+                           torch::TensorOptions().dtype(torch::kInt64).device(dev));
+  if (K == 0) return hist;
+  auto starts_cpu = starts.to(torch::kCPU).to(torch::kInt64);
+  auto counts_cpu = counts.to(torch::kCPU).to(torch::kInt64);
+  int64_t total_chunks = 0;
+  auto chunk_off = make_chunk_off(counts_cpu, HIST_ROWS_PER_WG, dev, &total_chunks);
+  if (total_chunks == 0) return hist;
+
+  // gather gpairs into segment order once (coalesced hist reads)
+  auto starts_acc = starts_cpu.accessor<int64_t, 1>();
+  auto counts_acc = counts_cpu.accessor<int64_t, 1>();
+  // Segments are contiguous [start, start+count) spans of ridx; gather the
+  // covering range [min_start, max_end) in one kernel.
+  int64_t min_start = INT64_MAX, max_end = 0;
+  for (int k = 0; k < K; ++k) {
+    min_start = std::min(min_start, starts_acc[k]);
+    max_end = std::max(max_end, starts_acc[k] + counts_acc[k]);
+  }
+  if (min_start == INT64_MAX) { min_start = 0; max_end = 0; }
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  int64_t span = max_end - min_start;
+  auto gpair_seg = torch::empty({std::max<int64_t>(span, 1), 2},
+                                gpair_q.options());
+  if (span > 0) {
+    int64_t blocks = std::min<int64_t>(ceil_div(span, 256), 8192);
+    hipLaunchKernelGGL(gather_gpair_kernel, dim3(blocks), dim3(256), 0,
+                       stream.stream(),
+                       (const longlong2*)gpair_q.data_ptr<int64_t>(),
+                       ridx.data_ptr<int32_t>() + min_start,
+                       (longlong2*)gpair_seg.data_ptr<int64_t>(), span);
+  }
+  // adjust starts so kernels index gpair_seg with (start - min_start)
+  auto starts_adj = starts_cpu - min_start;
+  auto sc_adj = cat_start_count(starts_adj, counts_cpu, dev);
+
+  // feature-block size: fit the LDS tile (fb * n_bins * 16 B) within the
+  // 64 KiB dynamic-LDS default so two workgroups co-reside per CU
+  int fb_size = (int)std::min<int64_t>(F, (64 * 1024) / (n_bins * 16));
+  if (fb_size < 1) fb_size = 1;
+  const int n_fb = (int)ceil_div(F, fb_size);
+  const size_t lds = (size_t)fb_size * n_bins * 2 * sizeof(long long);
+
+  // ridx pointer offset so seg indices align with gpair_seg
+  hipLaunchKernelGGL(build_histogram_kernel, dim3((uint32_t)total_chunks, n_fb),
+                     dim3(HIST_THREADS), lds, stream.stream(),
+                     bins.data_ptr<uint8_t>(),
+                     (const longlong2*)gpair_seg.data_ptr<int64_t>(),
+                     ridx.data_ptr<int32_t>() + min_start,
+                     sc_adj.data_ptr<int64_t>(), chunk_off.data_ptr<int64_t>(),
+                     reinterpret_cast<long long*>(hist.data_ptr<int64_t>()), K, F, (int)n_bins, fb_size);
+  return hist;
+}
+
+std::vector<torch::Tensor> find_splits(torch::Tensor hist, torch::Tensor parent_g,
+                                       torch::Tensor parent_h,
+                                       torch::Tensor feat_bins, double scale_g,
+                                       double scale_h, double lam, double alpha,
+                                       double gamma, double mcw) {
+  const int K = (int)hist.size(0);
+  const int F = (int)hist.size(1);
+  const int B = (int)hist.size(2);
+  auto dev = hist.device();
+  auto optsd = torch::TensorOptions().dtype(torch::kFloat64).device(dev);
+  auto optsi = torch::TensorOptions().dtype(torch::kInt32).device(dev);
+  auto optsl = torch::TensorOptions().dtype(torch::kInt64).device(dev);
+  auto optsb = torch::TensorOptions().dtype(torch::kUInt8).device(dev);
+  auto kf_gain = torch::empty({(int64_t)K * F}, optsd);
+  auto kf_bin = torch::empty({(int64_t)K * F}, optsi);
+  auto kf_dl = torch::empty({(int64_t)K * F}, optsb);
+  auto kf_lg = torch::empty({(int64_t)K * F}, optsl);
+  auto kf_lh = torch::empty({(int64_t)K * F}, optsl);
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  int64_t total = (int64_t)K * F;
+  hipLaunchKernelGGL(find_splits_kf_kernel, dim3((uint32_t)ceil_div(total, 256)),
+                     dim3(256), 0, stream.stream(),
+                     reinterpret_cast<const long long*>(hist.data_ptr<int64_t>()),
+                     reinterpret_cast<const long long*>(parent_g.data_ptr<int64_t>()),
+                     reinterpret_cast<const long long*>(parent_h.data_ptr<int64_t>()),
+                     feat_bins.data_ptr<int32_t>(), scale_g, scale_h, lam,
+                     alpha, gamma, mcw, kf_gain.data_ptr<double>(),
+                     kf_bin.data_ptr<int32_t>(), kf_dl.data_ptr<uint8_t>(),
+                     reinterpret_cast<long long*>(kf_lg.data_ptr<int64_t>()),
+                     reinterpret_cast<long long*>(kf_lh.data_ptr<int64_t>()), K, F, B);
+  auto out_gain = torch::empty({K}, torch::TensorOptions().dtype(torch::kFloat32).device(dev));
+  auto out_feat = torch::empty({K}, optsi);
+  auto out_bin = torch::empty({K}, optsi);
+  auto out_dl = torch::empty({K}, optsb);
+  auto out_lg = torch::empty({K}, optsl);
+  auto out_lh = torch::empty({K}, optsl);
+  hipLaunchKernelGGL(find_splits_reduce_kernel, dim3((uint32_t)ceil_div(K, 64)),
+                     dim3(64), 0, stream.stream(), kf_gain.data_ptr<double>(),
+                     kf_bin.data_ptr<int32_t>(), kf_dl.data_ptr<uint8_t>(),
+                     reinterpret_cast<const long long*>(kf_lg.data_ptr<int64_t>()),
+                     reinterpret_cast<const long long*>(kf_lh.data_ptr<int64_t>()),
+                     out_gain.data_ptr<float>(), out_feat.data_ptr<int32_t>(),
+                     out_bin.data_ptr<int32_t>(), out_dl.data_ptr<uint8_t>(),
+                     reinterpret_cast<long long*>(out_lg.data_ptr<int64_t>()),
+                     reinterpret_cast<long long*>(out_lh.data_ptr<int64_t>()), K, F);
+  return {out_gain, out_feat, out_bin, out_dl, out_lg, out_lh};
+}
+
+std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx,
+                                          torch::Tensor starts, torch::Tensor counts,
+                                          torch::Tensor split_feat,
+                                          torch::Tensor split_bin,
+                                          torch::Tensor default_left) {
+  const int K = (int)starts.size(0);
+  const int F = (int)bins.size(1);
+  auto dev = bins.device();
+  auto ridx_out = ridx.clone();
+  auto left_counts = torch::zeros({K}, torch::kInt64);
+  if (K == 0) return {ridx_out, left_counts};
+  auto starts_cpu = starts.to(torch::kCPU).to(torch::kInt64);
+  auto counts_cpu = counts.to(torch::kCPU).to(torch::kInt64);
+  int64_t total_chunks = 0;
+  auto chunk_off_cpu = torch::zeros({K + 1}, torch::kInt64);
+  {
+    auto acc = chunk_off_cpu.accessor<int64_t, 1>();
+    auto cacc = counts_cpu.accessor<int64_t, 1>();
+    for (int k = 0; k < K; ++k)
+      acc[k + 1] = acc[k] + (cacc[k] + PART_THREADS - 1) / PART_THREADS;
+    total_chunks = acc[K];
+  }
+  if (total_chunks == 0) return {ridx_out, left_counts};
+  auto chunk_off = chunk_off_cpu.to(dev);
+  auto sc = cat_start_count(starts_cpu, counts_cpu, dev);
+  auto sf = split_feat.to(dev).to(torch::kInt32);
+  auto sb = split_bin.to(dev).to(torch::kInt32);
+  auto dl = default_left.to(dev).to(torch::kUInt8);
+
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  auto block_counts = torch::empty({total_chunks},
+      torch::TensorOptions().dtype(torch::kInt32).device(dev));
+  hipLaunchKernelGGL(partition_count_kernel, dim3((uint32_t)total_chunks),
+                     dim3(PART_THREADS), 0, stream.stream(),
+                     bins.data_ptr<uint8_t>(), ridx.data_ptr<int32_t>(),
+                     sc.data_ptr<int64_t>(), chunk_off.data_ptr<int64_t>(),
+                     sf.data_ptr<int32_t>(), sb.data_ptr<int32_t>(),
+                     dl.data_ptr<uint8_t>(), block_counts.data_ptr<int32_t>(),
+                     K, F);
+  // per-node exclusive prefix of block counts
+  auto bc64 = block_counts.to(torch::kInt64);
+  auto csum = torch::cumsum(bc64, 0);
+  auto left_before = csum - bc64;  // global exclusive prefix
+  // subtract each node's base so prefix restarts per node; compute node
+  // totals as csum at node end minus base
+  auto chunk_off_acc = chunk_off_cpu.accessor<int64_t, 1>();
+  auto node_left_total_cpu = torch::zeros({K}, torch::kInt64);
+  {
+    auto csum_cpu = csum.to(torch::kCPU);
+    auto cs = csum_cpu.accessor<int64_t, 1>();
+    auto nl = node_left_total_cpu.accessor<int64_t, 1>();
+    auto lb_cpu = left_before.to(torch::kCPU);
+    auto lb = lb_cpu.accessor<int64_t, 1>();
+    // rebase left_before per node
+    for (int k = 0; k < K; ++k) {
+      int64_t c0 = chunk_off_acc[k], c1 = chunk_off_acc[k + 1];
+      if (c1 == c0) continue;
+      int64_t base = lb[c0];
+      nl[k] = cs[c1 - 1] - base;
+    }
+    // left_before rebasing done on CPU then moved back
+    auto lb_mut = lb_cpu.accessor<int64_t, 1>();
+    for (int k = 0; k < K; ++k) {
+      int64_t c0 = chunk_off_acc[k], c1 = chunk_off_acc[k + 1];
+      if (c1 == c0) continue;
+      int64_t base = lb_mut[c0];
+      for (int64_t c = c0; c < c1; ++c) lb_mut[c] -= base;
+    }
+    left_before = lb_cpu.to(dev);
+  }
+  auto node_left_total = node_left_total_cpu.to(dev);
+  hipLaunchKernelGGL(partition_scatter_kernel, dim3((uint32_t)total_chunks),
+                     dim3(PART_THREADS), 0, stream.stream(),
+                     bins.data_ptr<uint8_t>(), ridx.data_ptr<int32_t>(),
+                     ridx_out.data_ptr<int32_t>(), sc.data_ptr<int64_t>(),
+                     chunk_off.data_ptr<int64_t>(), sf.data_ptr<int32_t>(),
+                     sb.data_ptr<int32_t>(), dl.data_ptr<uint8_t>(),
+                     left_before.data_ptr<int64_t>(),
+                     node_left_total.data_ptr<int64_t>(), K, F);
+  return {ridx_out, node_left_total_cpu};
+}
+
+void predict_trees(torch::Tensor X, torch::Tensor feat, torch::Tensor thr,
+                   torch::Tensor left, torch::Tensor default_left,
+                   torch::Tensor value, torch::Tensor tree_ptr,
+                   torch::Tensor out, double tree_weight) {
+  int64_t n = X.size(0);
+  int F = (int)X.size(1);
+  int T = (int)tree_ptr.size(0) - 1;
+  if (n == 0 || T <= 0) return;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  int64_t blocks = std::min<int64_t>(ceil_div(n, 256), 8192);
+  hipLaunchKernelGGL(predict_trees_kernel, dim3(blocks), dim3(256), 0,
+                     stream.stream(), X.data_ptr<float>(),
+                     feat.data_ptr<int32_t>(), thr.data_ptr<float>(),
+                     left.data_ptr<int32_t>(), default_left.data_ptr<uint8_t>(),
+                     value.data_ptr<float>(), tree_ptr.data_ptr<int32_t>(),
+                     out.data_ptr<float>(), (float)tree_weight, n, F, T);
+}
+
+void update_margins(torch::Tensor margin, torch::Tensor ridx,
+                    torch::Tensor starts, torch::Tensor counts,
+                    torch::Tensor leaf_vals) {
+  const int K = (int)starts.size(0);
+  if (K == 0) return;
+  auto dev = margin.device();
+  auto starts_cpu = starts.to(torch::kCPU).to(torch::kInt64);
+  auto counts_cpu = counts.to(torch::kCPU).to(torch::kInt64);
+  auto chunk_off_cpu = torch::zeros({K + 1}, torch::kInt64);
+  int64_t total_chunks = 0;
+  {
+    auto acc = chunk_off_cpu.accessor<int64_t, 1>();
+    auto cacc = counts_cpu.accessor<int64_t, 1>();
+    for (int k = 0; k < K; ++k)
+      acc[k + 1] = acc[k] + (cacc[k] + PART_THREADS - 1) / PART_THREADS;
+    total_chunks = acc[K];
+  }
+  if (total_chunks == 0) return;
+  auto sc = cat_start_count(starts_cpu, counts_cpu, dev);
+  auto chunk_off = chunk_off_cpu.to(dev);
+  auto lv = leaf_vals.to(dev).to(torch::kFloat32);
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  hipLaunchKernelGGL(update_margins_kernel, dim3((uint32_t)total_chunks),
+                     dim3(PART_THREADS), 0, stream.stream(),
+                     margin.data_ptr<float>(), ridx.data_ptr<int32_t>(),
+                     sc.data_ptr<int64_t>(), chunk_off.data_ptr<int64_t>(),
+                     lv.data_ptr<float>(), K);
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("quantize_gpair", &quantize_gpair, "quantize gradient pairs");
+  m.def("bin_matrix", &bin_matrix, "bin feature matrix");
+  m.def("build_histogram", &build_histogram, "build gradient histograms");
+  m.def("find_splits", &find_splits, "best-split scan");
+  m.def("partition_rows", &partition_rows, "stable row partition");
+  m.def("predict_trees", &predict_trees, "tree-walk prediction");
+  m.def("update_margins", &update_margins, "leaf margin update");
+}

@@ -48,16 +48,30 @@ class HistogramCuts:
         )
 
 
+# below this row count a worker sends its exact per-value histogram, which
+# makes the merged cuts EXACTLY the global weighted quantiles - identical
+# for every world size / sharding (the distributed-equals-single contract)
+_EXACT_LIMIT = 1 << 16
+
+
 def _local_summary(col: torch.Tensor, n_points: int, seed: int):
     """(values, weights, min, max) summary of one feature column."""
     valid = col[~torch.isnan(col)]
     n = valid.numel()
     if n == 0:
         return None
+    if n <= _EXACT_LIMIT:
+        uniq, counts = torch.unique(valid, return_counts=True)
+        return (
+            uniq.cpu().numpy().astype(np.float64),
+            counts.cpu().numpy().astype(np.float64),
+            float(uniq[0]),
+            float(uniq[-1]),
+        )
     if n > _MAX_SAMPLE:
         # deterministic strided subsample (stable under re-runs)
         stride = n // _MAX_SAMPLE
-        valid = valid[:: stride][:_MAX_SAMPLE]
+        valid = valid[::stride][:_MAX_SAMPLE]
         n_rep = n
         n = valid.numel()
     else:

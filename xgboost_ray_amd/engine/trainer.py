@@ -202,8 +202,10 @@ class BoostingEngine:
                 if t.feat[nid] >= 0:
                     f = int(t.feat[nid])
                     lo, hi = cut_ptr[f], cut_ptr[f + 1]
+                    # stored thr == cut[b] for a "bin <= b" split, so the
+                    # bin index is exactly the position of thr in the cuts
                     b = np.searchsorted(cuts_np[lo:hi], t.thr[nid], side="left")
-                    split_bin[nid] = b - 1
+                    split_bin[nid] = b
             margin_view = (
                 self.margin if self.n_class == 1 else self.margin[:, cls]
             )

@@ -107,26 +107,28 @@ def partition_rows(bins, ridx, starts, counts, split_feat, split_bin, default_le
 
 def predict_trees(X, feat, thr, left, default_left, value, tree_ptr, out, tree_weight=1.0):
     dev = X.device
+    target = out if out.is_contiguous() else out.contiguous()
     _load().predict_trees(
-        X,
+        X.contiguous(),
         feat.to(dev),
         thr.to(dev),
         left.to(dev),
         default_left.to(dev),
         value.to(dev),
         tree_ptr.to(dev),
-        out,
+        target,
         float(tree_weight),
     )
+    if target is not out:
+        out.copy_(target)
     return out
 
 
 def update_margins(margin, ridx, starts, counts, leaf_values):
-    import numpy as np
-
     dev = margin.device
     lv = torch.as_tensor(leaf_values, dtype=torch.float32, device=dev)
-    _load().update_margins(
-        margin, ridx, starts.to(dev), counts.to(dev), lv
-    )
+    target = margin if margin.is_contiguous() else margin.contiguous()
+    _load().update_margins(target, ridx, starts.to(dev), counts.to(dev), lv)
+    if target is not margin:
+        margin.copy_(target)
     return margin
