@@ -32,7 +32,11 @@ if "--cpu-only" not in sys.argv:
                     sources=sources,
                     extra_compile_args={
                         "cxx": ["-O3", "-std=c++17"],
-                        "nvcc": ["-O3", "-std=c++17"],
+                        # -ffp-contract=off: the double-precision split
+                        # scan must be bitwise-identical to the CPU torch
+                        # oracle (FMA contraction flips near-tie splits);
+                        # the hot histogram path is integer and unaffected.
+                        "nvcc": ["-O3", "-std=c++17", "-ffp-contract=off"],
                     },
                 )
             ]
