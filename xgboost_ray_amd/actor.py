@@ -109,7 +109,10 @@ class _ActorWorker:
             shard = _materialize_shard(desc)
             self.shards[uid] = shard
             self.dist_callbacks.after_data_loading(self, desc)
-        return {uid: int(s["data"].shape[0]) for uid, s in self.shards.items()}
+        return {
+            uid: (int(s["data"].shape[0]) if s.get("data") is not None else -1)
+            for uid, s in self.shards.items()
+        }
 
     def train(
         self,
@@ -140,20 +143,36 @@ class _ActorWorker:
             device=self.device,
         )
         try:
-            t_x = torch.from_numpy(
-                np.ascontiguousarray(shard["data"], dtype=np.float32)
-            ).to(self.device)
-            dm = BinnedMatrix.build(
-                t_x,
-                label=_to_dev(shard.get("label"), self.device),
-                weight=_to_dev(shard.get("weight"), self.device),
-                base_margin=_to_dev(shard.get("base_margin"), self.device),
-                qid=_to_dev(shard.get("qid"), self.device, dtype=None),
-                max_bin=int(params.get("max_bin", 256)),
-                collective=coll,
-                seed=int(params.get("seed", 0) or 0),
-            )
-            del t_x
+            if shard.get("streaming"):
+                loader = shard["loader"]
+
+                def chunk_fn(loader=loader, shard=shard):
+                    return loader.iter_shards(
+                        shard["rank"], shard["num_actors"]
+                    )
+
+                dm = BinnedMatrix.build_streaming(
+                    chunk_fn,
+                    n_features=0,
+                    device=self.device,
+                    max_bin=int(params.get("max_bin", 256)),
+                    collective=coll,
+                )
+            else:
+                t_x = torch.from_numpy(
+                    np.ascontiguousarray(shard["data"], dtype=np.float32)
+                ).to(self.device)
+                dm = BinnedMatrix.build(
+                    t_x,
+                    label=_to_dev(shard.get("label"), self.device),
+                    weight=_to_dev(shard.get("weight"), self.device),
+                    base_margin=_to_dev(shard.get("base_margin"), self.device),
+                    qid=_to_dev(shard.get("qid"), self.device, dtype=None),
+                    max_bin=int(params.get("max_bin", 256)),
+                    collective=coll,
+                    seed=int(params.get("seed", 0) or 0),
+                )
+                del t_x
 
             eval_packs = []
             for uid, name in evals:
@@ -161,6 +180,12 @@ class _ActorWorker:
                     eval_packs.append(EvalPack(name=name, X=None))
                 else:
                     es = self.shards[uid]
+                    if es.get("streaming"):
+                        raise ValueError(
+                            "Streaming (RayDeviceQuantileDMatrix) eval sets "
+                            "are not supported - evaluate on the training "
+                            "matrix or an in-memory RayDMatrix."
+                        )
                     eval_packs.append(
                         EvalPack(
                             name=name,
@@ -219,7 +244,7 @@ class _ActorWorker:
             result = {
                 "bst": bst if return_bst else None,
                 "evals_result": evals_result,
-                "train_n": int(shard["data"].shape[0]),
+                "train_n": int(dm.n_rows),
             }
             self.dist_callbacks.after_train(self, result)
             return result
@@ -236,10 +261,21 @@ class _ActorWorker:
         shard = self.shards.get(data_uid)
         if shard is None:
             raise RuntimeError(f"Prediction data {data_uid} not loaded")
-        X = torch.from_numpy(
-            np.ascontiguousarray(shard["data"], np.float32)
-        ).to(self.device)
-        margin = bst.predict_margin_tensor(X)
+        if shard.get("streaming"):
+            loader = shard["loader"]
+            margins = []
+            for chunk in loader.iter_shards(shard["rank"], shard["num_actors"]):
+                Xc = torch.from_numpy(
+                    np.ascontiguousarray(chunk["data"], np.float32)
+                ).to(self.device)
+                margins.append(bst.predict_margin_tensor(Xc))
+                del Xc
+            margin = torch.cat(margins, dim=0)
+        else:
+            X = torch.from_numpy(
+                np.ascontiguousarray(shard["data"], np.float32)
+            ).to(self.device)
+            margin = bst.predict_margin_tensor(X)
         if kwargs.get("output_margin"):
             pred = margin.cpu().numpy()
         else:
@@ -329,6 +365,16 @@ def _materialize_shard(desc: dict) -> Dict[str, Any]:
             desc["num_actors"], None, rank=desc["rank"]
         )
         return shard
+    if desc["kind"] == "streaming":
+        # defer materialization: the training path streams file-by-file
+        # through the device sketch + binning kernels
+        return {
+            "streaming": True,
+            "loader": desc["loader"],
+            "rank": desc["rank"],
+            "num_actors": desc["num_actors"],
+            "data": None,
+        }
     if desc["kind"] == "inline":
         return desc["shard"]
     raise ValueError(f"Unknown shard descriptor kind: {desc['kind']}")

@@ -163,6 +163,55 @@ def build_cuts(
     )
 
 
+class SketchAccumulator:
+    """Streaming per-feature quantile sketch: per-chunk summaries merged at
+    finalize time (and across workers with one allgather). The MI355X
+    equivalent of XGBoost's DeviceQuantileDMatrix streaming sketch
+    (reference RayDataIter, matrix.py:127-196)."""
+
+    def __init__(self, n_features: int, max_bin: int):
+        self.F = n_features
+        self.max_bin = min(int(max_bin), 255)
+        self.summaries = [[] for _ in range(n_features)]
+        self.n_rows = 0
+
+    def push_chunk(self, X: torch.Tensor):
+        assert X.shape[1] == self.F
+        n_points = min(_SUMMARY_FACTOR * self.max_bin, 1 << 14)
+        for f in range(self.F):
+            s = _local_summary(X[:, f], n_points, 0)
+            if s is not None:
+                self.summaries[f].append(s)
+        self.n_rows += X.shape[0]
+
+    def finalize(self, collective: Optional[Collective] = None) -> HistogramCuts:
+        local = self.summaries
+        if collective is not None and collective.is_distributed:
+            gathered = collective.allgather_obj(local)
+        else:
+            gathered = [local]
+        cuts_list = []
+        for f in range(self.F):
+            parts = []
+            for worker in gathered:
+                parts.extend(worker[f])
+            cuts_list.append(_merge_to_cuts(parts, self.max_bin))
+        ptr = np.zeros(self.F + 1, dtype=np.int64)
+        for f in range(self.F):
+            ptr[f + 1] = ptr[f] + len(cuts_list[f])
+        flat = (
+            np.concatenate(cuts_list)
+            if ptr[-1] > 0
+            else np.zeros(0, dtype=np.float32)
+        )
+        max_bins = int(max((len(c) for c in cuts_list), default=1)) or 1
+        return HistogramCuts(
+            cuts_flat=torch.from_numpy(flat.astype(np.float32)),
+            cut_ptr=torch.from_numpy(ptr),
+            max_bins=max_bins,
+        )
+
+
 class BinnedMatrix:
     """Quantized training matrix: uint8 bins + labels/weights/margins."""
 
@@ -199,3 +248,69 @@ class BinnedMatrix:
         if cuts is None:
             cuts = build_cuts(X, max_bin, collective, seed)
         return cls(X, label, weight, base_margin, qid, cuts)
+
+    @classmethod
+    def build_streaming(
+        cls,
+        chunk_fn,
+        n_features: int,
+        device,
+        max_bin: int = 256,
+        collective: Optional[Collective] = None,
+    ) -> "BinnedMatrix":
+        """Out-of-core construction: two passes over a chunk iterator.
+
+        ``chunk_fn()`` returns a fresh iterator of shard dicts (numpy
+        arrays: data [+label/weight/base_margin/qid]). Pass 1 streams each
+        chunk through the device sketch; pass 2 bins chunk-by-chunk into
+        the preallocated uint8 matrix - the raw fp32 features are never
+        resident in full, so a 1B x 200 shard costs ~25 GB HBM per GPU
+        instead of ~800 GB (SURVEY.md #5 long-context note).
+        """
+        acc = None
+        metas = []
+        for chunk in chunk_fn():
+            Xc = torch.from_numpy(
+                np.ascontiguousarray(chunk["data"], dtype=np.float32)
+            ).to(device)
+            if acc is None:
+                acc = SketchAccumulator(Xc.shape[1], max_bin)
+            acc.push_chunk(Xc)
+            metas.append(Xc.shape[0])
+            del Xc
+        if acc is None:
+            acc = SketchAccumulator(n_features, max_bin)
+        cuts = acc.finalize(collective).to(device)
+        n = acc.n_rows
+
+        obj = cls.__new__(cls)
+        obj.cuts = cuts
+        obj.n_rows = n
+        obj.n_features = acc.F
+        obj.bins = torch.empty((n, acc.F), dtype=torch.uint8, device=device)
+        sides = {"label": [], "weight": [], "base_margin": [], "qid": []}
+        pos = 0
+        for chunk in chunk_fn():
+            Xc = torch.from_numpy(
+                np.ascontiguousarray(chunk["data"], dtype=np.float32)
+            ).to(device)
+            c = Xc.shape[0]
+            obj.bins[pos : pos + c] = ops.bin_matrix(
+                Xc, cuts.cuts_flat, cuts.cut_ptr
+            )
+            del Xc
+            pos += c
+            for key in sides:
+                v = chunk.get(key)
+                if v is not None:
+                    sides[key].append(np.ascontiguousarray(v))
+        for key, parts in sides.items():
+            if parts:
+                arr = np.concatenate(parts)
+                t = torch.from_numpy(
+                    arr.astype(np.float32) if key != "qid" else arr
+                ).to(device)
+                setattr(obj, key, t)
+            else:
+                setattr(obj, key, None)
+        return obj

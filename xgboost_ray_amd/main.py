@@ -177,7 +177,11 @@ def _shard_descriptors(
                 (
                     dm._uid,
                     {
-                        "kind": "distributed",
+                        "kind": (
+                            "streaming"
+                            if getattr(dm, "streaming", False)
+                            else "distributed"
+                        ),
                         "loader": dm.loader,
                         "num_actors": num_actors,
                         "rank": rank,

@@ -333,9 +333,7 @@ class _DistributedRayDMatrixLoader(_RayDMatrixLoader):
     def get_n(self):
         return len(self.data)
 
-    def load_data(self, num_actors: int, sharding: RayShardingMode, rank=None):
-        """Load only this rank's shard files; returns in-memory shard."""
-        assert rank is not None
+    def shard_indices(self, rank: int, num_actors: int):
         n_shards = len(self.data)
         if n_shards < num_actors:
             raise RuntimeError(
@@ -343,7 +341,22 @@ class _DistributedRayDMatrixLoader(_RayDMatrixLoader):
                 f"data source only has {n_shards} shards/files. Pass at "
                 f"least one file/partition per actor."
             )
-        indices = list(range(rank, n_shards, num_actors))
+        return list(range(rank, n_shards, num_actors))
+
+    def iter_shards(self, rank: int, num_actors: int):
+        """Yield one shard dict per assigned file/partition (streaming,
+        out-of-core construction path - the RayDeviceQuantileDMatrix
+        equivalent, reference matrix.py:127-196)."""
+        for i in self.shard_indices(rank, num_actors):
+            local_df = self.data_source.load_data(
+                self.data, ignore=self.ignore, indices=[i], **self.kwargs
+            )
+            yield self._to_shard(local_df)
+
+    def load_data(self, num_actors: int, sharding: RayShardingMode, rank=None):
+        """Load only this rank's shard files; returns in-memory shard."""
+        assert rank is not None
+        indices = self.shard_indices(rank, num_actors)
         local_df = self.data_source.load_data(
             self.data, ignore=self.ignore, indices=indices, **self.kwargs
         )
@@ -516,4 +529,13 @@ class RayQuantileDMatrix(RayDMatrix):
 
 
 class RayDeviceQuantileDMatrix(RayQuantileDMatrix):
-    """GPU streaming quantile matrix (reference matrix.py:1005-1033)."""
+    """GPU streaming (out-of-core) quantile matrix
+    (reference matrix.py:1005-1033 + RayDataIter matrix.py:127-196).
+
+    With a distributed (multi-file) data source, each actor streams its
+    files through the device sketch + binning kernels one at a time, so
+    only the 1-byte-per-cell binned matrix is ever fully resident -
+    1B x 200 fits in 8 x 288 GB HBM.
+    """
+
+    streaming = True
