@@ -63,7 +63,8 @@ __global__ void bin_matrix_kernel(const float* __restrict__ values,
                                   uint8_t* __restrict__ out,
                                   const float* __restrict__ cuts,
                                   const int64_t* __restrict__ cut_ptr,
-                                  int64_t n, int F, int total_cuts) {
+                                  int64_t n, int F, int total_cuts,
+                                  int64_t out_stride) {
   extern __shared__ float lds_cuts[];
   const float* C = cuts;
   if (LDS_CUTS) {
@@ -95,7 +96,7 @@ __global__ void bin_matrix_kernel(const float* __restrict__ values,
       if (l > nb - 1) l = nb - 1;
       b = (uint8_t)l;
     }
-    out[idx] = b;
+    out[(idx / F) * out_stride + f] = b;
   }
 }
 
@@ -124,14 +125,20 @@ __global__ void gather_gpair_kernel(const longlong2* __restrict__ gpair,
 #define HIST_THREADS 256
 #define HIST_ROWS_PER_WG 16384
 
+// VEC16: the binned matrix row stride is 16-byte aligned (padded layout,
+// pad bytes == 255) -> one uint4 load fetches a whole 16-feature block of
+// a row. Lane rotation spreads one wave-instruction's LDS atomics over 16
+// different features (max 4 lanes per feature histogram) so same-bin
+// serialization on skewed features drops ~16x.
+template <bool VEC16>
 __global__ __launch_bounds__(HIST_THREADS) void build_histogram_kernel(
-    const uint8_t* __restrict__ bins,        // [n_rows_total, F]
+    const uint8_t* __restrict__ bins,        // [n_rows_total, row_stride]
     const longlong2* __restrict__ gpair_seg, // [seg_total] segment order
     const int32_t* __restrict__ ridx,        // [seg_total]
     const int64_t* __restrict__ node_start,  // [K] segment starts
     const int64_t* __restrict__ chunk_off,   // [K+1] cumulative chunks
     long long* __restrict__ hist,            // [K, F, n_bins, 2]
-    int K, int F, int n_bins, int fb_size) {
+    int K, int F, int n_bins, int fb_size, int64_t row_stride) {
   // locate (node, chunk) from blockIdx.x via binary search on chunk_off
   int wg = blockIdx.x;
   int lo = 0, hi = K;
@@ -161,18 +168,37 @@ __global__ __launch_bounds__(HIST_THREADS) void build_histogram_kernel(
   int64_t row_hi = row_lo + HIST_ROWS_PER_WG;
   if (row_hi > node_count) row_hi = node_count;
 
+  const int lane = threadIdx.x & (WAVE - 1);
   for (int64_t i = row_lo + threadIdx.x; i < row_hi; i += blockDim.x) {
     const int64_t seg_i = seg_start + i;
     const longlong2 gp = gpair_seg[seg_i];
     const uint64_t r = (uint32_t)ridx[seg_i];
-    const uint8_t* rowb = bins + r * (uint64_t)F + f0;
-    #pragma unroll 4
-    for (int f = 0; f < fcount; ++f) {
-      const int b = rowb[f];
-      if (b != 255) {
-        unsigned long long* cell = &lds_hist[((size_t)f * n_bins + b) * 2];
-        atomicAdd(cell, (unsigned long long)gp.x);
-        atomicAdd(cell + 1, (unsigned long long)gp.y);
+    if (VEC16) {
+      // fb_size == 16 and row base 16B-aligned by construction
+      const uint4 packed =
+          *reinterpret_cast<const uint4*>(bins + r * row_stride + f0);
+      const uint8_t* bytes = reinterpret_cast<const uint8_t*>(&packed);
+      const int rot = lane & 15;
+      #pragma unroll
+      for (int ff = 0; ff < 16; ++ff) {
+        const int f = (ff + rot) & 15;
+        const int b = bytes[f];
+        if (b != 255) {
+          unsigned long long* cell = &lds_hist[((size_t)f * n_bins + b) * 2];
+          atomicAdd(cell, (unsigned long long)gp.x);
+          atomicAdd(cell + 1, (unsigned long long)gp.y);
+        }
+      }
+    } else {
+      const uint8_t* rowb = bins + r * row_stride + f0;
+      #pragma unroll 4
+      for (int f = 0; f < fcount; ++f) {
+        const int b = rowb[f];
+        if (b != 255) {
+          unsigned long long* cell = &lds_hist[((size_t)f * n_bins + b) * 2];
+          atomicAdd(cell, (unsigned long long)gp.x);
+          atomicAdd(cell + 1, (unsigned long long)gp.y);
+        }
       }
     }
   }
@@ -211,7 +237,11 @@ __device__ inline double calc_score(double G, double H, double lam, double alpha
   return denom > 0.0 ? G * G / denom : 0.0;
 }
 
-__global__ void find_splits_kf_kernel(
+// One WAVE per (node, feature): 64 lanes cooperatively stage the 256-bin
+// histogram into LDS (coalesced), then lane 0 runs the sequential scan
+// from LDS in the exact CPU FP order. ~100x the parallelism of a
+// thread-per-(k,f) scan at shallow depths, with identical numerics.
+__global__ __launch_bounds__(256) void find_splits_kf_kernel(
     const long long* __restrict__ hist,  // [K, F, B, 2]
     const long long* __restrict__ parent_g, const long long* __restrict__ parent_h,
     const int32_t* __restrict__ feat_bins, double scale_g, double scale_h,
@@ -222,11 +252,20 @@ __global__ void find_splits_kf_kernel(
     long long* __restrict__ out_lg,    // [K, F]
     long long* __restrict__ out_lh,    // [K, F]
     int K, int F, int B) {
-  int64_t kf = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  extern __shared__ long long lds_h[];  // [waves_per_block][B*2]
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int waves_per_block = blockDim.x / WAVE;
+  int64_t kf = (int64_t)blockIdx.x * waves_per_block + wave;
   if (kf >= (int64_t)K * F) return;
   const int k = (int)(kf / F);
   const int f = (int)(kf % F);
-  const long long* h = hist + ((size_t)k * F + f) * B * 2;
+  const long long* gh = hist + ((size_t)k * F + f) * B * 2;
+  long long* h = lds_h + (size_t)wave * B * 2;
+  for (int i = lane; i < B * 2; i += WAVE) h[i] = gh[i];
+  // wave-local LDS visibility: all lanes' stores above precede this wait
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  if (lane != 0) return;
   const int nb = feat_bins[f];
 
   const double Gp = (double)parent_g[k] / scale_g;
@@ -323,13 +362,18 @@ __global__ void find_splits_reduce_kernel(
 // ---------------------------------------------------------------------------
 #define PART_THREADS 256
 
-__device__ inline bool go_left_pred(const uint8_t* bins, uint64_t r, int F,
-                                    int feat, int split_bin, int dl) {
-  const int b = bins[r * (uint64_t)F + feat];
+__device__ inline bool go_left_pred(const uint8_t* bins, uint64_t r,
+                                    int64_t row_stride, int feat,
+                                    int split_bin, int dl) {
+  const int b = bins[r * (uint64_t)row_stride + feat];
   if (b == 255) return dl != 0;
   return b <= split_bin;
 }
 
+// Pass 1: evaluate the split predicate ONCE per row (a ~64B cache-line
+// gather from the binned matrix), cache it as one flag byte per row, and
+// count per-256-row-block lefts. Pass 2 then re-reads 1 B/row instead of
+// repeating the gather - halves the partition's memory cost.
 __global__ void partition_count_kernel(
     const uint8_t* __restrict__ bins, const int32_t* __restrict__ ridx,
     const int64_t* __restrict__ node_start,  // [K] then counts at [K..2K)
@@ -337,7 +381,8 @@ __global__ void partition_count_kernel(
     const int32_t* __restrict__ split_feat, const int32_t* __restrict__ split_bin,
     const uint8_t* __restrict__ default_left,
     int32_t* __restrict__ block_counts,  // [total_chunks]
-    int K, int F) {
+    uint8_t* __restrict__ flags,         // [n] go-left per segment position
+    int K, int64_t row_stride) {
   int wg = blockIdx.x;
   int lo = 0, hi = K;
   while (lo + 1 < hi) {
@@ -356,7 +401,8 @@ __global__ void partition_count_kernel(
   bool flag = false;
   if (i < count) {
     uint64_t r = (uint32_t)ridx[seg_start + i];
-    flag = go_left_pred(bins, r, F, feat, sbin, dl);
+    flag = go_left_pred(bins, r, row_stride, feat, sbin, dl);
+    flags[seg_start + i] = flag ? 1 : 0;
   }
   unsigned long long mask = __ballot(flag);
   __shared__ int wave_sums[PART_THREADS / WAVE];
@@ -371,15 +417,13 @@ __global__ void partition_count_kernel(
 }
 
 __global__ void partition_scatter_kernel(
-    const uint8_t* __restrict__ bins, const int32_t* __restrict__ ridx,
+    const uint8_t* __restrict__ flags, const int32_t* __restrict__ ridx,
     int32_t* __restrict__ ridx_out,
     const int64_t* __restrict__ node_start,  // [K] + counts at [K..2K)
     const int64_t* __restrict__ chunk_off,   // [K+1]
-    const int32_t* __restrict__ split_feat, const int32_t* __restrict__ split_bin,
-    const uint8_t* __restrict__ default_left,
     const int64_t* __restrict__ left_before,   // [total_chunks] excl. prefix within node
     const int64_t* __restrict__ node_left_total,  // [K]
-    int K, int F) {
+    int K) {
   int wg = blockIdx.x;
   int lo = 0, hi = K;
   while (lo + 1 < hi) {
@@ -391,8 +435,6 @@ __global__ void partition_scatter_kernel(
   const int64_t row_lo = chunk_in_node * PART_THREADS;
   const int64_t count = node_start[K + node];
   const int64_t seg_start = node_start[node];
-  const int feat = split_feat[node], sbin = split_bin[node];
-  const int dl = default_left[node];
 
   const int64_t i = row_lo + threadIdx.x;
   bool valid = i < count;
@@ -400,7 +442,7 @@ __global__ void partition_scatter_kernel(
   bool flag = false;
   if (valid) {
     rv = ridx[seg_start + i];
-    flag = go_left_pred(bins, (uint32_t)rv, F, feat, sbin, dl);
+    flag = flags[seg_start + i] != 0;
   }
   unsigned long long mask = __ballot(flag);
   const int lane = threadIdx.x & (WAVE - 1);
@@ -525,7 +567,12 @@ torch::Tensor bin_matrix(torch::Tensor values, torch::Tensor cuts_flat,
   TORCH_CHECK(values.is_cuda() && values.dim() == 2);
   int64_t n = values.size(0);
   int F = (int)values.size(1);
-  auto out = torch::empty({n, F}, values.options().dtype(torch::kUInt8));
+  // padded allocation: 16B-aligned row stride, pad bytes = 255 (missing)
+  // so the vectorized histogram path can read whole uint4 feature blocks
+  int64_t F_pad = ((int64_t)F + 15) / 16 * 16;
+  auto full = torch::full({n, F_pad}, 255,
+                          values.options().dtype(torch::kUInt8));
+  auto out = full.narrow(1, 0, F);
   if (n == 0) return out;
   int total_cuts = (int)cuts_flat.size(0);
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
@@ -534,13 +581,13 @@ torch::Tensor bin_matrix(torch::Tensor values, torch::Tensor cuts_flat,
   if (lds <= 64 * 1024) {
     hipLaunchKernelGGL((bin_matrix_kernel<true>), dim3(blocks), dim3(256), lds,
                        stream.stream(), values.data_ptr<float>(),
-                       out.data_ptr<uint8_t>(), cuts_flat.data_ptr<float>(),
-                       cut_ptr.data_ptr<int64_t>(), n, F, total_cuts);
+                       full.data_ptr<uint8_t>(), cuts_flat.data_ptr<float>(),
+                       cut_ptr.data_ptr<int64_t>(), n, F, total_cuts, F_pad);
   } else {
     hipLaunchKernelGGL((bin_matrix_kernel<false>), dim3(blocks), dim3(256), 0,
                        stream.stream(), values.data_ptr<float>(),
-                       out.data_ptr<uint8_t>(), cuts_flat.data_ptr<float>(),
-                       cut_ptr.data_ptr<int64_t>(), n, F, total_cuts);
+                       full.data_ptr<uint8_t>(), cuts_flat.data_ptr<float>(),
+                       cut_ptr.data_ptr<int64_t>(), n, F, total_cuts, F_pad);
   }
   return out;
 }
@@ -589,20 +636,40 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
   auto sc_adj = cat_start_count(starts_adj, counts_cpu, dev);
 
   // feature-block size: fit the LDS tile (fb * n_bins * 16 B) within the
-  // 64 KiB dynamic-LDS default so two workgroups co-reside per CU
-  int fb_size = (int)std::min<int64_t>(F, (64 * 1024) / (n_bins * 16));
+  // 64 KiB dynamic-LDS default so two workgroups co-reside per CU.
+  // Padded (16B-aligned row stride) matrices take the vectorized path:
+  // fb = 16 features per uint4 row load.
+  const int64_t row_stride = bins.stride(0);
+  const bool vec16 = (row_stride % 16 == 0) && n_bins <= 256;
+  int fb_size = vec16
+                    ? 16
+                    : (int)std::min<int64_t>(F, (64 * 1024) / (n_bins * 16));
   if (fb_size < 1) fb_size = 1;
   const int n_fb = (int)ceil_div(F, fb_size);
   const size_t lds = (size_t)fb_size * n_bins * 2 * sizeof(long long);
 
   // ridx pointer offset so seg indices align with gpair_seg
-  hipLaunchKernelGGL(build_histogram_kernel, dim3((uint32_t)total_chunks, n_fb),
-                     dim3(HIST_THREADS), lds, stream.stream(),
-                     bins.data_ptr<uint8_t>(),
-                     (const longlong2*)gpair_seg.data_ptr<int64_t>(),
-                     ridx.data_ptr<int32_t>() + min_start,
-                     sc_adj.data_ptr<int64_t>(), chunk_off.data_ptr<int64_t>(),
-                     reinterpret_cast<long long*>(hist.data_ptr<int64_t>()), K, F, (int)n_bins, fb_size);
+  if (vec16) {
+    hipLaunchKernelGGL((build_histogram_kernel<true>),
+                       dim3((uint32_t)total_chunks, n_fb),
+                       dim3(HIST_THREADS), lds, stream.stream(),
+                       bins.data_ptr<uint8_t>(),
+                       (const longlong2*)gpair_seg.data_ptr<int64_t>(),
+                       ridx.data_ptr<int32_t>() + min_start,
+                       sc_adj.data_ptr<int64_t>(), chunk_off.data_ptr<int64_t>(),
+                       reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
+                       K, F, (int)n_bins, fb_size, row_stride);
+  } else {
+    hipLaunchKernelGGL((build_histogram_kernel<false>),
+                       dim3((uint32_t)total_chunks, n_fb),
+                       dim3(HIST_THREADS), lds, stream.stream(),
+                       bins.data_ptr<uint8_t>(),
+                       (const longlong2*)gpair_seg.data_ptr<int64_t>(),
+                       ridx.data_ptr<int32_t>() + min_start,
+                       sc_adj.data_ptr<int64_t>(), chunk_off.data_ptr<int64_t>(),
+                       reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
+                       K, F, (int)n_bins, fb_size, row_stride);
+  }
   return hist;
 }
 
@@ -626,8 +693,11 @@ std::vector<torch::Tensor> find_splits(torch::Tensor hist, torch::Tensor parent_
   auto kf_lh = torch::empty({(int64_t)K * F}, optsl);
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   int64_t total = (int64_t)K * F;
-  hipLaunchKernelGGL(find_splits_kf_kernel, dim3((uint32_t)ceil_div(total, 256)),
-                     dim3(256), 0, stream.stream(),
+  const int waves_per_block = 4;
+  const size_t scan_lds = (size_t)waves_per_block * B * 2 * sizeof(long long);
+  hipLaunchKernelGGL(find_splits_kf_kernel,
+                     dim3((uint32_t)ceil_div(total, waves_per_block)),
+                     dim3(waves_per_block * WAVE), scan_lds, stream.stream(),
                      reinterpret_cast<const long long*>(hist.data_ptr<int64_t>()),
                      reinterpret_cast<const long long*>(parent_g.data_ptr<int64_t>()),
                      reinterpret_cast<const long long*>(parent_h.data_ptr<int64_t>()),
@@ -686,53 +756,44 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   auto block_counts = torch::empty({total_chunks},
       torch::TensorOptions().dtype(torch::kInt32).device(dev));
+  auto flags = torch::empty({ridx.size(0)},
+      torch::TensorOptions().dtype(torch::kUInt8).device(dev));
   hipLaunchKernelGGL(partition_count_kernel, dim3((uint32_t)total_chunks),
                      dim3(PART_THREADS), 0, stream.stream(),
                      bins.data_ptr<uint8_t>(), ridx.data_ptr<int32_t>(),
                      sc.data_ptr<int64_t>(), chunk_off.data_ptr<int64_t>(),
                      sf.data_ptr<int32_t>(), sb.data_ptr<int32_t>(),
                      dl.data_ptr<uint8_t>(), block_counts.data_ptr<int32_t>(),
-                     K, F);
-  // per-node exclusive prefix of block counts
+                     flags.data_ptr<uint8_t>(), K, bins.stride(0));
+  // per-node exclusive prefix of block counts (all on device: at 100M
+  // rows this is ~400K chunks - far too many for a host loop)
   auto bc64 = block_counts.to(torch::kInt64);
   auto csum = torch::cumsum(bc64, 0);
   auto left_before = csum - bc64;  // global exclusive prefix
-  // subtract each node's base so prefix restarts per node; compute node
-  // totals as csum at node end minus base
-  auto chunk_off_acc = chunk_off_cpu.accessor<int64_t, 1>();
-  auto node_left_total_cpu = torch::zeros({K}, torch::kInt64);
-  {
-    auto csum_cpu = csum.to(torch::kCPU);
-    auto cs = csum_cpu.accessor<int64_t, 1>();
-    auto nl = node_left_total_cpu.accessor<int64_t, 1>();
-    auto lb_cpu = left_before.to(torch::kCPU);
-    auto lb = lb_cpu.accessor<int64_t, 1>();
-    // rebase left_before per node
-    for (int k = 0; k < K; ++k) {
-      int64_t c0 = chunk_off_acc[k], c1 = chunk_off_acc[k + 1];
-      if (c1 == c0) continue;
-      int64_t base = lb[c0];
-      nl[k] = cs[c1 - 1] - base;
-    }
-    // left_before rebasing done on CPU then moved back
-    auto lb_mut = lb_cpu.accessor<int64_t, 1>();
-    for (int k = 0; k < K; ++k) {
-      int64_t c0 = chunk_off_acc[k], c1 = chunk_off_acc[k + 1];
-      if (c1 == c0) continue;
-      int64_t base = lb_mut[c0];
-      for (int64_t c = c0; c < c1; ++c) lb_mut[c] -= base;
-    }
-    left_before = lb_cpu.to(dev);
-  }
-  auto node_left_total = node_left_total_cpu.to(dev);
+  auto chunks_per_node_cpu =
+      chunk_off_cpu.slice(0, 1, K + 1) - chunk_off_cpu.slice(0, 0, K);
+  auto node_id = torch::repeat_interleave(
+                     torch::arange(K, torch::kInt64), chunks_per_node_cpu)
+                     .to(dev);
+  auto start_idx = chunk_off_cpu.slice(0, 0, K)
+                       .clamp_max(total_chunks - 1)
+                       .to(dev);
+  auto bases = left_before.index_select(0, start_idx);  // [K]
+  left_before = left_before - bases.index_select(0, node_id);
+  auto end_idx =
+      (chunk_off_cpu.slice(0, 1, K + 1) - 1).clamp_min(0).to(dev);
+  auto totals = csum.index_select(0, end_idx) - bases;  // [K]
+  auto nonzero = (chunks_per_node_cpu > 0).to(dev);
+  auto node_left_total =
+      torch::where(nonzero, totals, torch::zeros_like(totals));
+  auto node_left_total_cpu = node_left_total.to(torch::kCPU);
   hipLaunchKernelGGL(partition_scatter_kernel, dim3((uint32_t)total_chunks),
                      dim3(PART_THREADS), 0, stream.stream(),
-                     bins.data_ptr<uint8_t>(), ridx.data_ptr<int32_t>(),
+                     flags.data_ptr<uint8_t>(), ridx.data_ptr<int32_t>(),
                      ridx_out.data_ptr<int32_t>(), sc.data_ptr<int64_t>(),
-                     chunk_off.data_ptr<int64_t>(), sf.data_ptr<int32_t>(),
-                     sb.data_ptr<int32_t>(), dl.data_ptr<uint8_t>(),
+                     chunk_off.data_ptr<int64_t>(),
                      left_before.data_ptr<int64_t>(),
-                     node_left_total.data_ptr<int64_t>(), K, F);
+                     node_left_total.data_ptr<int64_t>(), K);
   return {ridx_out, node_left_total_cpu};
 }
 

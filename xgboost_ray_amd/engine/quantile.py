@@ -287,7 +287,18 @@ class BinnedMatrix:
         obj.cuts = cuts
         obj.n_rows = n
         obj.n_features = acc.F
-        obj.bins = torch.empty((n, acc.F), dtype=torch.uint8, device=device)
+        if device.type == "cuda":
+            # padded 16B-aligned row stride (vectorized histogram path);
+            # pad bytes = 255 (missing) are never accumulated
+            F_pad = (acc.F + 15) // 16 * 16
+            full = torch.full(
+                (n, F_pad), 255, dtype=torch.uint8, device=device
+            )
+            obj.bins = full[:, : acc.F]
+        else:
+            obj.bins = torch.empty(
+                (n, acc.F), dtype=torch.uint8, device=device
+            )
         sides = {"label": [], "weight": [], "base_margin": [], "qid": []}
         pos = 0
         for chunk in chunk_fn():
