@@ -1,0 +1,15 @@
+#!/bin/bash
+# CI examples runner (reference run_ci_examples.sh:25-46).
+set -e
+
+pushd "$(dirname "$0")/examples" >/dev/null
+
+for ex in simple.py simple_predict.py simple_objectstore.py \
+          train_on_parquet.py readme_sklearn_api.py; do
+    echo "=== examples/$ex ==="
+    python "$ex"
+done
+python higgs.py --rows 100000 --actors 2
+
+popd >/dev/null
+echo "CI EXAMPLES PASSED"

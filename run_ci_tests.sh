@@ -1,0 +1,18 @@
+#!/bin/bash
+# CI test runner (reference run_ci_tests.sh:23-38 protocol:
+# per-file pytest -x + a benchmark smoke run).
+set -e
+
+pushd "$(dirname "$0")" >/dev/null
+
+for f in tests/test_*.py; do
+    echo "=== $f ==="
+    python -m pytest -x -q -m "not gpu" "$f"
+done
+
+echo "=== benchmark smoke ==="
+python benchmarks/benchmark_cpu_gpu.py 2 10 20 --smoke-test
+python benchmarks/benchmark_ft.py --smoke-test
+
+popd >/dev/null
+echo "CI TESTS PASSED"

@@ -174,19 +174,40 @@ __global__ __launch_bounds__(HIST_THREADS) void build_histogram_kernel(
     const longlong2 gp = gpair_seg[seg_i];
     const uint64_t r = (uint32_t)ridx[seg_i];
     if (VEC16) {
-      // fb_size == 16 and row base 16B-aligned by construction
+      // fb_size == 16 and row base 16B-aligned by construction.
+      // Word-granular lane rotation: lane l processes its row's feature
+      // words in order (l&3), (l&3)+1, ... so one wave-instruction's LDS
+      // atomics spread over 4 feature groups (4x fewer same-address
+      // serializations on skewed features). The rotation is done with
+      // branchless selects on NAMED registers - a runtime-indexed byte
+      // array would go to scratch (5x slowdown).
       const uint4 packed =
           *reinterpret_cast<const uint4*>(bins + r * row_stride + f0);
-      const uint8_t* bytes = reinterpret_cast<const uint8_t*>(&packed);
-      const int rot = lane & 15;
+      const int r4 = lane & 3;
+      const uint32_t w0 = packed.x, w1 = packed.y, w2 = packed.z,
+                     w3 = packed.w;
+      const bool s1 = (r4 & 1) != 0, s2 = (r4 & 2) != 0;
+      const uint32_t t01 = s1 ? w1 : w0, t23 = s1 ? w3 : w2;
+      const uint32_t u01 = s1 ? w2 : w1, u23 = s1 ? w0 : w3;
+      const uint32_t rw0 = s2 ? t23 : t01;  // w[(0+r4)&3]
+      const uint32_t rw1 = s2 ? u23 : u01;  // w[(1+r4)&3]
+      const uint32_t rw2 = s2 ? t01 : t23;  // w[(2+r4)&3]
+      const uint32_t rw3 = s2 ? u01 : u23;  // w[(3+r4)&3]
+      const uint32_t rws[4] = {rw0, rw1, rw2, rw3};
       #pragma unroll
-      for (int ff = 0; ff < 16; ++ff) {
-        const int f = (ff + rot) & 15;
-        const int b = bytes[f];
-        if (b != 255) {
-          unsigned long long* cell = &lds_hist[((size_t)f * n_bins + b) * 2];
-          atomicAdd(cell, (unsigned long long)gp.x);
-          atomicAdd(cell + 1, (unsigned long long)gp.y);
+      for (int jj = 0; jj < 4; ++jj) {
+        const uint32_t w = rws[jj];
+        const int fw = ((jj + r4) & 3) * 4;
+        #pragma unroll
+        for (int kk = 0; kk < 4; ++kk) {
+          const int b = (w >> (8 * kk)) & 0xFF;
+          if (b != 255) {
+            const int f = fw + kk;
+            unsigned long long* cell =
+                &lds_hist[((size_t)f * n_bins + b) * 2];
+            atomicAdd(cell, (unsigned long long)gp.x);
+            atomicAdd(cell + 1, (unsigned long long)gp.y);
+          }
         }
       }
     } else {

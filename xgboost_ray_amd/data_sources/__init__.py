@@ -10,19 +10,37 @@ multi-file) and ObjectStore sources.
 from xgboost_ray_amd.data_sources.data_source import DataSource, RayFileType
 from xgboost_ray_amd.data_sources.numpy import Numpy
 from xgboost_ray_amd.data_sources.pandas import Pandas
+from xgboost_ray_amd.data_sources.modin import Modin
+from xgboost_ray_amd.data_sources.dask import Dask
+from xgboost_ray_amd.data_sources.partitioned import Partitioned
 from xgboost_ray_amd.data_sources.csv import CSV
 from xgboost_ray_amd.data_sources.parquet import Parquet
+from xgboost_ray_amd.data_sources.petastorm import Petastorm
 from xgboost_ray_amd.data_sources.object_store import ObjectStore
 
-data_sources = [Numpy, Pandas, CSV, Parquet, ObjectStore]
+data_sources = [
+    Numpy,
+    Pandas,
+    Modin,
+    Dask,
+    Partitioned,
+    CSV,
+    Parquet,
+    Petastorm,
+    ObjectStore,
+]
 
 __all__ = [
     "DataSource",
     "RayFileType",
     "Numpy",
     "Pandas",
+    "Modin",
+    "Dask",
+    "Partitioned",
     "CSV",
     "Parquet",
+    "Petastorm",
     "ObjectStore",
     "data_sources",
 ]

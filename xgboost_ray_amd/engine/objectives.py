@@ -73,6 +73,9 @@ class Logistic(Objective):
     name = "binary:logistic"
     default_metric = "logloss"
 
+    def __init__(self, scale_pos_weight: float = 1.0):
+        self.scale_pos_weight = float(scale_pos_weight)
+
     def prob_to_margin(self, base_score):
         base_score = min(max(base_score, _EPS), 1.0 - _EPS)
         return -math.log(1.0 / base_score - 1.0)
@@ -81,6 +84,11 @@ class Logistic(Objective):
         p = torch.sigmoid(margin)
         g = p - label
         h = torch.clamp(p * (1.0 - p), min=_EPS)
+        if self.scale_pos_weight != 1.0:
+            # XGBoost semantics: positive rows weighted by scale_pos_weight
+            w = 1.0 + (self.scale_pos_weight - 1.0) * label
+            g = g * w
+            h = h * w
         return self._apply_weight(g, h, weight)
 
     def transform_prediction(self, margin):
@@ -275,7 +283,9 @@ _REGISTRY = {
 }
 
 
-def get_objective(name_or_fn, num_class: int = 0) -> Objective:
+def get_objective(
+    name_or_fn, num_class: int = 0, scale_pos_weight: float = 1.0
+) -> Objective:
     if callable(name_or_fn):
         return CustomObjective(name_or_fn, num_class)
     name = name_or_fn or "reg:squarederror"
@@ -285,4 +295,7 @@ def get_objective(name_or_fn, num_class: int = 0) -> Objective:
         return SoftProb(num_class)
     if name not in _REGISTRY:
         raise ValueError(f"Unsupported objective: {name}")
-    return _REGISTRY[name]()
+    cls = _REGISTRY[name]
+    if issubclass(cls, Logistic) and scale_pos_weight != 1.0:
+        return cls(scale_pos_weight)
+    return cls()

@@ -45,6 +45,7 @@ class TrainParams:
     subsample: float = 1.0
     colsample_bytree: float = 1.0
     colsample_bylevel: float = 1.0
+    scale_pos_weight: float = 1.0
     base_score: Optional[float] = None
     num_class: int = 0
     eval_metric: List[str] = field(default_factory=list)
@@ -143,7 +144,9 @@ class BoostingEngine:
         self.rank = rank
         self.device = dtrain.bins.device
         obj_spec = custom_objective or self.p.objective
-        self.obj: Objective = get_objective(obj_spec, self.p.num_class)
+        self.obj: Objective = get_objective(
+            obj_spec, self.p.num_class, float(self.p.scale_pos_weight)
+        )
         self.n_class = max(1, self.p.num_class)
         if self.p.base_score is None:
             self.p.base_score = 0.5
