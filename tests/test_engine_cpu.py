@@ -188,6 +188,31 @@ class TestTraining:
         )
         assert res["train"]["logloss"][-1] < res["train"]["logloss"][0]
 
+    def test_colsample_bylevel_and_feature_weights(self):
+        dm, X, y = _binned()
+        dm.feature_weights = torch.tensor([5.0, 5.0, 1.0, 1.0, 1.0, 1.0])
+        res = {}
+        run_training(
+            {"objective": "binary:logistic", "max_depth": 4, "eta": 0.3,
+             "colsample_bylevel": 0.5, "colsample_bytree": 0.8,
+             "eval_metric": ["logloss"]},
+            dm, 10, evals=[EvalPack(name="train", X=None)], evals_result=res,
+        )
+        assert res["train"]["logloss"][-1] < res["train"]["logloss"][0]
+
+    def test_scale_pos_weight(self):
+        dm, X, y = _binned()
+        res1, res2 = {}, {}
+        for spw, res in ((1.0, res1), (5.0, res2)):
+            run_training(
+                {"objective": "binary:logistic", "max_depth": 4, "eta": 0.3,
+                 "scale_pos_weight": spw, "eval_metric": ["auc"]},
+                dm, 5, evals=[EvalPack(name="train", X=None)],
+                evals_result=res,
+            )
+        # different weighting must change the model
+        assert res1["train"]["auc"] != res2["train"]["auc"]
+
     def test_early_stopping(self):
         X, y = create_data(2000, 6, 0, "binary")
         Xv, yv = create_data(500, 6, 7, "binary")

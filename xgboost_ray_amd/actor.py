@@ -173,6 +173,10 @@ class _ActorWorker:
                     seed=int(params.get("seed", 0) or 0),
                 )
                 del t_x
+                if shard.get("feature_weights") is not None:
+                    dm.feature_weights = _to_dev(
+                        shard["feature_weights"], self.device
+                    )
 
             eval_packs = []
             for uid, name in evals:

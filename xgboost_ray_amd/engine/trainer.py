@@ -277,20 +277,61 @@ class BoostingEngine:
         mask = torch.rand(n, generator=gen) < self.p.subsample
         return torch.nonzero(mask.to(self.device)).flatten().to(torch.int32)
 
-    def _sample_features(self, it: int, cls: int) -> Optional[torch.Tensor]:
+    def _sample_features(
+        self, it: int, cls: int, level: Optional[int] = None
+    ) -> Optional[torch.Tensor]:
+        """Feature mask for this tree (level=None) or depth level.
+
+        Seeded identically on every rank (global seed, no rank term) so
+        all workers mask the same features. feature_weights bias the
+        selection (xgboost semantics)."""
         F = self.dtrain.n_features
-        if self.p.colsample_bytree >= 1.0:
+        frac = (
+            self.p.colsample_bytree if level is None else self.p.colsample_bylevel
+        )
+        if frac >= 1.0:
             return None
-        k = max(1, int(round(F * self.p.colsample_bytree)))
+        k = max(1, int(round(F * frac)))
         gen = torch.Generator(device="cpu")
-        gen.manual_seed((self.p.seed * 2654435761 + it * 97 + cls * 31 + 7) % (2**63))
-        perm = torch.randperm(F, generator=gen)[:k]
+        salt = 7 if level is None else 1009 + level
+        gen.manual_seed(
+            (self.p.seed * 2654435761 + it * 97 + cls * 31 + salt) % (2**63)
+        )
+        fw = getattr(self.dtrain, "feature_weights", None)
+        if fw is not None:
+            w = fw.cpu().double().clamp(min=0)
+            if float(w.sum()) > 0:
+                perm = torch.multinomial(
+                    w, k, replacement=False, generator=gen
+                )
+            else:
+                perm = torch.randperm(F, generator=gen)[:k]
+        else:
+            perm = torch.randperm(F, generator=gen)[:k]
         mask = torch.zeros(F, dtype=torch.bool)
         mask[perm] = True
         return mask.to(self.device)
 
     def _grow_tree(self, gpair: torch.Tensor, it: int, cls: int) -> Tree:
+        import os as _os
+        import time as _time
+
+        _prof = _os.environ.get("RXGB_PROFILE_ENGINE")
+        _t = {}
+
+        def _tick(name, _last=[None]):
+            if not _prof:
+                return
+            if self.device.type == "cuda":
+                torch.cuda.synchronize()
+            now = _time.perf_counter()
+            if _last[0] is not None:
+                _t[name] = _t.get(name, 0.0) + now - _last[0]
+            _last[0] = now
+
+        _tick(None)
         gq, scale_g, scale_h = self._quantize(gpair)
+        _tick("quantize")
         self._scale_g_cur = scale_g
         self._leaf_segs = []
         ridx = self._sample_rows(it, cls)
@@ -357,8 +398,10 @@ class BoostingEngine:
             hist = ops.build_histogram(
                 self.dtrain.bins, gq, ridx, starts, counts, self.n_bins
             )
+            _tick("hist")
             if self.coll.is_distributed:
                 self.coll.allreduce_(hist)
+            _tick("allreduce")
             for k, nd in enumerate(build_nodes):
                 nd.hist = hist[k]
             for big, small in derive_nodes:
@@ -375,10 +418,15 @@ class BoostingEngine:
                 [nd.sum_h for nd in frontier], dtype=torch.int64, device=self.device
             )
             fb = self.feat_bins
-            if feat_mask is not None:
+            mask = feat_mask
+            if self.p.colsample_bylevel < 1.0:
+                lvl_mask = self._sample_features(it, cls, level=depth)
+                mask = lvl_mask if mask is None else (mask & lvl_mask)
+            if mask is not None:
                 fb = torch.where(
-                    feat_mask, self.feat_bins, torch.zeros_like(self.feat_bins)
+                    mask, self.feat_bins, torch.zeros_like(self.feat_bins)
                 )
+            _tick("stack")
             best = ops.find_splits(
                 all_hist,
                 pg,
@@ -391,7 +439,9 @@ class BoostingEngine:
                 self.p.gamma,
                 self.p.min_child_weight,
             )
+            _tick("scan")
             gain = best["gain"].cpu().numpy()
+            _tick("scan_pull")
             bfeat = best["feature"].cpu().numpy()
             bbin = best["bin"].cpu().numpy()
             bdl = best["default_left"].cpu().numpy()
@@ -446,6 +496,7 @@ class BoostingEngine:
             scounts = torch.tensor(
                 [nd.count for nd in split_nodes], dtype=torch.int64
             )
+            _tick("tree_host")
             ridx, left_counts = ops.partition_rows(
                 self.dtrain.bins,
                 ridx,
@@ -455,6 +506,7 @@ class BoostingEngine:
                 torch.tensor(sb, dtype=torch.int32),
                 torch.tensor(sdl, dtype=torch.uint8),
             )
+            _tick("partition")
 
             new_frontier: List[_Node] = []
             for k, nd in enumerate(split_nodes):
@@ -503,6 +555,9 @@ class BoostingEngine:
                 np.asarray(leaf_vals, dtype=np.float32),
             )
         self._leaf_segs = []
+        _tick("margins")
+        if _prof:
+            print("PROF", {k: round(v*1000, 2) for k, v in _t.items()}, flush=True)
 
         tree = Tree(
             feat=np.asarray(feat_l, np.int32),

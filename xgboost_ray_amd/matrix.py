@@ -293,6 +293,9 @@ class _CentralRayDMatrixLoader(_RayDMatrixLoader):
                     actor_refs[key] = val
                 elif val is None:
                     actor_refs[key] = None
+                elif key == "feature_weights":
+                    # per-FEATURE array: never row-sharded, sent inline
+                    actor_refs[key] = np.asarray(val)
                 else:
                     actor_refs[key] = shm_store.put(val[idx])
             refs[actor_rank] = actor_refs
