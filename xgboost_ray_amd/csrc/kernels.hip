@@ -289,16 +289,20 @@ __global__ __launch_bounds__(256) void find_splits_kf_kernel(
   if (lane != 0) return;
   const int nb = feat_bins[f];
 
-  const double Gp = (double)parent_g[k] / scale_g;
-  const double Hp = (double)parent_h[k] / scale_h;
+  // multiply by the reciprocal (computed identically on the CPU oracle):
+  // f64 division is ~10x a multiply and sits on the sequential scan path
+  const double inv_g = 1.0 / scale_g;
+  const double inv_h = 1.0 / scale_h;
+  const double Gp = (double)parent_g[k] * inv_g;
+  const double Hp = (double)parent_h[k] * inv_h;
   const double parent_score = calc_score(Gp, Hp, lam, alpha);
 
   // feature totals (same order as cumsum's last element)
   double Gtot = 0.0, Htot = 0.0;
   long long Gtot_q = 0, Htot_q = 0;
   for (int b = 0; b < B; ++b) {
-    Gtot += (double)h[b * 2] / scale_g;
-    Htot += (double)h[b * 2 + 1] / scale_h;
+    Gtot += (double)h[b * 2] * inv_g;
+    Htot += (double)h[b * 2 + 1] * inv_h;
     Gtot_q += h[b * 2];
     Htot_q += h[b * 2 + 1];
   }
@@ -313,8 +317,8 @@ __global__ __launch_bounds__(256) void find_splits_kf_kernel(
     double GL = 0.0, HL = 0.0;
     long long GLq = 0, HLq = 0;
     for (int b = 0; b < nb - 1 && b < B; ++b) {
-      GL += (double)h[b * 2] / scale_g;
-      HL += (double)h[b * 2 + 1] / scale_h;
+      GL += (double)h[b * 2] * inv_g;
+      HL += (double)h[b * 2 + 1] * inv_h;
       GLq += h[b * 2];
       HLq += h[b * 2 + 1];
       double gl = dl ? GL + Gmiss : GL;

@@ -90,8 +90,13 @@ def find_splits(
     dev = hist.device
     # Dequantize to double for the scan: scan math must be identical on CPU
     # and GPU, and the quantized ints are exactly representable in f64.
-    g = hist[..., 0].double() / scale_g  # [K, F, B]
-    h = hist[..., 1].double() / scale_h
+    # Multiply by the (identically computed) reciprocal instead of dividing:
+    # f64 division is ~10x the cost of multiply on CDNA4 and the GPU kernel
+    # mirrors this exact op order.
+    inv_g = 1.0 / scale_g
+    inv_h = 1.0 / scale_h
+    g = hist[..., 0].double() * inv_g  # [K, F, B]
+    h = hist[..., 1].double() * inv_h
     gq = hist[..., 0]
     hq = hist[..., 1]
 
@@ -99,8 +104,8 @@ def find_splits(
     HL = torch.cumsum(h, dim=2)
     GLq = torch.cumsum(gq, dim=2)
     HLq = torch.cumsum(hq, dim=2)
-    Gp = (parent_g.double() / scale_g).view(K, 1, 1)
-    Hp = (parent_h.double() / scale_h).view(K, 1, 1)
+    Gp = (parent_g.double() * inv_g).view(K, 1, 1)
+    Hp = (parent_h.double() * inv_h).view(K, 1, 1)
     # Missing mass per (node, feature) = parent - feature total.
     Gmiss = Gp - GL[:, :, -1:].clone()
     Hmiss = Hp - HL[:, :, -1:].clone()
