@@ -66,6 +66,12 @@ class _ActorWorker:
     def _setup_device(self):
         import torch
 
+        # intra-actor thread parallelism (reference _set_omp_num_threads,
+        # main.py:355-362): torch CPU threads = cpus_per_actor
+        cpus = int(self.config.get("cpus_per_actor", 0) or 0)
+        if cpus > 0:
+            torch.set_num_threads(cpus)
+            os.environ["OMP_NUM_THREADS"] = str(cpus)
         use_gpu = self.config.get("use_gpu", False)
         if use_gpu:
             if not torch.cuda.is_available():

@@ -152,6 +152,7 @@ def _create_actor(
     stop_event,
     use_gpu: bool,
     distributed_callbacks,
+    cpus_per_actor: int = 0,
 ) -> ActorHandle:
     handle = ActorHandle(
         rank=rank,
@@ -160,6 +161,7 @@ def _create_actor(
         stop_event=stop_event,
         config={
             "use_gpu": use_gpu,
+            "cpus_per_actor": cpus_per_actor,
             "distributed_callbacks": distributed_callbacks,
         },
         gpu_id=(rank % max(1, torch.cuda.device_count())) if use_gpu else None,
@@ -254,6 +256,7 @@ def _train(
             state.stop_event,
             use_gpu,
             ray_params.distributed_callbacks,
+            cpus_per_actor=ray_params.cpus_per_actor,
         )
         state.actors[rank] = actor
     # start in parallel threads (spawn + torch import takes seconds)
