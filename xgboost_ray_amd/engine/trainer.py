@@ -338,9 +338,13 @@ class BoostingEngine:
         feat_mask = self._sample_features(it, cls)
         n_local = int(ridx.numel())
 
-        root_sum = gq[ridx.long()].sum(dim=0) if n_local else torch.zeros(
-            2, dtype=torch.int64, device=self.device
-        )
+        if n_local == 0:
+            root_sum = torch.zeros(2, dtype=torch.int64, device=self.device)
+        elif n_local == self.dtrain.n_rows:
+            # no row sampling: ridx is the identity - skip the 16B/row gather
+            root_sum = gq.sum(dim=0)
+        else:
+            root_sum = gq[ridx.long()].sum(dim=0)
         if self.coll.is_distributed:
             self.coll.allreduce_(root_sum)
         root = _Node(
