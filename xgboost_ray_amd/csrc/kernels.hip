@@ -344,13 +344,16 @@ __global__ __launch_bounds__(256) void find_splits_kf_kernel(
   out_lh[kf] = best.left_h;
 }
 
+// packs the per-node best split into ONE int64 [K, 6] row:
+// (f32 gain bits, feature, bin, default_left, left_g, left_h) so the
+// driver retrieves the whole depth's splits with a single D2H copy
+// (dozens of tiny pageable copies per depth were the training loop's
+// dominant host-side cost).
 __global__ void find_splits_reduce_kernel(
     const double* __restrict__ kf_gain, const int32_t* __restrict__ kf_bin,
     const uint8_t* __restrict__ kf_dl, const long long* __restrict__ kf_lg,
     const long long* __restrict__ kf_lh,
-    float* __restrict__ out_gain, int32_t* __restrict__ out_feat,
-    int32_t* __restrict__ out_bin, uint8_t* __restrict__ out_dl,
-    long long* __restrict__ out_lg, long long* __restrict__ out_lh,
+    long long* __restrict__ out_packed,  // [K, 6]
     int K, int F) {
   int k = blockIdx.x * blockDim.x + threadIdx.x;
   if (k >= K) return;
@@ -371,12 +374,14 @@ __global__ void find_splits_reduce_kernel(
       }
     }
   }
-  out_gain[k] = (float)bg;
-  out_feat[k] = bf;
-  out_bin[k] = bb;
-  out_dl[k] = (uint8_t)bdl;
-  out_lg[k] = blg;
-  out_lh[k] = blh;
+  long long* row = out_packed + (size_t)k * 6;
+  float gf = (float)bg;
+  row[0] = (long long)(int)__float_as_int(gf);
+  row[1] = bf;
+  row[2] = bb;
+  row[3] = bdl;
+  row[4] = blg;
+  row[5] = blh;
 }
 
 // ---------------------------------------------------------------------------
@@ -403,8 +408,8 @@ __global__ void partition_count_kernel(
     const uint8_t* __restrict__ bins, const int32_t* __restrict__ ridx,
     const int64_t* __restrict__ node_start,  // [K] then counts at [K..2K)
     const int64_t* __restrict__ chunk_off,   // [K+1]
-    const int32_t* __restrict__ split_feat, const int32_t* __restrict__ split_bin,
-    const uint8_t* __restrict__ default_left,
+    const int64_t* __restrict__ split_feat, const int64_t* __restrict__ split_bin,
+    const int64_t* __restrict__ default_left,
     int32_t* __restrict__ block_counts,  // [total_chunks]
     uint8_t* __restrict__ flags,         // [n] go-left per segment position
     int K, int64_t row_stride) {
@@ -419,8 +424,8 @@ __global__ void partition_count_kernel(
   const int64_t row_lo = chunk_in_node * PART_THREADS;
   const int64_t count = node_start[K + node];
   const int64_t seg_start = node_start[node];
-  const int feat = split_feat[node], sbin = split_bin[node];
-  const int dl = default_left[node];
+  const int feat = (int)split_feat[node], sbin = (int)split_bin[node];
+  const int dl = (int)default_left[node];
 
   const int64_t i = row_lo + threadIdx.x;
   bool flag = false;
@@ -553,9 +558,9 @@ __global__ void update_margins_kernel(
 // Host-side launchers / bindings
 // ===========================================================================
 
-static torch::Tensor make_chunk_off(const torch::Tensor& counts_cpu,
-                                    int64_t rows_per_chunk,
-                                    torch::Device dev, int64_t* total) {
+static torch::Tensor make_chunk_off_cpu(const torch::Tensor& counts_cpu,
+                                        int64_t rows_per_chunk,
+                                        int64_t* total) {
   const int64_t K = counts_cpu.size(0);
   auto off = torch::zeros({K + 1}, torch::kInt64);
   auto acc = off.accessor<int64_t, 1>();
@@ -564,7 +569,7 @@ static torch::Tensor make_chunk_off(const torch::Tensor& counts_cpu,
     acc[k + 1] = acc[k] + (cacc[k] + rows_per_chunk - 1) / rows_per_chunk;
   }
   *total = acc[K];
-  return off.to(dev);
+  return off;
 }
 
 static torch::Tensor cat_start_count(const torch::Tensor& starts_cpu,
@@ -630,7 +635,8 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
   auto starts_cpu = starts.to(torch::kCPU).to(torch::kInt64);
   auto counts_cpu = counts.to(torch::kCPU).to(torch::kInt64);
   int64_t total_chunks = 0;
-  auto chunk_off = make_chunk_off(counts_cpu, HIST_ROWS_PER_WG, dev, &total_chunks);
+  auto chunk_off_cpu =
+      make_chunk_off_cpu(counts_cpu, HIST_ROWS_PER_WG, &total_chunks);
   if (total_chunks == 0) return hist;
 
   // gather gpairs into segment order once (coalesced hist reads)
@@ -656,9 +662,14 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                        ridx.data_ptr<int32_t>() + min_start,
                        (longlong2*)gpair_seg.data_ptr<int64_t>(), span);
   }
-  // adjust starts so kernels index gpair_seg with (start - min_start)
+  // ONE H2D copy for all control data: [starts_adj(K) | counts(K) |
+  // chunk_off(K+1)] - tiny pageable copies around kernel launches were
+  // the training loop's dominant host cost.
   auto starts_adj = starts_cpu - min_start;
-  auto sc_adj = cat_start_count(starts_adj, counts_cpu, dev);
+  auto meta = torch::cat({starts_adj, counts_cpu, chunk_off_cpu}).to(dev);
+  int64_t* mp = meta.data_ptr<int64_t>();
+  int64_t* sc_adj_p = mp;          // starts at [0..K), counts at [K..2K)
+  int64_t* chunk_off_p = mp + 2 * K;
 
   // feature-block size: fit the LDS tile (fb * n_bins * 16 B) within the
   // 64 KiB dynamic-LDS default so two workgroups co-reside per CU.
@@ -681,7 +692,7 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                        bins.data_ptr<uint8_t>(),
                        (const longlong2*)gpair_seg.data_ptr<int64_t>(),
                        ridx.data_ptr<int32_t>() + min_start,
-                       sc_adj.data_ptr<int64_t>(), chunk_off.data_ptr<int64_t>(),
+                       sc_adj_p, chunk_off_p,
                        reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
                        K, F, (int)n_bins, fb_size, row_stride);
   } else {
@@ -691,7 +702,7 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                        bins.data_ptr<uint8_t>(),
                        (const longlong2*)gpair_seg.data_ptr<int64_t>(),
                        ridx.data_ptr<int32_t>() + min_start,
-                       sc_adj.data_ptr<int64_t>(), chunk_off.data_ptr<int64_t>(),
+                       sc_adj_p, chunk_off_p,
                        reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
                        K, F, (int)n_bins, fb_size, row_stride);
   }
@@ -731,22 +742,15 @@ std::vector<torch::Tensor> find_splits(torch::Tensor hist, torch::Tensor parent_
                      kf_bin.data_ptr<int32_t>(), kf_dl.data_ptr<uint8_t>(),
                      reinterpret_cast<long long*>(kf_lg.data_ptr<int64_t>()),
                      reinterpret_cast<long long*>(kf_lh.data_ptr<int64_t>()), K, F, B);
-  auto out_gain = torch::empty({K}, torch::TensorOptions().dtype(torch::kFloat32).device(dev));
-  auto out_feat = torch::empty({K}, optsi);
-  auto out_bin = torch::empty({K}, optsi);
-  auto out_dl = torch::empty({K}, optsb);
-  auto out_lg = torch::empty({K}, optsl);
-  auto out_lh = torch::empty({K}, optsl);
+  auto out_packed = torch::empty({K, 6}, optsl);
   hipLaunchKernelGGL(find_splits_reduce_kernel, dim3((uint32_t)ceil_div(K, 64)),
                      dim3(64), 0, stream.stream(), kf_gain.data_ptr<double>(),
                      kf_bin.data_ptr<int32_t>(), kf_dl.data_ptr<uint8_t>(),
                      reinterpret_cast<const long long*>(kf_lg.data_ptr<int64_t>()),
                      reinterpret_cast<const long long*>(kf_lh.data_ptr<int64_t>()),
-                     out_gain.data_ptr<float>(), out_feat.data_ptr<int32_t>(),
-                     out_bin.data_ptr<int32_t>(), out_dl.data_ptr<uint8_t>(),
-                     reinterpret_cast<long long*>(out_lg.data_ptr<int64_t>()),
-                     reinterpret_cast<long long*>(out_lh.data_ptr<int64_t>()), K, F);
-  return {out_gain, out_feat, out_bin, out_dl, out_lg, out_lh};
+                     reinterpret_cast<long long*>(out_packed.data_ptr<int64_t>()),
+                     K, F);
+  return {out_packed};
 }
 
 std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx,
@@ -772,11 +776,19 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
     total_chunks = acc[K];
   }
   if (total_chunks == 0) return {ridx_out, left_counts};
-  auto chunk_off = chunk_off_cpu.to(dev);
-  auto sc = cat_start_count(starts_cpu, counts_cpu, dev);
-  auto sf = split_feat.to(dev).to(torch::kInt32);
-  auto sb = split_bin.to(dev).to(torch::kInt32);
-  auto dl = default_left.to(dev).to(torch::kUInt8);
+  // ONE H2D copy: [starts(K) | counts(K) | chunk_off(K+1) | feat(K) |
+  // bin(K) | default_left(K)]
+  auto meta = torch::cat({starts_cpu, counts_cpu, chunk_off_cpu,
+                          split_feat.to(torch::kCPU).to(torch::kInt64),
+                          split_bin.to(torch::kCPU).to(torch::kInt64),
+                          default_left.to(torch::kCPU).to(torch::kInt64)})
+                  .to(dev);
+  int64_t* mp = meta.data_ptr<int64_t>();
+  int64_t* sc_p = mp;
+  int64_t* chunk_off_p = mp + 2 * K;
+  int64_t* sf_p = mp + 3 * K + 1;
+  int64_t* sb_p = mp + 4 * K + 1;
+  int64_t* dl_p = mp + 5 * K + 1;
 
   auto stream = at::cuda::getCurrentCUDAStream();
   auto block_counts = torch::empty({total_chunks},
@@ -786,9 +798,8 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
   hipLaunchKernelGGL(partition_count_kernel, dim3((uint32_t)total_chunks),
                      dim3(PART_THREADS), 0, stream.stream(),
                      bins.data_ptr<uint8_t>(), ridx.data_ptr<int32_t>(),
-                     sc.data_ptr<int64_t>(), chunk_off.data_ptr<int64_t>(),
-                     sf.data_ptr<int32_t>(), sb.data_ptr<int32_t>(),
-                     dl.data_ptr<uint8_t>(), block_counts.data_ptr<int32_t>(),
+                     sc_p, chunk_off_p, sf_p, sb_p, dl_p,
+                     block_counts.data_ptr<int32_t>(),
                      flags.data_ptr<uint8_t>(), K, bins.stride(0));
   // per-node exclusive prefix of block counts (all on device: at 100M
   // rows this is ~400K chunks - far too many for a host loop)
@@ -815,8 +826,7 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
   hipLaunchKernelGGL(partition_scatter_kernel, dim3((uint32_t)total_chunks),
                      dim3(PART_THREADS), 0, stream.stream(),
                      flags.data_ptr<uint8_t>(), ridx.data_ptr<int32_t>(),
-                     ridx_out.data_ptr<int32_t>(), sc.data_ptr<int64_t>(),
-                     chunk_off.data_ptr<int64_t>(),
+                     ridx_out.data_ptr<int32_t>(), sc_p, chunk_off_p,
                      left_before.data_ptr<int64_t>(),
                      node_left_total.data_ptr<int64_t>(), K);
   return {ridx_out, node_left_total_cpu};
@@ -858,14 +868,14 @@ void update_margins(torch::Tensor margin, torch::Tensor ridx,
     total_chunks = acc[K];
   }
   if (total_chunks == 0) return;
-  auto sc = cat_start_count(starts_cpu, counts_cpu, dev);
-  auto chunk_off = chunk_off_cpu.to(dev);
+  auto meta = torch::cat({starts_cpu, counts_cpu, chunk_off_cpu}).to(dev);
+  int64_t* mp = meta.data_ptr<int64_t>();
   auto lv = leaf_vals.to(dev).to(torch::kFloat32);
   auto stream = at::cuda::getCurrentCUDAStream();
   hipLaunchKernelGGL(update_margins_kernel, dim3((uint32_t)total_chunks),
                      dim3(PART_THREADS), 0, stream.stream(),
                      margin.data_ptr<float>(), ridx.data_ptr<int32_t>(),
-                     sc.data_ptr<int64_t>(), chunk_off.data_ptr<int64_t>(),
+                     mp, mp + 2 * K,
                      lv.data_ptr<float>(), K);
 }
 

@@ -415,12 +415,15 @@ class BoostingEngine:
 
             # ---- split scan over the whole frontier
             all_hist = torch.stack([nd.hist for nd in frontier])
-            pg = torch.tensor(
-                [nd.sum_g for nd in frontier], dtype=torch.int64, device=self.device
-            )
-            ph = torch.tensor(
-                [nd.sum_h for nd in frontier], dtype=torch.int64, device=self.device
-            )
+            # ONE H2D for both parent-sum vectors
+            psums = torch.from_numpy(
+                np.array(
+                    [[nd.sum_g for nd in frontier],
+                     [nd.sum_h for nd in frontier]],
+                    dtype=np.int64,
+                )
+            ).to(self.device)
+            pg, ph = psums[0], psums[1]
             fb = self.feat_bins
             mask = feat_mask
             if self.p.colsample_bylevel < 1.0:
@@ -444,13 +447,13 @@ class BoostingEngine:
                 self.p.min_child_weight,
             )
             _tick("scan")
-            gain = best["gain"].cpu().numpy()
+            gain = best["gain"]
             _tick("scan_pull")
-            bfeat = best["feature"].cpu().numpy()
-            bbin = best["bin"].cpu().numpy()
-            bdl = best["default_left"].cpu().numpy()
-            blg = best["left_g"].cpu().numpy()
-            blh = best["left_h"].cpu().numpy()
+            bfeat = best["feature"]
+            bbin = best["bin"]
+            bdl = best["default_left"]
+            blg = best["left_g"]
+            blh = best["left_h"]
 
             split_nodes: List[_Node] = []
             sf, sb, sdl = [], [], []

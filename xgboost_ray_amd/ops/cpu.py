@@ -170,8 +170,17 @@ def find_splits(
         best["left_g"] = torch.where(upd, lgq, best["left_g"])
         best["left_h"] = torch.where(upd, lhq, best["left_h"])
 
-    best["gain"] = best["gain"].float()
-    return best
+    # contract: outputs are host numpy arrays (the driver consumes them on
+    # the host; returning numpy makes the GPU path's single packed D2H
+    # transfer natural)
+    return {
+        "gain": best["gain"].float().cpu().numpy(),
+        "feature": best["feature"].cpu().numpy(),
+        "bin": best["bin"].cpu().numpy(),
+        "default_left": best["default_left"].cpu().numpy(),
+        "left_g": best["left_g"].cpu().numpy(),
+        "left_h": best["left_h"].cpu().numpy(),
+    }
 
 
 def partition_rows(bins, ridx, starts, counts, split_feat, split_bin, default_left):

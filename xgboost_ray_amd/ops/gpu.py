@@ -69,8 +69,10 @@ def find_splits(
     gamma,
     min_child_weight,
 ):
+    import numpy as np
+
     dev = hist.device
-    out = _load().find_splits(
+    (packed,) = _load().find_splits(
         hist,
         parent_g.to(dev),
         parent_h.to(dev),
@@ -82,13 +84,14 @@ def find_splits(
         float(gamma),
         float(min_child_weight),
     )
+    arr = packed.cpu().numpy()  # ONE D2H for the whole depth's splits
     return {
-        "gain": out[0],
-        "feature": out[1],
-        "bin": out[2],
-        "default_left": out[3],
-        "left_g": out[4],
-        "left_h": out[5],
+        "gain": arr[:, 0].astype(np.int32).view(np.float32),
+        "feature": arr[:, 1].astype(np.int32),
+        "bin": arr[:, 2].astype(np.int32),
+        "default_left": arr[:, 3].astype(np.uint8),
+        "left_g": arr[:, 4].copy(),
+        "left_h": arr[:, 5].copy(),
     }
 
 
