@@ -98,8 +98,18 @@ class AUC(Metric):
 
     def local_stats(self, margin, label, weight, qid, obj):
         p = torch.sigmoid(margin.double())
-        w = _w(label, weight)
         b = torch.clamp((p * _AUC_BINS).long(), max=_AUC_BINS - 1)
+        if weight is None:
+            # unweighted: integer scatter (native u64 atomics on GPU -
+            # the f64 scatter goes through CAS loops and is ~50x slower)
+            ones = torch.ones_like(b)
+            ypos = label > 0.5
+            pos = torch.zeros(_AUC_BINS, dtype=torch.int64, device=margin.device)
+            neg = torch.zeros(_AUC_BINS, dtype=torch.int64, device=margin.device)
+            pos.scatter_add_(0, b[ypos], ones[ypos])
+            neg.scatter_add_(0, b[~ypos], ones[~ypos])
+            return torch.cat([pos, neg]).double()
+        w = _w(label, weight)
         pos = torch.zeros(_AUC_BINS, dtype=torch.float64, device=margin.device)
         neg = torch.zeros(_AUC_BINS, dtype=torch.float64, device=margin.device)
         y = label.double()
