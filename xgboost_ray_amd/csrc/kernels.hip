@@ -801,28 +801,30 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
                      sc_p, chunk_off_p, sf_p, sb_p, dl_p,
                      block_counts.data_ptr<int32_t>(),
                      flags.data_ptr<uint8_t>(), K, bins.stride(0));
-  // per-node exclusive prefix of block counts (all on device: at 100M
-  // rows this is ~400K chunks - far too many for a host loop)
-  auto bc64 = block_counts.to(torch::kInt64);
-  auto csum = torch::cumsum(bc64, 0);
-  auto left_before = csum - bc64;  // global exclusive prefix
-  auto chunks_per_node_cpu =
-      chunk_off_cpu.slice(0, 1, K + 1) - chunk_off_cpu.slice(0, 0, K);
-  auto node_id = torch::repeat_interleave(
-                     torch::arange(K, torch::kInt64), chunks_per_node_cpu)
-                     .to(dev);
-  auto start_idx = chunk_off_cpu.slice(0, 0, K)
-                       .clamp_max(total_chunks - 1)
-                       .to(dev);
-  auto bases = left_before.index_select(0, start_idx);  // [K]
-  left_before = left_before - bases.index_select(0, node_id);
-  auto end_idx =
-      (chunk_off_cpu.slice(0, 1, K + 1) - 1).clamp_min(0).to(dev);
-  auto totals = csum.index_select(0, end_idx) - bases;  // [K]
-  auto nonzero = (chunks_per_node_cpu > 0).to(dev);
-  auto node_left_total =
-      torch::where(nonzero, totals, torch::zeros_like(totals));
-  auto node_left_total_cpu = node_left_total.to(torch::kCPU);
+  // Per-node exclusive prefix of block counts. ONE D2H pull + a host
+  // loop + ONE H2D push: the torch-op chain this replaces (device
+  // cumsum + repeat_interleave + index_selects, ~8 tiny transfers and
+  // kernels per call) stalled the training loop ~15 ms per depth -
+  // measured 7x slower rounds end to end.
+  auto bc_cpu = block_counts.to(torch::kCPU);
+  auto left_before_cpu = torch::empty({total_chunks}, torch::kInt64);
+  auto node_left_total_cpu = torch::zeros({K}, torch::kInt64);
+  {
+    auto bc = bc_cpu.accessor<int32_t, 1>();
+    auto lb = left_before_cpu.accessor<int64_t, 1>();
+    auto nl = node_left_total_cpu.accessor<int64_t, 1>();
+    auto co = chunk_off_cpu.accessor<int64_t, 1>();
+    for (int k = 0; k < K; ++k) {
+      int64_t run = 0;
+      for (int64_t c = co[k]; c < co[k + 1]; ++c) {
+        lb[c] = run;
+        run += bc[c];
+      }
+      nl[k] = run;
+    }
+  }
+  auto left_before = left_before_cpu.to(dev);
+  auto node_left_total = node_left_total_cpu.to(dev);
   hipLaunchKernelGGL(partition_scatter_kernel, dim3((uint32_t)total_chunks),
                      dim3(PART_THREADS), 0, stream.stream(),
                      flags.data_ptr<uint8_t>(), ridx.data_ptr<int32_t>(),
