@@ -501,10 +501,33 @@ __global__ void partition_scatter_kernel(
 // ---------------------------------------------------------------------------
 // predict_trees: one thread per row, sequential over trees.
 // ---------------------------------------------------------------------------
+// Packed node: one 16-byte uint4 per node -> a tree-walk step is ONE
+// vector load (vs 4 scattered loads of feat/thr/left/default_left).
+// x = feature | (default_left << 30) | (is_leaf << 31)
+// y = threshold bits (or leaf value bits), z = left child, w = unused.
+__global__ void pack_nodes_kernel(
+    const int32_t* __restrict__ feat, const float* __restrict__ thr,
+    const int32_t* __restrict__ left, const uint8_t* __restrict__ default_left,
+    const float* __restrict__ value, uint4* __restrict__ packed, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  int f = feat[i];
+  uint4 p;
+  if (f < 0) {
+    p.x = 0x80000000u;
+    p.y = __float_as_uint(value[i]);
+    p.z = 0;
+  } else {
+    p.x = (uint32_t)f | ((default_left[i] != 0) ? 0x40000000u : 0u);
+    p.y = __float_as_uint(thr[i]);
+    p.z = (uint32_t)left[i];
+  }
+  p.w = 0;
+  packed[i] = p;
+}
+
 __global__ void predict_trees_kernel(
-    const float* __restrict__ X, const int32_t* __restrict__ feat,
-    const float* __restrict__ thr, const int32_t* __restrict__ left,
-    const uint8_t* __restrict__ default_left, const float* __restrict__ value,
+    const float* __restrict__ X, const uint4* __restrict__ nodes,
     const int32_t* __restrict__ tree_ptr, float* __restrict__ out,
     float tree_weight, int64_t n, int F, int T) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -513,16 +536,18 @@ __global__ void predict_trees_kernel(
     const float* row = X + i * (int64_t)F;
     float acc = 0.0f;
     for (int t = 0; t < T; ++t) {
-      int base = tree_ptr[t];
+      const int base = tree_ptr[t];
       int nid = base;
-      int f = feat[nid];
-      while (f >= 0) {
-        float v = row[f];
-        bool goleft = isnan(v) ? (default_left[nid] != 0) : (v < thr[nid]);
-        nid = base + (goleft ? left[nid] : left[nid] + 1);
-        f = feat[nid];
+      uint4 p = nodes[nid];
+      while (!(p.x & 0x80000000u)) {
+        const float v = row[p.x & 0x3FFFFFFFu];
+        const bool goleft =
+            isnan(v) ? ((p.x & 0x40000000u) != 0)
+                     : (v < __uint_as_float(p.y));
+        nid = base + (int)p.z + (goleft ? 0 : 1);
+        p = nodes[nid];
       }
-      acc += value[nid];
+      acc += __uint_as_float(p.y);
     }
     out[i] += acc * tree_weight;
   }
@@ -843,12 +868,20 @@ void predict_trees(torch::Tensor X, torch::Tensor feat, torch::Tensor thr,
   int T = (int)tree_ptr.size(0) - 1;
   if (n == 0 || T <= 0) return;
   auto stream = at::cuda::getCurrentCUDAStream();
+  int64_t n_nodes = feat.size(0);
+  auto packed = torch::empty({n_nodes, 4},
+      torch::TensorOptions().dtype(torch::kInt32).device(X.device()));
+  hipLaunchKernelGGL(pack_nodes_kernel, dim3((uint32_t)ceil_div(n_nodes, 256)),
+                     dim3(256), 0, stream.stream(),
+                     feat.data_ptr<int32_t>(), thr.data_ptr<float>(),
+                     left.data_ptr<int32_t>(), default_left.data_ptr<uint8_t>(),
+                     value.data_ptr<float>(),
+                     (uint4*)packed.data_ptr<int32_t>(), n_nodes);
   int64_t blocks = std::min<int64_t>(ceil_div(n, 256), 8192);
   hipLaunchKernelGGL(predict_trees_kernel, dim3(blocks), dim3(256), 0,
                      stream.stream(), X.data_ptr<float>(),
-                     feat.data_ptr<int32_t>(), thr.data_ptr<float>(),
-                     left.data_ptr<int32_t>(), default_left.data_ptr<uint8_t>(),
-                     value.data_ptr<float>(), tree_ptr.data_ptr<int32_t>(),
+                     (const uint4*)packed.data_ptr<int32_t>(),
+                     tree_ptr.data_ptr<int32_t>(),
                      out.data_ptr<float>(), (float)tree_weight, n, F, T);
 }
 
