@@ -107,7 +107,17 @@ class TestRF:
         for cls in (RayXGBRFClassifier, RayXGBRFRegressor):
             model = cls(n_estimators=5, max_depth=4)
             model.fit(X, y, ray_params=RP)
-            assert model.get_booster().num_boosted_rounds() == 5
+            bst = model.get_booster()
+            # RF: ONE boosting round of n_estimators parallel trees
+            assert bst.num_boosted_rounds() == 1
+            assert len(bst.trees) == 5
+        # averaged forest predictions stay in a sane range
+        reg = RayXGBRFRegressor(n_estimators=10, max_depth=4)
+        Xr, yr = create_data(1500, 5, kind="reg")
+        reg.fit(Xr, yr, ray_params=RP)
+        pred = reg.get_booster().predict(Xr)
+        assert abs(pred.mean() - yr.mean()) < 0.5
+        assert np.corrcoef(pred, yr)[0, 1] > 0.6
 
 
 class TestRanker:

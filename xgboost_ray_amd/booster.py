@@ -140,8 +140,12 @@ class Booster:
             return 0.5
         return float(bs)
 
+    @property
+    def num_parallel_tree(self) -> int:
+        return max(1, int(self.params.get("num_parallel_tree", 1) or 1))
+
     def num_boosted_rounds(self) -> int:
-        k = max(1, self.num_class)
+        k = max(1, self.num_class) * self.num_parallel_tree
         return len(self.trees) // k
 
     def append_round(self, trees: Sequence[Tree], classes: Sequence[int]):
@@ -213,6 +217,7 @@ class Booster:
         if iteration_range is not None:
             lo, hi = iteration_range
             hi = min(hi, self.num_boosted_rounds()) if hi else self.num_boosted_rounds()
+        per_round = k * self.num_parallel_tree
 
         outs = []
         for cls in range(k):
@@ -220,7 +225,7 @@ class Booster:
             sel = [
                 i
                 for i in range(len(self.trees))
-                if self.tree_info[i] == cls and lo <= i // k < hi
+                if self.tree_info[i] == cls and lo <= i // per_round < hi
             ]
             if sel:
                 # contiguous runs share ptr structure; do it per selected tree
@@ -354,7 +359,7 @@ class Booster:
                 }
             )
         num_rounds = self.num_boosted_rounds()
-        k = max(1, self.num_class)
+        k = max(1, self.num_class) * self.num_parallel_tree
         doc = {
             "learner": {
                 "attributes": dict(self.attributes_),
@@ -364,7 +369,7 @@ class Booster:
                     "model": {
                         "gbtree_model_param": {
                             "num_trees": str(len(self.trees)),
-                            "num_parallel_tree": "1",
+                            "num_parallel_tree": str(self.num_parallel_tree),
                         },
                         "iteration_indptr": [i * k for i in range(num_rounds + 1)],
                         "tree_info": list(self.tree_info),

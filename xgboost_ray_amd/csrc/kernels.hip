@@ -579,6 +579,67 @@ __global__ void update_margins_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// lambdarank_grad: pairwise LambdaRank gradients, one workgroup per query
+// group; each thread owns docs i = tid, tid+256, ... and loops all other
+// docs j sequentially, so per-doc accumulation order is fixed ->
+// deterministic across runs (no atomics). NDCG variant weights each pair
+// by |delta NDCG| from precomputed ranks + idcg.
+// ---------------------------------------------------------------------------
+__global__ void lambdarank_kernel(
+    const float* __restrict__ margin, const float* __restrict__ label,
+    const int64_t* __restrict__ group_ptr,  // [G+1]
+    const int32_t* __restrict__ rank,       // [n] rank within group
+    const double* __restrict__ idcg,        // [G] (ndcg mode) or nullptr
+    float2* __restrict__ out,               // [n] (grad, hess)
+    int use_ndcg) {
+  const int g = blockIdx.x;
+  const int64_t s0 = group_ptr[g], s1 = group_ptr[g + 1];
+  const int len = (int)(s1 - s0);
+  const double inv_idcg =
+      (use_ndcg && idcg[g] > 0.0) ? 1.0 / idcg[g] : 0.0;
+  if (use_ndcg && inv_idcg == 0.0) {
+    for (int i = threadIdx.x; i < len; i += blockDim.x)
+      out[s0 + i] = make_float2(0.f, 0.f);
+    return;
+  }
+  for (int i = threadIdx.x; i < len; i += blockDim.x) {
+    const float yi = label[s0 + i];
+    const float mi = margin[s0 + i];
+    const double gain_i = use_ndcg ? (exp2((double)yi) - 1.0) : 0.0;
+    const double disc_i =
+        use_ndcg ? 1.0 / log2((double)rank[s0 + i] + 2.0) : 0.0;
+    double gacc = 0.0, hacc = 0.0;
+    for (int j = 0; j < len; ++j) {
+      const float yj = label[s0 + j];
+      if (yj == yi) continue;
+      const float mj = margin[s0 + j];
+      double w = 1.0;
+      if (use_ndcg) {
+        const double gain_j = exp2((double)yj) - 1.0;
+        const double disc_j = 1.0 / log2((double)rank[s0 + j] + 2.0);
+        w = fabs((gain_i - gain_j) * (disc_i - disc_j)) * inv_idcg;
+      }
+      if (yi > yj) {
+        // i should rank above j: pair (i, j)
+        const double rho = 1.0 / (1.0 + exp((double)(mi - mj)));
+        double hij = rho * (1.0 - rho);
+        if (hij < 1e-16) hij = 1e-16;
+        gacc += -rho * w;
+        hacc += hij * w;
+      } else {
+        // j should rank above i: pair (j, i), i is the loser
+        const double rho = 1.0 / (1.0 + exp((double)(mj - mi)));
+        double hij = rho * (1.0 - rho);
+        if (hij < 1e-16) hij = 1e-16;
+        gacc += rho * w;
+        hacc += hij * w;
+      }
+    }
+    out[s0 + i] = make_float2((float)gacc, (float)hacc);
+  }
+}
+
 // ===========================================================================
 // Host-side launchers / bindings
 // ===========================================================================
@@ -914,6 +975,24 @@ void update_margins(torch::Tensor margin, torch::Tensor ridx,
                      lv.data_ptr<float>(), K);
 }
 
+torch::Tensor lambdarank_grad(torch::Tensor margin, torch::Tensor label,
+                              torch::Tensor group_ptr, torch::Tensor rank,
+                              torch::Tensor idcg, bool use_ndcg) {
+  const int64_t n = margin.size(0);
+  const int G = (int)group_ptr.size(0) - 1;
+  auto out = torch::zeros({n, 2},
+      torch::TensorOptions().dtype(torch::kFloat32).device(margin.device()));
+  if (n == 0 || G <= 0) return out;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(lambdarank_kernel, dim3(G), dim3(256), 0,
+                     stream.stream(), margin.data_ptr<float>(),
+                     label.data_ptr<float>(), group_ptr.data_ptr<int64_t>(),
+                     rank.data_ptr<int32_t>(),
+                     use_ndcg ? idcg.data_ptr<double>() : nullptr,
+                     (float2*)out.data_ptr<float>(), use_ndcg ? 1 : 0);
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("quantize_gpair", &quantize_gpair, "quantize gradient pairs");
   m.def("bin_matrix", &bin_matrix, "bin feature matrix");
@@ -922,4 +1001,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("partition_rows", &partition_rows, "stable row partition");
   m.def("predict_trees", &predict_trees, "tree-walk prediction");
   m.def("update_margins", &update_margins, "leaf margin update");
+  m.def("lambdarank_grad", &lambdarank_grad, "pairwise lambdarank gradients");
 }

@@ -173,6 +173,8 @@ class LambdaRankBase(Objective):
         if qid is None:
             raise ValueError("rank objectives require qid/group information")
         device = margin.device
+        if device.type == "cuda":
+            return self._gradients_gpu(margin, label, weight, qid)
         g = torch.zeros_like(margin)
         h = torch.zeros_like(margin)
         qid = qid.to(device)
@@ -219,6 +221,58 @@ class LambdaRankBase(Objective):
             g[s:e] += -lam.sum(dim=1) + lam.sum(dim=0)
             h[s:e] += hess.sum(dim=1) + hess.sum(dim=0)
         return self._apply_weight(g, h, weight)
+
+    def _gradients_gpu(self, margin, label, weight, qid):
+        """HIP kernel path: one workgroup per query group, deterministic
+        per-doc accumulation (no atomics)."""
+        from xgboost_ray_amd.ops import gpu as gpu_ops
+
+        device = margin.device
+        qid = qid.to(device)
+        n = margin.shape[0]
+        change = torch.ones(n, dtype=torch.bool, device=device)
+        change[1:] = qid[1:] != qid[:-1]
+        group_id = torch.cumsum(change.long(), 0) - 1  # [n]
+        counts = torch.bincount(group_id)
+        group_ptr = torch.zeros(
+            len(counts) + 1, dtype=torch.int64, device=device
+        )
+        torch.cumsum(counts, 0, out=group_ptr[1:])
+
+        if self.ndcg_weighting:
+            # rank within group by descending margin: stable double argsort
+            order = torch.argsort(margin, descending=True, stable=True)
+            perm2 = torch.argsort(group_id[order], stable=True)
+            final = order[perm2]  # docs ordered by (group, margin desc)
+            pos = torch.empty(n, dtype=torch.int64, device=device)
+            pos[final] = torch.arange(n, device=device)
+            rank = (pos - group_ptr[group_id]).to(torch.int32)
+            # idcg per group from ideal (label-sorted) ordering
+            lorder = torch.argsort(label, descending=True, stable=True)
+            lperm2 = torch.argsort(group_id[lorder], stable=True)
+            lfinal = lorder[lperm2]
+            lpos = torch.empty(n, dtype=torch.int64, device=device)
+            lpos[lfinal] = torch.arange(n, device=device)
+            lrank = (lpos - group_ptr[group_id]).double()
+            gains = torch.pow(2.0, label.double()) - 1.0
+            disc = 1.0 / torch.log2(lrank + 2.0)
+            idcg = torch.zeros(len(counts), dtype=torch.float64, device=device)
+            idcg.scatter_add_(0, group_id, gains * disc)
+        else:
+            rank = torch.zeros(n, dtype=torch.int32, device=device)
+            idcg = torch.zeros(1, dtype=torch.float64, device=device)
+
+        gpair = gpu_ops.lambdarank_grad(
+            margin.contiguous().float(),
+            label.contiguous().float(),
+            group_ptr,
+            rank,
+            idcg,
+            self.ndcg_weighting,
+        )
+        if weight is not None:
+            gpair = gpair * weight.unsqueeze(1)
+        return gpair
 
 
 class RankPairwise(LambdaRankBase):

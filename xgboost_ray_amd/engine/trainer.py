@@ -43,6 +43,7 @@ class TrainParams:
     gamma: float = 0.0
     min_child_weight: float = 1.0
     subsample: float = 1.0
+    num_parallel_tree: int = 1
     colsample_bytree: float = 1.0
     colsample_bylevel: float = 1.0
     scale_pos_weight: float = 1.0
@@ -78,6 +79,7 @@ class TrainParams:
         p.max_bin = int(p.max_bin)
         p.num_class = int(p.num_class or 0)
         p.seed = int(p.seed or 0)
+        p.num_parallel_tree = max(1, int(p.num_parallel_tree or 1))
         return p
 
 
@@ -173,6 +175,7 @@ class BoostingEngine:
                 "num_feature": dtrain.n_features,
                 "max_depth": self.p.max_depth,
                 "eta": self.p.eta,
+                "num_parallel_tree": self.p.num_parallel_tree,
             }
         )
 
@@ -247,9 +250,15 @@ class BoostingEngine:
         trees, classes = [], []
         for cls in range(self.n_class):
             gp = gpair if self.n_class == 1 else gpair[:, cls, :]
-            tree = self._grow_tree(gp.contiguous(), it, cls)
-            trees.append(tree)
-            classes.append(cls)
+            # num_parallel_tree > 1: boosted random forest - k trees per
+            # round on the same gradients with independent row/column
+            # samples (XGBoost GBTree::BoostNewTrees semantics)
+            for ptree in range(self.p.num_parallel_tree):
+                tree = self._grow_tree(
+                    gp.contiguous(), it, cls, ptree=ptree
+                )
+                trees.append(tree)
+                classes.append(cls)
         self.booster.append_round(trees, classes)
         self.iteration += 1
         return trees
@@ -312,7 +321,9 @@ class BoostingEngine:
         mask[perm] = True
         return mask.to(self.device)
 
-    def _grow_tree(self, gpair: torch.Tensor, it: int, cls: int) -> Tree:
+    def _grow_tree(
+        self, gpair: torch.Tensor, it: int, cls: int, ptree: int = 0
+    ) -> Tree:
         import os as _os
         import time as _time
 
@@ -334,8 +345,8 @@ class BoostingEngine:
         _tick("quantize")
         self._scale_g_cur = scale_g
         self._leaf_segs = []
-        ridx = self._sample_rows(it, cls)
-        feat_mask = self._sample_features(it, cls)
+        ridx = self._sample_rows(it, cls + 101 * ptree)
+        feat_mask = self._sample_features(it, cls + 101 * ptree)
         n_local = int(ridx.numel())
 
         if n_local == 0:

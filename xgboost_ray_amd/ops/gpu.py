@@ -127,6 +127,12 @@ def predict_trees(X, feat, thr, left, default_left, value, tree_ptr, out, tree_w
     return out
 
 
+def lambdarank_grad(margin, label, group_ptr, rank, idcg, use_ndcg):
+    return _load().lambdarank_grad(
+        margin, label, group_ptr, rank, idcg, bool(use_ndcg)
+    )
+
+
 def update_margins(margin, ridx, starts, counts, leaf_values):
     dev = margin.device
     lv = torch.as_tensor(leaf_values, dtype=torch.float32, device=dev)

@@ -280,7 +280,7 @@ class _RayXGBModel(RayXGBMixin):
         self._Booster = train(
             params,
             train_dmatrix,
-            num_boost_round=self.n_estimators,
+            num_boost_round=self._num_boost_round(),
             evals=evals,
             evals_result=evals_result,
             ray_params=ray_params,
@@ -299,6 +299,9 @@ class _RayXGBModel(RayXGBMixin):
             self.best_iteration = self._Booster.best_iteration
             self.best_score = self._Booster.best_score
         return self
+
+    def _num_boost_round(self):
+        return self.n_estimators
 
     @property
     def feature_importances_(self):
@@ -347,7 +350,12 @@ class RayXGBRegressor(_RayXGBModel):
 
 
 class RayXGBRFRegressor(RayXGBRegressor):
-    """Random-forest-style regressor (reference sklearn.py:602-640)."""
+    """Random-forest regressor (reference sklearn.py:602-640).
+
+    XGBoost RF semantics: ONE boosting round of
+    ``num_parallel_tree = n_estimators`` trees, each grown on the same
+    gradients with independent row/column samples.
+    """
 
     def __init__(
         self, learning_rate=1.0, subsample=0.8, colsample_bytree=0.8,
@@ -357,6 +365,16 @@ class RayXGBRFRegressor(RayXGBRegressor):
             learning_rate=learning_rate, subsample=subsample,
             colsample_bytree=colsample_bytree, reg_lambda=reg_lambda, **kwargs,
         )
+
+    def get_xgb_params(self):
+        params = super().get_xgb_params()
+        params["num_parallel_tree"] = self.n_estimators
+        # average the forest: leaf weights scaled by 1/n_trees
+        params["eta"] = float(params.get("eta", 1.0)) / self.n_estimators
+        return params
+
+    def _num_boost_round(self):
+        return 1
 
 
 class RayXGBClassifier(_RayXGBModel):
@@ -442,7 +460,8 @@ class RayXGBClassifier(_RayXGBModel):
 
 
 class RayXGBRFClassifier(RayXGBClassifier):
-    """Random-forest-style classifier (reference sklearn.py:880-917)."""
+    """Random-forest classifier (reference sklearn.py:880-917): one round
+    of ``num_parallel_tree = n_estimators`` trees."""
 
     def __init__(
         self, learning_rate=1.0, subsample=0.8, colsample_bytree=0.8,
@@ -452,6 +471,15 @@ class RayXGBRFClassifier(RayXGBClassifier):
             learning_rate=learning_rate, subsample=subsample,
             colsample_bytree=colsample_bytree, reg_lambda=reg_lambda, **kwargs,
         )
+
+    def get_xgb_params(self):
+        params = super().get_xgb_params()
+        params["num_parallel_tree"] = self.n_estimators
+        params["eta"] = float(params.get("eta", 1.0)) / self.n_estimators
+        return params
+
+    def _num_boost_round(self):
+        return 1
 
 
 class RayXGBRanker(_RayXGBModel):
