@@ -36,17 +36,20 @@ static inline int64_t ceil_div(int64_t a, int64_t b) { return (a + b - 1) / b; }
 // ---------------------------------------------------------------------------
 // quantize_gpair: [n,2] f32 -> [n,2] i64 (round to nearest)
 // ---------------------------------------------------------------------------
+// int32 pairs: |q| <= 2^30 by scale construction, so int32 holds each
+// row exactly and int64 accumulators hold any sum - half the gradient
+// traffic of int64 pairs at identical numerics.
 __global__ void quantize_gpair_kernel(const float2* __restrict__ gpair,
-                                      longlong2* __restrict__ out,
+                                      int2* __restrict__ out,
                                       double scale_g, double scale_h,
                                       int64_t n) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (; i < n; i += stride) {
     float2 gp = gpair[i];
-    longlong2 q;
-    q.x = llrint((double)gp.x * scale_g);
-    q.y = llrint((double)gp.y * scale_h);
+    int2 q;
+    q.x = (int)llrint((double)gp.x * scale_g);
+    q.y = (int)llrint((double)gp.y * scale_h);
     out[i] = q;
   }
 }
@@ -105,9 +108,9 @@ __global__ void bin_matrix_kernel(const float* __restrict__ values,
 // (coalesced writes; the gather read hits L2/L3). Done once per depth so
 // the histogram kernel reads gradients coalesced per feature block.
 // ---------------------------------------------------------------------------
-__global__ void gather_gpair_kernel(const longlong2* __restrict__ gpair,
+__global__ void gather_gpair_kernel(const int2* __restrict__ gpair,
                                     const int32_t* __restrict__ ridx,
-                                    longlong2* __restrict__ out, int64_t n) {
+                                    int2* __restrict__ out, int64_t n) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (; i < n; i += stride) out[i] = gpair[(int64_t)(uint32_t)ridx[i]];
@@ -133,7 +136,7 @@ __global__ void gather_gpair_kernel(const longlong2* __restrict__ gpair,
 template <bool VEC16>
 __global__ __launch_bounds__(HIST_THREADS) void build_histogram_kernel(
     const uint8_t* __restrict__ bins,        // [n_rows_total, row_stride]
-    const longlong2* __restrict__ gpair_seg, // [seg_total] segment order
+    const int2* __restrict__ gpair_seg,      // [seg_total] segment order
     const int32_t* __restrict__ ridx,        // [seg_total]
     const int64_t* __restrict__ node_start,  // [K] segment starts
     const int64_t* __restrict__ chunk_off,   // [K+1] cumulative chunks
@@ -171,7 +174,8 @@ __global__ __launch_bounds__(HIST_THREADS) void build_histogram_kernel(
   const int lane = threadIdx.x & (WAVE - 1);
   for (int64_t i = row_lo + threadIdx.x; i < row_hi; i += blockDim.x) {
     const int64_t seg_i = seg_start + i;
-    const longlong2 gp = gpair_seg[seg_i];
+    const int2 gpi = gpair_seg[seg_i];
+    const longlong2 gp = {(long long)gpi.x, (long long)gpi.y};
     const uint64_t r = (uint32_t)ridx[seg_i];
     if (VEC16) {
       // fb_size == 16 and row base 16B-aligned by construction.
@@ -667,14 +671,14 @@ static torch::Tensor cat_start_count(const torch::Tensor& starts_cpu,
 
 torch::Tensor quantize_gpair(torch::Tensor gpair, double scale_g, double scale_h) {
   TORCH_CHECK(gpair.is_cuda() && gpair.dtype() == torch::kFloat32);
-  auto out = torch::empty_like(gpair, gpair.options().dtype(torch::kInt64));
+  auto out = torch::empty_like(gpair, gpair.options().dtype(torch::kInt32));
   int64_t n = gpair.size(0);
   if (n == 0) return out;
   auto stream = at::cuda::getCurrentCUDAStream();
   int64_t blocks = std::min<int64_t>(ceil_div(n, 256), 8192);
   hipLaunchKernelGGL(quantize_gpair_kernel, dim3(blocks), dim3(256), 0,
                      stream.stream(), (const float2*)gpair.data_ptr<float>(),
-                     (longlong2*)out.data_ptr<int64_t>(), scale_g, scale_h, n);
+                     (int2*)out.data_ptr<int32_t>(), scale_g, scale_h, n);
   return out;
 }
 
@@ -738,15 +742,17 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
   if (min_start == INT64_MAX) { min_start = 0; max_end = 0; }
   auto stream = at::cuda::getCurrentCUDAStream();
   int64_t span = max_end - min_start;
+  TORCH_CHECK(gpair_q.dtype() == torch::kInt32,
+              "gpair_q must be int32 packed pairs");
   auto gpair_seg = torch::empty({std::max<int64_t>(span, 1), 2},
                                 gpair_q.options());
   if (span > 0) {
     int64_t blocks = std::min<int64_t>(ceil_div(span, 256), 8192);
     hipLaunchKernelGGL(gather_gpair_kernel, dim3(blocks), dim3(256), 0,
                        stream.stream(),
-                       (const longlong2*)gpair_q.data_ptr<int64_t>(),
+                       (const int2*)gpair_q.data_ptr<int32_t>(),
                        ridx.data_ptr<int32_t>() + min_start,
-                       (longlong2*)gpair_seg.data_ptr<int64_t>(), span);
+                       (int2*)gpair_seg.data_ptr<int32_t>(), span);
   }
   // ONE H2D copy for all control data: [starts_adj(K) | counts(K) |
   // chunk_off(K+1)] - tiny pageable copies around kernel launches were
@@ -776,7 +782,7 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                        dim3((uint32_t)total_chunks, n_fb),
                        dim3(HIST_THREADS), lds, stream.stream(),
                        bins.data_ptr<uint8_t>(),
-                       (const longlong2*)gpair_seg.data_ptr<int64_t>(),
+                       (const int2*)gpair_seg.data_ptr<int32_t>(),
                        ridx.data_ptr<int32_t>() + min_start,
                        sc_adj_p, chunk_off_p,
                        reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
@@ -786,7 +792,7 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                        dim3((uint32_t)total_chunks, n_fb),
                        dim3(HIST_THREADS), lds, stream.stream(),
                        bins.data_ptr<uint8_t>(),
-                       (const longlong2*)gpair_seg.data_ptr<int64_t>(),
+                       (const int2*)gpair_seg.data_ptr<int32_t>(),
                        ridx.data_ptr<int32_t>() + min_start,
                        sc_adj_p, chunk_off_p,
                        reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),

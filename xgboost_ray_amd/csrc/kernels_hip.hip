@@ -36,17 +36,20 @@ static inline int64_t ceil_div(int64_t a, int64_t b) { return (a + b - 1) / b; }
 // ---------------------------------------------------------------------------
 // quantize_gpair: [n,2] f32 -> [n,2] i64 (round to nearest)
 // ---------------------------------------------------------------------------
+// int32 pairs: |q| <= 2^30 by scale construction, so int32 holds each
+// row exactly and int64 accumulators hold any sum - half the gradient
+// traffic of int64 pairs at identical numerics.
 __global__ void quantize_gpair_kernel(const float2* __restrict__ gpair,
-                                      longlong2* __restrict__ out,
+                                      int2* __restrict__ out,
                                       double scale_g, double scale_h,
                                       int64_t n) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (; i < n; i += stride) {
     float2 gp = gpair[i];
-    longlong2 q;
-    q.x = llrint((double)gp.x * scale_g);
-    q.y = llrint((double)gp.y * scale_h);
+    int2 q;
+    q.x = (int)llrint((double)gp.x * scale_g);
+    q.y = (int)llrint((double)gp.y * scale_h);
     out[i] = q;
   }
 }
@@ -105,9 +108,9 @@ __global__ void bin_matrix_kernel(const float* __restrict__ values,
 // (coalesced writes; the gather read hits L2/L3). Done once per depth so
 // the histogram kernel reads gradients coalesced per feature block.
 // ---------------------------------------------------------------------------
-__global__ void gather_gpair_kernel(const longlong2* __restrict__ gpair,
+__global__ void gather_gpair_kernel(const int2* __restrict__ gpair,
                                     const int32_t* __restrict__ ridx,
-                                    longlong2* __restrict__ out, int64_t n) {
+                                    int2* __restrict__ out, int64_t n) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (; i < n; i += stride) out[i] = gpair[(int64_t)(uint32_t)ridx[i]];
@@ -133,7 +136,7 @@ __global__ void gather_gpair_kernel(const longlong2* __restrict__ gpair,
 template <bool VEC16>
 __global__ __launch_bounds__(HIST_THREADS) void build_histogram_kernel(
     const uint8_t* __restrict__ bins,        // [n_rows_total, row_stride]
-    const longlong2* __restrict__ gpair_seg, // [seg_total] segment order
+    const int2* __restrict__ gpair_seg,      // [seg_total] segment order
     const int32_t* __restrict__ ridx,        // [seg_total]
     const int64_t* __restrict__ node_start,  // [K] segment starts
     const int64_t* __restrict__ chunk_off,   // [K+1] cumulative chunks
@@ -171,7 +174,8 @@ __global__ __launch_bounds__(HIST_THREADS) void build_histogram_kernel(
   const int lane = threadIdx.x & (WAVE - 1);
   for (int64_t i = row_lo + threadIdx.x; i < row_hi; i += blockDim.x) {
     const int64_t seg_i = seg_start + i;
-    const longlong2 gp = gpair_seg[seg_i];
+    const int2 gpi = gpair_seg[seg_i];
+    const longlong2 gp = {(long long)gpi.x, (long long)gpi.y};
     const uint64_t r = (uint32_t)ridx[seg_i];
     if (VEC16) {
       // fb_size == 16 and row base 16B-aligned by construction.
@@ -579,6 +583,67 @@ __global__ void update_margins_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// lambdarank_grad: pairwise LambdaRank gradients, one workgroup per query
+// group; each thread owns docs i = tid, tid+256, ... and loops all other
+// docs j sequentially, so per-doc accumulation order is fixed ->
+// deterministic across runs (no atomics). NDCG variant weights each pair
+// by |delta NDCG| from precomputed ranks + idcg.
+// ---------------------------------------------------------------------------
+__global__ void lambdarank_kernel(
+    const float* __restrict__ margin, const float* __restrict__ label,
+    const int64_t* __restrict__ group_ptr,  // [G+1]
+    const int32_t* __restrict__ rank,       // [n] rank within group
+    const double* __restrict__ idcg,        // [G] (ndcg mode) or nullptr
+    float2* __restrict__ out,               // [n] (grad, hess)
+    int use_ndcg) {
+  const int g = blockIdx.x;
+  const int64_t s0 = group_ptr[g], s1 = group_ptr[g + 1];
+  const int len = (int)(s1 - s0);
+  const double inv_idcg =
+      (use_ndcg && idcg[g] > 0.0) ? 1.0 / idcg[g] : 0.0;
+  if (use_ndcg && inv_idcg == 0.0) {
+    for (int i = threadIdx.x; i < len; i += blockDim.x)
+      out[s0 + i] = make_float2(0.f, 0.f);
+    return;
+  }
+  for (int i = threadIdx.x; i < len; i += blockDim.x) {
+    const float yi = label[s0 + i];
+    const float mi = margin[s0 + i];
+    const double gain_i = use_ndcg ? (exp2((double)yi) - 1.0) : 0.0;
+    const double disc_i =
+        use_ndcg ? 1.0 / log2((double)rank[s0 + i] + 2.0) : 0.0;
+    double gacc = 0.0, hacc = 0.0;
+    for (int j = 0; j < len; ++j) {
+      const float yj = label[s0 + j];
+      if (yj == yi) continue;
+      const float mj = margin[s0 + j];
+      double w = 1.0;
+      if (use_ndcg) {
+        const double gain_j = exp2((double)yj) - 1.0;
+        const double disc_j = 1.0 / log2((double)rank[s0 + j] + 2.0);
+        w = fabs((gain_i - gain_j) * (disc_i - disc_j)) * inv_idcg;
+      }
+      if (yi > yj) {
+        // i should rank above j: pair (i, j)
+        const double rho = 1.0 / (1.0 + exp((double)(mi - mj)));
+        double hij = rho * (1.0 - rho);
+        if (hij < 1e-16) hij = 1e-16;
+        gacc += -rho * w;
+        hacc += hij * w;
+      } else {
+        // j should rank above i: pair (j, i), i is the loser
+        const double rho = 1.0 / (1.0 + exp((double)(mj - mi)));
+        double hij = rho * (1.0 - rho);
+        if (hij < 1e-16) hij = 1e-16;
+        gacc += rho * w;
+        hacc += hij * w;
+      }
+    }
+    out[s0 + i] = make_float2((float)gacc, (float)hacc);
+  }
+}
+
 // ===========================================================================
 // Host-side launchers / bindings
 // ===========================================================================
@@ -606,14 +671,14 @@ static torch::Tensor cat_start_count(const torch::Tensor& starts_cpu,
 
 torch::Tensor quantize_gpair(torch::Tensor gpair, double scale_g, double scale_h) {
   TORCH_CHECK(gpair.is_cuda() && gpair.dtype() == torch::kFloat32);
-  auto out = torch::empty_like(gpair, gpair.options().dtype(torch::kInt64));
+  auto out = torch::empty_like(gpair, gpair.options().dtype(torch::kInt32));
   int64_t n = gpair.size(0);
   if (n == 0) return out;
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   int64_t blocks = std::min<int64_t>(ceil_div(n, 256), 8192);
   hipLaunchKernelGGL(quantize_gpair_kernel, dim3(blocks), dim3(256), 0,
                      stream.stream(), (const float2*)gpair.data_ptr<float>(),
-                     (longlong2*)out.data_ptr<int64_t>(), scale_g, scale_h, n);
+                     (int2*)out.data_ptr<int32_t>(), scale_g, scale_h, n);
   return out;
 }
 
@@ -677,15 +742,17 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
   if (min_start == INT64_MAX) { min_start = 0; max_end = 0; }
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   int64_t span = max_end - min_start;
+  TORCH_CHECK(gpair_q.dtype() == torch::kInt32,
+              "gpair_q must be int32 packed pairs");
   auto gpair_seg = torch::empty({std::max<int64_t>(span, 1), 2},
                                 gpair_q.options());
   if (span > 0) {
     int64_t blocks = std::min<int64_t>(ceil_div(span, 256), 8192);
     hipLaunchKernelGGL(gather_gpair_kernel, dim3(blocks), dim3(256), 0,
                        stream.stream(),
-                       (const longlong2*)gpair_q.data_ptr<int64_t>(),
+                       (const int2*)gpair_q.data_ptr<int32_t>(),
                        ridx.data_ptr<int32_t>() + min_start,
-                       (longlong2*)gpair_seg.data_ptr<int64_t>(), span);
+                       (int2*)gpair_seg.data_ptr<int32_t>(), span);
   }
   // ONE H2D copy for all control data: [starts_adj(K) | counts(K) |
   // chunk_off(K+1)] - tiny pageable copies around kernel launches were
@@ -715,7 +782,7 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                        dim3((uint32_t)total_chunks, n_fb),
                        dim3(HIST_THREADS), lds, stream.stream(),
                        bins.data_ptr<uint8_t>(),
-                       (const longlong2*)gpair_seg.data_ptr<int64_t>(),
+                       (const int2*)gpair_seg.data_ptr<int32_t>(),
                        ridx.data_ptr<int32_t>() + min_start,
                        sc_adj_p, chunk_off_p,
                        reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
@@ -725,7 +792,7 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                        dim3((uint32_t)total_chunks, n_fb),
                        dim3(HIST_THREADS), lds, stream.stream(),
                        bins.data_ptr<uint8_t>(),
-                       (const longlong2*)gpair_seg.data_ptr<int64_t>(),
+                       (const int2*)gpair_seg.data_ptr<int32_t>(),
                        ridx.data_ptr<int32_t>() + min_start,
                        sc_adj_p, chunk_off_p,
                        reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
@@ -914,6 +981,24 @@ void update_margins(torch::Tensor margin, torch::Tensor ridx,
                      lv.data_ptr<float>(), K);
 }
 
+torch::Tensor lambdarank_grad(torch::Tensor margin, torch::Tensor label,
+                              torch::Tensor group_ptr, torch::Tensor rank,
+                              torch::Tensor idcg, bool use_ndcg) {
+  const int64_t n = margin.size(0);
+  const int G = (int)group_ptr.size(0) - 1;
+  auto out = torch::zeros({n, 2},
+      torch::TensorOptions().dtype(torch::kFloat32).device(margin.device()));
+  if (n == 0 || G <= 0) return out;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  hipLaunchKernelGGL(lambdarank_kernel, dim3(G), dim3(256), 0,
+                     stream.stream(), margin.data_ptr<float>(),
+                     label.data_ptr<float>(), group_ptr.data_ptr<int64_t>(),
+                     rank.data_ptr<int32_t>(),
+                     use_ndcg ? idcg.data_ptr<double>() : nullptr,
+                     (float2*)out.data_ptr<float>(), use_ndcg ? 1 : 0);
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("quantize_gpair", &quantize_gpair, "quantize gradient pairs");
   m.def("bin_matrix", &bin_matrix, "bin feature matrix");
@@ -922,4 +1007,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("partition_rows", &partition_rows, "stable row partition");
   m.def("predict_trees", &predict_trees, "tree-walk prediction");
   m.def("update_margins", &update_margins, "leaf margin update");
+  m.def("lambdarank_grad", &lambdarank_grad, "pairwise lambdarank gradients");
 }

@@ -352,10 +352,10 @@ class BoostingEngine:
         if n_local == 0:
             root_sum = torch.zeros(2, dtype=torch.int64, device=self.device)
         elif n_local == self.dtrain.n_rows:
-            # no row sampling: ridx is the identity - skip the 16B/row gather
-            root_sum = gq.sum(dim=0)
+            # no row sampling: ridx is the identity - skip the per-row gather
+            root_sum = gq.sum(dim=0, dtype=torch.int64)
         else:
-            root_sum = gq[ridx.long()].sum(dim=0)
+            root_sum = gq[ridx.long()].sum(dim=0, dtype=torch.int64)
         if self.coll.is_distributed:
             self.coll.allreduce_(root_sum)
         root = _Node(

@@ -12,9 +12,11 @@ MISSING_BIN = 255
 
 
 def quantize_gpair(gpair: torch.Tensor, scale_g: float, scale_h: float) -> torch.Tensor:
-    out = torch.empty(gpair.shape, dtype=torch.int64, device=gpair.device)
-    out[:, 0] = torch.round(gpair[:, 0].double() * scale_g).to(torch.int64)
-    out[:, 1] = torch.round(gpair[:, 1].double() * scale_h).to(torch.int64)
+    # int32 packed pairs: |q| <= 2^30 by scale construction; accumulators
+    # are int64 so any sum is exact
+    out = torch.empty(gpair.shape, dtype=torch.int32, device=gpair.device)
+    out[:, 0] = torch.round(gpair[:, 0].double() * scale_g).to(torch.int32)
+    out[:, 1] = torch.round(gpair[:, 1].double() * scale_h).to(torch.int32)
     return out
 
 
@@ -57,8 +59,8 @@ def build_histogram(bins, gpair_q, ridx, starts, counts, n_bins):
         rb = bins[idx].long()  # [c, F]
         valid = rb != MISSING_BIN
         flat = foff.unsqueeze(0) + rb  # [c, F]
-        g = gpair_q[idx, 0].unsqueeze(1).expand_as(flat)
-        h = gpair_q[idx, 1].unsqueeze(1).expand_as(flat)
+        g = gpair_q[idx, 0].long().unsqueeze(1).expand_as(flat)
+        h = gpair_q[idx, 1].long().unsqueeze(1).expand_as(flat)
         hk = hist[k].view(F * n_bins, 2)
         fidx = flat[valid]
         hk[:, 0].scatter_add_(0, fidx, g[valid])
