@@ -423,6 +423,42 @@ class Booster:
             self.tree_info = [0] * len(self.trees)
         self._flat_cache = None
 
+    def get_score(self, fmap: str = "", importance_type: str = "weight"):
+        """Per-feature importance (xgboost Booster.get_score parity).
+
+        importance_type: weight (split count), gain, total_gain, cover,
+        total_cover.
+        """
+        counts: Dict[str, float] = {}
+        gains: Dict[str, float] = {}
+        covers: Dict[str, float] = {}
+
+        def fname(f):
+            if self.feature_names and f < len(self.feature_names):
+                return self.feature_names[f]
+            return f"f{f}"
+
+        for t in self.trees:
+            for nid in range(t.num_nodes):
+                f = int(t.feat[nid])
+                if f < 0:
+                    continue
+                key = fname(f)
+                counts[key] = counts.get(key, 0.0) + 1.0
+                gains[key] = gains.get(key, 0.0) + float(t.gain[nid])
+                covers[key] = covers.get(key, 0.0) + float(t.cover[nid])
+        if importance_type == "weight":
+            return counts
+        if importance_type == "total_gain":
+            return gains
+        if importance_type == "gain":
+            return {k: v / counts[k] for k, v in gains.items()}
+        if importance_type == "total_cover":
+            return covers
+        if importance_type == "cover":
+            return {k: v / counts[k] for k, v in covers.items()}
+        raise ValueError(f"Unknown importance_type: {importance_type}")
+
     def get_dump(self, fmap="", with_stats=False, dump_format="text"):
         out = []
         for t in self.trees:

@@ -17,6 +17,7 @@ from xgboost_ray_amd.data_sources.csv import CSV
 from xgboost_ray_amd.data_sources.parquet import Parquet
 from xgboost_ray_amd.data_sources.petastorm import Petastorm
 from xgboost_ray_amd.data_sources.object_store import ObjectStore
+from xgboost_ray_amd.data_sources.ray_dataset import RayDataset
 
 data_sources = [
     Numpy,
@@ -27,6 +28,7 @@ data_sources = [
     CSV,
     Parquet,
     Petastorm,
+    RayDataset,
     ObjectStore,
 ]
 

@@ -44,6 +44,11 @@ class RayXGBoostTrainingError(TrainingError):
     pass
 
 
+# reference-parity aliases (reference main.py error classes)
+RayActorError = ActorError
+RayXGBoostTrainingStopped = TrainingStoppedError
+
+
 class RayXGBoostActorAvailable(RuntimeError):
     """A replacement actor became available during elastic training
     (reference elastic.py:136-142)."""
