@@ -2,7 +2,9 @@
 # CI examples runner (reference run_ci_examples.sh:25-46).
 set -e
 
-pushd "$(dirname "$0")/examples" >/dev/null
+ROOT="$(cd "$(dirname "$0")" && pwd)"
+export PYTHONPATH="$ROOT:${PYTHONPATH:-}"
+pushd "$ROOT/examples" >/dev/null
 
 for ex in simple.py simple_predict.py simple_objectstore.py \
           train_on_parquet.py readme_sklearn_api.py; do

@@ -3,7 +3,9 @@
 # per-file pytest -x + a benchmark smoke run).
 set -e
 
-pushd "$(dirname "$0")" >/dev/null
+ROOT="$(cd "$(dirname "$0")" && pwd)"
+export PYTHONPATH="$ROOT:${PYTHONPATH:-}"
+pushd "$ROOT" >/dev/null
 
 for f in tests/test_*.py; do
     echo "=== $f ==="

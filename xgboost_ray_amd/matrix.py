@@ -509,6 +509,10 @@ class RayDMatrix:
         return shard
 
     def unload_data(self):
+        for actor_refs in self.refs.values():
+            for val in actor_refs.values():
+                if isinstance(val, shm_store.ObjectRef):
+                    shm_store.get_store().free(val)
         self.refs = {}
         self.loaded = False
 

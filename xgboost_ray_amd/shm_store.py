@@ -87,6 +87,18 @@ class ShmStore:
                 pass
         self._attached = {}
 
+    def free(self, ref: ObjectRef):
+        """Unlink one owned segment (matrix unload_data path)."""
+        for i, shm in enumerate(self._owned):
+            if shm.name == ref.shm_name:
+                try:
+                    shm.close()
+                    shm.unlink()
+                except Exception:
+                    pass
+                del self._owned[i]
+                return
+
     def shutdown(self):
         """Owner-side cleanup: unlink all segments this store created."""
         self.detach()
@@ -106,6 +118,9 @@ def get_store() -> ShmStore:
     global _GLOBAL_STORE
     if _GLOBAL_STORE is None:
         _GLOBAL_STORE = ShmStore()
+        import atexit
+
+        atexit.register(_GLOBAL_STORE.shutdown)
     return _GLOBAL_STORE
 
 
