@@ -132,6 +132,31 @@ def test_build_histogram_padded_exact(binned):
     torch.testing.assert_close(ref_ridx, out_ridx.cpu(), rtol=0, atol=0)
 
 
+def test_build_histogram_feature_ranges(binned):
+    """Chunked (f_range) builds into one output == single full build."""
+    gpu = _gpu_ops()
+    Xt, yt, cuts, bins_cpu = binned
+    n, F = bins_cpu.shape
+    F_pad = (F + 15) // 16 * 16
+    full = torch.full((n, F_pad), 255, dtype=torch.uint8, device="cuda")
+    full[:, :F] = bins_cpu.cuda()
+    bins_padded = full[:, :F]
+    gp = torch.stack([yt - 0.3, torch.rand(n) + 0.1], dim=1).float()
+    gq = cpu_ops.quantize_gpair(gp, 2.0**28, 2.0**27).cuda()
+    ridx = torch.arange(n, dtype=torch.int32, device="cuda")
+    starts, counts = _segments(n, 3)
+    ref = gpu.build_histogram(
+        bins_padded, gq, ridx, starts, counts, cuts.max_bins
+    )
+    out = torch.zeros_like(ref)
+    for f0, f1 in ((0, 16), (16, F)):
+        gpu.build_histogram(
+            bins_padded, gq, ridx, starts, counts, cuts.max_bins,
+            f_range=(f0, f1), out=out,
+        )
+    torch.testing.assert_close(ref, out, rtol=0, atol=0)
+
+
 def test_find_splits_bitwise(binned):
     gpu = _gpu_ops()
     Xt, yt, cuts, bins_cpu = binned

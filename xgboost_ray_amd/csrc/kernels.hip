@@ -141,7 +141,7 @@ __global__ __launch_bounds__(HIST_THREADS) void build_histogram_kernel(
     const int64_t* __restrict__ node_start,  // [K] segment starts
     const int64_t* __restrict__ chunk_off,   // [K+1] cumulative chunks
     long long* __restrict__ hist,            // [K, F, n_bins, 2]
-    int K, int F, int n_bins, int fb_size, int64_t row_stride) {
+    int K, int F, int n_bins, int fb_size, int64_t row_stride, int f_base) {
   // locate (node, chunk) from blockIdx.x via binary search on chunk_off
   int wg = blockIdx.x;
   int lo = 0, hi = K;
@@ -156,7 +156,7 @@ __global__ __launch_bounds__(HIST_THREADS) void build_histogram_kernel(
   const int64_t row_lo = chunk_in_node * HIST_ROWS_PER_WG;
 
   const int fb = blockIdx.y;
-  const int f0 = fb * fb_size;
+  const int f0 = f_base + fb * fb_size;
   const int fcount = fb_size < (F - f0) ? fb_size : (F - f0);
 
   extern __shared__ unsigned long long lds_hist[];  // [fb_size][n_bins][2]
@@ -714,13 +714,16 @@ torch::Tensor bin_matrix(torch::Tensor values, torch::Tensor cuts_flat,
 
 torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                               torch::Tensor ridx, torch::Tensor starts,
-                              torch::Tensor counts, int64_t n_bins) {
+                              torch::Tensor counts, int64_t n_bins,
+                              int64_t f_lo, int64_t f_hi,
+                              torch::Tensor hist) {
+  // builds features [f_lo, f_hi) into the caller-provided [K, F, n_bins, 2]
+  // histogram (zeroed by the caller); ranges let the driver overlap the
+  // RCCL AllReduce of one feature block with the build of the next.
   TORCH_CHECK(bins.is_cuda() && bins.dtype() == torch::kUInt8);
   const int K = (int)starts.size(0);
   const int F = (int)bins.size(1);
   auto dev = bins.device();
-  auto hist = torch::zeros({K, F, n_bins, 2},
-                           torch::TensorOptions().dtype(torch::kInt64).device(dev));
   if (K == 0) return hist;
   auto starts_cpu = starts.to(torch::kCPU).to(torch::kInt64);
   auto counts_cpu = counts.to(torch::kCPU).to(torch::kInt64);
@@ -773,7 +776,9 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                     ? 16
                     : (int)std::min<int64_t>(F, (64 * 1024) / (n_bins * 16));
   if (fb_size < 1) fb_size = 1;
-  const int n_fb = (int)ceil_div(F, fb_size);
+  TORCH_CHECK(f_lo % fb_size == 0 && f_lo < f_hi && f_hi <= F,
+              "feature range must align to the block size ", fb_size);
+  const int n_fb = (int)ceil_div(f_hi - f_lo, fb_size);
   const size_t lds = (size_t)fb_size * n_bins * 2 * sizeof(long long);
 
   // ridx pointer offset so seg indices align with gpair_seg
@@ -786,7 +791,7 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                        ridx.data_ptr<int32_t>() + min_start,
                        sc_adj_p, chunk_off_p,
                        reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
-                       K, F, (int)n_bins, fb_size, row_stride);
+                       K, F, (int)n_bins, fb_size, row_stride, (int)f_lo);
   } else {
     hipLaunchKernelGGL((build_histogram_kernel<false>),
                        dim3((uint32_t)total_chunks, n_fb),
@@ -796,7 +801,7 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                        ridx.data_ptr<int32_t>() + min_start,
                        sc_adj_p, chunk_off_p,
                        reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
-                       K, F, (int)n_bins, fb_size, row_stride);
+                       K, F, (int)n_bins, fb_size, row_stride, (int)f_lo);
   }
   return hist;
 }

@@ -410,12 +410,45 @@ class BoostingEngine:
             counts = torch.tensor(
                 [nd.count for nd in build_nodes], dtype=torch.int64
             )
-            hist = ops.build_histogram(
-                self.dtrain.bins, gq, ridx, starts, counts, self.n_bins
+            F = self.dtrain.n_features
+            overlap = (
+                self.coll.is_distributed
+                and self.device.type == "cuda"
+                and self.dtrain.bins.stride(0) % 16 == 0
+                and F > 16
             )
-            _tick("hist")
-            if self.coll.is_distributed:
-                self.coll.allreduce_(hist)
+            if overlap:
+                # overlap the RCCL AllReduce of each finished feature block
+                # with the build of the next (BASELINE north-star: side-
+                # stream collective over xGMI behind the histogram build)
+                hist = torch.zeros(
+                    (K, F, self.n_bins, 2),
+                    dtype=torch.int64,
+                    device=self.device,
+                )
+                n_chunks = min(4, (F + 31) // 32)
+                step = ((F + n_chunks - 1) // n_chunks + 15) // 16 * 16
+                pending = []
+                f0 = 0
+                while f0 < F:
+                    f1 = min(f0 + step, F)
+                    ops.build_histogram(
+                        self.dtrain.bins, gq, ridx, starts, counts,
+                        self.n_bins, f_range=(f0, f1), out=hist,
+                    )
+                    sl = hist[:, f0:f1].contiguous()
+                    pending.append((self.coll.allreduce_async(sl), sl, f0, f1))
+                    f0 = f1
+                for h, sl, c0, c1 in pending:
+                    h.wait()
+                    hist[:, c0:c1].copy_(sl)
+            else:
+                hist = ops.build_histogram(
+                    self.dtrain.bins, gq, ridx, starts, counts, self.n_bins
+                )
+                _tick("hist")
+                if self.coll.is_distributed:
+                    self.coll.allreduce_(hist)
             _tick("allreduce")
             for k, nd in enumerate(build_nodes):
                 nd.hist = hist[k]

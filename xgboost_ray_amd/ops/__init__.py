@@ -62,16 +62,22 @@ def bin_matrix(values, cuts_flat, cut_ptr):
     return _impl(values).bin_matrix(values, cuts_flat, cut_ptr)
 
 
-def build_histogram(bins, gpair_q, ridx, starts, counts, n_bins):
+def build_histogram(bins, gpair_q, ridx, starts, counts, n_bins,
+                    f_range=None, out=None):
     """Accumulate per-(node, feature, bin) int64 gradient-pair histograms.
 
-    bins: uint8 [n, F]; gpair_q: int64 [n, 2]; ridx: int32 [n] row index
-    array partitioned into contiguous per-node segments; starts/counts:
-    int64 [K] segment descriptors for the K nodes to build.
+    bins: uint8 [n, F]; gpair_q: int32 [n, 2] packed pairs; ridx: int32 [n]
+    row index array partitioned into contiguous per-node segments;
+    starts/counts: int64 [K] segment descriptors for the K nodes to build.
+    f_range=(f_lo, f_hi) builds only that feature slice (into ``out``,
+    which the caller allocates zeroed) so the driver can overlap the
+    AllReduce of one block with the build of the next.
     Returns int64 [K, F, n_bins, 2]. Missing values (bin==MISSING_BIN) are
     skipped; their mass is recovered as node_total - feature_sum.
     """
-    return _impl(bins).build_histogram(bins, gpair_q, ridx, starts, counts, n_bins)
+    return _impl(bins).build_histogram(
+        bins, gpair_q, ridx, starts, counts, n_bins, f_range, out
+    )
 
 
 def find_splits(

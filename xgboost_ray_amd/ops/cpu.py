@@ -46,17 +46,32 @@ def bin_matrix(values: torch.Tensor, cuts_flat: torch.Tensor, cut_ptr: torch.Ten
     return out
 
 
-def build_histogram(bins, gpair_q, ridx, starts, counts, n_bins):
+def build_histogram(bins, gpair_q, ridx, starts, counts, n_bins,
+                    f_range=None, out=None):
     K = len(starts)
     n, F = bins.shape
-    hist = torch.zeros((K, F, n_bins, 2), dtype=torch.int64, device=bins.device)
-    foff = torch.arange(F, dtype=torch.int64) * n_bins
+    hist = out
+    if hist is None:
+        hist = torch.zeros(
+            (K, F, n_bins, 2), dtype=torch.int64, device=bins.device
+        )
+    f_lo, f_hi = (0, F) if f_range is None else f_range
+    return _build_histogram_range(
+        bins, gpair_q, ridx, starts, counts, n_bins, f_lo, f_hi, hist
+    )
+
+
+def _build_histogram_range(bins, gpair_q, ridx, starts, counts, n_bins,
+                           f_lo, f_hi, hist):
+    K = len(starts)
+    n, F = bins.shape
+    foff = torch.arange(f_lo, f_hi, dtype=torch.int64) * n_bins
     for k in range(K):
         s, c = int(starts[k]), int(counts[k])
         if c == 0:
             continue
         idx = ridx[s : s + c].long()
-        rb = bins[idx].long()  # [c, F]
+        rb = bins[idx][:, f_lo:f_hi].long()  # [c, fr]
         valid = rb != MISSING_BIN
         flat = foff.unsqueeze(0) + rb  # [c, F]
         g = gpair_q[idx, 0].long().unsqueeze(1).expand_as(flat)

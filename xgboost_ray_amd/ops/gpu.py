@@ -45,15 +45,25 @@ def bin_matrix(values, cuts_flat, cut_ptr):
     )
 
 
-def build_histogram(bins, gpair_q, ridx, starts, counts, n_bins):
-    dev = bins.device
+def build_histogram(bins, gpair_q, ridx, starts, counts, n_bins,
+                    f_range=None, out=None):
+    K = len(starts)
+    F = bins.shape[1]
+    if out is None:
+        out = torch.zeros(
+            (K, F, int(n_bins), 2), dtype=torch.int64, device=bins.device
+        )
+    f_lo, f_hi = (0, F) if f_range is None else f_range
     return _load().build_histogram(
         bins,
         gpair_q,
         ridx,
-        starts.to(dev),
-        counts.to(dev),
+        starts,
+        counts,
         int(n_bins),
+        int(f_lo),
+        int(f_hi),
+        out,
     )
 
 
