@@ -132,26 +132,28 @@ def test_build_histogram_padded_exact(binned):
     torch.testing.assert_close(ref_ridx, out_ridx.cpu(), rtol=0, atol=0)
 
 
-def test_build_histogram_feature_ranges(binned):
+def test_build_histogram_feature_ranges():
     """Chunked (f_range) builds into one output == single full build."""
     gpu = _gpu_ops()
-    Xt, yt, cuts, bins_cpu = binned
-    n, F = bins_cpu.shape
+    torch.manual_seed(3)
+    n, F = 50_000, 40
     F_pad = (F + 15) // 16 * 16
     full = torch.full((n, F_pad), 255, dtype=torch.uint8, device="cuda")
-    full[:, :F] = bins_cpu.cuda()
+    full[:, :F] = torch.randint(
+        0, 200, (n, F), dtype=torch.uint8, device="cuda"
+    )
     bins_padded = full[:, :F]
-    gp = torch.stack([yt - 0.3, torch.rand(n) + 0.1], dim=1).float()
+    gp = torch.stack(
+        [torch.randn(n), torch.rand(n) + 0.1], dim=1
+    ).float()
     gq = cpu_ops.quantize_gpair(gp, 2.0**28, 2.0**27).cuda()
     ridx = torch.arange(n, dtype=torch.int32, device="cuda")
     starts, counts = _segments(n, 3)
-    ref = gpu.build_histogram(
-        bins_padded, gq, ridx, starts, counts, cuts.max_bins
-    )
+    ref = gpu.build_histogram(bins_padded, gq, ridx, starts, counts, 256)
     out = torch.zeros_like(ref)
-    for f0, f1 in ((0, 16), (16, F)):
+    for f0, f1 in ((0, 16), (16, 32), (32, F)):
         gpu.build_histogram(
-            bins_padded, gq, ridx, starts, counts, cuts.max_bins,
+            bins_padded, gq, ridx, starts, counts, 256,
             f_range=(f0, f1), out=out,
         )
     torch.testing.assert_close(ref, out, rtol=0, atol=0)
