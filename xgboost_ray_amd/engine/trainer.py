@@ -42,6 +42,7 @@ class TrainParams:
     reg_alpha: float = 0.0
     gamma: float = 0.0
     min_child_weight: float = 1.0
+    max_delta_step: float = 0.0
     subsample: float = 1.0
     num_parallel_tree: int = 1
     colsample_bytree: float = 1.0
@@ -630,6 +631,8 @@ class BoostingEngine:
         G = nd.sum_g / self._scale_g_cur
         H = nd.sum_h / scale_h
         w = _calc_weight(G, H, self.p.reg_lambda, self.p.reg_alpha)
+        if self.p.max_delta_step > 0:
+            w = max(-self.p.max_delta_step, min(self.p.max_delta_step, w))
         val_l[nd.nid] = self.p.eta * w
         cover_l[nd.nid] = H
         self._leaf_segs.append((nd.start, nd.count, self.p.eta * w))
