@@ -125,7 +125,7 @@ __global__ void gather_gpair_kernel(const int2* __restrict__ gpair,
 // atomics. Integer accumulation => order-independent, bitwise
 // deterministic (the checkpoint-determinism contract).
 // ---------------------------------------------------------------------------
-#define HIST_THREADS 256
+#define HIST_THREADS 512
 #define HIST_ROWS_PER_WG 16384
 
 // VEC16: the binned matrix row stride is 16-byte aligned (padded layout,
@@ -277,7 +277,13 @@ __global__ __launch_bounds__(256) void find_splits_kf_kernel(
     long long* __restrict__ out_lg,    // [K, F]
     long long* __restrict__ out_lh,    // [K, F]
     int K, int F, int B) {
-  extern __shared__ long long lds_h[];  // [waves_per_block][B*2]
+  // One wave per (node, feature). All 64 lanes stage the histogram into
+  // LDS AND precompute the dequantized doubles in parallel (each element
+  // rounds independently, so parallel precompute is bitwise-identical to
+  // the CPU oracle's elementwise multiply). Then the two missing-value
+  // directions scan on lanes 0 and 1 concurrently - the sequential
+  // dependent f64 chain was 52% issue-stall as a single-lane loop.
+  extern __shared__ long long lds_h[];  // per wave: [B*2 i64][B*2 f64]
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
   const int waves_per_block = blockDim.x / WAVE;
@@ -285,28 +291,32 @@ __global__ __launch_bounds__(256) void find_splits_kf_kernel(
   if (kf >= (int64_t)K * F) return;
   const int k = (int)(kf / F);
   const int f = (int)(kf % F);
-  const long long* gh = hist + ((size_t)k * F + f) * B * 2;
-  long long* h = lds_h + (size_t)wave * B * 2;
-  for (int i = lane; i < B * 2; i += WAVE) h[i] = gh[i];
-  // wave-local LDS visibility: all lanes' stores above precede this wait
-  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-  if (lane != 0) return;
-  const int nb = feat_bins[f];
-
-  // multiply by the reciprocal (computed identically on the CPU oracle):
-  // f64 division is ~10x a multiply and sits on the sequential scan path
   const double inv_g = 1.0 / scale_g;
   const double inv_h = 1.0 / scale_h;
+  const long long* gh = hist + ((size_t)k * F + f) * B * 2;
+  long long* h = lds_h + (size_t)wave * B * 4;
+  double* hd = reinterpret_cast<double*>(h + B * 2);
+  for (int i = lane; i < B * 2; i += WAVE) {
+    const long long v = gh[i];
+    h[i] = v;
+    hd[i] = (double)v * ((i & 1) ? inv_h : inv_g);
+  }
+  // wave-local LDS visibility: all lanes' stores above precede this wait
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  if (lane >= 2) return;
+  const int nb = feat_bins[f];
+
   const double Gp = (double)parent_g[k] * inv_g;
   const double Hp = (double)parent_h[k] * inv_h;
   const double parent_score = calc_score(Gp, Hp, lam, alpha);
 
-  // feature totals (same order as cumsum's last element)
+  // feature totals (same order as cumsum's last element); computed
+  // redundantly by both lanes in lockstep - cheaper than a broadcast
   double Gtot = 0.0, Htot = 0.0;
   long long Gtot_q = 0, Htot_q = 0;
   for (int b = 0; b < B; ++b) {
-    Gtot += (double)h[b * 2] * inv_g;
-    Htot += (double)h[b * 2 + 1] * inv_h;
+    Gtot += hd[b * 2];
+    Htot += hd[b * 2 + 1];
     Gtot_q += h[b * 2];
     Htot_q += h[b * 2 + 1];
   }
@@ -315,14 +325,15 @@ __global__ __launch_bounds__(256) void find_splits_kf_kernel(
   const long long Gmiss_q = parent_g[k] - Gtot_q;
   const long long Hmiss_q = parent_h[k] - Htot_q;
 
-  SplitCand best = {-1.0, 0, 0, 0, 0};
-  // evaluate default_left = 1 first, then 0; strict > keeps CPU tie-break
-  for (int dl = 1; dl >= 0; --dl) {
+  // lane 0 scans default_left=1, lane 1 scans default_left=0
+  const int dl = 1 - lane;
+  SplitCand best = {-1.0, 0, dl, 0, 0};
+  {
     double GL = 0.0, HL = 0.0;
     long long GLq = 0, HLq = 0;
     for (int b = 0; b < nb - 1 && b < B; ++b) {
-      GL += (double)h[b * 2] * inv_g;
-      HL += (double)h[b * 2 + 1] * inv_h;
+      GL += hd[b * 2];
+      HL += hd[b * 2 + 1];
       GLq += h[b * 2];
       HLq += h[b * 2 + 1];
       double gl = dl ? GL + Gmiss : GL;
@@ -335,12 +346,29 @@ __global__ __launch_bounds__(256) void find_splits_kf_kernel(
       if (gain > best.gain) {
         best.gain = gain;
         best.bin = b;
-        best.default_left = dl;
         best.left_g = dl ? GLq + Gmiss_q : GLq;
         best.left_h = dl ? HLq + Hmiss_q : HLq;
       }
     }
   }
+  // merge with the CPU oracle's preference: dl=0 wins only on a STRICTLY
+  // greater gain (dl=1 was evaluated first there)
+  const double g1 = __shfl(best.gain, 1);
+  const int b1 = __shfl(best.bin, 1);
+  const long long lg1 = __shfl(best.left_g, 1);
+  const long long lh1 = __shfl(best.left_h, 1);
+  if (lane != 0) return;
+  if (g1 > best.gain) {
+    best.gain = g1;
+    best.bin = b1;
+    best.default_left = 0;
+    best.left_g = lg1;
+    best.left_h = lh1;
+  } else {
+    best.default_left = 1;
+  }
+  // no-split sentinel keeps the CPU contract (gain -1, dl 0)
+  if (best.gain <= -1.0) best.default_left = 0;
   out_gain[kf] = best.gain;
   out_bin[kf] = best.bin;
   out_dl[kf] = (uint8_t)best.default_left;
@@ -395,6 +423,8 @@ __global__ void find_splits_reduce_kernel(
 // blocks are processed in segment order and lanes in row order.
 // ---------------------------------------------------------------------------
 #define PART_THREADS 256
+#define PART_ROWS_PER_THREAD 4
+#define PART_CHUNK (PART_THREADS * PART_ROWS_PER_THREAD)
 
 __device__ inline bool go_left_pred(const uint8_t* bins, uint64_t r,
                                     int64_t row_stride, int feat,
@@ -425,29 +455,46 @@ __global__ void partition_count_kernel(
   }
   const int node = lo;
   const int64_t chunk_in_node = wg - chunk_off[node];
-  const int64_t row_lo = chunk_in_node * PART_THREADS;
+  const int64_t row_lo = chunk_in_node * PART_CHUNK;
   const int64_t count = node_start[K + node];
   const int64_t seg_start = node_start[node];
   const int feat = (int)split_feat[node], sbin = (int)split_bin[node];
   const int dl = (int)default_left[node];
 
-  const int64_t i = row_lo + threadIdx.x;
-  bool flag = false;
-  if (i < count) {
-    uint64_t r = (uint32_t)ridx[seg_start + i];
-    flag = go_left_pred(bins, r, row_stride, feat, sbin, dl);
-    flags[seg_start + i] = flag ? 1 : 0;
+  // preload all stripes' row indices + bin gathers so 4 independent
+  // random-latency loads are in flight per thread (the kernel was 86%
+  // memory-parked at one row per thread)
+  int total = 0;
+  uint8_t bv[PART_ROWS_PER_THREAD];
+  bool valid[PART_ROWS_PER_THREAD];
+  #pragma unroll
+  for (int sstripe = 0; sstripe < PART_ROWS_PER_THREAD; ++sstripe) {
+    const int64_t i = row_lo + sstripe * PART_THREADS + threadIdx.x;
+    valid[sstripe] = i < count;
+    uint64_t r = valid[sstripe] ? (uint32_t)ridx[seg_start + i] : 0;
+    bv[sstripe] = bins[r * (uint64_t)row_stride + feat];
   }
-  unsigned long long mask = __ballot(flag);
   __shared__ int wave_sums[PART_THREADS / WAVE];
   const int wave_id = threadIdx.x / WAVE;
-  if ((threadIdx.x & (WAVE - 1)) == 0) wave_sums[wave_id] = __popcll(mask);
-  __syncthreads();
-  if (threadIdx.x == 0) {
-    int total = 0;
-    for (int w = 0; w < PART_THREADS / WAVE; ++w) total += wave_sums[w];
-    block_counts[wg] = total;
+  #pragma unroll
+  for (int sstripe = 0; sstripe < PART_ROWS_PER_THREAD; ++sstripe) {
+    const int64_t i = row_lo + sstripe * PART_THREADS + threadIdx.x;
+    bool flag = false;
+    if (valid[sstripe]) {
+      const int b = bv[sstripe];
+      flag = (b == 255) ? (dl != 0) : (b <= sbin);
+      flags[seg_start + i] = flag ? 1 : 0;
+    }
+    unsigned long long mask = __ballot(flag);
+    if ((threadIdx.x & (WAVE - 1)) == 0)
+      wave_sums[wave_id] = __popcll(mask);
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      for (int w = 0; w < PART_THREADS / WAVE; ++w) total += wave_sums[w];
+    }
+    __syncthreads();
   }
+  if (threadIdx.x == 0) block_counts[wg] = total;
 }
 
 __global__ void partition_scatter_kernel(
@@ -466,39 +513,52 @@ __global__ void partition_scatter_kernel(
   }
   const int node = lo;
   const int64_t chunk_in_node = wg - chunk_off[node];
-  const int64_t row_lo = chunk_in_node * PART_THREADS;
+  const int64_t row_lo = chunk_in_node * PART_CHUNK;
   const int64_t count = node_start[K + node];
   const int64_t seg_start = node_start[node];
-
-  const int64_t i = row_lo + threadIdx.x;
-  bool valid = i < count;
-  int32_t rv = 0;
-  bool flag = false;
-  if (valid) {
-    rv = ridx[seg_start + i];
-    flag = flags[seg_start + i] != 0;
-  }
-  unsigned long long mask = __ballot(flag);
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave_id = threadIdx.x / WAVE;
   __shared__ int wave_sums[PART_THREADS / WAVE];
-  if (lane == 0) wave_sums[wave_id] = __popcll(mask);
+  __shared__ int stripe_left_base;
+  if (threadIdx.x == 0) stripe_left_base = 0;
   __syncthreads();
-  int wave_left_before = 0;
-  for (int w = 0; w < wave_id; ++w) wave_left_before += wave_sums[w];
 
-  const int prefix_in_wave = __popcll(mask & ((lane == 0) ? 0ull : ((~0ull) >> (64 - lane))));
-  if (valid) {
-    const int64_t lbefore = left_before[wg];  // left rows in earlier chunks
-    if (flag) {
-      const int64_t pos = lbefore + wave_left_before + prefix_in_wave;
-      ridx_out[seg_start + pos] = rv;
-    } else {
-      // rights: position = node_left_total + (rows before me) - (lefts before me)
-      const int64_t rights_before =
-          (row_lo + threadIdx.x) - (lbefore + wave_left_before + prefix_in_wave);
-      ridx_out[seg_start + node_left_total[node] + rights_before] = rv;
+  #pragma unroll
+  for (int sstripe = 0; sstripe < PART_ROWS_PER_THREAD; ++sstripe) {
+    const int64_t i = row_lo + sstripe * PART_THREADS + threadIdx.x;
+    const bool valid = i < count;
+    int32_t rv = 0;
+    bool flag = false;
+    if (valid) {
+      rv = ridx[seg_start + i];
+      flag = flags[seg_start + i] != 0;
     }
+    unsigned long long mask = __ballot(flag);
+    if (lane == 0) wave_sums[wave_id] = __popcll(mask);
+    __syncthreads();
+    int wave_left_before = stripe_left_base;
+    for (int w = 0; w < wave_id; ++w) wave_left_before += wave_sums[w];
+
+    const int prefix_in_wave =
+        __popcll(mask & ((lane == 0) ? 0ull : ((~0ull) >> (64 - lane))));
+    if (valid) {
+      // lefts strictly before row i in the whole node segment: earlier
+      // chunks + earlier stripes of this chunk + this stripe's ballot
+      const int64_t my_left_prefix =
+          left_before[wg] + wave_left_before + prefix_in_wave;
+      if (flag) {
+        ridx_out[seg_start + my_left_prefix] = rv;
+      } else {
+        const int64_t rights_before = i - my_left_prefix;
+        ridx_out[seg_start + node_left_total[node] + rights_before] = rv;
+      }
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      for (int w = 0; w < PART_THREADS / WAVE; ++w)
+        stripe_left_base += wave_sums[w];
+    }
+    __syncthreads();
   }
 }
 
@@ -827,7 +887,7 @@ std::vector<torch::Tensor> find_splits(torch::Tensor hist, torch::Tensor parent_
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   int64_t total = (int64_t)K * F;
   const int waves_per_block = 4;
-  const size_t scan_lds = (size_t)waves_per_block * B * 2 * sizeof(long long);
+  const size_t scan_lds = (size_t)waves_per_block * B * 4 * sizeof(long long);
   hipLaunchKernelGGL(find_splits_kf_kernel,
                      dim3((uint32_t)ceil_div(total, waves_per_block)),
                      dim3(waves_per_block * WAVE), scan_lds, stream.stream(),
@@ -869,7 +929,7 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
     auto acc = chunk_off_cpu.accessor<int64_t, 1>();
     auto cacc = counts_cpu.accessor<int64_t, 1>();
     for (int k = 0; k < K; ++k)
-      acc[k + 1] = acc[k] + (cacc[k] + PART_THREADS - 1) / PART_THREADS;
+      acc[k + 1] = acc[k] + (cacc[k] + PART_CHUNK - 1) / PART_CHUNK;
     total_chunks = acc[K];
   }
   if (total_chunks == 0) return {ridx_out, left_counts};
