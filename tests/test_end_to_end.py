@@ -62,6 +62,27 @@ class TestJointTraining:
             preds.append(bst.predict(X, output_margin=True))
         np.testing.assert_allclose(preds[0], preds[1], rtol=1e-5, atol=1e-6)
 
+    def test_overlapped_allreduce_path(self, monkeypatch):
+        """The chunked build + async-AllReduce pipeline (the multi-GPU
+        hot path) must match the plain path exactly - forced on under
+        gloo via RXGB_FORCE_OVERLAP_ALLREDUCE."""
+        monkeypatch.setenv("RXGB_FORCE_OVERLAP_ALLREDUCE", "1")
+        X, y = create_data(3000, 40)
+        params = {"objective": "binary:logistic", "max_depth": 5, "eta": 0.3}
+        dtrain = RayDMatrix(X, label=y)
+        bst_overlap = train(
+            params, dtrain, 6, ray_params=RayParams(num_actors=2)
+        )
+        monkeypatch.delenv("RXGB_FORCE_OVERLAP_ALLREDUCE")
+        bst_plain = train(
+            params, RayDMatrix(X, label=y), 6,
+            ray_params=RayParams(num_actors=2),
+        )
+        np.testing.assert_array_equal(
+            bst_overlap.predict(X, output_margin=True),
+            bst_plain.predict(X, output_margin=True),
+        )
+
 
 class TestPredictions:
     def test_softprob_2d_combine(self):

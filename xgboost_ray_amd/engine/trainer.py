@@ -411,12 +411,16 @@ class BoostingEngine:
             counts = torch.tensor(
                 [nd.count for nd in build_nodes], dtype=torch.int64
             )
+            import os as _os2
+
             F = self.dtrain.n_features
-            overlap = (
-                self.coll.is_distributed
-                and self.device.type == "cuda"
-                and self.dtrain.bins.stride(0) % 16 == 0
-                and F > 16
+            overlap = self.coll.is_distributed and (
+                (
+                    self.device.type == "cuda"
+                    and self.dtrain.bins.stride(0) % 16 == 0
+                    and F > 16
+                )
+                or _os2.environ.get("RXGB_FORCE_OVERLAP_ALLREDUCE") == "1"
             )
             if overlap:
                 # overlap the RCCL AllReduce of each finished feature block
