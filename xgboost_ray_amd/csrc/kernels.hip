@@ -500,6 +500,8 @@ __global__ void partition_count_kernel(
 __global__ void partition_scatter_kernel(
     const uint8_t* __restrict__ flags, const int32_t* __restrict__ ridx,
     int32_t* __restrict__ ridx_out,
+    const int2* __restrict__ gseg,      // segment-ordered gradient pairs
+    int2* __restrict__ gseg_out,        // permuted alongside ridx
     const int64_t* __restrict__ node_start,  // [K] + counts at [K..2K)
     const int64_t* __restrict__ chunk_off,   // [K+1]
     const int64_t* __restrict__ left_before,   // [total_chunks] excl. prefix within node
@@ -528,9 +530,11 @@ __global__ void partition_scatter_kernel(
     const int64_t i = row_lo + sstripe * PART_THREADS + threadIdx.x;
     const bool valid = i < count;
     int32_t rv = 0;
+    int2 gv = {0, 0};
     bool flag = false;
     if (valid) {
       rv = ridx[seg_start + i];
+      if (gseg != nullptr) gv = gseg[seg_start + i];
       flag = flags[seg_start + i] != 0;
     }
     unsigned long long mask = __ballot(flag);
@@ -546,12 +550,11 @@ __global__ void partition_scatter_kernel(
       // chunks + earlier stripes of this chunk + this stripe's ballot
       const int64_t my_left_prefix =
           left_before[wg] + wave_left_before + prefix_in_wave;
-      if (flag) {
-        ridx_out[seg_start + my_left_prefix] = rv;
-      } else {
-        const int64_t rights_before = i - my_left_prefix;
-        ridx_out[seg_start + node_left_total[node] + rights_before] = rv;
-      }
+      const int64_t pos = flag
+                              ? my_left_prefix
+                              : node_left_total[node] + (i - my_left_prefix);
+      ridx_out[seg_start + pos] = rv;
+      if (gseg_out != nullptr) gseg_out[seg_start + pos] = gv;
     }
     __syncthreads();
     if (threadIdx.x == 0) {
@@ -776,7 +779,7 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                               torch::Tensor ridx, torch::Tensor starts,
                               torch::Tensor counts, int64_t n_bins,
                               int64_t f_lo, int64_t f_hi,
-                              torch::Tensor hist) {
+                              torch::Tensor hist, bool pregathered) {
   // builds features [f_lo, f_hi) into the caller-provided [K, F, n_bins, 2]
   // histogram (zeroed by the caller); ranges let the driver overlap the
   // RCCL AllReduce of one feature block with the build of the next.
@@ -807,15 +810,23 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
   int64_t span = max_end - min_start;
   TORCH_CHECK(gpair_q.dtype() == torch::kInt32,
               "gpair_q must be int32 packed pairs");
-  auto gpair_seg = torch::empty({std::max<int64_t>(span, 1), 2},
-                                gpair_q.options());
-  if (span > 0) {
-    int64_t blocks = std::min<int64_t>(ceil_div(span, 256), 8192);
-    hipLaunchKernelGGL(gather_gpair_kernel, dim3(blocks), dim3(256), 0,
-                       stream.stream(),
-                       (const int2*)gpair_q.data_ptr<int32_t>(),
-                       ridx.data_ptr<int32_t>() + min_start,
-                       (int2*)gpair_seg.data_ptr<int32_t>(), span);
+  torch::Tensor gpair_seg;
+  if (pregathered) {
+    // gpair_q is ALREADY in segment order aligned with ridx (the
+    // partition scatter permutes it) - no gather pass needed
+    gpair_seg = gpair_q;
+    min_start = 0;
+  } else {
+    gpair_seg = torch::empty({std::max<int64_t>(span, 1), 2},
+                             gpair_q.options());
+    if (span > 0) {
+      int64_t blocks = std::min<int64_t>(ceil_div(span, 256), 8192);
+      hipLaunchKernelGGL(gather_gpair_kernel, dim3(blocks), dim3(256), 0,
+                         stream.stream(),
+                         (const int2*)gpair_q.data_ptr<int32_t>(),
+                         ridx.data_ptr<int32_t>() + min_start,
+                         (int2*)gpair_seg.data_ptr<int32_t>(), span);
+    }
   }
   // ONE H2D copy for all control data: [starts_adj(K) | counts(K) |
   // chunk_off(K+1)] - tiny pageable copies around kernel launches were
@@ -914,13 +925,15 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
                                           torch::Tensor starts, torch::Tensor counts,
                                           torch::Tensor split_feat,
                                           torch::Tensor split_bin,
-                                          torch::Tensor default_left) {
+                                          torch::Tensor default_left,
+                                          torch::Tensor gseg) {
   const int K = (int)starts.size(0);
   const int F = (int)bins.size(1);
   auto dev = bins.device();
   auto ridx_out = ridx.clone();
+  auto gseg_out = gseg.clone();
   auto left_counts = torch::zeros({K}, torch::kInt64);
-  if (K == 0) return {ridx_out, left_counts};
+  if (K == 0) return {ridx_out, left_counts, gseg_out};
   auto starts_cpu = starts.to(torch::kCPU).to(torch::kInt64);
   auto counts_cpu = counts.to(torch::kCPU).to(torch::kInt64);
   int64_t total_chunks = 0;
@@ -932,7 +945,7 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
       acc[k + 1] = acc[k] + (cacc[k] + PART_CHUNK - 1) / PART_CHUNK;
     total_chunks = acc[K];
   }
-  if (total_chunks == 0) return {ridx_out, left_counts};
+  if (total_chunks == 0) return {ridx_out, left_counts, gseg_out};
   // ONE H2D copy: [starts(K) | counts(K) | chunk_off(K+1) | feat(K) |
   // bin(K) | default_left(K)]
   auto meta = torch::cat({starts_cpu, counts_cpu, chunk_off_cpu,
@@ -985,10 +998,15 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
   hipLaunchKernelGGL(partition_scatter_kernel, dim3((uint32_t)total_chunks),
                      dim3(PART_THREADS), 0, stream.stream(),
                      flags.data_ptr<uint8_t>(), ridx.data_ptr<int32_t>(),
-                     ridx_out.data_ptr<int32_t>(), sc_p, chunk_off_p,
+                     ridx_out.data_ptr<int32_t>(),
+                     gseg.numel() ? (const int2*)gseg.data_ptr<int32_t>()
+                                  : nullptr,
+                     gseg_out.numel() ? (int2*)gseg_out.data_ptr<int32_t>()
+                                      : nullptr,
+                     sc_p, chunk_off_p,
                      left_before.data_ptr<int64_t>(),
                      node_left_total.data_ptr<int64_t>(), K);
-  return {ridx_out, node_left_total_cpu};
+  return {ridx_out, node_left_total_cpu, gseg_out};
 }
 
 void predict_trees(torch::Tensor X, torch::Tensor feat, torch::Tensor thr,

@@ -347,6 +347,12 @@ class BoostingEngine:
         self._scale_g_cur = scale_g
         self._leaf_segs = []
         ridx = self._sample_rows(it, cls + 101 * ptree)
+        # segment-ordered gradient pairs aligned with ridx; the partition
+        # scatter permutes them each depth, so no per-depth gather pass
+        if ridx.numel() == self.dtrain.n_rows:
+            gseg = gq  # identity order (partition clones before writing)
+        else:
+            gseg = gq[ridx.long()].contiguous()
         feat_mask = self._sample_features(it, cls + 101 * ptree)
         n_local = int(ridx.numel())
 
@@ -438,8 +444,9 @@ class BoostingEngine:
                 while f0 < F:
                     f1 = min(f0 + step, F)
                     ops.build_histogram(
-                        self.dtrain.bins, gq, ridx, starts, counts,
+                        self.dtrain.bins, gseg, ridx, starts, counts,
                         self.n_bins, f_range=(f0, f1), out=hist,
+                        pregathered=True,
                     )
                     sl = hist[:, f0:f1].contiguous()
                     pending.append((self.coll.allreduce_async(sl), sl, f0, f1))
@@ -449,7 +456,8 @@ class BoostingEngine:
                     hist[:, c0:c1].copy_(sl)
             else:
                 hist = ops.build_histogram(
-                    self.dtrain.bins, gq, ridx, starts, counts, self.n_bins
+                    self.dtrain.bins, gseg, ridx, starts, counts,
+                    self.n_bins, pregathered=True,
                 )
                 _tick("hist")
                 if self.coll.is_distributed:
@@ -553,7 +561,7 @@ class BoostingEngine:
                 [nd.count for nd in split_nodes], dtype=torch.int64
             )
             _tick("tree_host")
-            ridx, left_counts = ops.partition_rows(
+            ridx, left_counts, gseg = ops.partition_rows(
                 self.dtrain.bins,
                 ridx,
                 sstarts,
@@ -561,6 +569,7 @@ class BoostingEngine:
                 torch.tensor(sf, dtype=torch.int32),
                 torch.tensor(sb, dtype=torch.int32),
                 torch.tensor(sdl, dtype=torch.uint8),
+                gpair_seg=gseg,
             )
             _tick("partition")
 

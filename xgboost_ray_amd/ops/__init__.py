@@ -63,7 +63,7 @@ def bin_matrix(values, cuts_flat, cut_ptr):
 
 
 def build_histogram(bins, gpair_q, ridx, starts, counts, n_bins,
-                    f_range=None, out=None):
+                    f_range=None, out=None, pregathered=False):
     """Accumulate per-(node, feature, bin) int64 gradient-pair histograms.
 
     bins: uint8 [n, F]; gpair_q: int32 [n, 2] packed pairs; ridx: int32 [n]
@@ -76,7 +76,8 @@ def build_histogram(bins, gpair_q, ridx, starts, counts, n_bins,
     skipped; their mass is recovered as node_total - feature_sum.
     """
     return _impl(bins).build_histogram(
-        bins, gpair_q, ridx, starts, counts, n_bins, f_range, out
+        bins, gpair_q, ridx, starts, counts, n_bins, f_range, out,
+        pregathered
     )
 
 
@@ -115,14 +116,19 @@ def find_splits(
     )
 
 
-def partition_rows(bins, ridx, starts, counts, split_feat, split_bin, default_left):
-    """Stable in-place partition of each node's ridx segment by its split.
+def partition_rows(bins, ridx, starts, counts, split_feat, split_bin,
+                   default_left, gpair_seg=None):
+    """Stable partition of each node's ridx segment by its split.
 
-    Rows with bin <= split_bin (or missing & default_left) go left.
-    Returns (ridx_out int32 [n], left_counts int64 [K]).
+    Rows with bin <= split_bin (or missing & default_left) go left. When
+    ``gpair_seg`` (int32 [n,2], segment-ordered gradient pairs) is given
+    it is permuted alongside ridx - fusing the next depth's gradient
+    gather into the scatter.
+    Returns (ridx_out, left_counts[, gpair_seg_out]).
     """
     return _impl(bins).partition_rows(
-        bins, ridx, starts, counts, split_feat, split_bin, default_left
+        bins, ridx, starts, counts, split_feat, split_bin, default_left,
+        gpair_seg
     )
 
 

@@ -47,7 +47,7 @@ def bin_matrix(values: torch.Tensor, cuts_flat: torch.Tensor, cut_ptr: torch.Ten
 
 
 def build_histogram(bins, gpair_q, ridx, starts, counts, n_bins,
-                    f_range=None, out=None):
+                    f_range=None, out=None, pregathered=False):
     K = len(starts)
     n, F = bins.shape
     hist = out
@@ -57,12 +57,13 @@ def build_histogram(bins, gpair_q, ridx, starts, counts, n_bins,
         )
     f_lo, f_hi = (0, F) if f_range is None else f_range
     return _build_histogram_range(
-        bins, gpair_q, ridx, starts, counts, n_bins, f_lo, f_hi, hist
+        bins, gpair_q, ridx, starts, counts, n_bins, f_lo, f_hi, hist,
+        pregathered
     )
 
 
 def _build_histogram_range(bins, gpair_q, ridx, starts, counts, n_bins,
-                           f_lo, f_hi, hist):
+                           f_lo, f_hi, hist, pregathered=False):
     K = len(starts)
     n, F = bins.shape
     foff = torch.arange(f_lo, f_hi, dtype=torch.int64) * n_bins
@@ -74,8 +75,9 @@ def _build_histogram_range(bins, gpair_q, ridx, starts, counts, n_bins,
         rb = bins[idx][:, f_lo:f_hi].long()  # [c, fr]
         valid = rb != MISSING_BIN
         flat = foff.unsqueeze(0) + rb  # [c, F]
-        g = gpair_q[idx, 0].long().unsqueeze(1).expand_as(flat)
-        h = gpair_q[idx, 1].long().unsqueeze(1).expand_as(flat)
+        gsrc = gpair_q[s : s + c] if pregathered else gpair_q[idx]
+        g = gsrc[:, 0].long().unsqueeze(1).expand_as(flat)
+        h = gsrc[:, 1].long().unsqueeze(1).expand_as(flat)
         hk = hist[k].view(F * n_bins, 2)
         fidx = flat[valid]
         hk[:, 0].scatter_add_(0, fidx, g[valid])
@@ -200,9 +202,11 @@ def find_splits(
     }
 
 
-def partition_rows(bins, ridx, starts, counts, split_feat, split_bin, default_left):
+def partition_rows(bins, ridx, starts, counts, split_feat, split_bin,
+                   default_left, gpair_seg=None):
     K = len(starts)
     out = ridx.clone()
+    gout = None if gpair_seg is None else gpair_seg.clone()
     left_counts = torch.zeros(K, dtype=torch.int64)
     for k in range(K):
         s, c = int(starts[k]), int(counts[k])
@@ -217,8 +221,13 @@ def partition_rows(bins, ridx, starts, counts, split_feat, split_bin, default_le
         else:
             go_left = go_left & ~miss
         out[s : s + c] = torch.cat([idx[go_left], idx[~go_left]])
+        if gout is not None:
+            gseg = gpair_seg[s : s + c]
+            gout[s : s + c] = torch.cat([gseg[go_left], gseg[~go_left]])
         left_counts[k] = int(go_left.sum())
-    return out, left_counts
+    if gout is None:
+        return out, left_counts
+    return out, left_counts, gout
 
 
 def predict_trees(X, feat, thr, left, default_left, value, tree_ptr, out, tree_weight=1.0):

@@ -46,7 +46,7 @@ def bin_matrix(values, cuts_flat, cut_ptr):
 
 
 def build_histogram(bins, gpair_q, ridx, starts, counts, n_bins,
-                    f_range=None, out=None):
+                    f_range=None, out=None, pregathered=False):
     K = len(starts)
     F = bins.shape[1]
     if out is None:
@@ -64,6 +64,7 @@ def build_histogram(bins, gpair_q, ridx, starts, counts, n_bins,
         int(f_lo),
         int(f_hi),
         out,
+        bool(pregathered),
     )
 
 
@@ -105,16 +106,19 @@ def find_splits(
     }
 
 
-def partition_rows(bins, ridx, starts, counts, split_feat, split_bin, default_left):
+def partition_rows(bins, ridx, starts, counts, split_feat, split_bin,
+                   default_left, gpair_seg=None):
     dev = bins.device
+    if gpair_seg is None:
+        gpair_seg = torch.zeros((0, 2), dtype=torch.int32, device=dev)
+        r, lc, _ = _load().partition_rows(
+            bins, ridx, starts, counts, split_feat, split_bin,
+            default_left, gpair_seg,
+        )
+        return r, lc
     return _load().partition_rows(
-        bins,
-        ridx,
-        starts.to(dev),
-        counts.to(dev),
-        split_feat.to(dev),
-        split_bin.to(dev),
-        default_left.to(dev),
+        bins, ridx, starts, counts, split_feat, split_bin, default_left,
+        gpair_seg,
     )
 
 
