@@ -148,6 +148,16 @@ class Booster:
         k = max(1, self.num_class) * self.num_parallel_tree
         return len(self.trees) // k
 
+    @property
+    def best_ntree_limit(self) -> int:
+        """Legacy xgboost attribute: trees up to the best iteration."""
+        it = (
+            self.best_iteration
+            if self.best_iteration is not None
+            else self.num_boosted_rounds() - 1
+        )
+        return (it + 1) * max(1, self.num_class) * self.num_parallel_tree
+
     def append_round(self, trees: Sequence[Tree], classes: Sequence[int]):
         self.trees.extend(trees)
         self.tree_info.extend(int(c) for c in classes)
