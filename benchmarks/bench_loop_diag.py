@@ -1,5 +1,11 @@
 """Diagnose in-loop stalls: time consecutive 5-round windows."""
 
+import os as _os
+import sys as _sys
+
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
+
+
 import time
 
 import torch

@@ -4,6 +4,12 @@ Times each engine op in isolation (20 reps, synced) to separate kernel
 time from host-side overhead. Run on an MI355X box.
 """
 
+import os as _os
+import sys as _sys
+
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
+
+
 import time
 
 import numpy as np

@@ -9,6 +9,12 @@ appends a CSV row to res.csv - the same protocol the reference uses
 (reference benchmark_cpu_gpu.py:101-106, 173-197).
 """
 
+import os as _os
+import sys as _sys
+
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
+
+
 import argparse
 import os
 import time

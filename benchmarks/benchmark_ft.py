@@ -10,6 +10,12 @@ Conditions (reference benchmark_ft.py:289-347):
 Prints per-condition wall time and final error.
 """
 
+import os as _os
+import sys as _sys
+
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
+
+
 import argparse
 import os
 import shutil

@@ -105,7 +105,8 @@ class _Node:
     count: int
     sum_g: int  # quantized
     sum_h: int
-    hist: Optional[torch.Tensor] = None  # [F, B, 2] int64 (this depth only)
+    slot: int = 0  # scan-slot in this depth's all_hist
+    parent_slot: int = 0  # parent's scan-slot in the previous depth
 
 
 class CallbackList:
@@ -383,14 +384,17 @@ class BoostingEngine:
         cut_ptr_cpu = self.dtrain.cuts.cut_ptr.cpu().numpy()
 
         frontier: List[_Node] = [root]
-        parent_hist: Dict[int, torch.Tensor] = {}
-        # map nid -> node bookkeeping for building
+        prev_all_hist: Optional[torch.Tensor] = None  # [prev_frontier, F, B, 2]
         for depth in range(self.p.max_depth):
             if not frontier:
                 break
-            # ---- decide build list: root, or per sibling pair the smaller
+            # ---- build order: built (smaller) children first, derived
+            # siblings after, so the built block is a CONTIGUOUS leading
+            # slice (one AllReduce, batched subtraction for the rest)
             build_nodes: List[_Node] = []
-            derive_nodes: List[Tuple[_Node, _Node]] = []  # (derive, built_sibling)
+            derive_big: List[_Node] = []
+            derive_parent_slot: List[int] = []
+            derive_sib_pos: List[int] = []
             if depth == 0:
                 build_nodes = [root]
             else:
@@ -398,19 +402,25 @@ class BoostingEngine:
                 while i < len(frontier):
                     a = frontier[i]
                     b = frontier[i + 1] if i + 1 < len(frontier) else None
-                    if b is not None and parent_l[a.nid] == parent_l[b.nid]:
+                    if b is not None and a.parent_slot == b.parent_slot:
                         # siblings: build the globally smaller one. Use the
                         # quantized hessian sum as the size proxy - it is
                         # identical on every rank (local counts are not).
                         small, big = (a, b) if a.sum_h <= b.sum_h else (b, a)
+                        derive_parent_slot.append(big.parent_slot)
+                        derive_sib_pos.append(len(build_nodes))
                         build_nodes.append(small)
-                        derive_nodes.append((big, small))
+                        derive_big.append(big)
                         i += 2
                     else:
                         build_nodes.append(a)
                         i += 1
+            order_nodes: List[_Node] = build_nodes + derive_big
+            for pos, nd in enumerate(order_nodes):
+                nd.slot = pos
 
             K = len(build_nodes)
+            nF = len(frontier)
             starts = torch.tensor(
                 [nd.start for nd in build_nodes], dtype=torch.int64
             )
@@ -420,6 +430,10 @@ class BoostingEngine:
             import os as _os2
 
             F = self.dtrain.n_features
+            all_hist = torch.zeros(
+                (nF, F, self.n_bins, 2), dtype=torch.int64, device=self.device
+            )
+            hist = all_hist[:K]  # built block (contiguous leading slice)
             overlap = self.coll.is_distributed and (
                 (
                     self.device.type == "cuda"
@@ -432,11 +446,6 @@ class BoostingEngine:
                 # overlap the RCCL AllReduce of each finished feature block
                 # with the build of the next (BASELINE north-star: side-
                 # stream collective over xGMI behind the histogram build)
-                hist = torch.zeros(
-                    (K, F, self.n_bins, 2),
-                    dtype=torch.int64,
-                    device=self.device,
-                )
                 n_chunks = min(4, (F + 31) // 32)
                 step = ((F + n_chunks - 1) // n_chunks + 15) // 16 * 16
                 pending = []
@@ -455,28 +464,34 @@ class BoostingEngine:
                     h.wait()
                     hist[:, c0:c1].copy_(sl)
             else:
-                hist = ops.build_histogram(
+                ops.build_histogram(
                     self.dtrain.bins, gseg, ridx, starts, counts,
-                    self.n_bins, pregathered=True,
+                    self.n_bins, out=hist, pregathered=True,
                 )
                 _tick("hist")
                 if self.coll.is_distributed:
                     self.coll.allreduce_(hist)
             _tick("allreduce")
-            for k, nd in enumerate(build_nodes):
-                nd.hist = hist[k]
-            for big, small in derive_nodes:
-                pid = parent_l[big.nid]
-                big.hist = parent_hist[pid] - small.hist
-            parent_hist = {}
+            if derive_big:
+                # sibling = parent - built, batched over all pairs
+                pslots = torch.tensor(
+                    derive_parent_slot, dtype=torch.int64, device=self.device
+                )
+                spos = torch.tensor(
+                    derive_sib_pos, dtype=torch.int64, device=self.device
+                )
+                torch.sub(
+                    prev_all_hist.index_select(0, pslots),
+                    all_hist.index_select(0, spos),
+                    out=all_hist[K:],
+                )
 
-            # ---- split scan over the whole frontier
-            all_hist = torch.stack([nd.hist for nd in frontier])
+            # ---- split scan over the whole frontier (scan-slot order)
             # ONE H2D for both parent-sum vectors
             psums = torch.from_numpy(
                 np.array(
-                    [[nd.sum_g for nd in frontier],
-                     [nd.sum_h for nd in frontier]],
+                    [[nd.sum_g for nd in order_nodes],
+                     [nd.sum_h for nd in order_nodes]],
                     dtype=np.int64,
                 )
             ).to(self.device)
@@ -515,7 +530,7 @@ class BoostingEngine:
             split_nodes: List[_Node] = []
             sf, sb, sdl = [], [], []
             children_meta = []
-            for k, nd in enumerate(frontier):
+            for k, nd in enumerate(order_nodes):
                 if gain[k] <= 0 or bfeat[k] < 0 or not np.isfinite(gain[k]):
                     self._finalize_leaf(nd, val_l, cover_l, scale_h)
                     continue
@@ -584,6 +599,7 @@ class BoostingEngine:
                     count=lcount,
                     sum_g=lg,
                     sum_h=lh,
+                    parent_slot=nd.slot,
                 )
                 rnode = _Node(
                     nid=lid + 1,
@@ -592,11 +608,11 @@ class BoostingEngine:
                     count=nd.count - lcount,
                     sum_g=nd.sum_g - lg,
                     sum_h=nd.sum_h - lh,
+                    parent_slot=nd.slot,
                 )
-                parent_hist[nd.nid] = nd.hist
-                nd.hist = None
                 new_frontier.extend([lnode, rnode])
             frontier = new_frontier
+            prev_all_hist = all_hist
 
         # remaining frontier nodes (max depth reached) become leaves
         final_segments = []
