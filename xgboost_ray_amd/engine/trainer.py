@@ -430,10 +430,11 @@ class BoostingEngine:
             import os as _os2
 
             F = self.dtrain.n_features
-            all_hist = torch.zeros(
+            all_hist = torch.empty(
                 (nF, F, self.n_bins, 2), dtype=torch.int64, device=self.device
             )
             hist = all_hist[:K]  # built block (contiguous leading slice)
+            hist.zero_()  # derived rows are fully overwritten by the sub
             overlap = self.coll.is_distributed and (
                 (
                     self.device.type == "cuda"
@@ -615,7 +616,6 @@ class BoostingEngine:
             prev_all_hist = all_hist
 
         # remaining frontier nodes (max depth reached) become leaves
-        final_segments = []
         for nd in frontier:
             self._finalize_leaf(nd, val_l, cover_l, scale_h)
         # leaf margin update: walk all leaves via the segment structure -
