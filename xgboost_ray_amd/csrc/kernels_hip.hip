@@ -22,6 +22,7 @@
 #include <ATen/hip/HIPContext.h>
 
 #include <cstdint>
+#include <cstdlib>
 #include <vector>
 
 #define WAVE 64
@@ -133,7 +134,7 @@ __global__ void gather_gpair_kernel(const int2* __restrict__ gpair,
 // a row. Lane rotation spreads one wave-instruction's LDS atomics over 16
 // different features (max 4 lanes per feature histogram) so same-bin
 // serialization on skewed features drops ~16x.
-template <bool VEC16>
+template <int VFB>  // 16: uint4 row loads; 8: uint2; 0: byte fallback
 __global__ __launch_bounds__(HIST_THREADS) void build_histogram_kernel(
     const uint8_t* __restrict__ bins,        // [n_rows_total, row_stride]
     const int2* __restrict__ gpair_seg,      // [seg_total] segment order
@@ -177,7 +178,7 @@ __global__ __launch_bounds__(HIST_THREADS) void build_histogram_kernel(
     const int2 gpi = gpair_seg[seg_i];
     const longlong2 gp = {(long long)gpi.x, (long long)gpi.y};
     const uint64_t r = (uint32_t)ridx[seg_i];
-    if (VEC16) {
+    if constexpr (VFB == 16) {
       // fb_size == 16 and row base 16B-aligned by construction.
       // Word-granular lane rotation: lane l processes its row's feature
       // words in order (l&3), (l&3)+1, ... so one wave-instruction's LDS
@@ -202,6 +203,30 @@ __global__ __launch_bounds__(HIST_THREADS) void build_histogram_kernel(
       for (int jj = 0; jj < 4; ++jj) {
         const uint32_t w = rws[jj];
         const int fw = ((jj + r4) & 3) * 4;
+        #pragma unroll
+        for (int kk = 0; kk < 4; ++kk) {
+          const int b = (w >> (8 * kk)) & 0xFF;
+          if (b != 255) {
+            const int f = fw + kk;
+            unsigned long long* cell =
+                &lds_hist[((size_t)f * n_bins + b) * 2];
+            atomicAdd(cell, (unsigned long long)gp.x);
+            atomicAdd(cell + 1, (unsigned long long)gp.y);
+          }
+        }
+      }
+    } else if constexpr (VFB == 8) {
+      const uint2 packed =
+          *reinterpret_cast<const uint2*>(bins + r * row_stride + f0);
+      const int r2 = lane & 1;
+      const uint32_t w0 = packed.x, w1 = packed.y;
+      const uint32_t rw0 = r2 ? w1 : w0;  // w[(0+r2)&1]
+      const uint32_t rw1 = r2 ? w0 : w1;  // w[(1+r2)&1]
+      const uint32_t rws[2] = {rw0, rw1};
+      #pragma unroll
+      for (int jj = 0; jj < 2; ++jj) {
+        const uint32_t w = rws[jj];
+        const int fw = ((jj + r2) & 1) * 4;
         #pragma unroll
         for (int kk = 0; kk < 4; ++kk) {
           const int b = (w >> (8 * kk)) & 0xFF;
@@ -847,14 +872,30 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                     ? 16
                     : (int)std::min<int64_t>(F, (64 * 1024) / (n_bins * 16));
   if (fb_size < 1) fb_size = 1;
+  // experiment knob: RXGB_HIST_FB=8 halves the LDS tile (4 workgroups/CU
+  // instead of 2) at the cost of 2x gradient re-reads per depth
+  if (const char* e = getenv("RXGB_HIST_FB")) {
+    int v = atoi(e);
+    if (v == 8 && vec16) fb_size = 8;
+  }
   TORCH_CHECK(f_lo % fb_size == 0 && f_lo < f_hi && f_hi <= F,
               "feature range must align to the block size ", fb_size);
   const int n_fb = (int)ceil_div(f_hi - f_lo, fb_size);
   const size_t lds = (size_t)fb_size * n_bins * 2 * sizeof(long long);
 
   // ridx pointer offset so seg indices align with gpair_seg
-  if (vec16) {
-    hipLaunchKernelGGL((build_histogram_kernel<true>),
+  if (vec16 && fb_size == 8) {
+    hipLaunchKernelGGL((build_histogram_kernel<8>),
+                       dim3((uint32_t)total_chunks, n_fb),
+                       dim3(HIST_THREADS), lds, stream.stream(),
+                       bins.data_ptr<uint8_t>(),
+                       (const int2*)gpair_seg.data_ptr<int32_t>(),
+                       ridx.data_ptr<int32_t>() + min_start,
+                       sc_adj_p, chunk_off_p,
+                       reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
+                       K, F, (int)n_bins, fb_size, row_stride, (int)f_lo);
+  } else if (vec16) {
+    hipLaunchKernelGGL((build_histogram_kernel<16>),
                        dim3((uint32_t)total_chunks, n_fb),
                        dim3(HIST_THREADS), lds, stream.stream(),
                        bins.data_ptr<uint8_t>(),
@@ -864,7 +905,7 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                        reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
                        K, F, (int)n_bins, fb_size, row_stride, (int)f_lo);
   } else {
-    hipLaunchKernelGGL((build_histogram_kernel<false>),
+    hipLaunchKernelGGL((build_histogram_kernel<0>),
                        dim3((uint32_t)total_chunks, n_fb),
                        dim3(HIST_THREADS), lds, stream.stream(),
                        bins.data_ptr<uint8_t>(),
