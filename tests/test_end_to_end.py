@@ -62,6 +62,21 @@ class TestJointTraining:
             preds.append(bst.predict(X, output_margin=True))
         np.testing.assert_allclose(preds[0], preds[1], rtol=1e-5, atol=1e-6)
 
+    def test_lossguide_distributed_equals_single(self):
+        """Leaf-wise growth is heap-ordered by globally identical gains,
+        so 2-actor == 1-actor bitwise."""
+        X, y = create_data(4096, 6)
+        params = {"objective": "binary:logistic", "grow_policy": "lossguide",
+                  "max_leaves": 24, "max_depth": 0, "eta": 0.3}
+        preds = []
+        for num_actors in (1, 2):
+            bst = train(
+                params, RayDMatrix(X, label=y), 5,
+                ray_params=RayParams(num_actors=num_actors),
+            )
+            preds.append(bst.predict(X, output_margin=True))
+        np.testing.assert_allclose(preds[0], preds[1], rtol=1e-6, atol=1e-7)
+
     def test_overlapped_allreduce_path(self, monkeypatch):
         """The chunked build + async-AllReduce pipeline (the multi-GPU
         hot path) must match the plain path exactly - forced on under

@@ -213,6 +213,30 @@ class TestTraining:
         # different weighting must change the model
         assert res1["train"]["auc"] != res2["train"]["auc"]
 
+    def test_lossguide_growth(self):
+        """grow_policy=lossguide: exact leaf budget, deterministic,
+        loss-decreasing, predict-consistent."""
+        dm, X, y = _binned()
+        res = {}
+        bst = run_training(
+            {"objective": "binary:logistic", "grow_policy": "lossguide",
+             "max_leaves": 15, "max_depth": 0, "eta": 0.3,
+             "eval_metric": ["logloss"]},
+            dm, 10, evals=[EvalPack(name="train", X=None)], evals_result=res,
+        )
+        t = bst.trees[0]
+        assert int((t.feat < 0).sum()) == 15
+        assert res["train"]["logloss"][-1] < res["train"]["logloss"][0] * 0.6
+        bst2 = run_training(
+            {"objective": "binary:logistic", "grow_policy": "lossguide",
+             "max_leaves": 15, "max_depth": 0, "eta": 0.3},
+            dm, 10,
+        )
+        np.testing.assert_array_equal(
+            bst.predict(X, output_margin=True),
+            bst2.predict(X, output_margin=True),
+        )
+
     def test_early_stopping(self):
         X, y = create_data(2000, 6, 0, "binary")
         Xv, yv = create_data(500, 6, 7, "binary")

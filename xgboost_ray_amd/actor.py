@@ -271,6 +271,7 @@ class _ActorWorker:
         shard = self.shards.get(data_uid)
         if shard is None:
             raise RuntimeError(f"Prediction data {data_uid} not loaded")
+        it_range = kwargs.get("iteration_range")
         if shard.get("streaming"):
             loader = shard["loader"]
             margins = []
@@ -278,14 +279,14 @@ class _ActorWorker:
                 Xc = torch.from_numpy(
                     np.ascontiguousarray(chunk["data"], np.float32)
                 ).to(self.device)
-                margins.append(bst.predict_margin_tensor(Xc))
+                margins.append(bst.predict_margin_tensor(Xc, it_range))
                 del Xc
             margin = torch.cat(margins, dim=0)
         else:
             X = torch.from_numpy(
                 np.ascontiguousarray(shard["data"], np.float32)
             ).to(self.device)
-            margin = bst.predict_margin_tensor(X)
+            margin = bst.predict_margin_tensor(X, it_range)
         if kwargs.get("output_margin"):
             pred = margin.cpu().numpy()
         else:

@@ -52,6 +52,8 @@ class TrainParams:
     num_class: int = 0
     eval_metric: List[str] = field(default_factory=list)
     seed: int = 0
+    grow_policy: str = "depthwise"
+    max_leaves: int = 0
     tree_method: str = "hist"
     verbosity: int = 1
     nthread: int = 0
@@ -347,6 +349,8 @@ class BoostingEngine:
         _tick("quantize")
         self._scale_g_cur = scale_g
         self._leaf_segs = []
+        if self.p.grow_policy == "lossguide":
+            return self._grow_tree_lossguide(gq, scale_g, scale_h, it, cls, ptree)
         ridx = self._sample_rows(it, cls + 101 * ptree)
         # segment-ordered gradient pairs aligned with ridx; the partition
         # scatter permutes them each depth, so no per-depth gather pass
@@ -753,6 +757,202 @@ class BoostingEngine:
                 ev.margin = pm.to(torch.float32) + (
                     ev.base_margin.reshape(shape) if ev.base_margin is not None else 0
                 )
+
+    def _grow_tree_lossguide(self, gq, scale_g, scale_h, it, cls, ptree):
+        """Leaf-wise (best-gain-first) growth - grow_policy=lossguide with
+        max_leaves (XGBoost/LightGBM semantics). Each expansion partitions
+        one leaf segment and histograms the smaller child (sibling by
+        subtraction); the expansion order is driven by the globally
+        identical scan gains, so all ranks grow the same tree."""
+        import heapq
+
+        ridx = self._sample_rows(it, cls + 101 * ptree)
+        if ridx.numel() == self.dtrain.n_rows:
+            gseg = gq
+        else:
+            gseg = gq[ridx.long()].contiguous()
+        n_local = int(ridx.numel())
+        if n_local == 0:
+            root_sum = torch.zeros(2, dtype=torch.int64, device=self.device)
+        elif n_local == self.dtrain.n_rows:
+            root_sum = gq.sum(dim=0, dtype=torch.int64)
+        else:
+            root_sum = gq[ridx.long()].sum(dim=0, dtype=torch.int64)
+        if self.coll.is_distributed:
+            self.coll.allreduce_(root_sum)
+
+        feat_mask = self._sample_features(it, cls + 101 * ptree)
+        fb = self.feat_bins
+        if feat_mask is not None:
+            fb = torch.where(
+                feat_mask, self.feat_bins, torch.zeros_like(self.feat_bins)
+            )
+        cuts_flat_cpu = self.dtrain.cuts.cuts_flat.cpu().numpy()
+        cut_ptr_cpu = self.dtrain.cuts.cut_ptr.cpu().numpy()
+        max_leaves = self.p.max_leaves or (1 << 30)
+        max_depth = self.p.max_depth if self.p.max_depth > 0 else 64
+
+        feat_l, thr_l = [-1], [0.0]
+        left_l, dl_l, val_l, gain_l, cover_l = [-1], [0], [0.0], [0.0], [0.0]
+        parent_l = [-1]
+
+        def scan_nodes(hists, sums):
+            best = ops.find_splits(
+                hists,
+                torch.tensor([sg for sg, sh in sums], dtype=torch.int64,
+                             device=self.device),
+                torch.tensor([sh for sg, sh in sums], dtype=torch.int64,
+                             device=self.device),
+                fb, scale_g, scale_h, self.p.reg_lambda, self.p.reg_alpha,
+                self.p.gamma, self.p.min_child_weight,
+            )
+            return best
+
+        # root
+        root_hist = ops.build_histogram(
+            self.dtrain.bins, gseg, ridx,
+            torch.tensor([0], dtype=torch.int64),
+            torch.tensor([n_local], dtype=torch.int64),
+            self.n_bins, pregathered=True,
+        )
+        if self.coll.is_distributed:
+            self.coll.allreduce_(root_hist)
+        rbest = scan_nodes(root_hist, [(int(root_sum[0]), int(root_sum[1]))])
+        heap = []
+        counter = [0]
+
+        def push(nid, depth, start, count, sg, sh, hist, best_k):
+            gain = float(best_k["gain"])
+            rec = {
+                "nid": nid, "depth": depth, "start": start, "count": count,
+                "sg": sg, "sh": sh, "hist": hist,
+                "feature": int(best_k["feature"]), "bin": int(best_k["bin"]),
+                "dl": int(best_k["default_left"]),
+                "lg": int(best_k["left_g"]), "lh": int(best_k["left_h"]),
+            }
+            if gain > 0 and np.isfinite(gain) and rec["feature"] >= 0:
+                heapq.heappush(heap, (-gain, counter[0], rec))
+                counter[0] += 1
+            else:
+                self._finalize_leaf_rec(rec, val_l, cover_l, scale_h)
+
+        def best_row(best, k):
+            return {key: best[key][k] for key in best}
+
+        push(0, 0, 0, n_local, int(root_sum[0]), int(root_sum[1]),
+             root_hist[0], best_row(rbest, 0))
+        n_leaves = 1
+        while heap and n_leaves < max_leaves:
+            neg_gain, _cnt, rec = heapq.heappop(heap)
+            nid = rec["nid"]
+            f, b = rec["feature"], rec["bin"]
+            thr = float(cuts_flat_cpu[cut_ptr_cpu[f] + b])
+            lid = len(feat_l)
+            feat_l[nid] = f
+            thr_l[nid] = thr
+            left_l[nid] = lid
+            dl_l[nid] = rec["dl"]
+            gain_l[nid] = -neg_gain
+            cover_l[nid] = rec["sh"] / scale_h
+            for _i in range(2):
+                feat_l.append(-1)
+                thr_l.append(0.0)
+                left_l.append(-1)
+                dl_l.append(0)
+                val_l.append(0.0)
+                gain_l.append(0.0)
+                cover_l.append(0.0)
+                parent_l.append(nid)
+            ridx, left_counts, gseg = ops.partition_rows(
+                self.dtrain.bins, ridx,
+                torch.tensor([rec["start"]], dtype=torch.int64),
+                torch.tensor([rec["count"]], dtype=torch.int64),
+                torch.tensor([f], dtype=torch.int32),
+                torch.tensor([b], dtype=torch.int32),
+                torch.tensor([rec["dl"]], dtype=torch.uint8),
+                gpair_seg=gseg,
+            )
+            lcount = int(left_counts[0])
+            lsums = (rec["lg"], rec["lh"])
+            rsums = (rec["sg"] - rec["lg"], rec["sh"] - rec["lh"])
+            lseg = (rec["start"], lcount)
+            rseg = (rec["start"] + lcount, rec["count"] - lcount)
+            children = [
+                (lid, rec["depth"] + 1) + lseg + lsums,
+                (lid + 1, rec["depth"] + 1) + rseg + rsums,
+            ]
+            # build the (globally) smaller child, derive the sibling
+            small_i = 0 if lsums[1] <= rsums[1] else 1
+            sc = children[small_i]
+            sh_hist = ops.build_histogram(
+                self.dtrain.bins, gseg, ridx,
+                torch.tensor([sc[2]], dtype=torch.int64),
+                torch.tensor([sc[3]], dtype=torch.int64),
+                self.n_bins, pregathered=True,
+            )
+            if self.coll.is_distributed:
+                self.coll.allreduce_(sh_hist)
+            other_hist = rec["hist"] - sh_hist[0]
+            rec["hist"] = None
+            if small_i == 0:
+                hists = torch.stack([sh_hist[0], other_hist])
+            else:
+                hists = torch.stack([other_hist, sh_hist[0]])
+            cbest = scan_nodes(
+                hists, [(children[0][4], children[0][5]),
+                        (children[1][4], children[1][5])]
+            )
+            for ci, ch in enumerate(children):
+                if ch[1] >= max_depth:
+                    self._finalize_leaf_rec(
+                        {"start": ch[2], "count": ch[3], "sg": ch[4],
+                         "sh": ch[5]}, val_l, cover_l, scale_h, nid=ch[0],
+                    )
+                else:
+                    push(ch[0], ch[1], ch[2], ch[3], ch[4], ch[5],
+                         hists[ci], best_row(cbest, ci))
+            n_leaves += 1
+
+        # drain remaining candidates as leaves
+        while heap:
+            _, _, rec = heapq.heappop(heap)
+            self._finalize_leaf_rec(rec, val_l, cover_l, scale_h)
+
+        leaf_starts = [st for (st, c, v) in self._leaf_segs]
+        leaf_counts = [c for (st, c, v) in self._leaf_segs]
+        leaf_vals = [v for (st, c, v) in self._leaf_segs]
+        if leaf_starts:
+            ops.update_margins(
+                self.margin if self.n_class == 1 else self.margin[:, cls],
+                ridx,
+                torch.tensor(leaf_starts, dtype=torch.int64),
+                torch.tensor(leaf_counts, dtype=torch.int64),
+                np.asarray(leaf_vals, dtype=np.float32),
+            )
+        self._leaf_segs = []
+        return Tree(
+            feat=np.asarray(feat_l, np.int32),
+            thr=np.asarray(thr_l, np.float32),
+            left=np.asarray(left_l, np.int32),
+            default_left=np.asarray(dl_l, np.uint8),
+            value=np.asarray(val_l, np.float32),
+            gain=np.asarray(gain_l, np.float32),
+            cover=np.asarray(cover_l, np.float32),
+            parent=np.asarray(
+                [2147483647 if pp < 0 else pp for pp in parent_l], np.int32
+            ),
+        )
+
+    def _finalize_leaf_rec(self, rec, val_l, cover_l, scale_h, nid=None):
+        nid = rec["nid"] if nid is None else nid
+        G = rec["sg"] / self._scale_g_cur
+        H = rec["sh"] / scale_h
+        w = _calc_weight(G, H, self.p.reg_lambda, self.p.reg_alpha)
+        if self.p.max_delta_step > 0:
+            w = max(-self.p.max_delta_step, min(self.p.max_delta_step, w))
+        val_l[nid] = self.p.eta * w
+        cover_l[nid] = H
+        self._leaf_segs.append((rec["start"], rec["count"], self.p.eta * w))
 
     # quantization scale of the current tree (set in _grow_tree via _quantize)
     _scale_g_cur: float = 1.0
