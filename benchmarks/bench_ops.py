@@ -109,6 +109,33 @@ def main():
         lambda: torch.sigmoid(m),
     )
 
+    # serving: packed-node tree-walk over a realistic 100-tree depth-8 model
+    from xgboost_ray_amd.engine.trainer import run_training
+
+    small = BinnedMatrix.build(
+        torch.randn(500_000, F, device=dev, generator=gen),
+        label=(torch.rand(500_000, device=dev, generator=gen) > 0.5).float(),
+        max_bin=256,
+    )
+    bst = run_training(
+        {"objective": "binary:logistic", "max_depth": 8, "eta": 0.1}, small, 100
+    )
+    Xp = torch.randn(2_000_000, F, device=dev, generator=gen)
+    flat = bst._flat_trees(dev)
+    out = torch.zeros(2_000_000, device=dev)
+    from xgboost_ray_amd.ops import gpu as gops
+
+    dt = timeit(
+        "predict 2M x 100 trees",
+        lambda: gops.predict_trees(
+            Xp, flat["feat"], flat["thr"], flat["left"],
+            flat["default_left"], flat["value"], flat["tree_ptr"], out,
+        ),
+        reps=10,
+    )
+    print(f"serving throughput: {2_000_000 / dt * 1000 / 1e6:.1f} M rows/s "
+          f"(100 trees, depth 8)")
+
 
 if __name__ == "__main__":
     main()

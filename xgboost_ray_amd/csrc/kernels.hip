@@ -127,7 +127,15 @@ __global__ void gather_gpair_kernel(const int2* __restrict__ gpair,
 // deterministic (the checkpoint-determinism contract).
 // ---------------------------------------------------------------------------
 #define HIST_THREADS 512
-#define HIST_ROWS_PER_WG 16384
+static int hist_rows_per_wg() {
+  static int v = [] {
+    const char* e = getenv("RXGB_HIST_ROWS");
+    int r = e ? atoi(e) : 16384;
+    return (r >= 1024 && r <= 262144) ? r : 16384;
+  }();
+  return v;
+}
+#define HIST_ROWS_PER_WG hist_rows_per_wg()
 
 // VEC16: the binned matrix row stride is 16-byte aligned (padded layout,
 // pad bytes == 255) -> one uint4 load fetches a whole 16-feature block of
@@ -142,7 +150,8 @@ __global__ __launch_bounds__(HIST_THREADS) void build_histogram_kernel(
     const int64_t* __restrict__ node_start,  // [K] segment starts
     const int64_t* __restrict__ chunk_off,   // [K+1] cumulative chunks
     long long* __restrict__ hist,            // [K, F, n_bins, 2]
-    int K, int F, int n_bins, int fb_size, int64_t row_stride, int f_base) {
+    int K, int F, int n_bins, int fb_size, int64_t row_stride, int f_base,
+    int rows_per_wg) {
   // locate (node, chunk) from blockIdx.x via binary search on chunk_off
   int wg = blockIdx.x;
   int lo = 0, hi = K;
@@ -154,7 +163,7 @@ __global__ __launch_bounds__(HIST_THREADS) void build_histogram_kernel(
   const int64_t chunk_in_node = wg - chunk_off[node];
   const int64_t seg_start = node_start[node];
   // rows of this chunk within the node's segment
-  const int64_t row_lo = chunk_in_node * HIST_ROWS_PER_WG;
+  const int64_t row_lo = chunk_in_node * (int64_t)rows_per_wg;
 
   const int fb = blockIdx.y;
   const int f0 = f_base + fb * fb_size;
@@ -169,7 +178,7 @@ __global__ __launch_bounds__(HIST_THREADS) void build_histogram_kernel(
   // (segment order), bin reads are per-row contiguous byte runs.
   // counts are stored at node_start[K..2K) (cat_start_count layout)
   const int64_t node_count = node_start[K + node];
-  int64_t row_hi = row_lo + HIST_ROWS_PER_WG;
+  int64_t row_hi = row_lo + rows_per_wg;
   if (row_hi > node_count) row_hi = node_count;
 
   const int lane = threadIdx.x & (WAVE - 1);
@@ -893,7 +902,8 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                        ridx.data_ptr<int32_t>() + min_start,
                        sc_adj_p, chunk_off_p,
                        reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
-                       K, F, (int)n_bins, fb_size, row_stride, (int)f_lo);
+                       K, F, (int)n_bins, fb_size, row_stride, (int)f_lo,
+                       HIST_ROWS_PER_WG);
   } else if (vec16) {
     hipLaunchKernelGGL((build_histogram_kernel<16>),
                        dim3((uint32_t)total_chunks, n_fb),
@@ -903,7 +913,8 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                        ridx.data_ptr<int32_t>() + min_start,
                        sc_adj_p, chunk_off_p,
                        reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
-                       K, F, (int)n_bins, fb_size, row_stride, (int)f_lo);
+                       K, F, (int)n_bins, fb_size, row_stride, (int)f_lo,
+                       HIST_ROWS_PER_WG);
   } else {
     hipLaunchKernelGGL((build_histogram_kernel<0>),
                        dim3((uint32_t)total_chunks, n_fb),
@@ -913,7 +924,8 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                        ridx.data_ptr<int32_t>() + min_start,
                        sc_adj_p, chunk_off_p,
                        reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
-                       K, F, (int)n_bins, fb_size, row_stride, (int)f_lo);
+                       K, F, (int)n_bins, fb_size, row_stride, (int)f_lo,
+                       HIST_ROWS_PER_WG);
   }
   return hist;
 }
