@@ -361,3 +361,56 @@ class TestMetrics:
         w = torch.tensor([3.0, 1.0])
         v = m.finalize(m.local_stats(margin, label, w, None, None))
         assert abs(v - np.sqrt((3 * 1 + 9) / 4)) < 1e-9
+
+
+class TestMonotone:
+    def test_monotone_increasing_constraint(self):
+        rng = np.random.RandomState(0)
+        n = 6000
+        X = rng.rand(n, 3).astype(np.float32)
+        # f0 has a NON-monotone true effect; the constraint must force a
+        # monotone-increasing model response anyway
+        y = (np.sin(X[:, 0] * 6) + X[:, 1] + 0.1 * rng.randn(n)).astype(
+            np.float32
+        )
+        dm = BinnedMatrix.build(
+            torch.from_numpy(X), label=torch.from_numpy(y), max_bin=64
+        )
+        bst = run_training(
+            {"objective": "reg:squarederror", "max_depth": 5, "eta": 0.3,
+             "monotone_constraints": "(1,0,0)"},
+            dm, 20,
+        )
+        # sweep f0 with the other features fixed
+        grid = np.zeros((50, 3), dtype=np.float32)
+        grid[:, 0] = np.linspace(0, 1, 50)
+        grid[:, 1] = 0.5
+        grid[:, 2] = 0.5
+        pred = bst.predict(grid, output_margin=True)
+        assert (np.diff(pred) >= -1e-6).all(), "response must be monotone"
+        # and without the constraint it must NOT be monotone (sanity)
+        bst2 = run_training(
+            {"objective": "reg:squarederror", "max_depth": 5, "eta": 0.3},
+            dm, 20,
+        )
+        pred2 = bst2.predict(grid, output_margin=True)
+        assert not (np.diff(pred2) >= -1e-6).all()
+
+    def test_monotone_decreasing(self):
+        rng = np.random.RandomState(1)
+        n = 4000
+        X = rng.rand(n, 2).astype(np.float32)
+        y = (-X[:, 0] + 0.5 * X[:, 1] + 0.05 * rng.randn(n)).astype(np.float32)
+        dm = BinnedMatrix.build(
+            torch.from_numpy(X), label=torch.from_numpy(y), max_bin=64
+        )
+        bst = run_training(
+            {"objective": "reg:squarederror", "max_depth": 4, "eta": 0.3,
+             "monotone_constraints": [-1, 0]},
+            dm, 15,
+        )
+        grid = np.zeros((40, 2), dtype=np.float32)
+        grid[:, 0] = np.linspace(0, 1, 40)
+        grid[:, 1] = 0.5
+        pred = bst.predict(grid, output_margin=True)
+        assert (np.diff(pred) <= 1e-6).all()

@@ -321,3 +321,29 @@ def test_full_training_gpu_deterministic(binned):
         outs.append(bst.predict(Xt[:20000].numpy(), output_margin=True))
         del dm
     np.testing.assert_array_equal(outs[0], outs[1])
+
+
+@pytest.mark.gpu
+def test_monotone_training_gpu():
+    from xgboost_ray_amd.engine.quantile import BinnedMatrix
+    from xgboost_ray_amd.engine.trainer import run_training
+
+    rng = np.random.RandomState(0)
+    n = 100_000
+    X = rng.rand(n, 3).astype(np.float32)
+    y = (np.sin(X[:, 0] * 6) + X[:, 1] + 0.1 * rng.randn(n)).astype(np.float32)
+    dm = BinnedMatrix.build(
+        torch.from_numpy(X).cuda(), label=torch.from_numpy(y).cuda(),
+        max_bin=128,
+    )
+    bst = run_training(
+        {"objective": "reg:squarederror", "max_depth": 6, "eta": 0.3,
+         "monotone_constraints": "(1,0,0)"},
+        dm, 20,
+    )
+    grid = np.zeros((100, 3), dtype=np.float32)
+    grid[:, 0] = np.linspace(0, 1, 100)
+    grid[:, 1] = 0.5
+    grid[:, 2] = 0.5
+    pred = bst.predict(grid, output_margin=True)
+    assert (np.diff(pred) >= -1e-6).all()

@@ -300,11 +300,23 @@ __device__ inline double calc_score(double G, double H, double lam, double alpha
 // histogram into LDS (coalesced), then lane 0 runs the sequential scan
 // from LDS in the exact CPU FP order. ~100x the parallelism of a
 // thread-per-(k,f) scan at shallow depths, with identical numerics.
+__device__ inline double calc_weight_d(double G, double H, double lam,
+                                       double alpha) {
+  if (alpha > 0.0) {
+    double t = fabs(G) - alpha;
+    G = t > 0.0 ? copysign(t, G) : 0.0;
+  }
+  double denom = H + lam;
+  return denom > 0.0 ? -G / denom : 0.0;
+}
+
 __global__ __launch_bounds__(256) void find_splits_kf_kernel(
     const long long* __restrict__ hist,  // [K, F, B, 2]
     const long long* __restrict__ parent_g, const long long* __restrict__ parent_h,
     const int32_t* __restrict__ feat_bins, double scale_g, double scale_h,
     double lam, double alpha, double gamma, double mcw,
+    const int8_t* __restrict__ mono,    // [F] or nullptr
+    const double* __restrict__ bounds,  // [K, 2] node weight bounds
     double* __restrict__ out_gain,     // [K, F]
     int32_t* __restrict__ out_bin,     // [K, F]
     uint8_t* __restrict__ out_dl,      // [K, F]
@@ -374,6 +386,18 @@ __global__ __launch_bounds__(256) void find_splits_kf_kernel(
       double hl = dl ? HL + Hmiss : HL;
       double gr = Gp - gl, hr = Hp - hl;
       if (hl < mcw || hr < mcw) continue;
+      if (mono != nullptr) {
+        const int c = mono[f];
+        if (c != 0) {
+          const double blo = bounds[(size_t)k * 2];
+          const double bup = bounds[(size_t)k * 2 + 1];
+          double wl = calc_weight_d(gl, hl, lam, alpha);
+          double wr = calc_weight_d(gr, hr, lam, alpha);
+          wl = fmin(fmax(wl, blo), bup);
+          wr = fmin(fmax(wr, blo), bup);
+          if (c > 0 ? (wl > wr) : (wl < wr)) continue;
+        }
+      }
       double gain = 0.5 * (calc_score(gl, hl, lam, alpha) +
                            calc_score(gr, hr, lam, alpha) - parent_score) -
                     gamma;
@@ -934,7 +958,8 @@ std::vector<torch::Tensor> find_splits(torch::Tensor hist, torch::Tensor parent_
                                        torch::Tensor parent_h,
                                        torch::Tensor feat_bins, double scale_g,
                                        double scale_h, double lam, double alpha,
-                                       double gamma, double mcw) {
+                                       double gamma, double mcw,
+                                       torch::Tensor mono, torch::Tensor bounds) {
   const int K = (int)hist.size(0);
   const int F = (int)hist.size(1);
   const int B = (int)hist.size(2);
@@ -959,7 +984,10 @@ std::vector<torch::Tensor> find_splits(torch::Tensor hist, torch::Tensor parent_
                      reinterpret_cast<const long long*>(parent_g.data_ptr<int64_t>()),
                      reinterpret_cast<const long long*>(parent_h.data_ptr<int64_t>()),
                      feat_bins.data_ptr<int32_t>(), scale_g, scale_h, lam,
-                     alpha, gamma, mcw, kf_gain.data_ptr<double>(),
+                     alpha, gamma, mcw,
+                     mono.numel() ? mono.data_ptr<int8_t>() : nullptr,
+                     bounds.numel() ? bounds.data_ptr<double>() : nullptr,
+                     kf_gain.data_ptr<double>(),
                      kf_bin.data_ptr<int32_t>(), kf_dl.data_ptr<uint8_t>(),
                      reinterpret_cast<long long*>(kf_lg.data_ptr<int64_t>()),
                      reinterpret_cast<long long*>(kf_lh.data_ptr<int64_t>()), K, F, B);

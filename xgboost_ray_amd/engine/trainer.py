@@ -54,6 +54,7 @@ class TrainParams:
     seed: int = 0
     grow_policy: str = "depthwise"
     max_leaves: int = 0
+    monotone_constraints: object = None
     tree_method: str = "hist"
     verbosity: int = 1
     nthread: int = 0
@@ -83,6 +84,11 @@ class TrainParams:
         p.num_class = int(p.num_class or 0)
         p.seed = int(p.seed or 0)
         p.num_parallel_tree = max(1, int(p.num_parallel_tree or 1))
+        if isinstance(p.monotone_constraints, str):
+            txt = p.monotone_constraints.strip().strip("()")
+            p.monotone_constraints = [
+                int(x) for x in txt.split(",") if x.strip()
+            ]
         return p
 
 
@@ -109,6 +115,8 @@ class _Node:
     sum_h: int
     slot: int = 0  # scan-slot in this depth's all_hist
     parent_slot: int = 0  # parent's scan-slot in the previous depth
+    w_lower: float = -math.inf  # monotone weight bounds
+    w_upper: float = math.inf
 
 
 class CallbackList:
@@ -170,6 +178,19 @@ class BoostingEngine:
             self.obj.validate_label(dtrain.label)
         self.feat_bins = dtrain.cuts.feat_bins().to(self.device)
         self.n_bins = dtrain.cuts.max_bins
+        self.mono = None
+        mc = self.p.monotone_constraints
+        if mc:
+            arr = np.zeros(dtrain.n_features, dtype=np.int8)
+            for i, c in enumerate(list(mc)[: dtrain.n_features]):
+                arr[i] = int(c)
+            if arr.any():
+                self.mono = torch.from_numpy(arr).to(self.device)
+                if self.p.grow_policy == "lossguide":
+                    raise ValueError(
+                        "monotone_constraints are not supported with "
+                        "grow_policy=lossguide yet"
+                    )
         self.iteration = 0
         self.booster = Booster(
             params={
@@ -511,6 +532,12 @@ class BoostingEngine:
                     mask, self.feat_bins, torch.zeros_like(self.feat_bins)
                 )
             _tick("stack")
+            mono_bounds = None
+            if self.mono is not None:
+                mono_bounds = torch.tensor(
+                    [[nd.w_lower, nd.w_upper] for nd in order_nodes],
+                    dtype=torch.float64,
+                )
             best = ops.find_splits(
                 all_hist,
                 pg,
@@ -522,6 +549,8 @@ class BoostingEngine:
                 self.p.reg_alpha,
                 self.p.gamma,
                 self.p.min_child_weight,
+                monotone=self.mono,
+                bounds=mono_bounds,
             )
             _tick("scan")
             gain = best["gain"]
@@ -605,6 +634,8 @@ class BoostingEngine:
                     sum_g=lg,
                     sum_h=lh,
                     parent_slot=nd.slot,
+                    w_lower=nd.w_lower,
+                    w_upper=nd.w_upper,
                 )
                 rnode = _Node(
                     nid=lid + 1,
@@ -614,7 +645,28 @@ class BoostingEngine:
                     sum_g=nd.sum_g - lg,
                     sum_h=nd.sum_h - lh,
                     parent_slot=nd.slot,
+                    w_lower=nd.w_lower,
+                    w_upper=nd.w_upper,
                 )
+                if self.mono is not None:
+                    c = int(self.mono[sf[k]])
+                    if c != 0:
+                        wl = _clamp(
+                            _calc_weight(lg / scale_g, lh / scale_h,
+                                         self.p.reg_lambda, self.p.reg_alpha),
+                            nd.w_lower, nd.w_upper)
+                        wr = _clamp(
+                            _calc_weight((nd.sum_g - lg) / scale_g,
+                                         (nd.sum_h - lh) / scale_h,
+                                         self.p.reg_lambda, self.p.reg_alpha),
+                            nd.w_lower, nd.w_upper)
+                        mid = 0.5 * (wl + wr)
+                        if c > 0:
+                            lnode.w_upper = min(lnode.w_upper, mid)
+                            rnode.w_lower = max(rnode.w_lower, mid)
+                        else:
+                            lnode.w_lower = max(lnode.w_lower, mid)
+                            rnode.w_upper = min(rnode.w_upper, mid)
                 new_frontier.extend([lnode, rnode])
             frontier = new_frontier
             prev_all_hist = all_hist
@@ -664,6 +716,8 @@ class BoostingEngine:
         G = nd.sum_g / self._scale_g_cur
         H = nd.sum_h / scale_h
         w = _calc_weight(G, H, self.p.reg_lambda, self.p.reg_alpha)
+        if self.mono is not None:
+            w = _clamp(w, nd.w_lower, nd.w_upper)
         if self.p.max_delta_step > 0:
             w = max(-self.p.max_delta_step, min(self.p.max_delta_step, w))
         val_l[nd.nid] = self.p.eta * w
@@ -956,6 +1010,10 @@ class BoostingEngine:
 
     # quantization scale of the current tree (set in _grow_tree via _quantize)
     _scale_g_cur: float = 1.0
+
+
+def _clamp(v: float, lo: float, up: float) -> float:
+    return max(lo, min(up, v))
 
 
 def _calc_weight(G: float, H: float, reg_lambda: float, reg_alpha: float) -> float:

@@ -92,6 +92,8 @@ def find_splits(
     reg_alpha,
     gamma,
     min_child_weight,
+    monotone=None,
+    bounds=None,
 ):
     """Best-split scan over histograms.
 
@@ -113,6 +115,8 @@ def find_splits(
         reg_alpha,
         gamma,
         min_child_weight,
+        monotone,
+        bounds,
     )
 
 

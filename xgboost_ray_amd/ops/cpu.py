@@ -93,6 +93,13 @@ def _calc_score(G, H, reg_lambda, reg_alpha):
     return torch.where(denom > 0, G * G / denom, torch.zeros_like(G))
 
 
+def _calc_weight_t(G, H, reg_lambda, reg_alpha):
+    if reg_alpha > 0:
+        G = torch.sign(G) * torch.clamp(G.abs() - reg_alpha, min=0.0)
+    denom = H + reg_lambda
+    return torch.where(denom > 0, -G / denom, torch.zeros_like(G))
+
+
 def find_splits(
     hist,
     parent_g,
@@ -104,6 +111,8 @@ def find_splits(
     reg_alpha,
     gamma,
     min_child_weight,
+    monotone=None,
+    bounds=None,
 ):
     K, F, B, _ = hist.shape
     dev = hist.device
@@ -155,6 +164,22 @@ def find_splits(
             & (hl >= min_child_weight)
             & (hr >= min_child_weight)
         )
+        if monotone is not None:
+            # monotone constraints: clamp child weights into the node's
+            # bound interval, then require the constrained ordering
+            c = monotone.to(dev).view(1, F, 1)
+            lo = bounds[:, 0].to(dev).view(K, 1, 1)
+            up = bounds[:, 1].to(dev).view(K, 1, 1)
+            wl = torch.clamp(
+                _calc_weight_t(gl, hl, reg_lambda, reg_alpha), lo, up
+            )
+            wr = torch.clamp(
+                _calc_weight_t(gr, hr, reg_lambda, reg_alpha), lo, up
+            )
+            ok = ok & torch.where(
+                c > 0, wl <= wr, torch.where(c < 0, wl >= wr,
+                                             torch.ones_like(ok)),
+            )
         gain = (
             0.5
             * (

@@ -79,10 +79,15 @@ def find_splits(
     reg_alpha,
     gamma,
     min_child_weight,
+    monotone=None,
+    bounds=None,
 ):
     import numpy as np
 
     dev = hist.device
+    if monotone is None:
+        monotone = torch.zeros(0, dtype=torch.int8, device=dev)
+        bounds = torch.zeros((0, 2), dtype=torch.float64, device=dev)
     (packed,) = _load().find_splits(
         hist,
         parent_g.to(dev),
@@ -94,6 +99,8 @@ def find_splits(
         float(reg_alpha),
         float(gamma),
         float(min_child_weight),
+        monotone.to(dev).to(torch.int8),
+        bounds.to(dev).to(torch.float64),
     )
     arr = packed.cpu().numpy()  # ONE D2H for the whole depth's splits
     return {
