@@ -47,3 +47,49 @@ def test_train_gpu_fault_restart(tmp_path):
                                lock_dir=str(tmp_path))],
     )
     assert bst.num_boosted_rounds() == 12
+
+
+def test_streaming_train_gpu(tmp_path):
+    """Out-of-core path (RayDeviceQuantileDMatrix) through GPU actors."""
+    import pandas as pd
+
+    from xgboost_ray_amd import RayDeviceQuantileDMatrix
+
+    rng = np.random.RandomState(0)
+    files = []
+    for i in range(4):
+        Xc = rng.randn(100_000, 10).astype(np.float32)
+        yc = ((Xc[:, 0] + Xc[:, 1]) > 0).astype(np.float32)
+        df = pd.DataFrame(Xc, columns=[f"f{j}" for j in range(10)])
+        df["label"] = yc
+        path = str(tmp_path / f"s{i}.parquet")
+        df.to_parquet(path)
+        files.append(path)
+    res = {}
+    dm = RayDeviceQuantileDMatrix(files, label="label")
+    bst = train(
+        {"objective": "binary:logistic", "tree_method": "gpu_hist",
+         "max_depth": 6, "eval_metric": ["auc"]},
+        dm,
+        8,
+        evals=[(dm, "train")],
+        evals_result=res,
+        ray_params=RayParams(num_actors=1, gpus_per_actor=1),
+    )
+    assert bst.num_boosted_rounds() == 8
+    assert res["train"]["auc"][-1] > 0.9
+
+
+def test_deep_tree_gpu():
+    """depth 12 / 4096-node frontiers exercise the chunked machinery."""
+    X, y = create_data(300_000, 10)
+    res = {}
+    train(
+        {"objective": "binary:logistic", "tree_method": "gpu_hist",
+         "max_depth": 12, "eta": 0.3, "eval_metric": ["logloss"]},
+        RayDMatrix(X, label=y), 4,
+        evals_result=res,
+        evals=[(RayDMatrix(X, label=y), "train")],
+        ray_params=RayParams(num_actors=1, gpus_per_actor=1),
+    )
+    assert res["train"]["logloss"][-1] < 0.3
