@@ -128,10 +128,13 @@ def main():
     w = np.linspace(-1, 1, args.cols)
     y = ((X @ w + 0.5 * rng.randn(args.rows)) > 0).astype(np.float32)
 
-    for condition in (
-        "fewer_workers", "non_elastic", "elastic_no_comeback",
-        "elastic_comeback",
-    ):
+    conditions = ["fewer_workers", "non_elastic"]
+    if args.affected > 0:
+        conditions += ["elastic_no_comeback", "elastic_comeback"]
+    else:
+        print("affected=0: skipping elastic conditions "
+              "(max_failed_actors must be > 0)")
+    for condition in conditions:
         out = run_condition(
             condition, X, y, args.workers, args.rounds, args.affected,
             args.gpu,
