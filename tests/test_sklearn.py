@@ -159,3 +159,26 @@ class TestSklearnInterop:
         clf.fit(dm, None, ray_params=RP)
         pred = clf.predict(RayDMatrix(X), ray_params=RP)
         assert len(pred) == 800
+
+
+class TestSklearnEcosystem:
+    def test_grid_search_cv(self):
+        from sklearn.model_selection import GridSearchCV
+
+        X, y = create_data(600, 4)
+        clf = RayXGBClassifier(n_estimators=3, max_depth=3, n_jobs=1)
+        gs = GridSearchCV(
+            clf, {"max_depth": [2, 3]}, cv=2, scoring="accuracy"
+        )
+        gs.fit(X, y)
+        assert gs.best_params_["max_depth"] in (2, 3)
+        assert gs.best_score_ > 0.8
+
+    def test_cross_val_score(self):
+        from sklearn.model_selection import cross_val_score
+
+        X, y = create_data(600, 4, kind="reg")
+        reg = RayXGBRegressor(n_estimators=4, max_depth=3, n_jobs=1)
+        scores = cross_val_score(reg, X, y, cv=2, scoring="r2")
+        assert len(scores) == 2
+        assert scores.mean() > 0.5
