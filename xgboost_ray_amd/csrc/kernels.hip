@@ -651,30 +651,60 @@ __global__ void pack_nodes_kernel(
   packed[i] = p;
 }
 
+// R consecutive rows per thread: their tree walks interleave, so R
+// independent node loads are in flight per step (the walk is L2-latency
+// bound; one chain per thread leaves the memory pipe idle).
+template <int R>
 __global__ void predict_trees_kernel(
     const float* __restrict__ X, const uint4* __restrict__ nodes,
     const int32_t* __restrict__ tree_ptr, float* __restrict__ out,
     float tree_weight, int64_t n, int F, int T) {
-  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (; i < n; i += stride) {
-    const float* row = X + i * (int64_t)F;
-    float acc = 0.0f;
+  int64_t grp = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t gstride = (int64_t)gridDim.x * blockDim.x;
+  const int64_t n_grp = (n + R - 1) / R;
+  for (; grp < n_grp; grp += gstride) {
+    const int64_t i0 = grp * R;
+    float acc[R];
+    bool valid[R];
+    #pragma unroll
+    for (int r = 0; r < R; ++r) {
+      acc[r] = 0.0f;
+      valid[r] = (i0 + r) < n;
+    }
     for (int t = 0; t < T; ++t) {
       const int base = tree_ptr[t];
-      int nid = base;
-      uint4 p = nodes[nid];
-      while (!(p.x & 0x80000000u)) {
-        const float v = row[p.x & 0x3FFFFFFFu];
-        const bool goleft =
-            isnan(v) ? ((p.x & 0x40000000u) != 0)
-                     : (v < __uint_as_float(p.y));
-        nid = base + (int)p.z + (goleft ? 0 : 1);
-        p = nodes[nid];
+      uint4 p[R];
+      bool act[R];
+      #pragma unroll
+      for (int r = 0; r < R; ++r) {
+        act[r] = valid[r];
+        p[r] = nodes[base];
       }
-      acc += __uint_as_float(p.y);
+      bool any = true;
+      while (any) {
+        any = false;
+        #pragma unroll
+        for (int r = 0; r < R; ++r) {
+          if (!act[r]) continue;
+          if (p[r].x & 0x80000000u) {
+            acc[r] += __uint_as_float(p[r].y);
+            act[r] = false;
+          } else {
+            const float v = X[(i0 + r) * (int64_t)F +
+                              (p[r].x & 0x3FFFFFFFu)];
+            const bool goleft =
+                isnan(v) ? ((p[r].x & 0x40000000u) != 0)
+                         : (v < __uint_as_float(p[r].y));
+            p[r] = nodes[base + (int)p[r].z + (goleft ? 0 : 1)];
+            any = true;
+          }
+        }
+      }
     }
-    out[i] += acc * tree_weight;
+    #pragma unroll
+    for (int r = 0; r < R; ++r) {
+      if (valid[r]) out[i0 + r] += acc[r] * tree_weight;
+    }
   }
 }
 
@@ -1108,8 +1138,8 @@ void predict_trees(torch::Tensor X, torch::Tensor feat, torch::Tensor thr,
                      left.data_ptr<int32_t>(), default_left.data_ptr<uint8_t>(),
                      value.data_ptr<float>(),
                      (uint4*)packed.data_ptr<int32_t>(), n_nodes);
-  int64_t blocks = std::min<int64_t>(ceil_div(n, 256), 8192);
-  hipLaunchKernelGGL(predict_trees_kernel, dim3(blocks), dim3(256), 0,
+  int64_t blocks = std::min<int64_t>(ceil_div(ceil_div(n, 4), 256), 8192);
+  hipLaunchKernelGGL((predict_trees_kernel<4>), dim3(blocks), dim3(256), 0,
                      stream.stream(), X.data_ptr<float>(),
                      (const uint4*)packed.data_ptr<int32_t>(),
                      tree_ptr.data_ptr<int32_t>(),
