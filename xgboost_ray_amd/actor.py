@@ -12,7 +12,6 @@ engine checks it in an after-iteration callback, so no thread is needed).
 
 import os
 import pickle
-import sys
 import threading
 import time
 import traceback
@@ -131,7 +130,6 @@ class _ActorWorker:
     ) -> Dict[str, Any]:
         import torch
 
-        from xgboost_ray_amd.booster import Booster
         from xgboost_ray_amd.engine.collective import Collective
         from xgboost_ray_amd.engine.quantile import BinnedMatrix
         from xgboost_ray_amd.engine.trainer import EvalPack, run_training

@@ -9,7 +9,7 @@ XGBoost JSON model schema so models interoperate with stock XGBoost
 """
 
 import json
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional, Sequence, Union
 
 import numpy as np

@@ -11,7 +11,7 @@ multi-node sources keep working unchanged.
 """
 
 from collections import defaultdict
-from typing import Any, Dict, Optional, Sequence
+from typing import Any, Dict, Sequence
 
 
 def get_actor_rank_ips(actors: Sequence) -> Dict[int, str]:

@@ -4,7 +4,7 @@ Single files or lists of files; with a list, ``indices`` selects which
 files an actor loads (file-level distributed sharding).
 """
 
-from typing import Any, List, Optional, Sequence, Union
+from typing import Any, Optional, Sequence, Union
 
 import pandas as pd
 

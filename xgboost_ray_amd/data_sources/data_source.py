@@ -1,7 +1,7 @@
 """DataSource abstraction (reference data_sources/data_source.py:22-155)."""
 
 from enum import Enum
-from typing import Any, Dict, List, Optional, Sequence, Tuple, Union
+from typing import Any, Dict, List, Optional, Sequence, Tuple
 
 import pandas as pd
 

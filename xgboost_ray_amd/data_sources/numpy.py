@@ -1,6 +1,6 @@
 """Numpy data source (reference data_sources/numpy.py:13-33)."""
 
-from typing import Any, List, Optional, Sequence
+from typing import Any, Optional, Sequence
 
 import numpy as np
 import pandas as pd

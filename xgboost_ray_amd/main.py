@@ -23,7 +23,7 @@ import threading
 import time
 import warnings
 from dataclasses import dataclass, field
-from typing import Any, Callable, Dict, List, Optional, Sequence, Tuple, Union
+from typing import Any, Dict, List, Optional, Sequence, Tuple, Union
 
 import numpy as np
 import torch
@@ -37,7 +37,7 @@ from xgboost_ray_amd.actor import (
 from xgboost_ray_amd.booster import Booster
 from xgboost_ray_amd.env import ENV
 from xgboost_ray_amd.matrix import RayDMatrix, RayShardingMode, combine_data
-from xgboost_ray_amd.util import Future, find_free_port, wait_futures
+from xgboost_ray_amd.util import Future, find_free_port
 
 
 class RayXGBoostTrainingError(TrainingError):

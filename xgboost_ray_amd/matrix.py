@@ -18,11 +18,7 @@ import numpy as np
 import pandas as pd
 
 from xgboost_ray_amd import shm_store
-from xgboost_ray_amd.data_sources import (
-    DataSource,
-    RayFileType,
-    data_sources,
-)
+from xgboost_ray_amd.data_sources import RayFileType, data_sources
 from xgboost_ray_amd.data_sources.object_store import ObjectStore
 
 Data = Union[str, List[str], np.ndarray, pd.DataFrame, pd.Series]

@@ -7,7 +7,7 @@ mapping (reference sklearn.py:341-355) are preserved.
 """
 
 import warnings
-from typing import Any, Dict, List, Optional, Tuple, Union
+from typing import Dict, Optional, Union
 
 import numpy as np
 

@@ -8,7 +8,7 @@ queues are kernel objects shared with actor processes at spawn.
 import socket
 import threading
 import time
-from typing import Callable, List, Optional
+from typing import List, Optional
 
 
 def find_free_port() -> int:
