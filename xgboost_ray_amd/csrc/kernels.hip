@@ -142,6 +142,41 @@ static int hist_rows_per_wg() {
 // a row. Lane rotation spreads one wave-instruction's LDS atomics over 16
 // different features (max 4 lanes per feature histogram) so same-bin
 // serialization on skewed features drops ~16x.
+// One row's 16-feature uint4 -> LDS tile atomics with word-granular lane
+// rotation (shared by the single-block and multi-block kernels below).
+__device__ __forceinline__ void hist_accum_row16(
+    unsigned long long* lds_hist, const uint8_t* __restrict__ bins,
+    uint64_t r, int64_t row_stride, int f0, int n_bins, int lane,
+    longlong2 gp) {
+  const uint4 packed =
+      *reinterpret_cast<const uint4*>(bins + r * row_stride + f0);
+  const int r4 = lane & 3;
+  const uint32_t w0 = packed.x, w1 = packed.y, w2 = packed.z, w3 = packed.w;
+  const bool s1 = (r4 & 1) != 0, s2 = (r4 & 2) != 0;
+  const uint32_t t01 = s1 ? w1 : w0, t23 = s1 ? w3 : w2;
+  const uint32_t u01 = s1 ? w2 : w1, u23 = s1 ? w0 : w3;
+  const uint32_t rw0 = s2 ? t23 : t01;  // w[(0+r4)&3]
+  const uint32_t rw1 = s2 ? u23 : u01;  // w[(1+r4)&3]
+  const uint32_t rw2 = s2 ? t01 : t23;  // w[(2+r4)&3]
+  const uint32_t rw3 = s2 ? u01 : u23;  // w[(3+r4)&3]
+  const uint32_t rws[4] = {rw0, rw1, rw2, rw3};
+  #pragma unroll
+  for (int jj = 0; jj < 4; ++jj) {
+    const uint32_t w = rws[jj];
+    const int fw = ((jj + r4) & 3) * 4;
+    #pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+      const int b = (w >> (8 * kk)) & 0xFF;
+      if (b != 255) {
+        const int f = fw + kk;
+        unsigned long long* cell = &lds_hist[((size_t)f * n_bins + b) * 2];
+        atomicAdd(cell, (unsigned long long)gp.x);
+        atomicAdd(cell + 1, (unsigned long long)gp.y);
+      }
+    }
+  }
+}
+
 template <int VFB>  // 16: uint4 row loads; 8: uint2; 0: byte fallback
 __global__ __launch_bounds__(HIST_THREADS) void build_histogram_kernel(
     const uint8_t* __restrict__ bins,        // [n_rows_total, row_stride]
@@ -195,35 +230,7 @@ __global__ __launch_bounds__(HIST_THREADS) void build_histogram_kernel(
       // serializations on skewed features). The rotation is done with
       // branchless selects on NAMED registers - a runtime-indexed byte
       // array would go to scratch (5x slowdown).
-      const uint4 packed =
-          *reinterpret_cast<const uint4*>(bins + r * row_stride + f0);
-      const int r4 = lane & 3;
-      const uint32_t w0 = packed.x, w1 = packed.y, w2 = packed.z,
-                     w3 = packed.w;
-      const bool s1 = (r4 & 1) != 0, s2 = (r4 & 2) != 0;
-      const uint32_t t01 = s1 ? w1 : w0, t23 = s1 ? w3 : w2;
-      const uint32_t u01 = s1 ? w2 : w1, u23 = s1 ? w0 : w3;
-      const uint32_t rw0 = s2 ? t23 : t01;  // w[(0+r4)&3]
-      const uint32_t rw1 = s2 ? u23 : u01;  // w[(1+r4)&3]
-      const uint32_t rw2 = s2 ? t01 : t23;  // w[(2+r4)&3]
-      const uint32_t rw3 = s2 ? u01 : u23;  // w[(3+r4)&3]
-      const uint32_t rws[4] = {rw0, rw1, rw2, rw3};
-      #pragma unroll
-      for (int jj = 0; jj < 4; ++jj) {
-        const uint32_t w = rws[jj];
-        const int fw = ((jj + r4) & 3) * 4;
-        #pragma unroll
-        for (int kk = 0; kk < 4; ++kk) {
-          const int b = (w >> (8 * kk)) & 0xFF;
-          if (b != 255) {
-            const int f = fw + kk;
-            unsigned long long* cell =
-                &lds_hist[((size_t)f * n_bins + b) * 2];
-            atomicAdd(cell, (unsigned long long)gp.x);
-            atomicAdd(cell + 1, (unsigned long long)gp.y);
-          }
-        }
-      }
+      hist_accum_row16(lds_hist, bins, r, row_stride, f0, n_bins, lane, gp);
     } else if constexpr (VFB == 8) {
       const uint2 packed =
           *reinterpret_cast<const uint2*>(bins + r * row_stride + f0);
@@ -271,6 +278,89 @@ __global__ __launch_bounds__(HIST_THREADS) void build_histogram_kernel(
     if (v != 0ull) {
       atomicAdd((unsigned long long*)&ghist[i], v);
     }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// build_histogram_multifb: multi-feature-block variant for wide matrices.
+//
+// At F=200 (13 blocks of 16) the baseline kernel re-reads gpair_seg (8 B)
+// + ridx (4 B) per row per block: 156 B/row/depth where only 12 B are
+// needed. Here ONE workgroup loops over every feature block of its range,
+// keeping its R rows/thread of gradient pairs + row indices in registers
+// across blocks. Occupancy is LDS-bound anyway (64 KiB tile -> 2 WGs/CU
+// = 4 waves/SIMD), so up to ~128 VGPRs the caching is free.
+// Same LDS tile, same lane rotation, same int64 atomics => bitwise
+// identical histograms to the baseline kernel.
+// ---------------------------------------------------------------------------
+template <int R>  // rows cached per thread; rows_per_wg = R * HIST_THREADS
+__global__ __launch_bounds__(HIST_THREADS) void build_histogram_multifb_kernel(
+    const uint8_t* __restrict__ bins,        // [n_rows_total, row_stride]
+    const int2* __restrict__ gpair_seg,      // [seg_total] segment order
+    const int32_t* __restrict__ ridx,        // [seg_total]
+    const int64_t* __restrict__ node_start,  // [K] starts, [K..2K) counts
+    const int64_t* __restrict__ chunk_off,   // [K+1] cumulative chunks
+    long long* __restrict__ hist,            // [K, F, n_bins, 2]
+    int K, int F, int n_bins, int n_fb, int64_t row_stride, int f_base) {
+  int wg = blockIdx.x;
+  int lo = 0, hi = K;
+  while (lo + 1 < hi) {
+    int m = (lo + hi) >> 1;
+    if (chunk_off[m] <= wg) lo = m; else hi = m;
+  }
+  const int node = lo;
+  const int64_t chunk_in_node = wg - chunk_off[node];
+  const int64_t seg_start = node_start[node];
+  const int64_t row_lo = chunk_in_node * (int64_t)(R * HIST_THREADS);
+  const int64_t node_count = node_start[K + node];
+  int64_t row_hi = row_lo + (int64_t)(R * HIST_THREADS);
+  if (row_hi > node_count) row_hi = node_count;
+
+  // Cache this thread's rows once. Strided assignment (i = row_lo + tid +
+  // rr*512) keeps the loads coalesced per wave; validity is a prefix in
+  // rr, so the block loop can predicate on rr < nvalid.
+  int2 gp_reg[R];
+  uint32_t rx_reg[R];
+  int nvalid = 0;
+  #pragma unroll
+  for (int rr = 0; rr < R; ++rr) {
+    const int64_t i = row_lo + threadIdx.x + (int64_t)rr * HIST_THREADS;
+    if (i < row_hi) {
+      const int64_t s = seg_start + i;
+      gp_reg[rr] = gpair_seg[s];
+      rx_reg[rr] = (uint32_t)ridx[s];
+      nvalid = rr + 1;
+    }
+  }
+
+  extern __shared__ unsigned long long lds_hist[];
+  const int tile_full = 16 * n_bins * 2;
+  const int lane = threadIdx.x & (WAVE - 1);
+  for (int blk = 0; blk < n_fb; ++blk) {
+    const int f0 = f_base + blk * 16;
+    for (int i = threadIdx.x; i < tile_full; i += blockDim.x)
+      lds_hist[i] = 0ull;
+    __syncthreads();
+    #pragma unroll
+    for (int rr = 0; rr < R; ++rr) {
+      if (rr < nvalid) {
+        const longlong2 gp = {(long long)gp_reg[rr].x,
+                              (long long)gp_reg[rr].y};
+        hist_accum_row16(lds_hist, bins, (uint64_t)rx_reg[rr], row_stride,
+                         f0, n_bins, lane, gp);
+      }
+    }
+    __syncthreads();
+    // merge only real features (partial last block: pad bins==255 were
+    // skipped in accumulation, but hist rows past F-1 must not be touched)
+    const int fcount = 16 < (F - f0) ? 16 : (F - f0);
+    const int tile = fcount * n_bins * 2;
+    long long* ghist = hist + (((size_t)node * F + f0) * n_bins) * 2;
+    for (int i = threadIdx.x; i < tile; i += blockDim.x) {
+      const unsigned long long v = lds_hist[i];
+      if (v != 0ull) atomicAdd((unsigned long long*)&ghist[i], v);
+    }
+    __syncthreads();
   }
 }
 
@@ -878,9 +968,35 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
   if (K == 0) return hist;
   auto starts_cpu = starts.to(torch::kCPU).to(torch::kInt64);
   auto counts_cpu = counts.to(torch::kCPU).to(torch::kInt64);
+
+  // feature-block config decides the kernel flavor, which decides the
+  // row-chunk size, so resolve it before computing chunk offsets.
+  const int64_t row_stride0 = bins.stride(0);
+  const bool vec16_pre = (row_stride0 % 16 == 0) && n_bins <= 256;
+  int fb_pre = 16;
+  if (const char* e = getenv("RXGB_HIST_FB")) {
+    if (atoi(e) == 8) fb_pre = 8;
+  }
+  const int n_fb_pre =
+      vec16_pre ? (int)ceil_div(f_hi - f_lo, (int64_t)fb_pre) : 1;
+  // multifb: one WG sweeps all feature blocks with register-cached
+  // gpairs/ridx (see kernel comment). Default on for multi-block ranges;
+  // RXGB_HIST_MULTIFB=0 disables, RXGB_HIST_MULTIFB_R in {8,16,32}.
+  bool multifb = vec16_pre && fb_pre == 16 && n_fb_pre >= 2;
+  if (const char* e = getenv("RXGB_HIST_MULTIFB")) {
+    if (atoi(e) == 0) multifb = false;
+  }
+  int mfb_r = 16;
+  if (const char* e = getenv("RXGB_HIST_MULTIFB_R")) {
+    int v = atoi(e);
+    if (v == 8 || v == 16 || v == 32) mfb_r = v;
+  }
+  const int rows_per_wg =
+      multifb ? mfb_r * HIST_THREADS : HIST_ROWS_PER_WG;
+
   int64_t total_chunks = 0;
   auto chunk_off_cpu =
-      make_chunk_off_cpu(counts_cpu, HIST_ROWS_PER_WG, &total_chunks);
+      make_chunk_off_cpu(counts_cpu, rows_per_wg, &total_chunks);
   if (total_chunks == 0) return hist;
 
   // gather gpairs into segment order once (coalesced hist reads)
@@ -929,25 +1045,39 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
   // 64 KiB dynamic-LDS default so two workgroups co-reside per CU.
   // Padded (16B-aligned row stride) matrices take the vectorized path:
   // fb = 16 features per uint4 row load.
-  const int64_t row_stride = bins.stride(0);
-  const bool vec16 = (row_stride % 16 == 0) && n_bins <= 256;
+  const int64_t row_stride = row_stride0;
+  const bool vec16 = vec16_pre;
   int fb_size = vec16
                     ? 16
                     : (int)std::min<int64_t>(F, (64 * 1024) / (n_bins * 16));
   if (fb_size < 1) fb_size = 1;
   // experiment knob: RXGB_HIST_FB=8 halves the LDS tile (4 workgroups/CU
   // instead of 2) at the cost of 2x gradient re-reads per depth
-  if (const char* e = getenv("RXGB_HIST_FB")) {
-    int v = atoi(e);
-    if (v == 8 && vec16) fb_size = 8;
-  }
+  if (vec16 && fb_pre == 8) fb_size = 8;
   TORCH_CHECK(f_lo % fb_size == 0 && f_lo < f_hi && f_hi <= F,
               "feature range must align to the block size ", fb_size);
   const int n_fb = (int)ceil_div(f_hi - f_lo, fb_size);
   const size_t lds = (size_t)fb_size * n_bins * 2 * sizeof(long long);
 
   // ridx pointer offset so seg indices align with gpair_seg
-  if (vec16 && fb_size == 8) {
+  if (multifb) {
+    auto launch_mfb = [&](auto rc) {
+      hipLaunchKernelGGL((build_histogram_multifb_kernel<decltype(rc)::value>),
+                         dim3((uint32_t)total_chunks), dim3(HIST_THREADS),
+                         lds, stream.stream(), bins.data_ptr<uint8_t>(),
+                         (const int2*)gpair_seg.data_ptr<int32_t>(),
+                         ridx.data_ptr<int32_t>() + min_start, sc_adj_p,
+                         chunk_off_p,
+                         reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
+                         K, F, (int)n_bins, n_fb, row_stride, (int)f_lo);
+    };
+    if (mfb_r == 8)
+      launch_mfb(std::integral_constant<int, 8>{});
+    else if (mfb_r == 32)
+      launch_mfb(std::integral_constant<int, 32>{});
+    else
+      launch_mfb(std::integral_constant<int, 16>{});
+  } else if (vec16 && fb_size == 8) {
     hipLaunchKernelGGL((build_histogram_kernel<8>),
                        dim3((uint32_t)total_chunks, n_fb),
                        dim3(HIST_THREADS), lds, stream.stream(),
@@ -957,7 +1087,7 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                        sc_adj_p, chunk_off_p,
                        reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
                        K, F, (int)n_bins, fb_size, row_stride, (int)f_lo,
-                       HIST_ROWS_PER_WG);
+                       rows_per_wg);
   } else if (vec16) {
     hipLaunchKernelGGL((build_histogram_kernel<16>),
                        dim3((uint32_t)total_chunks, n_fb),
@@ -968,7 +1098,7 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                        sc_adj_p, chunk_off_p,
                        reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
                        K, F, (int)n_bins, fb_size, row_stride, (int)f_lo,
-                       HIST_ROWS_PER_WG);
+                       rows_per_wg);
   } else {
     hipLaunchKernelGGL((build_histogram_kernel<0>),
                        dim3((uint32_t)total_chunks, n_fb),
@@ -979,7 +1109,7 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                        sc_adj_p, chunk_off_p,
                        reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
                        K, F, (int)n_bins, fb_size, row_stride, (int)f_lo,
-                       HIST_ROWS_PER_WG);
+                       rows_per_wg);
   }
   return hist;
 }
