@@ -982,14 +982,19 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
   // multifb: one WG sweeps all feature blocks with register-cached
   // gpairs/ridx (see kernel comment). Default on for multi-block ranges;
   // RXGB_HIST_MULTIFB=0 disables, RXGB_HIST_MULTIFB_R in {8,16,32}.
-  bool multifb = vec16_pre && fb_pre == 16 && n_fb_pre >= 2;
+  // n_fb >= 3: at 2 blocks (e.g. HIGGS F=28) the re-read saving measured
+  // below the per-WG block-sweep overhead (8.55 vs 8.41 ms/round); from 3
+  // blocks up it wins (100M x 200: 138.7 vs 155.6 ms/round at R=8).
+  bool multifb = vec16_pre && fb_pre == 16 && n_fb_pre >= 3;
   if (const char* e = getenv("RXGB_HIST_MULTIFB")) {
     if (atoi(e) == 0) multifb = false;
   }
-  int mfb_r = 16;
+  // R=8 (4096 rows/WG) measured best at 100M x 200: finer chunks load-
+  // balance deep depths better than R=16/32, and occupancy is LDS-bound.
+  int mfb_r = 8;
   if (const char* e = getenv("RXGB_HIST_MULTIFB_R")) {
     int v = atoi(e);
-    if (v == 8 || v == 16 || v == 32) mfb_r = v;
+    if (v == 4 || v == 8 || v == 16 || v == 32) mfb_r = v;
   }
   const int rows_per_wg =
       multifb ? mfb_r * HIST_THREADS : HIST_ROWS_PER_WG;
@@ -1071,12 +1076,14 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                          reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
                          K, F, (int)n_bins, n_fb, row_stride, (int)f_lo);
     };
-    if (mfb_r == 8)
-      launch_mfb(std::integral_constant<int, 8>{});
+    if (mfb_r == 4)
+      launch_mfb(std::integral_constant<int, 4>{});
+    else if (mfb_r == 16)
+      launch_mfb(std::integral_constant<int, 16>{});
     else if (mfb_r == 32)
       launch_mfb(std::integral_constant<int, 32>{});
     else
-      launch_mfb(std::integral_constant<int, 16>{});
+      launch_mfb(std::integral_constant<int, 8>{});
   } else if (vec16 && fb_size == 8) {
     hipLaunchKernelGGL((build_histogram_kernel<8>),
                        dim3((uint32_t)total_chunks, n_fb),
