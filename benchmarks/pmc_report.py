@@ -33,12 +33,21 @@ if cn:
     for i, n in cur.execute(f"SELECT {idc}, {nmc} FROM {cn[0]}"):
         name_by_id[i] = n
 pcols = cols[pmc_t]
-did = "dispatch_id" if "dispatch_id" in pcols else pcols[0]
-cid = "counter_id" if "counter_id" in pcols else [c for c in pcols if "counter" in c][0]
+kcols = cols[kd]
+print("# kd cols:", kcols)
+cid = ("pmc_id" if "pmc_id" in pcols else
+       "counter_id" if "counter_id" in pcols else
+       [c for c in pcols if "counter" in c or "pmc" in c][0])
 val = "value" if "value" in pcols else pcols[-1]
+if "event_id" in pcols and "event_id" in kcols:
+    join = f"p.event_id = d.event_id"
+elif "dispatch_id" in pcols:
+    join = f"p.dispatch_id = d.id"
+else:
+    join = f"p.{pcols[0]} = d.id"
 rows = cur.execute(f"""
   SELECT sym.display_name, p.{cid}, SUM(p.{val})
-  FROM {pmc_t} p JOIN {kd} d ON p.{did} = d.id
+  FROM {pmc_t} p JOIN {kd} d ON {join}
   JOIN {ks} sym ON d.kernel_id = sym.id
   GROUP BY 1, 2""").fetchall()
 durs = dict(cur.execute(f"""
