@@ -144,10 +144,16 @@ static int hist_rows_per_wg() {
 // serialization on skewed features drops ~16x.
 // One row's 16-feature uint4 -> LDS tile atomics with word-granular lane
 // rotation (shared by the single-block and multi-block kernels below).
+//
+// LDS layout is SoA: g-plane at [f*n_bins + b], h-plane at
+// [16*n_bins + f*n_bins + b]. Interleaved (g,h) pairs put the bank index
+// at (b*4)%64 -> only bin%16 distinguishes banks; the split doubles the
+// spread to bin%32 (PMC: conflict/active 0.74 interleaved).
 __device__ __forceinline__ void hist_accum_row16(
     unsigned long long* lds_hist, const uint8_t* __restrict__ bins,
     uint64_t r, int64_t row_stride, int f0, int n_bins, int lane,
     longlong2 gp) {
+  const int hplane = 16 * n_bins;
   const uint4 packed =
       *reinterpret_cast<const uint4*>(bins + r * row_stride + f0);
   const int r4 = lane & 3;
@@ -169,9 +175,9 @@ __device__ __forceinline__ void hist_accum_row16(
       const int b = (w >> (8 * kk)) & 0xFF;
       if (b != 255) {
         const int f = fw + kk;
-        unsigned long long* cell = &lds_hist[((size_t)f * n_bins + b) * 2];
+        unsigned long long* cell = &lds_hist[(size_t)f * n_bins + b];
         atomicAdd(cell, (unsigned long long)gp.x);
-        atomicAdd(cell + 1, (unsigned long long)gp.y);
+        atomicAdd(cell + hplane, (unsigned long long)gp.y);
       }
     }
   }
@@ -205,7 +211,9 @@ __global__ __launch_bounds__(HIST_THREADS) void build_histogram_kernel(
   const int fcount = fb_size < (F - f0) ? fb_size : (F - f0);
 
   extern __shared__ unsigned long long lds_hist[];  // [fb_size][n_bins][2]
-  const int tile = fcount * n_bins * 2;
+  // VFB==16 uses the SoA g/h-plane layout (see hist_accum_row16): zero
+  // both full planes even for a partial last block.
+  const int tile = (VFB == 16 ? 16 : fcount) * n_bins * 2;
   for (int i = threadIdx.x; i < tile; i += blockDim.x) lds_hist[i] = 0ull;
   __syncthreads();
 
@@ -270,13 +278,24 @@ __global__ __launch_bounds__(HIST_THREADS) void build_histogram_kernel(
   }
   __syncthreads();
 
-  // merge LDS tile into the global histogram
+  // merge LDS tile into the global (interleaved [.,2]) histogram
   long long* ghist =
       hist + (((size_t)node * F + f0) * n_bins) * 2;
-  for (int i = threadIdx.x; i < tile; i += blockDim.x) {
-    const unsigned long long v = lds_hist[i];
-    if (v != 0ull) {
-      atomicAdd((unsigned long long*)&ghist[i], v);
+  if constexpr (VFB == 16) {
+    const int cells = fcount * n_bins;
+    const int hplane = 16 * n_bins;
+    for (int i = threadIdx.x; i < cells; i += blockDim.x) {
+      const unsigned long long g = lds_hist[i];
+      const unsigned long long h = lds_hist[hplane + i];
+      if (g) atomicAdd((unsigned long long*)&ghist[2 * i], g);
+      if (h) atomicAdd((unsigned long long*)&ghist[2 * i + 1], h);
+    }
+  } else {
+    for (int i = threadIdx.x; i < tile; i += blockDim.x) {
+      const unsigned long long v = lds_hist[i];
+      if (v != 0ull) {
+        atomicAdd((unsigned long long*)&ghist[i], v);
+      }
     }
   }
 }
@@ -352,13 +371,17 @@ __global__ __launch_bounds__(HIST_THREADS) void build_histogram_multifb_kernel(
     }
     __syncthreads();
     // merge only real features (partial last block: pad bins==255 were
-    // skipped in accumulation, but hist rows past F-1 must not be touched)
+    // skipped in accumulation, but hist rows past F-1 must not be touched);
+    // LDS is SoA g/h planes, global stays interleaved [.,2]
     const int fcount = 16 < (F - f0) ? 16 : (F - f0);
-    const int tile = fcount * n_bins * 2;
+    const int cells = fcount * n_bins;
+    const int hplane = 16 * n_bins;
     long long* ghist = hist + (((size_t)node * F + f0) * n_bins) * 2;
-    for (int i = threadIdx.x; i < tile; i += blockDim.x) {
-      const unsigned long long v = lds_hist[i];
-      if (v != 0ull) atomicAdd((unsigned long long*)&ghist[i], v);
+    for (int i = threadIdx.x; i < cells; i += blockDim.x) {
+      const unsigned long long g = lds_hist[i];
+      const unsigned long long h = lds_hist[hplane + i];
+      if (g) atomicAdd((unsigned long long*)&ghist[2 * i], g);
+      if (h) atomicAdd((unsigned long long*)&ghist[2 * i + 1], h);
     }
     __syncthreads();
   }
