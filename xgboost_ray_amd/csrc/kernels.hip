@@ -282,13 +282,13 @@ __global__ __launch_bounds__(HIST_THREADS) void build_histogram_kernel(
   long long* ghist =
       hist + (((size_t)node * F + f0) * n_bins) * 2;
   if constexpr (VFB == 16) {
-    const int cells = fcount * n_bins;
+    // re-interleave while merging: consecutive lanes write consecutive
+    // global u64s (a strided 2i/2i+1 pattern doubled global transactions)
+    const int n2 = fcount * n_bins * 2;
     const int hplane = 16 * n_bins;
-    for (int i = threadIdx.x; i < cells; i += blockDim.x) {
-      const unsigned long long g = lds_hist[i];
-      const unsigned long long h = lds_hist[hplane + i];
-      if (g) atomicAdd((unsigned long long*)&ghist[2 * i], g);
-      if (h) atomicAdd((unsigned long long*)&ghist[2 * i + 1], h);
+    for (int j = threadIdx.x; j < n2; j += blockDim.x) {
+      const unsigned long long v = lds_hist[(j & 1) * hplane + (j >> 1)];
+      if (v) atomicAdd((unsigned long long*)&ghist[j], v);
     }
   } else {
     for (int i = threadIdx.x; i < tile; i += blockDim.x) {
@@ -354,6 +354,7 @@ __global__ __launch_bounds__(HIST_THREADS) void build_histogram_multifb_kernel(
 
   extern __shared__ unsigned long long lds_hist[];
   const int tile_full = 16 * n_bins * 2;
+  const int hplane = 16 * n_bins;
   const int lane = threadIdx.x & (WAVE - 1);
   for (int blk = 0; blk < n_fb; ++blk) {
     const int f0 = f_base + blk * 16;
@@ -374,14 +375,11 @@ __global__ __launch_bounds__(HIST_THREADS) void build_histogram_multifb_kernel(
     // skipped in accumulation, but hist rows past F-1 must not be touched);
     // LDS is SoA g/h planes, global stays interleaved [.,2]
     const int fcount = 16 < (F - f0) ? 16 : (F - f0);
-    const int cells = fcount * n_bins;
-    const int hplane = 16 * n_bins;
+    const int n2 = fcount * n_bins * 2;
     long long* ghist = hist + (((size_t)node * F + f0) * n_bins) * 2;
-    for (int i = threadIdx.x; i < cells; i += blockDim.x) {
-      const unsigned long long g = lds_hist[i];
-      const unsigned long long h = lds_hist[hplane + i];
-      if (g) atomicAdd((unsigned long long*)&ghist[2 * i], g);
-      if (h) atomicAdd((unsigned long long*)&ghist[2 * i + 1], h);
+    for (int j = threadIdx.x; j < n2; j += blockDim.x) {
+      const unsigned long long v = lds_hist[(j & 1) * hplane + (j >> 1)];
+      if (v) atomicAdd((unsigned long long*)&ghist[j], v);
     }
     __syncthreads();
   }
