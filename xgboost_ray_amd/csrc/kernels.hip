@@ -592,8 +592,9 @@ __global__ void find_splits_reduce_kernel(
 // blocks are processed in segment order and lanes in row order.
 // ---------------------------------------------------------------------------
 #define PART_THREADS 256
-#define PART_ROWS_PER_THREAD 4
+#define PART_ROWS_PER_THREAD 8
 #define PART_CHUNK (PART_THREADS * PART_ROWS_PER_THREAD)
+#define PART_WAVES (PART_THREADS / WAVE)
 
 __device__ inline bool go_left_pred(const uint8_t* bins, uint64_t r,
                                     int64_t row_stride, int feat,
@@ -630,10 +631,11 @@ __global__ void partition_count_kernel(
   const int feat = (int)split_feat[node], sbin = (int)split_bin[node];
   const int dl = (int)default_left[node];
 
-  // preload all stripes' row indices + bin gathers so 4 independent
+  // preload all stripes' row indices + bin gathers so 8 independent
   // random-latency loads are in flight per thread (the kernel was 86%
-  // memory-parked at one row per thread)
-  int total = 0;
+  // memory-parked at one row per thread), accumulate the per-wave ballot
+  // popcounts in a register, and reduce with ONE barrier at the end (the
+  // per-stripe sync ping-pong serialized the old 4-stripe version).
   uint8_t bv[PART_ROWS_PER_THREAD];
   bool valid[PART_ROWS_PER_THREAD];
   #pragma unroll
@@ -643,8 +645,10 @@ __global__ void partition_count_kernel(
     uint64_t r = valid[sstripe] ? (uint32_t)ridx[seg_start + i] : 0;
     bv[sstripe] = bins[r * (uint64_t)row_stride + feat];
   }
-  __shared__ int wave_sums[PART_THREADS / WAVE];
+  __shared__ int wave_sums[PART_WAVES];
   const int wave_id = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  int wsum = 0;
   #pragma unroll
   for (int sstripe = 0; sstripe < PART_ROWS_PER_THREAD; ++sstripe) {
     const int64_t i = row_lo + sstripe * PART_THREADS + threadIdx.x;
@@ -654,16 +658,15 @@ __global__ void partition_count_kernel(
       flag = (b == 255) ? (dl != 0) : (b <= sbin);
       flags[seg_start + i] = flag ? 1 : 0;
     }
-    unsigned long long mask = __ballot(flag);
-    if ((threadIdx.x & (WAVE - 1)) == 0)
-      wave_sums[wave_id] = __popcll(mask);
-    __syncthreads();
-    if (threadIdx.x == 0) {
-      for (int w = 0; w < PART_THREADS / WAVE; ++w) total += wave_sums[w];
-    }
-    __syncthreads();
+    wsum += __popcll(__ballot(flag));
   }
-  if (threadIdx.x == 0) block_counts[wg] = total;
+  if (lane == 0) wave_sums[wave_id] = wsum;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    int total = 0;
+    for (int w = 0; w < PART_WAVES; ++w) total += wave_sums[w];
+    block_counts[wg] = total;
+  }
 }
 
 __global__ void partition_scatter_kernel(
@@ -689,48 +692,54 @@ __global__ void partition_scatter_kernel(
   const int64_t seg_start = node_start[node];
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave_id = threadIdx.x / WAVE;
-  __shared__ int wave_sums[PART_THREADS / WAVE];
-  __shared__ int stripe_left_base;
-  if (threadIdx.x == 0) stripe_left_base = 0;
+
+  // Load every stripe's row data upfront (3 coalesced streams, all in
+  // flight), publish every (stripe, wave) ballot popcount to LDS, ONE
+  // barrier, then compute each row's exclusive left-prefix from the LDS
+  // table and write. No cross-stripe serialization.
+  int32_t rv[PART_ROWS_PER_THREAD];
+  int2 gv[PART_ROWS_PER_THREAD];
+  bool fl[PART_ROWS_PER_THREAD];
+  bool valid[PART_ROWS_PER_THREAD];
+  unsigned long long masks[PART_ROWS_PER_THREAD];
+  __shared__ int wave_sums[PART_ROWS_PER_THREAD][PART_WAVES];
+  #pragma unroll
+  for (int s = 0; s < PART_ROWS_PER_THREAD; ++s) {
+    const int64_t i = row_lo + s * PART_THREADS + threadIdx.x;
+    valid[s] = i < count;
+    rv[s] = 0;
+    gv[s] = {0, 0};
+    fl[s] = false;
+    if (valid[s]) {
+      rv[s] = ridx[seg_start + i];
+      if (gseg != nullptr) gv[s] = gseg[seg_start + i];
+      fl[s] = flags[seg_start + i] != 0;
+    }
+    masks[s] = __ballot(fl[s]);
+    if (lane == 0) wave_sums[s][wave_id] = __popcll(masks[s]);
+  }
   __syncthreads();
 
+  int stripe_base = 0;  // lefts in earlier stripes of this chunk
   #pragma unroll
-  for (int sstripe = 0; sstripe < PART_ROWS_PER_THREAD; ++sstripe) {
-    const int64_t i = row_lo + sstripe * PART_THREADS + threadIdx.x;
-    const bool valid = i < count;
-    int32_t rv = 0;
-    int2 gv = {0, 0};
-    bool flag = false;
-    if (valid) {
-      rv = ridx[seg_start + i];
-      if (gseg != nullptr) gv = gseg[seg_start + i];
-      flag = flags[seg_start + i] != 0;
-    }
-    unsigned long long mask = __ballot(flag);
-    if (lane == 0) wave_sums[wave_id] = __popcll(mask);
-    __syncthreads();
-    int wave_left_before = stripe_left_base;
-    for (int w = 0; w < wave_id; ++w) wave_left_before += wave_sums[w];
-
-    const int prefix_in_wave =
-        __popcll(mask & ((lane == 0) ? 0ull : ((~0ull) >> (64 - lane))));
-    if (valid) {
+  for (int s = 0; s < PART_ROWS_PER_THREAD; ++s) {
+    const int64_t i = row_lo + s * PART_THREADS + threadIdx.x;
+    int wave_left_before = stripe_base;
+    for (int w = 0; w < wave_id; ++w) wave_left_before += wave_sums[s][w];
+    const int prefix_in_wave = __popcll(
+        masks[s] & ((lane == 0) ? 0ull : ((~0ull) >> (64 - lane))));
+    if (valid[s]) {
       // lefts strictly before row i in the whole node segment: earlier
       // chunks + earlier stripes of this chunk + this stripe's ballot
       const int64_t my_left_prefix =
           left_before[wg] + wave_left_before + prefix_in_wave;
-      const int64_t pos = flag
+      const int64_t pos = fl[s]
                               ? my_left_prefix
                               : node_left_total[node] + (i - my_left_prefix);
-      ridx_out[seg_start + pos] = rv;
-      if (gseg_out != nullptr) gseg_out[seg_start + pos] = gv;
+      ridx_out[seg_start + pos] = rv[s];
+      if (gseg_out != nullptr) gseg_out[seg_start + pos] = gv[s];
     }
-    __syncthreads();
-    if (threadIdx.x == 0) {
-      for (int w = 0; w < PART_THREADS / WAVE; ++w)
-        stripe_left_base += wave_sums[w];
-    }
-    __syncthreads();
+    for (int w = 0; w < PART_WAVES; ++w) stripe_base += wave_sums[s][w];
   }
 }
 
