@@ -616,7 +616,9 @@ __global__ void partition_count_kernel(
     const int64_t* __restrict__ default_left,
     int32_t* __restrict__ block_counts,  // [total_chunks]
     uint8_t* __restrict__ flags,         // [n] go-left per segment position
-    int K, int64_t row_stride) {
+    int K, int64_t row_stride,
+    const uint8_t* __restrict__ bins_T,  // [F, n] column-major copy or null
+    int64_t n_rows_total) {
   int wg = blockIdx.x;
   int lo = 0, hi = K;
   while (lo + 1 < hi) {
@@ -638,12 +640,18 @@ __global__ void partition_count_kernel(
   // per-stripe sync ping-pong serialized the old 4-stripe version).
   uint8_t bv[PART_ROWS_PER_THREAD];
   bool valid[PART_ROWS_PER_THREAD];
+  // column-major copy: the gather touches a dense per-feature column
+  // (ridx ascending within a segment -> near-sequential lines) instead
+  // of one 64 B row-major line per row (64x the line traffic).
+  const uint8_t* colbase =
+      bins_T ? bins_T + (int64_t)feat * n_rows_total : nullptr;
   #pragma unroll
   for (int sstripe = 0; sstripe < PART_ROWS_PER_THREAD; ++sstripe) {
     const int64_t i = row_lo + sstripe * PART_THREADS + threadIdx.x;
     valid[sstripe] = i < count;
     uint64_t r = valid[sstripe] ? (uint32_t)ridx[seg_start + i] : 0;
-    bv[sstripe] = bins[r * (uint64_t)row_stride + feat];
+    bv[sstripe] = colbase ? colbase[r]
+                          : bins[r * (uint64_t)row_stride + feat];
   }
   __shared__ int wave_sums[PART_WAVES];
   const int wave_id = threadIdx.x / WAVE;
@@ -1012,10 +1020,10 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
   // multifb: one WG sweeps all feature blocks with register-cached
   // gpairs/ridx (see kernel comment). Default on for multi-block ranges;
   // RXGB_HIST_MULTIFB=0 disables, RXGB_HIST_MULTIFB_R in {8,16,32}.
-  // n_fb >= 3: at 2 blocks (e.g. HIGGS F=28) the re-read saving measured
-  // below the per-WG block-sweep overhead (8.55 vs 8.41 ms/round); from 3
-  // blocks up it wins (100M x 200: 138.7 vs 155.6 ms/round at R=8).
-  bool multifb = vec16_pre && fb_pre == 16 && n_fb_pre >= 3;
+  // any multi-block range: with the SoA LDS layout + interleaved merge
+  // it wins at 2 blocks too (HIGGS 8.18 vs 8.31 ms/round) and by 11% at
+  // 13 blocks (100M x 200: 138.7 vs 155.6 ms/round at R=8).
+  bool multifb = vec16_pre && fb_pre == 16 && n_fb_pre >= 2;
   if (const char* e = getenv("RXGB_HIST_MULTIFB")) {
     if (atoi(e) == 0) multifb = false;
   }
@@ -1204,7 +1212,8 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
                                           torch::Tensor split_feat,
                                           torch::Tensor split_bin,
                                           torch::Tensor default_left,
-                                          torch::Tensor gseg) {
+                                          torch::Tensor gseg,
+                                          torch::Tensor bins_t) {
   const int K = (int)starts.size(0);
   const int F = (int)bins.size(1);
   auto dev = bins.device();
@@ -1248,7 +1257,9 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
                      bins.data_ptr<uint8_t>(), ridx.data_ptr<int32_t>(),
                      sc_p, chunk_off_p, sf_p, sb_p, dl_p,
                      block_counts.data_ptr<int32_t>(),
-                     flags.data_ptr<uint8_t>(), K, bins.stride(0));
+                     flags.data_ptr<uint8_t>(), K, bins.stride(0),
+                     bins_t.numel() ? bins_t.data_ptr<uint8_t>() : nullptr,
+                     bins.size(0));
   // Per-node exclusive prefix of block counts. ONE D2H pull + a host
   // loop + ONE H2D push: the torch-op chain this replaces (device
   // cumsum + repeat_interleave + index_selects, ~8 tiny transfers and

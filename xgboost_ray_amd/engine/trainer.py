@@ -619,6 +619,7 @@ class BoostingEngine:
                 torch.tensor(sb, dtype=torch.int32),
                 torch.tensor(sdl, dtype=torch.uint8),
                 gpair_seg=gseg,
+                bins_t=getattr(self.dtrain, "bins_t", None),
             )
             _tick("partition")
 
@@ -925,6 +926,7 @@ class BoostingEngine:
                 torch.tensor([b], dtype=torch.int32),
                 torch.tensor([rec["dl"]], dtype=torch.uint8),
                 gpair_seg=gseg,
+                bins_t=getattr(self.dtrain, "bins_t", None),
             )
             lcount = int(left_counts[0])
             lsums = (rec["lg"], rec["lh"])

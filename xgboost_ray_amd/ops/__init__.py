@@ -121,7 +121,7 @@ def find_splits(
 
 
 def partition_rows(bins, ridx, starts, counts, split_feat, split_bin,
-                   default_left, gpair_seg=None):
+                   default_left, gpair_seg=None, bins_t=None):
     """Stable partition of each node's ridx segment by its split.
 
     Rows with bin <= split_bin (or missing & default_left) go left. When
@@ -132,7 +132,7 @@ def partition_rows(bins, ridx, starts, counts, split_feat, split_bin,
     """
     return _impl(bins).partition_rows(
         bins, ridx, starts, counts, split_feat, split_bin, default_left,
-        gpair_seg
+        gpair_seg, bins_t
     )
 
 

@@ -228,7 +228,7 @@ def find_splits(
 
 
 def partition_rows(bins, ridx, starts, counts, split_feat, split_bin,
-                   default_left, gpair_seg=None):
+                   default_left, gpair_seg=None, bins_t=None):
     K = len(starts)
     out = ridx.clone()
     gout = None if gpair_seg is None else gpair_seg.clone()

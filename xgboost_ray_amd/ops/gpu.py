@@ -114,18 +114,20 @@ def find_splits(
 
 
 def partition_rows(bins, ridx, starts, counts, split_feat, split_bin,
-                   default_left, gpair_seg=None):
+                   default_left, gpair_seg=None, bins_t=None):
     dev = bins.device
+    if bins_t is None:
+        bins_t = torch.zeros(0, dtype=torch.uint8, device=dev)
     if gpair_seg is None:
         gpair_seg = torch.zeros((0, 2), dtype=torch.int32, device=dev)
         r, lc, _ = _load().partition_rows(
             bins, ridx, starts, counts, split_feat, split_bin,
-            default_left, gpair_seg,
+            default_left, gpair_seg, bins_t,
         )
         return r, lc
     return _load().partition_rows(
         bins, ridx, starts, counts, split_feat, split_bin, default_left,
-        gpair_seg,
+        gpair_seg, bins_t,
     )
 
 
