@@ -23,6 +23,7 @@
 
 #include <cstdint>
 #include <cstdlib>
+#include <cstring>
 #include <vector>
 
 #define WAVE 64
@@ -382,6 +383,74 @@ __global__ __launch_bounds__(HIST_THREADS) void build_histogram_multifb_kernel(
       if (v) atomicAdd((unsigned long long*)&ghist[j], v);
     }
     __syncthreads();
+  }
+}
+
+// ---------------------------------------------------------------------------
+// build_histogram_xcd: XCD-swizzled (chunk, block) grid for wide matrices.
+//
+// The multifb kernel re-fetches each row's bin lines once per feature
+// block over a long window - no L2 survival. Here each (row-chunk,
+// feature-block) pair is its own workgroup again, but the flat index is
+// swizzled so ALL n_fb blocks of one chunk land on the SAME XCD
+// (consecutive blockIdx round-robin over the 8 XCDs; stride-8 indices
+// share one): the chunk's bin lines are fetched once into that XCD's L2
+// and every 16 B uint4 of each 64 B line is consumed by a co-resident
+// workgroup. gpair_seg/ridx re-reads hit L2 the same way (the 13 blocks
+// read the same 48 KB of segment data).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(HIST_THREADS) void build_histogram_xcd_kernel(
+    const uint8_t* __restrict__ bins,        // [n_rows_total, row_stride]
+    const int2* __restrict__ gpair_seg,      // [seg_total] segment order
+    const int32_t* __restrict__ ridx,        // [seg_total]
+    const int64_t* __restrict__ node_start,  // [K] starts, [K..2K) counts
+    const int64_t* __restrict__ chunk_off,   // [K+1] cumulative chunks
+    long long* __restrict__ hist,            // [K, F, n_bins, 2]
+    int K, int F, int n_bins, int n_fb, int64_t row_stride, int f_base,
+    int rows_per_wg, int total_chunks) {
+  const int flat = blockIdx.x;
+  const int octet = flat / (8 * n_fb);
+  const int rem = flat % (8 * n_fb);
+  const int fb = rem / 8;
+  const int wg = octet * 8 + (rem % 8);  // row-chunk id
+  if (wg >= total_chunks) return;        // tail of the last chunk-octet
+
+  int lo = 0, hi = K;
+  while (lo + 1 < hi) {
+    int m = (lo + hi) >> 1;
+    if (chunk_off[m] <= wg) lo = m; else hi = m;
+  }
+  const int node = lo;
+  const int64_t chunk_in_node = wg - chunk_off[node];
+  const int64_t seg_start = node_start[node];
+  const int64_t row_lo = chunk_in_node * (int64_t)rows_per_wg;
+  const int f0 = f_base + fb * 16;
+  const int fcount = 16 < (F - f0) ? 16 : (F - f0);
+
+  extern __shared__ unsigned long long lds_hist[];
+  const int hplane = 16 * n_bins;
+  for (int i = threadIdx.x; i < 2 * hplane; i += blockDim.x)
+    lds_hist[i] = 0ull;
+  __syncthreads();
+
+  const int64_t node_count = node_start[K + node];
+  int64_t row_hi = row_lo + rows_per_wg;
+  if (row_hi > node_count) row_hi = node_count;
+  const int lane = threadIdx.x & (WAVE - 1);
+  for (int64_t i = row_lo + threadIdx.x; i < row_hi; i += blockDim.x) {
+    const int64_t seg_i = seg_start + i;
+    const int2 gpi = gpair_seg[seg_i];
+    const longlong2 gp = {(long long)gpi.x, (long long)gpi.y};
+    hist_accum_row16(lds_hist, bins, (uint32_t)ridx[seg_i], row_stride, f0,
+                     n_bins, lane, gp);
+  }
+  __syncthreads();
+
+  const int n2 = fcount * n_bins * 2;
+  long long* ghist = hist + (((size_t)node * F + f0) * n_bins) * 2;
+  for (int j = threadIdx.x; j < n2; j += blockDim.x) {
+    const unsigned long long v = lds_hist[(j & 1) * hplane + (j >> 1)];
+    if (v) atomicAdd((unsigned long long*)&ghist[j], v);
   }
 }
 
@@ -1027,6 +1096,16 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
   if (const char* e = getenv("RXGB_HIST_MULTIFB")) {
     if (atoi(e) == 0) multifb = false;
   }
+  // RXGB_HIST_MODE=xcd: co-resident (chunk, block) grid variant (A/B)
+  bool xcd_mode = false;
+  if (const char* e = getenv("RXGB_HIST_MODE")) {
+    if (strcmp(e, "xcd") == 0 && multifb) { xcd_mode = true; multifb = false; }
+  }
+  int xcd_rows = 4096;
+  if (const char* e = getenv("RXGB_HIST_XCD_ROWS")) {
+    int v = atoi(e);
+    if (v >= 512 && v <= 65536) xcd_rows = v;
+  }
   // R=8 (4096 rows/WG) measured best at 100M x 200: finer chunks load-
   // balance deep depths better than R=16/32, and occupancy is LDS-bound.
   int mfb_r = 8;
@@ -1034,8 +1113,9 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
     int v = atoi(e);
     if (v == 4 || v == 8 || v == 16 || v == 32) mfb_r = v;
   }
-  const int rows_per_wg =
-      multifb ? mfb_r * HIST_THREADS : HIST_ROWS_PER_WG;
+  const int rows_per_wg = multifb ? mfb_r * HIST_THREADS
+                          : xcd_mode ? xcd_rows
+                                     : HIST_ROWS_PER_WG;
 
   int64_t total_chunks = 0;
   auto chunk_off_cpu =
@@ -1103,7 +1183,18 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
   const size_t lds = (size_t)fb_size * n_bins * 2 * sizeof(long long);
 
   // ridx pointer offset so seg indices align with gpair_seg
-  if (multifb) {
+  if (xcd_mode) {
+    const int64_t grid = ceil_div(total_chunks, 8) * 8 * (int64_t)n_fb;
+    hipLaunchKernelGGL((build_histogram_xcd_kernel), dim3((uint32_t)grid),
+                       dim3(HIST_THREADS), lds, stream.stream(),
+                       bins.data_ptr<uint8_t>(),
+                       (const int2*)gpair_seg.data_ptr<int32_t>(),
+                       ridx.data_ptr<int32_t>() + min_start, sc_adj_p,
+                       chunk_off_p,
+                       reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
+                       K, F, (int)n_bins, n_fb, row_stride, (int)f_lo,
+                       rows_per_wg, (int)total_chunks);
+  } else if (multifb) {
     auto launch_mfb = [&](auto rc) {
       hipLaunchKernelGGL((build_histogram_multifb_kernel<decltype(rc)::value>),
                          dim3((uint32_t)total_chunks), dim3(HIST_THREADS),
