@@ -1097,7 +1097,7 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
     if (atoi(e) == 0) multifb = false;
   }
   // Default mode for multi-block ranges: XCD-swizzled co-resident grid
-  // (hist 102 -> 58 ms at 100M x 200; HIGGS 8.31 -> 7.32 ms/round).
+  // (hist 102 -> 56 ms at 100M x 200; HIGGS 8.31 -> 7.25 ms/round).
   // RXGB_HIST_MODE=multifb selects the register-cached block-sweep
   // variant instead; =base the original (chunk, block) 2-D grid.
   bool xcd_mode = multifb;
@@ -1106,7 +1106,7 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
     else if (strcmp(e, "base") == 0) { xcd_mode = false; multifb = false; }
   }
   if (xcd_mode) multifb = false;
-  int xcd_rows = 8192;
+  int xcd_rows = 16384;
   if (const char* e = getenv("RXGB_HIST_XCD_ROWS")) {
     int v = atoi(e);
     if (v >= 512 && v <= 65536) xcd_rows = v;
