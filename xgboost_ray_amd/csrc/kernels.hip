@@ -1082,6 +1082,38 @@ torch::Tensor bin_matrix(torch::Tensor values, torch::Tensor cuts_flat,
   return out;
 }
 
+
+// ---------------------------------------------------------------------------
+// Pinned staging: pageable H2D/D2H around every per-depth launch blocks the
+// host in the driver's staging path. Each call site owns a cached pinned
+// buffer; an event guards against overwriting a buffer whose async H2D is
+// still in flight (can happen in the chunked-overlap histogram path where
+// several builds are enqueued back-to-back with no intervening sync).
+// ---------------------------------------------------------------------------
+struct PinnedStager {
+  torch::Tensor buf;
+  hipEvent_t ev = nullptr;
+  bool pending = false;
+  torch::Tensor get(int64_t n, torch::Dtype dt = torch::kInt64) {
+    if (pending) {
+      hipEventSynchronize(ev);
+      pending = false;
+    }
+    if (!buf.defined() || buf.numel() < n || buf.scalar_type() != dt) {
+      int64_t cap = 4096;
+      while (cap < n) cap *= 2;
+      buf = torch::empty({cap},
+                         torch::TensorOptions().dtype(dt).pinned_memory(true));
+    }
+    return buf.narrow(0, 0, n);
+  }
+  void mark(hipStream_t s) {
+    if (!ev) (void)hipEventCreateWithFlags(&ev, hipEventDisableTiming);
+    (void)hipEventRecord(ev, s);
+    pending = true;
+  }
+};
+
 torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                               torch::Tensor ridx, torch::Tensor starts,
                               torch::Tensor counts, int64_t n_bins,
@@ -1186,7 +1218,11 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
   // chunk_off(K+1)] - tiny pageable copies around kernel launches were
   // the training loop's dominant host cost.
   auto starts_adj = starts_cpu - min_start;
-  auto meta = torch::cat({starts_adj, counts_cpu, chunk_off_cpu}).to(dev);
+  static thread_local PinnedStager hist_meta_stager;
+  auto meta_cpu = hist_meta_stager.get(2 * K + K + 1);
+  torch::cat_out(meta_cpu, {starts_adj, counts_cpu, chunk_off_cpu});
+  auto meta = meta_cpu.to(dev, /*non_blocking=*/true);
+  hist_meta_stager.mark(stream.stream());
   int64_t* mp = meta.data_ptr<int64_t>();
   int64_t* sc_adj_p = mp;          // starts at [0..K), counts at [K..2K)
   int64_t* chunk_off_p = mp + 2 * K;
@@ -1322,7 +1358,12 @@ std::vector<torch::Tensor> find_splits(torch::Tensor hist, torch::Tensor parent_
                      reinterpret_cast<const long long*>(kf_lh.data_ptr<int64_t>()),
                      reinterpret_cast<long long*>(out_packed.data_ptr<int64_t>()),
                      K, F);
-  return {out_packed};
+  // pinned D2H: the caller consumes (copies out of) the result
+  // immediately, so returning a view of the cached buffer is safe
+  static thread_local PinnedStager split_out_stager;
+  auto out_cpu = split_out_stager.get((int64_t)K * 6).view({K, 6});
+  out_cpu.copy_(out_packed);
+  return {out_cpu};
 }
 
 std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx,
@@ -1352,12 +1393,16 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
   }
   if (total_chunks == 0) return {ridx_out, left_counts, gseg_out};
   // ONE H2D copy: [starts(K) | counts(K) | chunk_off(K+1) | feat(K) |
-  // bin(K) | default_left(K)]
-  auto meta = torch::cat({starts_cpu, counts_cpu, chunk_off_cpu,
-                          split_feat.to(torch::kCPU).to(torch::kInt64),
-                          split_bin.to(torch::kCPU).to(torch::kInt64),
-                          default_left.to(torch::kCPU).to(torch::kInt64)})
-                  .to(dev);
+  // bin(K) | default_left(K)] - staged through a pinned buffer
+  auto stream0 = at::cuda::getCurrentCUDAStream();
+  static thread_local PinnedStager part_meta_stager;
+  auto meta_cpu = part_meta_stager.get(5 * K + 1);
+  torch::cat_out(meta_cpu, {starts_cpu, counts_cpu, chunk_off_cpu,
+                            split_feat.to(torch::kCPU).to(torch::kInt64),
+                            split_bin.to(torch::kCPU).to(torch::kInt64),
+                            default_left.to(torch::kCPU).to(torch::kInt64)});
+  auto meta = meta_cpu.to(dev, /*non_blocking=*/true);
+  part_meta_stager.mark(stream0.stream());
   int64_t* mp = meta.data_ptr<int64_t>();
   int64_t* sc_p = mp;
   int64_t* chunk_off_p = mp + 2 * K;
@@ -1383,12 +1428,16 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
   // cumsum + repeat_interleave + index_selects, ~8 tiny transfers and
   // kernels per call) stalled the training loop ~15 ms per depth -
   // measured 7x slower rounds end to end.
-  auto bc_cpu = block_counts.to(torch::kCPU);
-  auto left_before_cpu = torch::empty({total_chunks}, torch::kInt64);
+  // pinned D2H pull (no pageable staging bounce), then build
+  // [left_before | node_left_total] in one pinned buffer -> ONE H2D
+  static thread_local PinnedStager bc_stager, lb_stager;
+  auto bc_cpu = bc_stager.get(total_chunks, torch::kInt32);
+  bc_cpu.copy_(block_counts);  // synchronous: host consumes it next
+  auto lbnl_cpu = lb_stager.get(total_chunks + K);
   auto node_left_total_cpu = torch::zeros({K}, torch::kInt64);
   {
     auto bc = bc_cpu.accessor<int32_t, 1>();
-    auto lb = left_before_cpu.accessor<int64_t, 1>();
+    auto lb = lbnl_cpu.accessor<int64_t, 1>();
     auto nl = node_left_total_cpu.accessor<int64_t, 1>();
     auto co = chunk_off_cpu.accessor<int64_t, 1>();
     for (int k = 0; k < K; ++k) {
@@ -1398,10 +1447,13 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
         run += bc[c];
       }
       nl[k] = run;
+      lb[total_chunks + k] = run;
     }
   }
-  auto left_before = left_before_cpu.to(dev);
-  auto node_left_total = node_left_total_cpu.to(dev);
+  auto lbnl = lbnl_cpu.to(dev, /*non_blocking=*/true);
+  lb_stager.mark(stream0.stream());
+  auto left_before = lbnl.narrow(0, 0, total_chunks);
+  auto node_left_total = lbnl.narrow(0, total_chunks, K);
   hipLaunchKernelGGL(partition_scatter_kernel, dim3((uint32_t)total_chunks),
                      dim3(PART_THREADS), 0, stream.stream(),
                      flags.data_ptr<uint8_t>(), ridx.data_ptr<int32_t>(),

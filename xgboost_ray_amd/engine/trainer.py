@@ -157,6 +157,10 @@ class BoostingEngine:
         self.coll = collective or Collective()
         self.rank = rank
         self.device = dtrain.bins.device
+        # pinned double-use staging buffers for tiny per-depth H2D pushes
+        # (pageable copies block the host in the driver's staging path);
+        # each slot is drained by a stream-ordered D2H pull before reuse
+        self._pin_bufs = {}
         obj_spec = custom_objective or self.p.objective
         self.obj: Objective = get_objective(
             obj_spec, self.p.num_class, float(self.p.scale_pos_weight)
@@ -346,6 +350,20 @@ class BoostingEngine:
         mask[perm] = True
         return mask.to(self.device)
 
+
+    def _stage_i64(self, arr, slot):
+        """Pinned H2D staging for a small int64 numpy vector (GPU only)."""
+        arr = np.ascontiguousarray(arr, dtype=np.int64).reshape(-1)
+        n = arr.size
+        buf = self._pin_bufs.get(slot)
+        if buf is None or buf.numel() < n:
+            cap = max(4096, 1 << max(0, (n - 1)).bit_length())
+            buf = torch.empty(cap, dtype=torch.int64, pin_memory=True)
+            self._pin_bufs[slot] = buf
+        view = buf[:n]
+        view.copy_(torch.from_numpy(arr))
+        return view.to(self.device, non_blocking=True)
+
     def _grow_tree(
         self, gpair: torch.Tensor, it: int, cls: int, ptree: int = 0
     ) -> Tree:
@@ -500,12 +518,16 @@ class BoostingEngine:
             _tick("allreduce")
             if derive_big:
                 # sibling = parent - built, batched over all pairs
-                pslots = torch.tensor(
-                    derive_parent_slot, dtype=torch.int64, device=self.device
-                )
-                spos = torch.tensor(
-                    derive_sib_pos, dtype=torch.int64, device=self.device
-                )
+                if self.device.type == "cuda":
+                    pslots = self._stage_i64(
+                        np.asarray(derive_parent_slot), "pslots"
+                    )
+                    spos = self._stage_i64(np.asarray(derive_sib_pos), "spos")
+                else:
+                    pslots = torch.tensor(
+                        derive_parent_slot, dtype=torch.int64
+                    )
+                    spos = torch.tensor(derive_sib_pos, dtype=torch.int64)
                 torch.sub(
                     prev_all_hist.index_select(0, pslots),
                     all_hist.index_select(0, spos),
@@ -514,14 +536,18 @@ class BoostingEngine:
 
             # ---- split scan over the whole frontier (scan-slot order)
             # ONE H2D for both parent-sum vectors
-            psums = torch.from_numpy(
-                np.array(
-                    [[nd.sum_g for nd in order_nodes],
-                     [nd.sum_h for nd in order_nodes]],
-                    dtype=np.int64,
-                )
-            ).to(self.device)
-            pg, ph = psums[0], psums[1]
+            psums_np = np.array(
+                [[nd.sum_g for nd in order_nodes],
+                 [nd.sum_h for nd in order_nodes]],
+                dtype=np.int64,
+            )
+            if self.device.type == "cuda":
+                psums = self._stage_i64(psums_np, "psums")
+                KK = psums_np.shape[1]
+                pg, ph = psums[:KK], psums[KK : 2 * KK]
+            else:
+                psums = torch.from_numpy(psums_np)
+                pg, ph = psums[0], psums[1]
             fb = self.feat_bins
             mask = feat_mask
             if self.p.colsample_bylevel < 1.0:
