@@ -525,16 +525,19 @@ __global__ __launch_bounds__(256) void find_splits_kf_kernel(
     long long* __restrict__ out_lg,    // [K, F]
     long long* __restrict__ out_lh,    // [K, F]
     int K, int F, int B) {
-  // One wave per (node, feature). All 64 lanes stage the histogram into
-  // LDS AND precompute the dequantized doubles in parallel (each element
-  // rounds independently, so parallel precompute is bitwise-identical to
-  // the CPU oracle's elementwise multiply). Then the two missing-value
-  // directions scan on lanes 0 and 1 concurrently - the sequential
-  // dependent f64 chain was 52% issue-stall as a single-lane loop.
-  extern __shared__ long long lds_h[];  // per wave: [B*2 i64][B*2 f64]
-  const int wave = threadIdx.x / WAVE;
-  const int lane = threadIdx.x % WAVE;
-  const int waves_per_block = blockDim.x / WAVE;
+  // One 32-lane HALF-wave per (node, feature): each hardware wave
+  // carries two independent (k, f) scans, doubling scan throughput per
+  // wave with the FP order unchanged. The 32 lanes stage the histogram
+  // into LDS AND precompute the dequantized doubles in parallel (each
+  // element rounds independently, so parallel precompute is
+  // bitwise-identical to the CPU oracle's elementwise multiply). Then
+  // the two missing-value directions scan on lanes 0 and 1 of the half
+  // concurrently - the sequential dependent f64 chain was 52%
+  // issue-stall as a single-lane loop.
+  extern __shared__ long long lds_h[];  // per half-wave: [B*2 i64][B*2 f64]
+  const int wave = threadIdx.x / 32;    // half-wave id within the block
+  const int lane = threadIdx.x % 32;
+  const int waves_per_block = blockDim.x / 32;
   int64_t kf = (int64_t)blockIdx.x * waves_per_block + wave;
   if (kf >= (int64_t)K * F) return;
   const int k = (int)(kf / F);
@@ -544,7 +547,7 @@ __global__ __launch_bounds__(256) void find_splits_kf_kernel(
   const long long* gh = hist + ((size_t)k * F + f) * B * 2;
   long long* h = lds_h + (size_t)wave * B * 4;
   double* hd = reinterpret_cast<double*>(h + B * 2);
-  for (int i = lane; i < B * 2; i += WAVE) {
+  for (int i = lane; i < B * 2; i += 32) {
     const long long v = gh[i];
     h[i] = v;
     hd[i] = (double)v * ((i & 1) ? inv_h : inv_g);
@@ -613,10 +616,10 @@ __global__ __launch_bounds__(256) void find_splits_kf_kernel(
   }
   // merge with the CPU oracle's preference: dl=0 wins only on a STRICTLY
   // greater gain (dl=1 was evaluated first there)
-  const double g1 = __shfl(best.gain, 1);
-  const int b1 = __shfl(best.bin, 1);
-  const long long lg1 = __shfl(best.left_g, 1);
-  const long long lh1 = __shfl(best.left_h, 1);
+  const double g1 = __shfl(best.gain, 1, 32);
+  const int b1 = __shfl(best.bin, 1, 32);
+  const long long lg1 = __shfl(best.left_g, 1, 32);
+  const long long lh1 = __shfl(best.left_h, 1, 32);
   if (lane != 0) return;
   if (g1 > best.gain) {
     best.gain = g1;
@@ -1334,11 +1337,13 @@ std::vector<torch::Tensor> find_splits(torch::Tensor hist, torch::Tensor parent_
   auto kf_lh = torch::empty({(int64_t)K * F}, optsl);
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   int64_t total = (int64_t)K * F;
+  // 4 half-waves of 32 lanes per 128-thread block: each hardware wave
+  // runs two independent (k, f) scans
   const int waves_per_block = 4;
   const size_t scan_lds = (size_t)waves_per_block * B * 4 * sizeof(long long);
   hipLaunchKernelGGL(find_splits_kf_kernel,
                      dim3((uint32_t)ceil_div(total, waves_per_block)),
-                     dim3(waves_per_block * WAVE), scan_lds, stream.stream(),
+                     dim3(waves_per_block * 32), scan_lds, stream.stream(),
                      reinterpret_cast<const long long*>(hist.data_ptr<int64_t>()),
                      reinterpret_cast<const long long*>(parent_g.data_ptr<int64_t>()),
                      reinterpret_cast<const long long*>(parent_h.data_ptr<int64_t>()),
