@@ -933,6 +933,11 @@ __global__ void predict_trees_kernel(
 // ---------------------------------------------------------------------------
 // update_margins: margin[ridx[seg]] += leaf_val[node]
 // ---------------------------------------------------------------------------
+// 8 rows per thread: the margin update is a random-line RMW per row
+// (leaf rows are ~n_leaves apart in the dense margin vector); one row
+// per thread left the kernel latency-parked at 3.4x the write floor.
+#define MARGIN_ROWS_PER_THREAD 8
+#define MARGIN_CHUNK (PART_THREADS * MARGIN_ROWS_PER_THREAD)
 __global__ void update_margins_kernel(
     float* __restrict__ margin, const int32_t* __restrict__ ridx,
     const int64_t* __restrict__ node_start,  // [K] + counts at [K..2K)
@@ -946,13 +951,21 @@ __global__ void update_margins_kernel(
   }
   const int node = lo;
   const int64_t chunk_in_node = wg - chunk_off[node];
-  const int64_t row_lo = chunk_in_node * PART_THREADS;
+  const int64_t row_lo = chunk_in_node * MARGIN_CHUNK;
   const int64_t count = node_start[K + node];
   const int64_t seg_start = node_start[node];
   const float v = leaf_vals[node];
-  const int64_t i = row_lo + threadIdx.x;
-  if (i < count) {
-    margin[(uint32_t)ridx[seg_start + i]] += v;
+  int32_t rv[MARGIN_ROWS_PER_THREAD];
+  bool valid[MARGIN_ROWS_PER_THREAD];
+  #pragma unroll
+  for (int s = 0; s < MARGIN_ROWS_PER_THREAD; ++s) {
+    const int64_t i = row_lo + s * PART_THREADS + threadIdx.x;
+    valid[s] = i < count;
+    rv[s] = valid[s] ? ridx[seg_start + i] : 0;
+  }
+  #pragma unroll
+  for (int s = 0; s < MARGIN_ROWS_PER_THREAD; ++s) {
+    if (valid[s]) margin[(uint32_t)rv[s]] += v;
   }
 }
 
@@ -1515,11 +1528,16 @@ void update_margins(torch::Tensor margin, torch::Tensor ridx,
     auto acc = chunk_off_cpu.accessor<int64_t, 1>();
     auto cacc = counts_cpu.accessor<int64_t, 1>();
     for (int k = 0; k < K; ++k)
-      acc[k + 1] = acc[k] + (cacc[k] + PART_THREADS - 1) / PART_THREADS;
+      acc[k + 1] = acc[k] + (cacc[k] + MARGIN_CHUNK - 1) / MARGIN_CHUNK;
     total_chunks = acc[K];
   }
   if (total_chunks == 0) return;
-  auto meta = torch::cat({starts_cpu, counts_cpu, chunk_off_cpu}).to(dev);
+  auto stream0 = at::cuda::getCurrentCUDAStream();
+  static thread_local PinnedStager margin_meta_stager;
+  auto meta_cpu = margin_meta_stager.get(3 * K + 1);
+  torch::cat_out(meta_cpu, {starts_cpu, counts_cpu, chunk_off_cpu});
+  auto meta = meta_cpu.to(dev, /*non_blocking=*/true);
+  margin_meta_stager.mark(stream0.stream());
   int64_t* mp = meta.data_ptr<int64_t>();
   auto lv = leaf_vals.to(dev).to(torch::kFloat32);
   auto stream = at::cuda::getCurrentCUDAStream();
