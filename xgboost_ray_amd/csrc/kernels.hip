@@ -418,7 +418,8 @@ __global__ __launch_bounds__(HIST_THREADS) void build_histogram_multifb_kernel(
 // workgroup. gpair_seg/ridx re-reads hit L2 the same way (the 13 blocks
 // read the same 48 KB of segment data).
 // ---------------------------------------------------------------------------
-__global__ __launch_bounds__(HIST_THREADS) void build_histogram_xcd_kernel(
+template <int THREADS>
+__global__ __launch_bounds__(THREADS) void build_histogram_xcd_kernel(
     const uint8_t* __restrict__ bins,        // [n_rows_total, row_stride]
     const int2* __restrict__ gpair_seg,      // [seg_total] segment order
     const int32_t* __restrict__ ridx,        // [seg_total]
@@ -1264,15 +1265,27 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
   // ridx pointer offset so seg indices align with gpair_seg
   if (xcd_mode) {
     const int64_t grid = ceil_div(total_chunks, 8) * 8 * (int64_t)n_fb;
-    hipLaunchKernelGGL((build_histogram_xcd_kernel), dim3((uint32_t)grid),
-                       dim3(HIST_THREADS), lds, stream.stream(),
-                       bins.data_ptr<uint8_t>(),
-                       (const int2*)gpair_seg.data_ptr<int32_t>(),
-                       ridx.data_ptr<int32_t>() + min_start, sc_adj_p,
-                       chunk_off_p,
-                       reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
-                       K, F, (int)n_bins, n_fb, row_stride, (int)f_lo,
-                       rows_per_wg, (int)total_chunks);
+    // 1024-thread WGs double waves/SIMD (2 WGs/CU within 160 KB LDS
+    // either way); RXGB_HIST_XCD_THREADS=512 selects the narrow variant
+    int xt = 1024;
+    if (const char* e = getenv("RXGB_HIST_XCD_THREADS")) {
+      if (atoi(e) == 512) xt = 512;
+    }
+    auto launch_xcd = [&](auto tc) {
+      hipLaunchKernelGGL((build_histogram_xcd_kernel<decltype(tc)::value>),
+                         dim3((uint32_t)grid), dim3(decltype(tc)::value),
+                         lds, stream.stream(), bins.data_ptr<uint8_t>(),
+                         (const int2*)gpair_seg.data_ptr<int32_t>(),
+                         ridx.data_ptr<int32_t>() + min_start, sc_adj_p,
+                         chunk_off_p,
+                         reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
+                         K, F, (int)n_bins, n_fb, row_stride, (int)f_lo,
+                         rows_per_wg, (int)total_chunks);
+    };
+    if (xt == 512)
+      launch_xcd(std::integral_constant<int, 512>{});
+    else
+      launch_xcd(std::integral_constant<int, 1024>{});
   } else if (multifb) {
     auto launch_mfb = [&](auto rc) {
       hipLaunchKernelGGL((build_histogram_multifb_kernel<decltype(rc)::value>),
