@@ -1266,10 +1266,13 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
   if (xcd_mode) {
     const int64_t grid = ceil_div(total_chunks, 8) * 8 * (int64_t)n_fb;
     // 1024-thread WGs double waves/SIMD (2 WGs/CU within 160 KB LDS
-    // either way); RXGB_HIST_XCD_THREADS=512 selects the narrow variant
-    int xt = 1024;
+    // either way) and win on huge matrices (100M: hist 55.6 -> 53.9 ms)
+    // but lose ~3% on 11M-row shapes (deep-depth tail waste), so gate on
+    // matrix size; RXGB_HIST_XCD_THREADS=512/1024 overrides.
+    int xt = bins.size(0) > (int64_t)32 * 1024 * 1024 ? 1024 : 512;
     if (const char* e = getenv("RXGB_HIST_XCD_THREADS")) {
-      if (atoi(e) == 512) xt = 512;
+      int v = atoi(e);
+      if (v == 512 || v == 1024) xt = v;
     }
     auto launch_xcd = [&](auto tc) {
       hipLaunchKernelGGL((build_histogram_xcd_kernel<decltype(tc)::value>),
