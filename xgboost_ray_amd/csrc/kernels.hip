@@ -150,13 +150,10 @@ static int hist_rows_per_wg() {
 // [16*n_bins + f*n_bins + b]. Interleaved (g,h) pairs put the bank index
 // at (b*4)%64 -> only bin%16 distinguishes banks; the split doubles the
 // spread to bin%32 (PMC: conflict/active 0.74 interleaved).
-__device__ __forceinline__ void hist_accum_row16(
-    unsigned long long* lds_hist, const uint8_t* __restrict__ bins,
-    uint64_t r, int64_t row_stride, int f0, int n_bins, int lane,
+__device__ __forceinline__ void hist_accum_packed16(
+    unsigned long long* lds_hist, uint4 packed, int n_bins, int lane,
     longlong2 gp) {
   const int hplane = 16 * n_bins;
-  const uint4 packed =
-      *reinterpret_cast<const uint4*>(bins + r * row_stride + f0);
   const int r4 = lane & 3;
   const uint32_t w0 = packed.x, w1 = packed.y, w2 = packed.z, w3 = packed.w;
   const bool s1 = (r4 & 1) != 0, s2 = (r4 & 2) != 0;
@@ -195,6 +192,15 @@ __device__ __forceinline__ void hist_accum_row16(
       }
     }
   }
+}
+
+__device__ __forceinline__ void hist_accum_row16(
+    unsigned long long* lds_hist, const uint8_t* __restrict__ bins,
+    uint64_t r, int64_t row_stride, int f0, int n_bins, int lane,
+    longlong2 gp) {
+  const uint4 packed =
+      *reinterpret_cast<const uint4*>(bins + r * row_stride + f0);
+  hist_accum_packed16(lds_hist, packed, n_bins, lane, gp);
 }
 
 template <int VFB>  // 16: uint4 row loads; 8: uint2; 0: byte fallback
@@ -457,12 +463,31 @@ __global__ __launch_bounds__(THREADS) void build_histogram_xcd_kernel(
   int64_t row_hi = row_lo + rows_per_wg;
   if (row_hi > node_count) row_hi = node_count;
   const int lane = threadIdx.x & (WAVE - 1);
-  for (int64_t i = row_lo + threadIdx.x; i < row_hi; i += blockDim.x) {
+  // 2-row software pipeline: both rows' random uint4 gathers are in
+  // flight before either LDS-atomic burst starts (SQ_WAIT_ANY was 17x
+  // SQ_BUSY with one outstanding gather per thread)
+  for (int64_t i = row_lo + threadIdx.x; i < row_hi;
+       i += 2 * (int64_t)blockDim.x) {
+    const int64_t i2 = i + blockDim.x;
     const int64_t seg_i = seg_start + i;
     const int2 gpi = gpair_seg[seg_i];
-    const longlong2 gp = {(long long)gpi.x, (long long)gpi.y};
-    hist_accum_row16(lds_hist, bins, (uint32_t)ridx[seg_i], row_stride, f0,
-                     n_bins, lane, gp);
+    const uint64_t r1 = (uint32_t)ridx[seg_i];
+    const uint4 p1 = *reinterpret_cast<const uint4*>(
+        bins + r1 * row_stride + f0);
+    const bool v2 = i2 < row_hi;
+    int2 gpi2 = {0, 0};
+    uint4 p2 = {0, 0, 0, 0};
+    if (v2) {
+      const int64_t seg_i2 = seg_start + i2;
+      gpi2 = gpair_seg[seg_i2];
+      const uint64_t r2 = (uint32_t)ridx[seg_i2];
+      p2 = *reinterpret_cast<const uint4*>(bins + r2 * row_stride + f0);
+    }
+    hist_accum_packed16(lds_hist, p1, n_bins, lane,
+                        {(long long)gpi.x, (long long)gpi.y});
+    if (v2)
+      hist_accum_packed16(lds_hist, p2, n_bins, lane,
+                          {(long long)gpi2.x, (long long)gpi2.y});
   }
   __syncthreads();
 
