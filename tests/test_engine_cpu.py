@@ -414,3 +414,72 @@ class TestMonotone:
         grid[:, 1] = 0.5
         pred = bst.predict(grid, output_margin=True)
         assert (np.diff(pred) <= 1e-6).all()
+
+
+def _paths_respect_constraints(bst, sets):
+    """Every root->node path must satisfy: each split feature shares a
+    constraint set with ALL features used above it on the path."""
+    sets = [frozenset(s) for s in sets]
+    for t in bst.trees:
+        stack = [(0, frozenset())]
+        while stack:
+            nid, path = stack.pop()
+            f = int(t.feat[nid])
+            if f < 0:
+                continue
+            if path:
+                ok = any(path | {f} <= cs for cs in sets)
+                if not ok:
+                    return False
+            lid = int(t.left[nid])
+            stack.append((lid, path | {f}))
+            stack.append((lid + 1, path | {f}))
+    return True
+
+
+class TestInteractionConstraints:
+    def _data(self, seed=0, n=8000):
+        rng = np.random.RandomState(seed)
+        X = rng.rand(n, 6).astype(np.float32)
+        # strong pairwise interactions across the constraint boundary so
+        # an unconstrained tree would freely mix all features
+        y = (
+            X[:, 0] * X[:, 3]
+            + X[:, 1] * X[:, 4]
+            + X[:, 2] * X[:, 5]
+            + 0.05 * rng.randn(n)
+        ).astype(np.float32)
+        return X, y
+
+    def test_paths_confined_to_sets(self):
+        X, y = self._data()
+        dm = BinnedMatrix.build(
+            torch.from_numpy(X), label=torch.from_numpy(y), max_bin=64
+        )
+        sets = [[0, 1, 2], [3, 4, 5]]
+        bst = run_training(
+            {"objective": "reg:squarederror", "max_depth": 5, "eta": 0.3,
+             "interaction_constraints": sets},
+            dm, 10,
+        )
+        assert _paths_respect_constraints(bst, sets)
+        # sanity: without constraints the same data mixes features
+        bst2 = run_training(
+            {"objective": "reg:squarederror", "max_depth": 5, "eta": 0.3},
+            dm, 10,
+        )
+        assert not _paths_respect_constraints(bst2, sets)
+
+    def test_string_spec_and_lossguide(self):
+        X, y = self._data(seed=1)
+        dm = BinnedMatrix.build(
+            torch.from_numpy(X), label=torch.from_numpy(y), max_bin=64
+        )
+        sets = [[0, 1], [2, 3], [4, 5]]
+        bst = run_training(
+            {"objective": "reg:squarederror", "eta": 0.3,
+             "grow_policy": "lossguide", "max_leaves": 24, "max_depth": 6,
+             "interaction_constraints": "[[0, 1], [2, 3], [4, 5]]"},
+            dm, 8,
+        )
+        assert _paths_respect_constraints(bst, sets)

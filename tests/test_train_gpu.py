@@ -93,3 +93,46 @@ def test_deep_tree_gpu():
         ray_params=RayParams(num_actors=1, gpus_per_actor=1),
     )
     assert res["train"]["logloss"][-1] < 0.3
+
+
+@pytest.mark.gpu
+def test_interaction_constraints_gpu():
+    """GPU-trained trees obey interaction constraints (scan-kernel gate)."""
+    import torch
+    from tests.test_engine_cpu import _paths_respect_constraints
+    from xgboost_ray_amd.engine.quantile import BinnedMatrix
+    from xgboost_ray_amd.engine.trainer import run_training
+
+    rng = np.random.RandomState(3)
+    n = 20000
+    X = rng.rand(n, 6).astype(np.float32)
+    y = (
+        X[:, 0] * X[:, 3] + X[:, 1] * X[:, 4] + X[:, 2] * X[:, 5]
+        + 0.05 * rng.randn(n)
+    ).astype(np.float32)
+    dm = BinnedMatrix.build(
+        torch.from_numpy(X).cuda(),
+        label=torch.from_numpy(y).cuda(),
+        max_bin=64,
+    )
+    sets = [[0, 1, 2], [3, 4, 5]]
+    bst = run_training(
+        {"objective": "reg:squarederror", "max_depth": 5, "eta": 0.3,
+         "interaction_constraints": sets},
+        dm, 10,
+    )
+    assert _paths_respect_constraints(bst, sets)
+
+    # and bitwise CPU == GPU with the constraint active
+    dm_cpu = BinnedMatrix.build(
+        torch.from_numpy(X), label=torch.from_numpy(y), max_bin=64
+    )
+    bst_cpu = run_training(
+        {"objective": "reg:squarederror", "max_depth": 5, "eta": 0.3,
+         "interaction_constraints": sets},
+        dm_cpu, 10,
+    )
+    for tg, tc in zip(bst.trees, bst_cpu.trees):
+        assert np.array_equal(tg.feat, tc.feat)
+        assert np.array_equal(tg.thr, tc.thr)
+        assert np.array_equal(tg.value, tc.value)

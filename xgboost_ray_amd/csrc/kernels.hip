@@ -545,6 +545,7 @@ __global__ __launch_bounds__(256) void find_splits_kf_kernel(
     double lam, double alpha, double gamma, double mcw,
     const int8_t* __restrict__ mono,    // [F] or nullptr
     const double* __restrict__ bounds,  // [K, 2] node weight bounds
+    const uint8_t* __restrict__ allowed,  // [K, F] interaction gate or null
     double* __restrict__ out_gain,     // [K, F]
     int32_t* __restrict__ out_bin,     // [K, F]
     uint8_t* __restrict__ out_dl,      // [K, F]
@@ -568,6 +569,18 @@ __global__ __launch_bounds__(256) void find_splits_kf_kernel(
   if (kf >= (int64_t)K * F) return;
   const int k = (int)(kf / F);
   const int f = (int)(kf % F);
+  if (allowed != nullptr && allowed[kf] == 0) {
+    // interaction constraints: disallowed (node, feature) -> the same
+    // no-split sentinel the CPU oracle produces (gain -1, dl 0)
+    if (lane == 0) {
+      out_gain[kf] = -1.0;
+      out_bin[kf] = 0;
+      out_dl[kf] = 0;
+      out_lg[kf] = 0;
+      out_lh[kf] = 0;
+    }
+    return;
+  }
   const double inv_g = 1.0 / scale_g;
   const double inv_h = 1.0 / scale_h;
   const long long* gh = hist + ((size_t)k * F + f) * B * 2;
@@ -1375,7 +1388,8 @@ std::vector<torch::Tensor> find_splits(torch::Tensor hist, torch::Tensor parent_
                                        torch::Tensor feat_bins, double scale_g,
                                        double scale_h, double lam, double alpha,
                                        double gamma, double mcw,
-                                       torch::Tensor mono, torch::Tensor bounds) {
+                                       torch::Tensor mono, torch::Tensor bounds,
+                                       torch::Tensor allowed) {
   const int K = (int)hist.size(0);
   const int F = (int)hist.size(1);
   const int B = (int)hist.size(2);
@@ -1405,6 +1419,7 @@ std::vector<torch::Tensor> find_splits(torch::Tensor hist, torch::Tensor parent_
                      alpha, gamma, mcw,
                      mono.numel() ? mono.data_ptr<int8_t>() : nullptr,
                      bounds.numel() ? bounds.data_ptr<double>() : nullptr,
+                     allowed.numel() ? allowed.data_ptr<uint8_t>() : nullptr,
                      kf_gain.data_ptr<double>(),
                      kf_bin.data_ptr<int32_t>(), kf_dl.data_ptr<uint8_t>(),
                      reinterpret_cast<long long*>(kf_lg.data_ptr<int64_t>()),

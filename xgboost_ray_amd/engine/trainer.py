@@ -55,6 +55,7 @@ class TrainParams:
     grow_policy: str = "depthwise"
     max_leaves: int = 0
     monotone_constraints: object = None
+    interaction_constraints: object = None
     tree_method: str = "hist"
     verbosity: int = 1
     nthread: int = 0
@@ -89,6 +90,12 @@ class TrainParams:
             p.monotone_constraints = [
                 int(x) for x in txt.split(",") if x.strip()
             ]
+        if isinstance(p.interaction_constraints, str):
+            import ast
+
+            p.interaction_constraints = ast.literal_eval(
+                p.interaction_constraints
+            )
         return p
 
 
@@ -117,6 +124,7 @@ class _Node:
     parent_slot: int = 0  # parent's scan-slot in the previous depth
     w_lower: float = -math.inf  # monotone weight bounds
     w_upper: float = math.inf
+    path_feats: tuple = ()  # split features on the root path (sorted)
 
 
 class CallbackList:
@@ -195,6 +203,15 @@ class BoostingEngine:
                         "monotone_constraints are not supported with "
                         "grow_policy=lossguide yet"
                     )
+        # interaction constraints: list of feature-index groups
+        # (reference defers to xgboost's `interaction_constraints`;
+        # semantics: a split feature must share a constraint set with
+        # every feature already used on the node's root path)
+        self.interaction_sets = None
+        ic = self.p.interaction_constraints
+        if ic:
+            self.interaction_sets = [frozenset(int(f) for f in g)
+                                     for g in ic]
         self.iteration = 0
         self.booster = Booster(
             params={
@@ -363,6 +380,33 @@ class BoostingEngine:
         view = buf[:n]
         view.copy_(torch.from_numpy(arr))
         return view.to(self.device, non_blocking=True)
+
+
+    def _allowed_mask(self, paths):
+        """[K, F] uint8 gate from interaction constraints (None if off).
+
+        ``paths``: per node, the tuple of split features on its root
+        path. A feature is allowed iff some constraint set contains the
+        whole path and the feature (xgboost semantics; at the root every
+        feature may open a path).
+        """
+        if self.interaction_sets is None:
+            return None
+        F = self.dtrain.n_features
+        m = np.zeros((len(paths), F), np.uint8)
+        for i, path in enumerate(paths):
+            pf = set(path)
+            if not pf:
+                m[i, :] = 1  # root: any feature may open the path
+            else:
+                allow = set()
+                for cs in self.interaction_sets:
+                    if pf <= cs:
+                        allow |= cs
+                for f in allow:
+                    if 0 <= f < F:
+                        m[i, f] = 1
+        return torch.from_numpy(m)
 
     def _grow_tree(
         self, gpair: torch.Tensor, it: int, cls: int, ptree: int = 0
@@ -577,6 +621,9 @@ class BoostingEngine:
                 self.p.min_child_weight,
                 monotone=self.mono,
                 bounds=mono_bounds,
+                allowed=self._allowed_mask(
+                    [nd.path_feats for nd in order_nodes]
+                ),
             )
             _tick("scan")
             gain = best["gain"]
@@ -653,6 +700,11 @@ class BoostingEngine:
             for k, nd in enumerate(split_nodes):
                 lid, lg, lh = children_meta[k]
                 lcount = int(left_counts[k])
+                child_path = ()
+                if self.interaction_sets is not None:
+                    child_path = tuple(
+                        sorted(set(nd.path_feats) | {int(sf[k])})
+                    )
                 lnode = _Node(
                     nid=lid,
                     depth=depth + 1,
@@ -663,6 +715,7 @@ class BoostingEngine:
                     parent_slot=nd.slot,
                     w_lower=nd.w_lower,
                     w_upper=nd.w_upper,
+                    path_feats=child_path,
                 )
                 rnode = _Node(
                     nid=lid + 1,
@@ -674,6 +727,7 @@ class BoostingEngine:
                     parent_slot=nd.slot,
                     w_lower=nd.w_lower,
                     w_upper=nd.w_upper,
+                    path_feats=child_path,
                 )
                 if self.mono is not None:
                     c = int(self.mono[sf[k]])
@@ -877,7 +931,7 @@ class BoostingEngine:
         left_l, dl_l, val_l, gain_l, cover_l = [-1], [0], [0.0], [0.0], [0.0]
         parent_l = [-1]
 
-        def scan_nodes(hists, sums):
+        def scan_nodes(hists, sums, paths=None):
             best = ops.find_splits(
                 hists,
                 torch.tensor([sg for sg, sh in sums], dtype=torch.int64,
@@ -886,6 +940,9 @@ class BoostingEngine:
                              device=self.device),
                 fb, scale_g, scale_h, self.p.reg_lambda, self.p.reg_alpha,
                 self.p.gamma, self.p.min_child_weight,
+                allowed=self._allowed_mask(paths)
+                if paths is not None and self.interaction_sets is not None
+                else None,
             )
             return best
 
@@ -898,15 +955,16 @@ class BoostingEngine:
         )
         if self.coll.is_distributed:
             self.coll.allreduce_(root_hist)
-        rbest = scan_nodes(root_hist, [(int(root_sum[0]), int(root_sum[1]))])
+        rbest = scan_nodes(root_hist, [(int(root_sum[0]), int(root_sum[1]))],
+                           paths=[()])
         heap = []
         counter = [0]
 
-        def push(nid, depth, start, count, sg, sh, hist, best_k):
+        def push(nid, depth, start, count, sg, sh, hist, best_k, path=()):
             gain = float(best_k["gain"])
             rec = {
                 "nid": nid, "depth": depth, "start": start, "count": count,
-                "sg": sg, "sh": sh, "hist": hist,
+                "sg": sg, "sh": sh, "hist": hist, "path": path,
                 "feature": int(best_k["feature"]), "bin": int(best_k["bin"]),
                 "dl": int(best_k["default_left"]),
                 "lg": int(best_k["left_g"]), "lh": int(best_k["left_h"]),
@@ -980,9 +1038,11 @@ class BoostingEngine:
                 hists = torch.stack([sh_hist[0], other_hist])
             else:
                 hists = torch.stack([other_hist, sh_hist[0]])
+            child_path = tuple(sorted(set(rec.get("path", ())) | {f}))
             cbest = scan_nodes(
                 hists, [(children[0][4], children[0][5]),
-                        (children[1][4], children[1][5])]
+                        (children[1][4], children[1][5])],
+                paths=[child_path, child_path],
             )
             for ci, ch in enumerate(children):
                 if ch[1] >= max_depth:
@@ -992,7 +1052,7 @@ class BoostingEngine:
                     )
                 else:
                     push(ch[0], ch[1], ch[2], ch[3], ch[4], ch[5],
-                         hists[ci], best_row(cbest, ci))
+                         hists[ci], best_row(cbest, ci), path=child_path)
             n_leaves += 1
 
         # drain remaining candidates as leaves

@@ -94,6 +94,7 @@ def find_splits(
     min_child_weight,
     monotone=None,
     bounds=None,
+    allowed=None,
 ):
     """Best-split scan over histograms.
 
@@ -117,6 +118,7 @@ def find_splits(
         min_child_weight,
         monotone,
         bounds,
+        allowed,
     )
 
 

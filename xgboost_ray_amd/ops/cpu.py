@@ -113,6 +113,7 @@ def find_splits(
     min_child_weight,
     monotone=None,
     bounds=None,
+    allowed=None,
 ):
     K, F, B, _ = hist.shape
     dev = hist.device
@@ -164,6 +165,9 @@ def find_splits(
             & (hl >= min_child_weight)
             & (hr >= min_child_weight)
         )
+        if allowed is not None:
+            # interaction constraints: per-(node, feature) gate
+            ok = ok & (allowed.to(dev) != 0).view(K, F, 1)
         if monotone is not None:
             # monotone constraints: clamp child weights into the node's
             # bound interval, then require the constrained ordering

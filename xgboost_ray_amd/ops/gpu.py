@@ -81,6 +81,7 @@ def find_splits(
     min_child_weight,
     monotone=None,
     bounds=None,
+    allowed=None,
 ):
     import numpy as np
 
@@ -88,6 +89,10 @@ def find_splits(
     if monotone is None:
         monotone = torch.zeros(0, dtype=torch.int8, device=dev)
         bounds = torch.zeros((0, 2), dtype=torch.float64, device=dev)
+    if allowed is None:
+        allowed = torch.zeros(0, dtype=torch.uint8, device=dev)
+    else:
+        allowed = allowed.to(dev).to(torch.uint8).contiguous()
     (packed,) = _load().find_splits(
         hist,
         parent_g.to(dev),
@@ -101,6 +106,7 @@ def find_splits(
         float(min_child_weight),
         monotone.to(dev).to(torch.int8),
         bounds.to(dev).to(torch.float64),
+        allowed,
     )
     arr = packed.cpu().numpy()  # ONE D2H for the whole depth's splits
     return {
