@@ -101,14 +101,19 @@ class AUC(Metric):
         b = torch.clamp((p * _AUC_BINS).long(), max=_AUC_BINS - 1)
         if weight is None:
             # unweighted: integer scatter (native u64 atomics on GPU -
-            # the f64 scatter goes through CAS loops and is ~50x slower)
-            ones = torch.ones_like(b)
-            ypos = label > 0.5
-            pos = torch.zeros(_AUC_BINS, dtype=torch.int64, device=margin.device)
-            neg = torch.zeros(_AUC_BINS, dtype=torch.int64, device=margin.device)
-            pos.scatter_add_(0, b[ypos], ones[ypos])
-            neg.scatter_add_(0, b[~ypos], ones[~ypos])
-            return torch.cat([pos, neg]).double()
+            # the f64 scatter goes through CAS loops and is ~50x slower).
+            # One fused scatter into [neg plane | pos plane]: boolean
+            # fancy-indexing (b[ypos]) costs a nonzero + sort + two
+            # gathers per eval round.
+            idx = b + (label > 0.5).long() * _AUC_BINS
+            hist = torch.zeros(
+                2 * _AUC_BINS, dtype=torch.int64, device=margin.device
+            )
+            src = torch.ones(
+                (), dtype=torch.int64, device=margin.device
+            ).expand(idx.shape)
+            hist.scatter_add_(0, idx, src)
+            return torch.cat([hist[_AUC_BINS:], hist[:_AUC_BINS]]).double()
         w = _w(label, weight)
         pos = torch.zeros(_AUC_BINS, dtype=torch.float64, device=margin.device)
         neg = torch.zeros(_AUC_BINS, dtype=torch.float64, device=margin.device)
