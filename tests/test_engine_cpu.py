@@ -483,3 +483,57 @@ class TestInteractionConstraints:
             dm, 8,
         )
         assert _paths_respect_constraints(bst, sets)
+
+
+class TestExtraMetricsAndPredict:
+    def test_aucpr_metric(self):
+        from xgboost_ray_amd.engine.metrics import get_metric
+
+        rng = np.random.RandomState(0)
+        n = 20000
+        score = torch.from_numpy(rng.randn(n).astype(np.float32))
+        label = torch.from_numpy(
+            (score.numpy() + rng.randn(n) > 0).astype(np.float32)
+        )
+        m = get_metric("aucpr")
+        v = m.finalize(m.local_stats(score, label, None, None, None))
+        from sklearn.metrics import average_precision_score
+        import scipy.special as sp
+
+        ref = average_precision_score(label.numpy(), sp.expit(score.numpy()))
+        assert abs(v - ref) < 0.01
+
+    def test_aucpr_in_training(self):
+        from xgboost_ray_amd.engine.trainer import EvalPack
+
+        dm, _, _ = _binned(kind="binary")
+        res = {}
+        bst = run_training(
+            {"objective": "binary:logistic", "max_depth": 3,
+             "eval_metric": ["aucpr"]},
+            dm, 5, evals=[EvalPack(name="train", X=None)],
+            evals_result=res,
+        )
+        assert "aucpr" in res.get("train", {})
+        vals = res["train"]["aucpr"]
+        assert len(vals) == 5 and 0.0 <= vals[-1] <= 1.0
+        assert vals[-1] >= vals[0] - 0.05  # improves (or holds) with rounds
+
+    def test_pred_leaf(self):
+        rng = np.random.RandomState(0)
+        X = rng.rand(1500, 5).astype(np.float32)
+        y = (X[:, 0] + X[:, 1] > 1).astype(np.float32)
+        dm = BinnedMatrix.build(
+            torch.from_numpy(X), label=torch.from_numpy(y), max_bin=32
+        )
+        bst = run_training(
+            {"objective": "binary:logistic", "max_depth": 4}, dm, 4
+        )
+        leaves = bst.predict(X, pred_leaf=True)
+        assert leaves.shape == (1500, 4)
+        for ti, t in enumerate(bst.trees):
+            assert (t.feat[leaves[:, ti]] < 0).all()
+        # iteration_range subsets trees
+        l2 = bst.predict(X, pred_leaf=True, iteration_range=(1, 3))
+        assert l2.shape == (1500, 2)
+        assert np.array_equal(l2[:, 0], leaves[:, 1])

@@ -269,6 +269,7 @@ class Booster:
         output_margin: bool = False,
         iteration_range=None,
         validate_features: bool = True,
+        pred_leaf: bool = False,
         **kwargs,
     ) -> np.ndarray:
         from xgboost_ray_amd.engine.objectives import get_objective
@@ -279,6 +280,8 @@ class Booster:
         else:
             X = _as_float32_matrix(data)
             bm = None
+        if pred_leaf:
+            return self.predict_leaf(X, iteration_range)
         Xt = torch.from_numpy(np.ascontiguousarray(X, dtype=np.float32))
         margin = self.predict_margin_tensor(Xt, iteration_range)
         if bm is not None:
@@ -289,6 +292,38 @@ class Booster:
             return margin.cpu().numpy()
         obj = get_objective(self.objective, self.num_class)
         return obj.transform_prediction(margin).cpu().numpy()
+
+
+    def predict_leaf(self, X, iteration_range=None) -> np.ndarray:
+        """Leaf index per (row, tree) - xgboost ``pred_leaf=True``."""
+        X = _as_float32_matrix(X)
+        lo, hi = 0, self.num_boosted_rounds()
+        if iteration_range is not None:
+            lo, hi = iteration_range
+            hi = hi or self.num_boosted_rounds()
+        k = max(1, self.num_class) * self.num_parallel_tree
+        trees = self.trees[lo * k : hi * k]
+        n = X.shape[0]
+        out = np.zeros((n, len(trees)), dtype=np.int32)
+        for ti, t in enumerate(trees):
+            cur = np.zeros(n, dtype=np.int32)
+            # walk all rows level-synchronously (vectorized numpy)
+            while True:
+                f = t.feat[cur]
+                inner = f >= 0
+                if not inner.any():
+                    break
+                rows = np.nonzero(inner)[0]
+                fv = X[rows, f[rows]]
+                left = t.left[cur[rows]]
+                go_left = fv < t.thr[cur[rows]]
+                miss = np.isnan(fv)
+                go_left = np.where(
+                    miss, t.default_left[cur[rows]].astype(bool), go_left
+                )
+                cur[rows] = np.where(go_left, left, left + 1)
+            out[:, ti] = cur
+        return out
 
     # -- persistence -------------------------------------------------------
     def __getstate__(self):

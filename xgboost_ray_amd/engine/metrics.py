@@ -132,6 +132,35 @@ class AUC(Metric):
         return float(auc / denom) if denom > 0 else 0.5
 
 
+class AUCPR(Metric):
+    """Binned PR AUC: precision-recall area via the same score bins as AUC
+    (xgboost `aucpr`; reference passes eval_metric through, SURVEY #2.3)."""
+
+    name = "aucpr"
+    higher_better = True
+
+    def local_stats(self, margin, label, weight, qid, obj):
+        return AUC.local_stats(self, margin, label, weight, qid, obj)
+
+    def finalize(self, s):
+        pos, neg = s[:_AUC_BINS], s[_AUC_BINS:]
+        tot_pos = pos.sum()
+        if float(tot_pos) <= 0:
+            return 0.0
+        # descending-score sweep: cumulative TP/FP from the top bin down
+        tp = torch.flip(torch.cumsum(torch.flip(pos, [0]), 0), [0])
+        fp = torch.flip(torch.cumsum(torch.flip(neg, [0]), 0), [0])
+        recall = tp / tot_pos
+        prec = tp / torch.clamp(tp + fp, min=1e-300)
+        # integrate precision d(recall) over bins, descending threshold:
+        # recall increases as the threshold drops
+        r = torch.flip(recall, [0])
+        p = torch.flip(prec, [0])
+        dr = torch.diff(r, prepend=torch.zeros(1, dtype=r.dtype,
+                                               device=r.device))
+        return float((p * dr).sum())
+
+
 class MLogLoss(Metric):
     name = "mlogloss"
 
@@ -228,6 +257,7 @@ def get_metric(name: str) -> Metric:
         "logloss": LogLoss,
         "error": BinaryError,
         "auc": AUC,
+        "aucpr": AUCPR,
         "mlogloss": MLogLoss,
         "merror": MError,
         "ndcg": NDCG,
