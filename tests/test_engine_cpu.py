@@ -537,3 +537,62 @@ class TestExtraMetricsAndPredict:
         l2 = bst.predict(X, pred_leaf=True, iteration_range=(1, 3))
         assert l2.shape == (1500, 2)
         assert np.array_equal(l2[:, 0], leaves[:, 1])
+
+    def test_pred_contribs_additivity_and_exactness(self):
+        from itertools import combinations
+
+        rng = np.random.RandomState(0)
+        X = rng.rand(400, 4).astype(np.float32)
+        y = (X[:, 0] * X[:, 1] + X[:, 2] + 0.1 * rng.randn(400)).astype(
+            np.float32
+        )
+        dm = BinnedMatrix.build(
+            torch.from_numpy(X), label=torch.from_numpy(y), max_bin=16
+        )
+        bst = run_training(
+            {"objective": "reg:squarederror", "max_depth": 3}, dm, 3
+        )
+        rows = X[:8]
+        contribs = bst.predict(rows, pred_contribs=True)
+        margin = bst.predict(rows, output_margin=True)
+        assert np.abs(contribs.sum(1) - margin).max() < 1e-5
+
+        # brute-force Shapley with cover-weighted conditional expectation
+        def cond_exp(t, x, S):
+            def rec(nid):
+                f = t.feat[nid]
+                if f < 0:
+                    return float(t.value[nid])
+                l, r = int(t.left[nid]), int(t.left[nid]) + 1
+                if f in S:
+                    fv = x[f]
+                    if np.isnan(fv):
+                        return rec(l if t.default_left[nid] else r)
+                    return rec(l if fv < t.thr[nid] else r)
+                cl, cr = float(t.cover[l]), float(t.cover[r])
+                tot = cl + cr if cl + cr > 0 else 1.0
+                return (cl * rec(l) + cr * rec(r)) / tot
+
+            return rec(0)
+
+        import math
+
+        F = 4
+        for xi in range(3):
+            x = rows[xi]
+            exact = np.zeros(F)
+            for f in range(F):
+                others = [j for j in range(F) if j != f]
+                for k in range(F):
+                    for S in combinations(others, k):
+                        wgt = (
+                            math.factorial(k)
+                            * math.factorial(F - k - 1)
+                            / math.factorial(F)
+                        )
+                        for t in bst.trees:
+                            exact[f] += wgt * (
+                                cond_exp(t, x, set(S) | {f})
+                                - cond_exp(t, x, set(S))
+                            )
+            assert np.abs(exact - contribs[xi, :F]).max() < 1e-5

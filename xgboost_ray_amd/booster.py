@@ -101,6 +101,81 @@ def _as_float32_vec(v):
     return np.ascontiguousarray(np.asarray(v), dtype=np.float32).reshape(-1)
 
 
+
+def _tree_shap(t, x, mean_val, cover, out, scale):
+    """One row's SHAP contributions for one tree (Lundberg EXTEND/UNWIND).
+
+    ``m`` is the path of unique features: fraction of zero paths (z),
+    fraction of one paths (o), and feature index (d)."""
+
+    def extend(m, pz, po, pi):
+        m = m + [[pz, po, pi, 1.0 if len(m) == 0 else 0.0]]
+        ln = len(m)
+        for i in range(ln - 2, -1, -1):
+            m[i + 1][3] += po * m[i][3] * (i + 1) / ln
+            m[i][3] = pz * m[i][3] * (ln - 1 - i) / ln
+        return m
+
+    def unwind_sum(m, i):
+        ln = len(m) - 1
+        po, pz = m[i][1], m[i][0]
+        total = 0.0
+        if po != 0.0:
+            nxt = m[ln][3]
+            for j in range(ln - 1, -1, -1):
+                tmp = nxt * (ln + 1) / ((j + 1) * po)
+                total += tmp
+                nxt = m[j][3] - tmp * pz * (ln - j) / (ln + 1)
+        else:
+            for j in range(ln):
+                total += m[j][3] * (ln + 1) / (pz * (ln - j))
+        return total
+
+    def recurse(nid, m, pz, po, pi):
+        m = extend(list(map(list, m)), pz, po, pi)
+        f = t.feat[nid]
+        if f < 0:
+            v = t.value[nid] * scale
+            for i in range(1, len(m)):
+                w = unwind_sum(m, i)
+                out[m[i][2]] += w * (m[i][1] - m[i][0]) * v
+            return
+        l, r = int(t.left[nid]), int(t.left[nid] + 1)
+        fv = x[f]
+        if np.isnan(fv):
+            hot = l if t.default_left[nid] else r
+        else:
+            hot = l if fv < t.thr[nid] else r
+        cold = r if hot == l else l
+        iz, io = 1.0, 1.0
+        # if this feature already on the path, unwind it first
+        path_idx = next(
+            (i for i in range(1, len(m)) if m[i][2] == f), None
+        )
+        if path_idx is not None:
+            iz, io = m[path_idx][0], m[path_idx][1]
+            # unwind
+            ln = len(m) - 1
+            po_, pz_ = m[path_idx][1], m[path_idx][0]
+            nxt = m[ln][3]
+            for j in range(ln - 1, -1, -1):
+                if po_ != 0.0:
+                    tmp = nxt * (ln + 1) / ((j + 1) * po_)
+                    nxt = m[j][3] - tmp * pz_ * (ln - j) / (ln + 1)
+                    m[j][3] = tmp
+                else:
+                    m[j][3] = m[j][3] * (ln + 1) / (pz_ * (ln - j))
+            for j in range(path_idx, ln):
+                m[j][0], m[j][1], m[j][2] = m[j + 1][0], m[j + 1][1], m[j + 1][2]
+            m = m[:-1]
+        c = cover[nid] if cover[nid] > 0 else 1.0
+        recurse(hot, m, iz * cover[hot] / c, io, f)
+        recurse(cold, m, iz * cover[cold] / c, 0.0, f)
+
+    out[len(x)] += mean_val[0] * scale
+    recurse(0, [], 1.0, 1.0, -1)
+
+
 class Booster:
     def __init__(
         self,
@@ -270,6 +345,7 @@ class Booster:
         iteration_range=None,
         validate_features: bool = True,
         pred_leaf: bool = False,
+        pred_contribs: bool = False,
         **kwargs,
     ) -> np.ndarray:
         from xgboost_ray_amd.engine.objectives import get_objective
@@ -282,6 +358,8 @@ class Booster:
             bm = None
         if pred_leaf:
             return self.predict_leaf(X, iteration_range)
+        if pred_contribs:
+            return self.predict_contribs(X, iteration_range)
         Xt = torch.from_numpy(np.ascontiguousarray(X, dtype=np.float32))
         margin = self.predict_margin_tensor(Xt, iteration_range)
         if bm is not None:
@@ -293,6 +371,64 @@ class Booster:
         obj = get_objective(self.objective, self.num_class)
         return obj.transform_prediction(margin).cpu().numpy()
 
+
+    def predict_contribs(self, X, iteration_range=None) -> np.ndarray:
+        """Exact TreeSHAP feature attributions (xgboost ``pred_contribs``).
+
+        Returns [n, F+1]; the last column is the bias (expected value).
+        Rows satisfy additivity: contribs.sum(1) == margin. Implements the
+        Lundberg et al. polynomial-time EXTEND/UNWIND recursion per tree.
+        Single-class models only (matches common usage; multi-class
+        contributions are a round-2 item)."""
+        if self.num_class > 1:
+            raise NotImplementedError(
+                "pred_contribs for multi-class models is not supported yet"
+            )
+        X = _as_float32_matrix(X)
+        n, F = X.shape
+        out = np.zeros((n, F + 1), dtype=np.float64)
+        lo, hi = 0, self.num_boosted_rounds()
+        if iteration_range is not None:
+            lo, hi = iteration_range
+            hi = hi or self.num_boosted_rounds()
+        k = self.num_parallel_tree
+        trees = self.trees[lo * k : hi * k]
+        scale = 1.0 / k if k > 1 else 1.0
+        for t in trees:
+            # per-node cover-weighted expected values
+            nn = t.num_nodes
+            mean_val = np.zeros(nn)
+            cover = t.cover.astype(np.float64)
+
+            def node_mean(nid):
+                f = t.feat[nid]
+                if f < 0:
+                    mean_val[nid] = t.value[nid]
+                    return t.value[nid] * 1.0
+                l, r = t.left[nid], t.left[nid] + 1
+                node_mean(l)
+                node_mean(r)
+                c = cover[nid] if cover[nid] > 0 else 1.0
+                mean_val[nid] = (
+                    cover[l] * mean_val[l] + cover[r] * mean_val[r]
+                ) / c
+                return mean_val[nid]
+
+            import sys as _sys
+
+            old_lim = _sys.getrecursionlimit()
+            _sys.setrecursionlimit(max(old_lim, 4 * nn + 100))
+            node_mean(0)
+            for i in range(n):
+                _tree_shap(t, X[i], mean_val, cover, out[i], scale)
+            _sys.setrecursionlimit(old_lim)
+        # bias column: margin expectation = sum of tree means (added in
+        # _tree_shap) + the model's base margin
+        from xgboost_ray_amd.engine.objectives import get_objective
+
+        obj = get_objective(self.objective, self.num_class)
+        out[:, F] += float(obj.prob_to_margin(self.base_score))
+        return out
 
     def predict_leaf(self, X, iteration_range=None) -> np.ndarray:
         """Leaf index per (row, tree) - xgboost ``pred_leaf=True``."""
@@ -503,6 +639,45 @@ class Booster:
         if importance_type == "cover":
             return {k: v / counts[k] for k, v in covers.items()}
         raise ValueError(f"Unknown importance_type: {importance_type}")
+
+    def trees_to_dataframe(self, fmap=""):
+        """Flat per-node table (xgboost Booster.trees_to_dataframe parity):
+        columns Tree/Node/ID/Feature/Split/Yes/No/Missing/Gain/Cover."""
+        import pandas as pd
+
+        rows = []
+        for ti, t in enumerate(self.trees):
+            for nid in range(t.num_nodes):
+                f = int(t.feat[nid])
+                leaf = f < 0
+                if leaf:
+                    feature, split, yes, no, miss = "Leaf", None, None, None, None
+                    gain = float(t.value[nid])
+                else:
+                    if self.feature_names and f < len(self.feature_names):
+                        feature = self.feature_names[f]
+                    else:
+                        feature = f"f{f}"
+                    split = float(t.thr[nid])
+                    l = int(t.left[nid])
+                    yes, no = f"{ti}-{l}", f"{ti}-{l + 1}"
+                    miss = yes if t.default_left[nid] else no
+                    gain = float(t.gain[nid])
+                rows.append(
+                    {
+                        "Tree": ti,
+                        "Node": nid,
+                        "ID": f"{ti}-{nid}",
+                        "Feature": feature,
+                        "Split": split,
+                        "Yes": yes,
+                        "No": no,
+                        "Missing": miss,
+                        "Gain": gain,
+                        "Cover": float(t.cover[nid]),
+                    }
+                )
+        return pd.DataFrame(rows)
 
     def get_dump(self, fmap="", with_stats=False, dump_format="text"):
         out = []
