@@ -243,3 +243,36 @@ class TestDistributedCallbacks:
                 "init", "before_load", "after_load",
                 "before_train", "after_train",
             ]
+
+
+def test_predict_modes_distributed():
+    """pred_leaf / pred_contribs through the 2-actor predict path."""
+    import numpy as np
+
+    from xgboost_ray_amd import RayDMatrix, RayParams, predict, train
+
+    rng = np.random.RandomState(0)
+    X = rng.rand(2000, 5).astype(np.float32)
+    y = (X[:, 0] + X[:, 1] > 1).astype(np.float32)
+    bst = train(
+        {"objective": "binary:logistic", "max_depth": 3},
+        RayDMatrix(X, y),
+        num_boost_round=3,
+        ray_params=RayParams(num_actors=2),
+    )
+    leaves = predict(
+        bst, RayDMatrix(X), ray_params=RayParams(num_actors=2),
+        pred_leaf=True,
+    )
+    assert leaves.shape == (2000, 3)
+    for ti, t in enumerate(bst.trees):
+        assert (t.feat[leaves[:, ti].astype(int)] < 0).all()
+    contribs = predict(
+        bst, RayDMatrix(X), ray_params=RayParams(num_actors=2),
+        pred_contribs=True,
+    )
+    margin = predict(
+        bst, RayDMatrix(X), ray_params=RayParams(num_actors=2),
+        output_margin=True,
+    )
+    assert np.abs(contribs.sum(1) - margin).max() < 1e-4

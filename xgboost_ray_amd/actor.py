@@ -270,6 +270,20 @@ class _ActorWorker:
         if shard is None:
             raise RuntimeError(f"Prediction data {data_uid} not loaded")
         it_range = kwargs.get("iteration_range")
+        if kwargs.get("pred_leaf") or kwargs.get("pred_contribs"):
+            # leaf-index / SHAP modes run per shard on the CPU tree walk
+            if shard.get("streaming"):
+                parts = [
+                    bst.predict(chunk["data"], **kwargs)
+                    for chunk in shard["loader"].iter_shards(
+                        shard["rank"], shard["num_actors"]
+                    )
+                ]
+                out = np.concatenate(parts, axis=0)
+            else:
+                out = bst.predict(shard["data"], **kwargs)
+            self.dist_callbacks.after_predict(self, out)
+            return out
         if shard.get("streaming"):
             loader = shard["loader"]
             margins = []
