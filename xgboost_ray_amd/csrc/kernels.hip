@@ -729,14 +729,6 @@ __global__ void find_splits_reduce_kernel(
 #define PART_CHUNK (PART_THREADS * PART_ROWS_PER_THREAD)
 #define PART_WAVES (PART_THREADS / WAVE)
 
-__device__ inline bool go_left_pred(const uint8_t* bins, uint64_t r,
-                                    int64_t row_stride, int feat,
-                                    int split_bin, int dl) {
-  const int b = bins[r * (uint64_t)row_stride + feat];
-  if (b == 255) return dl != 0;
-  return b <= split_bin;
-}
-
 // Pass 1: evaluate the split predicate ONCE per row (a ~64B cache-line
 // gather from the binned matrix), cache it as one flag byte per row, and
 // count per-256-row-block lefts. Pass 2 then re-reads 1 B/row instead of
