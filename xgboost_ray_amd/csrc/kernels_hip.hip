@@ -538,6 +538,7 @@ __device__ inline double calc_weight_d(double G, double H, double lam,
   return denom > 0.0 ? -G / denom : 0.0;
 }
 
+#define SCAN_SEG 16
 __global__ __launch_bounds__(256) void find_splits_kf_kernel(
     const long long* __restrict__ hist,  // [K, F, B, 2]
     const long long* __restrict__ parent_g, const long long* __restrict__ parent_h,
@@ -552,19 +553,20 @@ __global__ __launch_bounds__(256) void find_splits_kf_kernel(
     long long* __restrict__ out_lg,    // [K, F]
     long long* __restrict__ out_lh,    // [K, F]
     int K, int F, int B) {
-  // One 32-lane HALF-wave per (node, feature): each hardware wave
-  // carries two independent (k, f) scans, doubling scan throughput per
-  // wave with the FP order unchanged. The 32 lanes stage the histogram
+  // One SCAN_SEG-lane segment per (node, feature): each hardware wave
+  // carries 64/SCAN_SEG independent (k, f) scans, multiplying the
+  // interleaved f64 dependency chains per wave (the scan is
+  // chain-latency bound) with the FP order unchanged. The 32 lanes stage the histogram
   // into LDS AND precompute the dequantized doubles in parallel (each
   // element rounds independently, so parallel precompute is
   // bitwise-identical to the CPU oracle's elementwise multiply). Then
   // the two missing-value directions scan on lanes 0 and 1 of the half
   // concurrently - the sequential dependent f64 chain was 52%
   // issue-stall as a single-lane loop.
-  extern __shared__ long long lds_h[];  // per half-wave: [B*2 i64][B*2 f64]
-  const int wave = threadIdx.x / 32;    // half-wave id within the block
-  const int lane = threadIdx.x % 32;
-  const int waves_per_block = blockDim.x / 32;
+  extern __shared__ long long lds_h[];  // per segment: [B*2 i64][B*2 f64]
+  const int wave = threadIdx.x / SCAN_SEG;  // scan-segment id in the block
+  const int lane = threadIdx.x % SCAN_SEG;
+  const int waves_per_block = blockDim.x / SCAN_SEG;
   int64_t kf = (int64_t)blockIdx.x * waves_per_block + wave;
   if (kf >= (int64_t)K * F) return;
   const int k = (int)(kf / F);
@@ -586,7 +588,7 @@ __global__ __launch_bounds__(256) void find_splits_kf_kernel(
   const long long* gh = hist + ((size_t)k * F + f) * B * 2;
   long long* h = lds_h + (size_t)wave * B * 4;
   double* hd = reinterpret_cast<double*>(h + B * 2);
-  for (int i = lane; i < B * 2; i += 32) {
+  for (int i = lane; i < B * 2; i += SCAN_SEG) {
     const long long v = gh[i];
     h[i] = v;
     hd[i] = (double)v * ((i & 1) ? inv_h : inv_g);
@@ -655,10 +657,10 @@ __global__ __launch_bounds__(256) void find_splits_kf_kernel(
   }
   // merge with the CPU oracle's preference: dl=0 wins only on a STRICTLY
   // greater gain (dl=1 was evaluated first there)
-  const double g1 = __shfl(best.gain, 1, 32);
-  const int b1 = __shfl(best.bin, 1, 32);
-  const long long lg1 = __shfl(best.left_g, 1, 32);
-  const long long lh1 = __shfl(best.left_h, 1, 32);
+  const double g1 = __shfl(best.gain, 1, SCAN_SEG);
+  const int b1 = __shfl(best.bin, 1, SCAN_SEG);
+  const long long lg1 = __shfl(best.left_g, 1, SCAN_SEG);
+  const long long lh1 = __shfl(best.left_h, 1, SCAN_SEG);
   if (lane != 0) return;
   if (g1 > best.gain) {
     best.gain = g1;
@@ -1397,13 +1399,13 @@ std::vector<torch::Tensor> find_splits(torch::Tensor hist, torch::Tensor parent_
   auto kf_lh = torch::empty({(int64_t)K * F}, optsl);
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   int64_t total = (int64_t)K * F;
-  // 4 half-waves of 32 lanes per 128-thread block: each hardware wave
-  // runs two independent (k, f) scans
-  const int waves_per_block = 4;
+  // 8 segments of 16 lanes per 128-thread block: each hardware wave
+  // runs four independent (k, f) scans
+  const int waves_per_block = 8;
   const size_t scan_lds = (size_t)waves_per_block * B * 4 * sizeof(long long);
   hipLaunchKernelGGL(find_splits_kf_kernel,
                      dim3((uint32_t)ceil_div(total, waves_per_block)),
-                     dim3(waves_per_block * 32), scan_lds, stream.stream(),
+                     dim3(waves_per_block * 16), scan_lds, stream.stream(),
                      reinterpret_cast<const long long*>(hist.data_ptr<int64_t>()),
                      reinterpret_cast<const long long*>(parent_g.data_ptr<int64_t>()),
                      reinterpret_cast<const long long*>(parent_h.data_ptr<int64_t>()),
