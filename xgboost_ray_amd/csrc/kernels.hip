@@ -538,7 +538,7 @@ __device__ inline double calc_weight_d(double G, double H, double lam,
   return denom > 0.0 ? -G / denom : 0.0;
 }
 
-#define SCAN_SEG 16
+#define SCAN_SEG 8
 __global__ __launch_bounds__(256) void find_splits_kf_kernel(
     const long long* __restrict__ hist,  // [K, F, B, 2]
     const long long* __restrict__ parent_g, const long long* __restrict__ parent_h,
@@ -1405,7 +1405,8 @@ std::vector<torch::Tensor> find_splits(torch::Tensor hist, torch::Tensor parent_
   const size_t scan_lds = (size_t)waves_per_block * B * 4 * sizeof(long long);
   hipLaunchKernelGGL(find_splits_kf_kernel,
                      dim3((uint32_t)ceil_div(total, waves_per_block)),
-                     dim3(waves_per_block * 16), scan_lds, stream.stream(),
+                     dim3(waves_per_block * SCAN_SEG), scan_lds,
+                     stream.stream(),
                      reinterpret_cast<const long long*>(hist.data_ptr<int64_t>()),
                      reinterpret_cast<const long long*>(parent_g.data_ptr<int64_t>()),
                      reinterpret_cast<const long long*>(parent_h.data_ptr<int64_t>()),
