@@ -804,6 +804,48 @@ __global__ void partition_count_kernel(
   }
 }
 
+
+// Per-node exclusive prefix of per-chunk left counts, on device: one WG
+// per node walks its chunk range in 256-wide blocks with an LDS
+// Hillis-Steele scan. Replaces a 100 KB D2H pull + host loop + 200 KB
+// H2D push per depth (and their stream syncs) with one kernel and a
+// K-element total pull.
+__global__ void partition_prefix_kernel(
+    const int32_t* __restrict__ block_counts,  // [total_chunks]
+    const int64_t* __restrict__ chunk_off,     // [K+1]
+    int64_t* __restrict__ left_before,         // [total_chunks]
+    int64_t* __restrict__ node_left_total,     // [K]
+    int K) {
+  const int k = blockIdx.x;
+  if (k >= K) return;
+  const int64_t c0 = chunk_off[k], c1 = chunk_off[k + 1];
+  __shared__ int64_t sc[256];
+  __shared__ int64_t sc2[256];
+  int64_t run = 0;
+  for (int64_t base = c0; base < c1; base += 256) {
+    const int nin = (int)min((int64_t)256, c1 - base);
+    const int64_t v =
+        threadIdx.x < nin ? (int64_t)block_counts[base + threadIdx.x] : 0;
+    sc[threadIdx.x] = v;
+    __syncthreads();
+    // inclusive Hillis-Steele scan over 256 entries (LDS ping-pong)
+    int64_t* src = sc;
+    int64_t* dst = sc2;
+    for (int off = 1; off < 256; off <<= 1) {
+      const int64_t x = src[threadIdx.x];
+      dst[threadIdx.x] =
+          threadIdx.x >= off ? x + src[threadIdx.x - off] : x;
+      __syncthreads();
+      int64_t* t = src; src = dst; dst = t;
+    }
+    if (threadIdx.x < nin)
+      left_before[base + threadIdx.x] = run + src[threadIdx.x] - v;
+    run += src[255];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) node_left_total[k] = run;
+}
+
 __global__ void partition_scatter_kernel(
     const uint8_t* __restrict__ flags, const int32_t* __restrict__ ridx,
     int32_t* __restrict__ ridx_out,
@@ -1497,32 +1539,21 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
   // cumsum + repeat_interleave + index_selects, ~8 tiny transfers and
   // kernels per call) stalled the training loop ~15 ms per depth -
   // measured 7x slower rounds end to end.
-  // pinned D2H pull (no pageable staging bounce), then build
-  // [left_before | node_left_total] in one pinned buffer -> ONE H2D
-  static thread_local PinnedStager bc_stager, lb_stager;
-  auto bc_cpu = bc_stager.get(total_chunks, torch::kInt32);
-  bc_cpu.copy_(block_counts);  // synchronous: host consumes it next
-  auto lbnl_cpu = lb_stager.get(total_chunks + K);
-  auto node_left_total_cpu = torch::zeros({K}, torch::kInt64);
-  {
-    auto bc = bc_cpu.accessor<int32_t, 1>();
-    auto lb = lbnl_cpu.accessor<int64_t, 1>();
-    auto nl = node_left_total_cpu.accessor<int64_t, 1>();
-    auto co = chunk_off_cpu.accessor<int64_t, 1>();
-    for (int k = 0; k < K; ++k) {
-      int64_t run = 0;
-      for (int64_t c = co[k]; c < co[k + 1]; ++c) {
-        lb[c] = run;
-        run += bc[c];
-      }
-      nl[k] = run;
-      lb[total_chunks + k] = run;
-    }
-  }
-  auto lbnl = lbnl_cpu.to(dev, /*non_blocking=*/true);
-  lb_stager.mark(stream0.stream());
-  auto left_before = lbnl.narrow(0, 0, total_chunks);
-  auto node_left_total = lbnl.narrow(0, total_chunks, K);
+  // device-side rebase: per-node exclusive prefix of block counts; the
+  // host only pulls the K node totals it needs for tree bookkeeping
+  auto left_before = torch::empty({total_chunks},
+      torch::TensorOptions().dtype(torch::kInt64).device(dev));
+  auto node_left_total = torch::empty({K},
+      torch::TensorOptions().dtype(torch::kInt64).device(dev));
+  hipLaunchKernelGGL(partition_prefix_kernel, dim3((uint32_t)K), dim3(256),
+                     0, stream0.stream(),
+                     block_counts.data_ptr<int32_t>(), chunk_off_p,
+                     left_before.data_ptr<int64_t>(),
+                     node_left_total.data_ptr<int64_t>(), K);
+  static thread_local PinnedStager nl_stager;
+  auto node_left_total_cpu_pin = nl_stager.get(K);
+  node_left_total_cpu_pin.copy_(node_left_total);  // sync: host needs it
+  auto node_left_total_cpu = node_left_total_cpu_pin.clone();
   hipLaunchKernelGGL(partition_scatter_kernel, dim3((uint32_t)total_chunks),
                      dim3(PART_THREADS), 0, stream.stream(),
                      flags.data_ptr<uint8_t>(), ridx.data_ptr<int32_t>(),
