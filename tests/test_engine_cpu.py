@@ -596,3 +596,22 @@ class TestExtraMetricsAndPredict:
                                 - cond_exp(t, x, set(S))
                             )
             assert np.abs(exact - contribs[xi, :F]).max() < 1e-5
+
+    def test_pred_contribs_multiclass(self):
+        rng = np.random.RandomState(1)
+        X = rng.rand(500, 4).astype(np.float32)
+        y = np.digitize(
+            X[:, 0] * 3 + X[:, 1] * 2,
+            np.quantile(X[:, 0] * 3 + X[:, 1] * 2, [0.33, 0.66]),
+        ).astype(np.float32)
+        dm = BinnedMatrix.build(
+            torch.from_numpy(X), label=torch.from_numpy(y), max_bin=16
+        )
+        bst = run_training(
+            {"objective": "multi:softprob", "num_class": 3, "max_depth": 3},
+            dm, 4,
+        )
+        c = bst.predict(X[:30], pred_contribs=True)
+        m = bst.predict(X[:30], output_margin=True)
+        assert c.shape == (30, 3, 5)
+        assert np.abs(c.sum(-1) - m).max() < 1e-5

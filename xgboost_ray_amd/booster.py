@@ -375,53 +375,40 @@ class Booster:
     def predict_contribs(self, X, iteration_range=None) -> np.ndarray:
         """Exact TreeSHAP feature attributions (xgboost ``pred_contribs``).
 
-        Returns [n, F+1]; the last column is the bias (expected value).
-        Rows satisfy additivity: contribs.sum(1) == margin. Implements the
-        Lundberg et al. polynomial-time EXTEND/UNWIND recursion per tree.
-        Single-class models only (matches common usage; multi-class
-        contributions are a round-2 item)."""
-        if self.num_class > 1:
-            raise NotImplementedError(
-                "pred_contribs for multi-class models is not supported yet"
-            )
+        Returns [n, F+1] (multi-class: [n, num_class, F+1]); the last
+        column is the bias (expected value). Rows satisfy additivity:
+        contribs.sum(-1) == margin. Implements the Lundberg et al.
+        polynomial-time EXTEND/UNWIND recursion per tree."""
         X = _as_float32_matrix(X)
         n, F = X.shape
-        out = np.zeros((n, F + 1), dtype=np.float64)
         lo, hi = 0, self.num_boosted_rounds()
         if iteration_range is not None:
             lo, hi = iteration_range
             hi = hi or self.num_boosted_rounds()
+        if self.num_class > 1:
+            from xgboost_ray_amd.engine.objectives import get_objective
+
+            obj = get_objective(self.objective, self.num_class)
+            base = float(obj.prob_to_margin(self.base_score))
+            outc = np.full(
+                (n, self.num_class, F + 1), 0.0, dtype=np.float64
+            )
+            k = max(1, self.num_class) * self.num_parallel_tree
+            for ti in range(lo * k, min(hi * k, len(self.trees))):
+                cls = self.tree_info[ti]
+                t = self.trees[ti]
+                self._shap_one_tree(
+                    t, X, outc[:, cls, :], 1.0 / self.num_parallel_tree
+                    if self.num_parallel_tree > 1 else 1.0,
+                )
+            outc[:, :, F] += base
+            return outc
+        out = np.zeros((n, F + 1), dtype=np.float64)
         k = self.num_parallel_tree
         trees = self.trees[lo * k : hi * k]
         scale = 1.0 / k if k > 1 else 1.0
         for t in trees:
-            # per-node cover-weighted expected values
-            nn = t.num_nodes
-            mean_val = np.zeros(nn)
-            cover = t.cover.astype(np.float64)
-
-            def node_mean(nid):
-                f = t.feat[nid]
-                if f < 0:
-                    mean_val[nid] = t.value[nid]
-                    return t.value[nid] * 1.0
-                l, r = t.left[nid], t.left[nid] + 1
-                node_mean(l)
-                node_mean(r)
-                c = cover[nid] if cover[nid] > 0 else 1.0
-                mean_val[nid] = (
-                    cover[l] * mean_val[l] + cover[r] * mean_val[r]
-                ) / c
-                return mean_val[nid]
-
-            import sys as _sys
-
-            old_lim = _sys.getrecursionlimit()
-            _sys.setrecursionlimit(max(old_lim, 4 * nn + 100))
-            node_mean(0)
-            for i in range(n):
-                _tree_shap(t, X[i], mean_val, cover, out[i], scale)
-            _sys.setrecursionlimit(old_lim)
+            self._shap_one_tree(t, X, out, scale)
         # bias column: margin expectation = sum of tree means (added in
         # _tree_shap) + the model's base margin
         from xgboost_ray_amd.engine.objectives import get_objective
@@ -429,6 +416,36 @@ class Booster:
         obj = get_objective(self.objective, self.num_class)
         out[:, F] += float(obj.prob_to_margin(self.base_score))
         return out
+
+
+    @staticmethod
+    def _shap_one_tree(t, X, out, scale):
+        """Accumulate one tree's SHAP contributions into out [n, F+1]."""
+        nn = t.num_nodes
+        mean_val = np.zeros(nn)
+        cover = t.cover.astype(np.float64)
+
+        def node_mean(nid):
+            f = t.feat[nid]
+            if f < 0:
+                mean_val[nid] = t.value[nid]
+                return
+            l, r = t.left[nid], t.left[nid] + 1
+            node_mean(l)
+            node_mean(r)
+            c = cover[nid] if cover[nid] > 0 else 1.0
+            mean_val[nid] = (
+                cover[l] * mean_val[l] + cover[r] * mean_val[r]
+            ) / c
+
+        import sys as _sys
+
+        old_lim = _sys.getrecursionlimit()
+        _sys.setrecursionlimit(max(old_lim, 4 * nn + 100))
+        node_mean(0)
+        for i in range(X.shape[0]):
+            _tree_shap(t, X[i], mean_val, cover, out[i], scale)
+        _sys.setrecursionlimit(old_lim)
 
     def predict_leaf(self, X, iteration_range=None) -> np.ndarray:
         """Leaf index per (row, tree) - xgboost ``pred_leaf=True``."""
