@@ -276,3 +276,65 @@ def test_predict_modes_distributed():
         output_margin=True,
     )
     assert np.abs(contribs.sum(1) - margin).max() < 1e-4
+
+
+def test_boost_from_prediction():
+    """base_margin continuation == continued training (reference
+    test_xgboost_api.py test_boost_from_prediction_hist semantics)."""
+    import numpy as np
+
+    from xgboost_ray_amd import RayDMatrix, RayParams, predict, train
+
+    rng = np.random.RandomState(0)
+    X = rng.rand(4000, 6).astype(np.float32)
+    y = (X[:, 0] + X[:, 1] * 2 > 1.2).astype(np.float32)
+    rp = RayParams(num_actors=2)
+    params = {"objective": "binary:logistic", "max_depth": 4, "eta": 0.3}
+
+    bst_a = train(params, RayDMatrix(X, y), num_boost_round=4, ray_params=rp)
+    margin_a = predict(
+        bst_a, RayDMatrix(X), ray_params=rp, output_margin=True
+    )
+    bst_b = train(
+        params, RayDMatrix(X, y, base_margin=margin_a),
+        num_boost_round=4, ray_params=rp,
+    )
+    margin_b = predict(
+        bst_b, RayDMatrix(X, base_margin=margin_a), ray_params=rp,
+        output_margin=True,
+    )
+
+    bst_full = train(
+        params, RayDMatrix(X, y), num_boost_round=8, ray_params=rp
+    )
+    margin_full = predict(
+        bst_full, RayDMatrix(X), ray_params=rp, output_margin=True
+    )
+    # not bitwise (per-tree quantization scales differ) but numerically
+    # the same boosting trajectory
+    assert np.allclose(margin_b, margin_full, atol=1e-3)
+
+
+def test_eval_set_weights():
+    """Weighted eval metrics flow through the distributed eval path."""
+    import numpy as np
+
+    from xgboost_ray_amd import RayDMatrix, RayParams, train
+
+    rng = np.random.RandomState(1)
+    X = rng.rand(3000, 5).astype(np.float32)
+    y = (X[:, 0] > 0.5).astype(np.float32)
+    w = np.where(y > 0, 2.0, 1.0).astype(np.float32)
+    res = {}
+    train(
+        {"objective": "binary:logistic", "max_depth": 3,
+         "eval_metric": ["logloss"]},
+        RayDMatrix(X, y),
+        num_boost_round=3,
+        ray_params=RayParams(num_actors=2),
+        evals=[(RayDMatrix(X, y, weight=w), "weighted")],
+        evals_result=res,
+    )
+    assert "weighted" in res and "logloss" in res["weighted"]
+    vals = res["weighted"]["logloss"]
+    assert len(vals) == 3 and vals[-1] < vals[0]

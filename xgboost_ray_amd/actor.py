@@ -291,7 +291,15 @@ class _ActorWorker:
                 Xc = torch.from_numpy(
                     np.ascontiguousarray(chunk["data"], np.float32)
                 ).to(self.device)
-                margins.append(bst.predict_margin_tensor(Xc, it_range))
+                mg = bst.predict_margin_tensor(Xc, it_range)
+                bm = chunk.get("base_margin")
+                if bm is not None:
+                    mg = mg + torch.from_numpy(
+                        np.ascontiguousarray(bm, np.float32)
+                    ).to(mg.device).reshape(
+                        mg.shape[0], *([1] * (mg.dim() - 1))
+                    )
+                margins.append(mg)
                 del Xc
             margin = torch.cat(margins, dim=0)
         else:
@@ -299,6 +307,15 @@ class _ActorWorker:
                 np.ascontiguousarray(shard["data"], np.float32)
             ).to(self.device)
             margin = bst.predict_margin_tensor(X, it_range)
+            # xgboost semantics: a base_margin on the prediction matrix
+            # is added to the raw margin before any transform
+            bm = shard.get("base_margin")
+            if bm is not None:
+                margin = margin + torch.from_numpy(
+                    np.ascontiguousarray(bm, np.float32)
+                ).to(margin.device).reshape(
+                    margin.shape[0], *([1] * (margin.dim() - 1))
+                )
         if kwargs.get("output_margin"):
             pred = margin.cpu().numpy()
         else:
