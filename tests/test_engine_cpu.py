@@ -615,3 +615,31 @@ class TestExtraMetricsAndPredict:
         m = bst.predict(X[:30], output_margin=True)
         assert c.shape == (30, 3, 5)
         assert np.abs(c.sum(-1) - m).max() < 1e-5
+
+    def test_pred_interactions(self):
+        rng = np.random.RandomState(0)
+        X = rng.rand(1000, 4).astype(np.float32)
+        y = (
+            4 * (X[:, 0] > 0.5) * (X[:, 1] > 0.5) + X[:, 2]
+        ).astype(np.float32)
+        dm = BinnedMatrix.build(
+            torch.from_numpy(X), label=torch.from_numpy(y), max_bin=32
+        )
+        bst = run_training(
+            {"objective": "reg:squarederror", "max_depth": 4}, dm, 5
+        )
+        rows = X[:10]
+        M = bst.predict(rows, pred_interactions=True)
+        c = bst.predict(rows, pred_contribs=True)
+        m = bst.predict(rows, output_margin=True)
+        assert M.shape == (10, 5, 5)
+        assert np.abs(M.sum(2) - c).max() < 1e-6       # rows -> SHAP
+        assert np.abs(M.sum((1, 2)) - m).max() < 1e-5  # total -> margin
+        assert np.abs(
+            M[:, :4, :4] - np.transpose(M[:, :4, :4], (0, 2, 1))
+        ).max() == 0.0                                 # symmetric
+        off = np.abs(M[:, :4, :4].copy())
+        for i in range(4):
+            off[:, i, i] = 0
+        # interaction mass concentrates on the true interacting pair
+        assert off[:, 0, 1].mean() > 10 * (off[:, 2, 3].mean() + 1e-12)

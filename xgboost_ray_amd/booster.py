@@ -102,11 +102,40 @@ def _as_float32_vec(v):
 
 
 
-def _tree_shap(t, x, mean_val, cover, out, scale):
+def _fill_node_means(t, mean_val, cover):
+    """Cover-weighted expected leaf value per node (iterative postorder)."""
+    order = []
+    stack = [0]
+    while stack:
+        nid = stack.pop()
+        order.append(nid)
+        if t.feat[nid] >= 0:
+            stack.append(int(t.left[nid]))
+            stack.append(int(t.left[nid]) + 1)
+    for nid in reversed(order):
+        f = t.feat[nid]
+        if f < 0:
+            mean_val[nid] = t.value[nid]
+        else:
+            l, r = int(t.left[nid]), int(t.left[nid]) + 1
+            c = cover[nid] if cover[nid] > 0 else 1.0
+            mean_val[nid] = (
+                cover[l] * mean_val[l] + cover[r] * mean_val[r]
+            ) / c
+
+
+def _tree_shap(t, x, mean_val, cover, out, scale, condition=0,
+               condition_feature=-1):
     """One row's SHAP contributions for one tree (Lundberg EXTEND/UNWIND).
 
     ``m`` is the path of unique features: fraction of zero paths (z),
-    fraction of one paths (o), and feature index (d)."""
+    fraction of one paths (o), and feature index (d). ``condition`` +1
+    fixes ``condition_feature`` present (its splits always follow x),
+    -1 fixes it absent (cover-weighted mixture over its branches; the
+    mixture weight travels as ``cf``) - the two passes that build SHAP
+    interaction values. Conditioned splits never extend the path, so the
+    conditioned feature receives no credit.
+    """
 
     def extend(m, pz, po, pi):
         m = m + [[pz, po, pi, 1.0 if len(m) == 0 else 0.0]]
@@ -131,11 +160,16 @@ def _tree_shap(t, x, mean_val, cover, out, scale):
                 total += m[j][3] * (ln + 1) / (pz * (ln - j))
         return total
 
-    def recurse(nid, m, pz, po, pi):
-        m = extend(list(map(list, m)), pz, po, pi)
+    SKIP = -2  # sentinel: enter child without extending the path
+
+    def recurse(nid, m, pz, po, pi, cf):
+        if cf == 0.0:
+            return
+        if pi != SKIP:
+            m = extend(list(map(list, m)), pz, po, pi)
         f = t.feat[nid]
         if f < 0:
-            v = t.value[nid] * scale
+            v = t.value[nid] * scale * cf
             for i in range(1, len(m)):
                 w = unwind_sum(m, i)
                 out[m[i][2]] += w * (m[i][1] - m[i][0]) * v
@@ -147,14 +181,21 @@ def _tree_shap(t, x, mean_val, cover, out, scale):
         else:
             hot = l if fv < t.thr[nid] else r
         cold = r if hot == l else l
+        c = cover[nid] if cover[nid] > 0 else 1.0
+        if condition != 0 and f == condition_feature:
+            if condition > 0:
+                recurse(hot, m, 0.0, 0.0, SKIP, cf)
+            else:
+                recurse(hot, m, 0.0, 0.0, SKIP, cf * cover[hot] / c)
+                recurse(cold, m, 0.0, 0.0, SKIP, cf * cover[cold] / c)
+            return
         iz, io = 1.0, 1.0
-        # if this feature already on the path, unwind it first
         path_idx = next(
             (i for i in range(1, len(m)) if m[i][2] == f), None
         )
         if path_idx is not None:
+            m = list(map(list, m))
             iz, io = m[path_idx][0], m[path_idx][1]
-            # unwind
             ln = len(m) - 1
             po_, pz_ = m[path_idx][1], m[path_idx][0]
             nxt = m[ln][3]
@@ -168,12 +209,12 @@ def _tree_shap(t, x, mean_val, cover, out, scale):
             for j in range(path_idx, ln):
                 m[j][0], m[j][1], m[j][2] = m[j + 1][0], m[j + 1][1], m[j + 1][2]
             m = m[:-1]
-        c = cover[nid] if cover[nid] > 0 else 1.0
-        recurse(hot, m, iz * cover[hot] / c, io, f)
-        recurse(cold, m, iz * cover[cold] / c, 0.0, f)
+        recurse(hot, m, iz * cover[hot] / c, io, f, cf)
+        recurse(cold, m, iz * cover[cold] / c, 0.0, f, cf)
 
-    out[len(x)] += mean_val[0] * scale
-    recurse(0, [], 1.0, 1.0, -1)
+    if condition == 0:
+        out[len(x)] += mean_val[0] * scale
+    recurse(0, [], 1.0, 1.0, -1, 1.0)
 
 
 class Booster:
@@ -346,6 +387,7 @@ class Booster:
         validate_features: bool = True,
         pred_leaf: bool = False,
         pred_contribs: bool = False,
+        pred_interactions: bool = False,
         **kwargs,
     ) -> np.ndarray:
         from xgboost_ray_amd.engine.objectives import get_objective
@@ -360,6 +402,8 @@ class Booster:
             return self.predict_leaf(X, iteration_range)
         if pred_contribs:
             return self.predict_contribs(X, iteration_range)
+        if pred_interactions:
+            return self.predict_interactions(X, iteration_range)
         Xt = torch.from_numpy(np.ascontiguousarray(X, dtype=np.float32))
         margin = self.predict_margin_tensor(Xt, iteration_range)
         if bm is not None:
@@ -421,31 +465,60 @@ class Booster:
     @staticmethod
     def _shap_one_tree(t, X, out, scale):
         """Accumulate one tree's SHAP contributions into out [n, F+1]."""
-        nn = t.num_nodes
-        mean_val = np.zeros(nn)
+        mean_val = np.zeros(t.num_nodes)
         cover = t.cover.astype(np.float64)
-
-        def node_mean(nid):
-            f = t.feat[nid]
-            if f < 0:
-                mean_val[nid] = t.value[nid]
-                return
-            l, r = t.left[nid], t.left[nid] + 1
-            node_mean(l)
-            node_mean(r)
-            c = cover[nid] if cover[nid] > 0 else 1.0
-            mean_val[nid] = (
-                cover[l] * mean_val[l] + cover[r] * mean_val[r]
-            ) / c
-
-        import sys as _sys
-
-        old_lim = _sys.getrecursionlimit()
-        _sys.setrecursionlimit(max(old_lim, 4 * nn + 100))
-        node_mean(0)
+        _fill_node_means(t, mean_val, cover)
         for i in range(X.shape[0]):
             _tree_shap(t, X[i], mean_val, cover, out[i], scale)
-        _sys.setrecursionlimit(old_lim)
+
+    def predict_interactions(self, X, iteration_range=None) -> np.ndarray:
+        """SHAP interaction values (xgboost ``pred_interactions``).
+
+        [n, F+1, F+1]: off-diagonal [i, j] is the interaction between
+        features i and j ((shap_j | i present) - (shap_j | i absent))/2,
+        the diagonal holds main effects so each row i sums to the
+        feature's SHAP value, and [F, F] is the bias - the whole matrix
+        sums to the margin. Single-class models; O(F) TreeSHAP passes
+        per row, so intended for explanation-sized batches."""
+        if self.num_class > 1:
+            raise NotImplementedError(
+                "pred_interactions for multi-class models is not supported"
+            )
+        X = _as_float32_matrix(X)
+        n, F = X.shape
+        contribs = self.predict_contribs(X, iteration_range)
+        lo, hi = 0, self.num_boosted_rounds()
+        if iteration_range is not None:
+            lo, hi = iteration_range
+            hi = hi or self.num_boosted_rounds()
+        k = self.num_parallel_tree
+        trees = self.trees[lo * k : hi * k]
+        scale = 1.0 / k if k > 1 else 1.0
+        M = np.zeros((n, F + 1, F + 1), dtype=np.float64)
+        on = np.zeros((n, F + 1), dtype=np.float64)
+        off = np.zeros((n, F + 1), dtype=np.float64)
+        for i in range(F):
+            on[:] = 0.0
+            off[:] = 0.0
+            for t in trees:
+                nn = t.num_nodes
+                mean_val = np.zeros(nn)
+                cover = t.cover.astype(np.float64)
+                _fill_node_means(t, mean_val, cover)
+                for r in range(n):
+                    _tree_shap(t, X[r], mean_val, cover, on[r], scale,
+                               condition=1, condition_feature=i)
+                    _tree_shap(t, X[r], mean_val, cover, off[r], scale,
+                               condition=-1, condition_feature=i)
+            M[:, i, :F] = (on[:, :F] - off[:, :F]) / 2.0
+            M[:, i, i] = 0.0
+        # symmetrize then set diagonals so each row sums to the SHAP value
+        M[:, :F, :F] = (M[:, :F, :F] + np.transpose(M[:, :F, :F],
+                                                    (0, 2, 1))) / 2.0
+        for i in range(F):
+            M[:, i, i] = contribs[:, i] - M[:, i, :F].sum(axis=1)
+        M[:, F, F] = contribs[:, F]
+        return M
 
     def predict_leaf(self, X, iteration_range=None) -> np.ndarray:
         """Leaf index per (row, tree) - xgboost ``pred_leaf=True``."""
