@@ -338,3 +338,26 @@ def test_eval_set_weights():
     assert "weighted" in res and "logloss" in res["weighted"]
     vals = res["weighted"]["logloss"]
     assert len(vals) == 3 and vals[-1] < vals[0]
+
+
+def test_gblinear_distributed():
+    """booster=gblinear through 2 actors == single-actor training."""
+    import numpy as np
+
+    from xgboost_ray_amd import RayDMatrix, RayParams, predict, train
+
+    rng = np.random.RandomState(0)
+    X = rng.randn(6000, 5).astype(np.float32)
+    w_true = np.array([1.0, -2.0, 0.5, 0.0, 3.0], np.float32)
+    y = (X @ w_true + 0.3).astype(np.float32)
+    params = {"objective": "reg:squarederror", "booster": "gblinear",
+              "eta": 0.5, "lambda": 0.0, "base_score": 0.0}
+    bst2 = train(params, RayDMatrix(X, y), num_boost_round=30,
+                 ray_params=RayParams(num_actors=2))
+    w2 = bst2.linear_weights[:, 0]
+    assert np.abs(w2[:5] - w_true).max() < 0.05
+    bst1 = train(params, RayDMatrix(X, y), num_boost_round=30,
+                 ray_params=RayParams(num_actors=1))
+    assert np.allclose(bst1.linear_weights, bst2.linear_weights, atol=1e-4)
+    pred = predict(bst2, RayDMatrix(X), ray_params=RayParams(num_actors=2))
+    assert np.abs(pred - (X @ w2[:5] + w2[5])).max() < 1e-3

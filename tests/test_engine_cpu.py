@@ -643,3 +643,83 @@ class TestExtraMetricsAndPredict:
             off[:, i, i] = 0
         # interaction mass concentrates on the true interacting pair
         assert off[:, 0, 1].mean() > 10 * (off[:, 2, 3].mean() + 1e-12)
+
+
+class TestGBLinear:
+    def _linear_data(self, n=20000, seed=0):
+        rng = np.random.RandomState(seed)
+        X = rng.randn(n, 6).astype(np.float32)
+        true_w = np.array([2.0, -1.0, 0.5, 0, 0, 3.0], np.float32)
+        y = (X @ true_w + 0.7 + 0.05 * rng.randn(n)).astype(np.float32)
+        return X, y, true_w
+
+    def test_gblinear_recovers_weights(self):
+        X, y, true_w = self._linear_data()
+        dm = BinnedMatrix.build(
+            torch.from_numpy(X), label=torch.from_numpy(y),
+            max_bin=64, keep_raw=True,
+        )
+        bst = run_training(
+            {"objective": "reg:squarederror", "booster": "gblinear",
+             "eta": 0.5, "lambda": 0.0, "base_score": 0.0},
+            dm, 40,
+        )
+        w = bst.linear_weights[:, 0]
+        assert np.abs(w[:6] - true_w).max() < 0.05
+        pred = bst.predict(X[:200], output_margin=True)
+        assert np.abs(pred - (X[:200] @ w[:6] + w[6])).max() < 1e-4
+
+    def test_gblinear_l1_sparsifies(self):
+        X, y, _ = self._linear_data(seed=1)
+        dm = BinnedMatrix.build(
+            torch.from_numpy(X), label=torch.from_numpy(y),
+            max_bin=64, keep_raw=True,
+        )
+        bst = run_training(
+            {"objective": "reg:squarederror", "booster": "gblinear",
+             "eta": 0.5, "lambda": 0.0, "alpha": 500.0,
+             "base_score": 0.0},
+            dm, 40,
+        )
+        w = bst.linear_weights[:6, 0]
+        # the two zero-coefficient features must be driven to (near) zero
+        assert np.abs(w[3]) < 1e-2 and np.abs(w[4]) < 1e-2
+        # strong features survive the penalty
+        assert abs(w[5]) > 1.0
+
+    def test_gblinear_json_roundtrip_and_resume(self):
+        import json
+
+        from xgboost_ray_amd.booster import Booster
+
+        X, y, _ = self._linear_data(seed=2)
+        dm = BinnedMatrix.build(
+            torch.from_numpy(X), label=torch.from_numpy(y),
+            max_bin=64, keep_raw=True,
+        )
+        params = {"objective": "reg:squarederror", "booster": "gblinear",
+                  "eta": 0.5, "base_score": 0.0}
+        b1 = run_training(params, dm, 10)
+        b2 = Booster()
+        b2._from_json_dict(json.loads(b1.save_raw().decode()))
+        assert np.allclose(b2.linear_weights, b1.linear_weights)
+        assert b2.num_boosted_rounds() == 10
+        # resume: 10 + 10 == 20 straight
+        b3 = run_training(params, dm, 10, xgb_model=b2)
+        b20 = run_training(params, dm, 20)
+        # margins are recomputed in fp32 from the restored weights, so
+        # resume matches continuous training to fp32 accumulation noise
+        assert np.allclose(
+            b3.linear_weights, b20.linear_weights, atol=1e-5
+        )
+
+    def test_gblinear_requires_raw(self):
+        X, y, _ = self._linear_data(n=2000)
+        dm = BinnedMatrix.build(
+            torch.from_numpy(X), label=torch.from_numpy(y), max_bin=64
+        )
+        with pytest.raises(ValueError, match="keep_raw"):
+            run_training(
+                {"objective": "reg:squarederror", "booster": "gblinear"},
+                dm, 2,
+            )

@@ -182,3 +182,22 @@ class TestSklearnEcosystem:
         scores = cross_val_score(reg, X, y, cv=2, scoring="r2")
         assert len(scores) == 2
         assert scores.mean() > 0.5
+
+
+def test_sklearn_gblinear():
+    """booster='gblinear' through the sklearn estimator (reference
+    test_sklearn_api_gblinear)."""
+    import numpy as np
+
+    from xgboost_ray_amd import RayParams, RayXGBRegressor
+
+    rng = np.random.RandomState(0)
+    X = rng.randn(4000, 4).astype(np.float32)
+    y = (X @ np.array([1.0, -1.0, 2.0, 0.0], np.float32)).astype(np.float32)
+    reg = RayXGBRegressor(
+        booster="gblinear", n_estimators=30, learning_rate=0.5,
+        reg_lambda=0.0, base_score=0.0,
+    )
+    reg.fit(X, y, ray_params=RayParams(num_actors=2))
+    pred = reg.predict(X[:500], ray_params=RayParams(num_actors=2))
+    assert np.abs(pred - y[:500]).max() < 0.1

@@ -148,6 +148,11 @@ class _ActorWorker:
         )
         try:
             if shard.get("streaming"):
+                if params.get("booster") == "gblinear":
+                    raise ValueError(
+                        "booster=gblinear is not supported with streaming "
+                        "(RayDeviceQuantileDMatrix) training data"
+                    )
                 loader = shard["loader"]
 
                 def chunk_fn(loader=loader, shard=shard):
@@ -175,6 +180,7 @@ class _ActorWorker:
                     max_bin=int(params.get("max_bin", 256)),
                     collective=coll,
                     seed=int(params.get("seed", 0) or 0),
+                    keep_raw=params.get("booster") == "gblinear",
                 )
                 del t_x
                 if shard.get("feature_weights") is not None:

@@ -234,6 +234,9 @@ class Booster:
         self.feature_names: Optional[List[str]] = self.params.pop(
             "feature_names", None
         )
+        # gblinear state: [F+1, num_group] float64, bias row last
+        self.linear_weights = None
+        self.linear_rounds = 0
         self._flat_cache = None
 
     # -- core info ---------------------------------------------------------
@@ -261,6 +264,8 @@ class Booster:
         return max(1, int(self.params.get("num_parallel_tree", 1) or 1))
 
     def num_boosted_rounds(self) -> int:
+        if self.params.get("booster") == "gblinear":
+            return int(self.linear_rounds)
         k = max(1, self.num_class) * self.num_parallel_tree
         return len(self.trees) // k
 
@@ -337,6 +342,14 @@ class Booster:
         n = X.shape[0]
         k = max(1, self.num_class)
         base = obj.prob_to_margin(self.base_score)
+        if self.params.get("booster") == "gblinear":
+            # linear model: margin = X @ w + bias (+ base); iteration_range
+            # has no meaning for a single weight vector
+            w = torch.from_numpy(
+                np.asarray(self.linear_weights, dtype=np.float32)
+            ).to(X.device)
+            m = X @ w[:-1] + w[-1] + base
+            return m[:, 0] if k == 1 else m
         flat = self._flat_trees(X.device)
 
         lo, hi = 0, self.num_boosted_rounds()
@@ -631,28 +644,43 @@ class Booster:
             )
         num_rounds = self.num_boosted_rounds()
         k = max(1, self.num_class) * self.num_parallel_tree
+        if self.params.get("booster") == "gblinear":
+            # xgboost gblinear schema: flat weights, bias block last
+            gb = {
+                "model": {
+                    "weights": np.asarray(self.linear_weights)
+                    .astype(float).reshape(-1).tolist(),
+                },
+                "name": "gblinear",
+            }
+        else:
+            gb = {
+                "model": {
+                    "gbtree_model_param": {
+                        "num_trees": str(len(self.trees)),
+                        "num_parallel_tree": str(self.num_parallel_tree),
+                    },
+                    "iteration_indptr": [i * k for i in range(num_rounds + 1)],
+                    "tree_info": list(self.tree_info),
+                    "trees": trees_json,
+                },
+                "name": "gbtree",
+            }
         doc = {
             "learner": {
                 "attributes": dict(self.attributes_),
                 "feature_names": self.feature_names or [],
                 "feature_types": [],
-                "gradient_booster": {
-                    "model": {
-                        "gbtree_model_param": {
-                            "num_trees": str(len(self.trees)),
-                            "num_parallel_tree": str(self.num_parallel_tree),
-                        },
-                        "iteration_indptr": [i * k for i in range(num_rounds + 1)],
-                        "tree_info": list(self.tree_info),
-                        "trees": trees_json,
-                    },
-                    "name": "gbtree",
-                },
+                "gradient_booster": gb,
                 "learner_model_param": learner_param,
                 "objective": _objective_json(self.objective),
             },
             "version": [2, 1, 0],
         }
+        if self.params.get("booster") == "gblinear":
+            doc["learner"]["attributes"]["gblinear_rounds"] = str(
+                self.linear_rounds
+            )
         return doc
 
     def _from_json_dict(self, doc: Dict):
@@ -667,6 +695,19 @@ class Booster:
         self.feature_names = learner.get("feature_names") or None
         self.attributes_ = dict(learner.get("attributes", {}))
         model = learner["gradient_booster"]["model"]
+        if learner["gradient_booster"].get("name") == "gblinear":
+            self.params["booster"] = "gblinear"
+            F = int(lp.get("num_feature", 0))
+            k = max(1, int(lp.get("num_class", 0) or 0) or 1)
+            w = np.asarray(model["weights"], dtype=np.float64)
+            self.linear_weights = w.reshape(F + 1, k)
+            self.linear_rounds = int(
+                self.attributes_.pop("gblinear_rounds", 0) or 0
+            )
+            self.trees = []
+            self.tree_info = []
+            self._flat_cache = None
+            return
         self.tree_info = [int(x) for x in model.get("tree_info", [])]
         self.trees = []
         for tj in model["trees"]:

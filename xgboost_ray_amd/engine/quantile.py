@@ -244,6 +244,7 @@ class BinnedMatrix:
         self.cuts = cuts
         self.bins = ops.bin_matrix(X, cuts.cuts_flat, cuts.cut_ptr)
         self.bins_t = _make_bins_t(self.bins)
+        self.raw_X = None  # kept only for the gblinear booster
         self.n_rows, self.n_features = X.shape
         self.label = label
         self.weight = weight
@@ -261,11 +262,15 @@ class BinnedMatrix:
         max_bin: int = 256,
         collective: Optional[Collective] = None,
         cuts: Optional[HistogramCuts] = None,
+        keep_raw: bool = False,
         seed: int = 0,
     ) -> "BinnedMatrix":
         if cuts is None:
             cuts = build_cuts(X, max_bin, collective, seed)
-        return cls(X, label, weight, base_margin, qid, cuts)
+        obj = cls(X, label, weight, base_margin, qid, cuts)
+        if keep_raw:
+            obj.raw_X = X.contiguous()
+        return obj
 
     @classmethod
     def build_streaming(
@@ -343,4 +348,5 @@ class BinnedMatrix:
             else:
                 setattr(obj, key, None)
         obj.bins_t = _make_bins_t(obj.bins)
+        obj.raw_X = None
         return obj

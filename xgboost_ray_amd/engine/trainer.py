@@ -57,6 +57,7 @@ class TrainParams:
     monotone_constraints: object = None
     interaction_constraints: object = None
     tree_method: str = "hist"
+    booster: str = "gbtree"
     verbosity: int = 1
     nthread: int = 0
     disable_default_eval_metric: bool = False
@@ -212,6 +213,23 @@ class BoostingEngine:
         if ic:
             self.interaction_sets = [frozenset(int(f) for f in g)
                                      for g in ic]
+        self.lin_w = None
+        if self.p.booster == "gblinear":
+            raw = getattr(dtrain, "raw_X", None)
+            if raw is None:
+                raise ValueError(
+                    "booster=gblinear needs the raw feature matrix - "
+                    "build the BinnedMatrix with keep_raw=True (streaming "
+                    "matrices are not supported for gblinear)"
+                )
+            self.lin_w = torch.zeros(
+                (dtrain.n_features + 1, self.n_class),
+                dtype=torch.float64, device=self.device,
+            )
+        elif self.p.booster not in ("gbtree", ""):
+            raise ValueError(
+                f"Unsupported booster: {self.p.booster!r} (gbtree/gblinear)"
+            )
         self.iteration = 0
         self.booster = Booster(
             params={
@@ -241,6 +259,19 @@ class BoostingEngine:
         )
         if self.dtrain.base_margin is not None:
             self.margin = self.margin + self.dtrain.base_margin.reshape(shape)
+        if getattr(model, "linear_weights", None) is not None:
+            # gblinear resume: restore weights, recompute margins exactly
+            self.lin_w = torch.from_numpy(
+                np.asarray(model.linear_weights, np.float64)
+            ).to(self.device).reshape(self.dtrain.n_features + 1,
+                                      self.n_class)
+            X = self.dtrain.raw_X
+            w32 = self.lin_w.float()
+            m = X @ w32[:-1] + w32[-1]
+            self.margin = (
+                self.margin + (m[:, 0] if self.n_class == 1 else m)
+            )
+            return
         self._replay_trees_binned(model)
 
     def _replay_trees_binned(self, model: Booster):
@@ -293,6 +324,10 @@ class BoostingEngine:
         gpair = self.obj.gradients(
             self.margin, label, self.dtrain.weight, self.dtrain.qid
         )
+        if self.p.booster == "gblinear":
+            self._update_linear(gpair)
+            self.iteration += 1
+            return []
         trees, classes = [], []
         for cls in range(self.n_class):
             gp = gpair if self.n_class == 1 else gpair[:, cls, :]
@@ -308,6 +343,65 @@ class BoostingEngine:
         self.booster.append_round(trees, classes)
         self.iteration += 1
         return trees
+
+
+    def _update_linear(self, gpair):
+        """One gblinear round: preconditioned full-batch coordinate step.
+
+        Deterministic replacement for xgboost's gblinear updaters
+        (shotgun is non-deterministic; this is a damped Jacobi update
+        with the same per-coordinate Newton preconditioner, L2 on
+        weights and L1 soft-thresholding, eta-damped). Distributed: one
+        fp64 AllReduce of [X^T g | (X*X)^T h | sum_g | sum_h] per class.
+        """
+        X = self.dtrain.raw_X
+        n, F = X.shape
+        lam = float(self.p.reg_lambda)
+        alpha = float(self.p.reg_alpha)
+        eta = float(self.p.eta)
+        deltas = []
+        for cls in range(self.n_class):
+            gp = gpair if self.n_class == 1 else gpair[:, cls, :]
+            # chunked fp64 accumulation: fixed chunk order -> identical
+            # sums on every rank layout of the same data
+            sg = torch.zeros(F, dtype=torch.float64, device=self.device)
+            sh = torch.zeros(F, dtype=torch.float64, device=self.device)
+            sb = torch.zeros(2, dtype=torch.float64, device=self.device)
+            CH = 1 << 20
+            for s0 in range(0, n, CH):
+                Xc = X[s0 : s0 + CH].double()
+                gc = gp[s0 : s0 + CH, 0].double()
+                hc = gp[s0 : s0 + CH, 1].double()
+                sg += Xc.t() @ gc
+                sh += (Xc * Xc).t() @ hc
+                sb[0] += gc.sum()
+                sb[1] += hc.sum()
+            if self.coll.is_distributed:
+                packed = torch.cat([sg, sh, sb])
+                self.coll.allreduce_(packed)
+                sg, sh, sb = packed[:F], packed[F : 2 * F], packed[2 * F :]
+            w = self.lin_w[:F, cls]
+            denom = sh + lam
+            full = w - (sg + lam * w) / denom
+            if alpha > 0.0:
+                thr = alpha / denom
+                full = torch.sign(full) * torch.clamp(
+                    full.abs() - thr, min=0.0
+                )
+            dw = eta * (full - w)
+            db = eta * (-sb[0] / (sb[1] + lam))
+            self.lin_w[:F, cls] += dw
+            self.lin_w[F, cls] += db
+            dwf = dw.float()
+            mv = self.margin if self.n_class == 1 else self.margin[:, cls]
+            for s0 in range(0, n, CH):
+                mv[s0 : s0 + CH] += X[s0 : s0 + CH] @ dwf
+            mv += float(db)
+            deltas.append((dwf, float(db)))
+        self._lin_deltas = deltas
+        self.booster.linear_weights = self.lin_w.cpu().numpy()
+        self.booster.linear_rounds = self.iteration + 1
+        self.booster.params["booster"] = "gblinear"
 
     def _quantize(self, gpair: torch.Tensor) -> Tuple[torch.Tensor, float, float]:
         mx = torch.stack(
@@ -866,6 +960,16 @@ class BoostingEngine:
 
     def update_eval_margins(self, evals: Sequence[EvalPack], trees, classes):
         """Incrementally add the new round's trees to eval-set margins."""
+        if self.p.booster == "gblinear":
+            for ev in evals:
+                if ev.X is None:
+                    continue
+                for cls, (dwf, db) in enumerate(self._lin_deltas):
+                    mv = (ev.margin if self.n_class == 1
+                          else ev.margin[:, cls])
+                    mv += ev.X @ dwf.to(ev.X.device)
+                    mv += db
+            return
         for ev in evals:
             if ev.X is None:
                 continue
@@ -1156,7 +1260,7 @@ def run_training(
         if cb.before_iteration(engine.booster, it, log):
             break
         trees = engine.update()
-        classes = engine.booster.tree_info[-len(trees):]
+        classes = engine.booster.tree_info[-len(trees):] if trees else []
         engine.update_eval_margins(evals, trees, classes)
         if evals:
             results = engine.eval_sets(evals, feval)
