@@ -170,3 +170,27 @@ def test_lossguide_and_sampling_gpu():
     for ta, tb in zip(bst.trees, bst2.trees):
         assert np.array_equal(ta.thr, tb.thr)
         assert np.array_equal(ta.value, tb.value)
+
+
+@pytest.mark.gpu
+def test_gblinear_gpu():
+    """gblinear on device tensors (chunked fp64 GEMV path)."""
+    import torch
+    from xgboost_ray_amd.engine.quantile import BinnedMatrix
+    from xgboost_ray_amd.engine.trainer import run_training
+
+    rng = np.random.RandomState(0)
+    X = rng.randn(200000, 6).astype(np.float32)
+    w_true = np.array([2.0, -1.0, 0.5, 0, 0, 3.0], np.float32)
+    y = (X @ w_true + 0.7).astype(np.float32)
+    dm = BinnedMatrix.build(
+        torch.from_numpy(X).cuda(), label=torch.from_numpy(y).cuda(),
+        max_bin=64, keep_raw=True,
+    )
+    bst = run_training(
+        {"objective": "reg:squarederror", "booster": "gblinear",
+         "eta": 0.5, "lambda": 0.0, "base_score": 0.0},
+        dm, 40,
+    )
+    w = bst.linear_weights[:, 0]
+    assert np.abs(w[:6] - w_true).max() < 0.05
