@@ -361,3 +361,24 @@ def test_gblinear_distributed():
     assert np.allclose(bst1.linear_weights, bst2.linear_weights, atol=1e-4)
     pred = predict(bst2, RayDMatrix(X), ray_params=RayParams(num_actors=2))
     assert np.abs(pred - (X @ w2[:5] + w2[5])).max() < 1e-3
+
+
+def test_dart_distributed_equals_single():
+    """booster=dart: dropout RNG is (seed, iteration)-keyed, so the
+    distributed model equals the single-actor model bitwise."""
+    import numpy as np
+
+    from xgboost_ray_amd import RayDMatrix, RayParams, train
+
+    rng = np.random.RandomState(0)
+    X = rng.rand(6000, 5).astype(np.float32)
+    y = (X[:, 0] + X[:, 1] + 0.1 * rng.randn(6000)).astype(np.float32)
+    params = {"objective": "reg:squarederror", "booster": "dart",
+              "rate_drop": 0.3, "max_depth": 4, "eta": 0.3, "seed": 7}
+    b1 = train(params, RayDMatrix(X, y), num_boost_round=10,
+               ray_params=RayParams(num_actors=1))
+    b2 = train(params, RayDMatrix(X, y), num_boost_round=10,
+               ray_params=RayParams(num_actors=2))
+    for ta, tb in zip(b1.trees, b2.trees):
+        assert np.array_equal(ta.thr, tb.thr)
+        assert np.array_equal(ta.value, tb.value)

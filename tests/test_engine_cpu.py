@@ -723,3 +723,57 @@ class TestGBLinear:
                 {"objective": "reg:squarederror", "booster": "gblinear"},
                 dm, 2,
             )
+
+
+class TestDart:
+    def _dm(self, seed=0):
+        rng = np.random.RandomState(seed)
+        X = rng.rand(8000, 6).astype(np.float32)
+        y = (np.sin(X[:, 0] * 5) + X[:, 1]
+             + 0.1 * rng.randn(8000)).astype(np.float32)
+        return X, y, BinnedMatrix.build(
+            torch.from_numpy(X), label=torch.from_numpy(y), max_bin=64
+        )
+
+    def test_dart_basics(self):
+        X, y, dm = self._dm()
+        base = {"objective": "reg:squarederror", "max_depth": 4,
+                "eta": 0.3, "seed": 3}
+        b_gb = run_training(dict(base), dm, 15)
+        b_d = run_training(dict(base, booster="dart", rate_drop=0.3),
+                           dm, 15)
+        b_d2 = run_training(dict(base, booster="dart", rate_drop=0.3),
+                            dm, 15)
+        b_d0 = run_training(dict(base, booster="dart", rate_drop=0.0),
+                            dm, 15)
+        p = X[:300]
+        assert not np.allclose(b_gb.predict(p), b_d.predict(p))
+        assert np.array_equal(b_d.predict(p), b_d2.predict(p))
+        assert np.allclose(b_d0.predict(p), b_gb.predict(p))
+        rmse = np.sqrt(np.mean((b_d.predict(X) - y) ** 2))
+        assert rmse < 0.5
+
+    def test_dart_margin_state_consistent(self):
+        from xgboost_ray_amd.engine.trainer import BoostingEngine
+
+        X, y, dm = self._dm(seed=1)
+        eng = BoostingEngine(
+            {"objective": "reg:squarederror", "booster": "dart",
+             "rate_drop": 0.4, "max_depth": 3, "eta": 0.3, "seed": 1}, dm)
+        for _ in range(12):
+            eng.update()
+        pred = eng.booster.predict(X, output_margin=True)
+        assert np.abs(eng.margin.numpy() - pred).max() < 1e-4
+
+    def test_dart_save_load(self, tmp_path):
+        from xgboost_ray_amd.booster import Booster
+
+        X, y, dm = self._dm(seed=2)
+        b = run_training(
+            {"objective": "reg:squarederror", "booster": "dart",
+             "rate_drop": 0.3, "max_depth": 3, "eta": 0.3}, dm, 8)
+        f = str(tmp_path / "dart.json")
+        b.save_model(f)
+        b2 = Booster()
+        b2.load_model(f)
+        assert np.allclose(b2.predict(X[:200]), b.predict(X[:200]))

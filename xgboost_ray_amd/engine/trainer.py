@@ -58,6 +58,11 @@ class TrainParams:
     interaction_constraints: object = None
     tree_method: str = "hist"
     booster: str = "gbtree"
+    # DART (booster="dart") dropout parameters (xgboost semantics)
+    rate_drop: float = 0.0
+    skip_drop: float = 0.0
+    one_drop: bool = False
+    normalize_type: str = "tree"
     verbosity: int = 1
     nthread: int = 0
     disable_default_eval_metric: bool = False
@@ -226,9 +231,10 @@ class BoostingEngine:
                 (dtrain.n_features + 1, self.n_class),
                 dtype=torch.float64, device=self.device,
             )
-        elif self.p.booster not in ("gbtree", ""):
+        elif self.p.booster not in ("gbtree", "dart", ""):
             raise ValueError(
-                f"Unsupported booster: {self.p.booster!r} (gbtree/gblinear)"
+                f"Unsupported booster: {self.p.booster!r} "
+                "(gbtree/dart/gblinear)"
             )
         self.iteration = 0
         self.booster = Booster(
@@ -321,6 +327,9 @@ class BoostingEngine:
     def update(self):
         it = self.iteration
         label = self.dtrain.label
+        dart_ctx = None
+        if self.p.booster == "dart":
+            dart_ctx = self._dart_dropout(it)
         gpair = self.obj.gradients(
             self.margin, label, self.dtrain.weight, self.dtrain.qid
         )
@@ -341,6 +350,8 @@ class BoostingEngine:
                 trees.append(tree)
                 classes.append(cls)
         self.booster.append_round(trees, classes)
+        if dart_ctx is not None:
+            self._dart_commit(dart_ctx, trees, classes)
         self.iteration += 1
         return trees
 
@@ -402,6 +413,86 @@ class BoostingEngine:
         self.booster.linear_weights = self.lin_w.cpu().numpy()
         self.booster.linear_rounds = self.iteration + 1
         self.booster.params["booster"] = "gblinear"
+
+
+    # -- DART (dropout) ----------------------------------------------------
+    def _dart_predict_trees(self, tree_ids):
+        """Margin contribution of the given trees on the training rows."""
+        contrib = torch.zeros_like(self.margin)
+        for ti in tree_ids:
+            t = self.booster.trees[ti]
+            cls = self.booster.tree_info[ti]
+            mv = contrib if self.n_class == 1 else contrib[:, cls]
+            self._add_tree_margin_raw(t, mv)
+        return contrib
+
+    def _add_tree_margin_raw(self, t, margin_view):
+        """Add one tree's predictions via the binned matrix (exact)."""
+        cuts = self.dtrain.cuts
+        cut_ptr = cuts.cut_ptr.cpu().numpy()
+        cuts_np = cuts.cuts_flat.cpu().numpy()
+        split_bin = np.zeros(t.num_nodes, dtype=np.int32)
+        for nid in range(t.num_nodes):
+            if t.feat[nid] >= 0:
+                f = int(t.feat[nid])
+                lo, hi = cut_ptr[f], cut_ptr[f + 1]
+                split_bin[nid] = np.searchsorted(
+                    cuts_np[lo:hi], t.thr[nid], side="left"
+                )
+        self._add_tree_margin_binned(t, split_bin, margin_view)
+
+    def _dart_dropout(self, it):
+        """Pick the dropped-tree set and remove it from the margins.
+
+        Deterministic: seeded by (seed, iteration), identical on every
+        rank. Returns (dropped_ids, dropped_contrib) or a no-drop marker.
+        """
+        n_trees = len(self.booster.trees)
+        rng = np.random.RandomState(
+            (int(self.p.seed) * 77003 + it * 9973) % (2**31)
+        )
+        if n_trees == 0 or rng.rand() < float(self.p.skip_drop):
+            return ([], None)
+        drop = np.nonzero(rng.rand(n_trees) < float(self.p.rate_drop))[0]
+        if drop.size == 0 and self.p.one_drop:
+            drop = np.array([int(rng.randint(n_trees))])
+        if drop.size == 0:
+            return ([], None)
+        dropped = [int(d) for d in drop]
+        contrib = self._dart_predict_trees(dropped)
+        self.margin -= contrib
+        return (dropped, contrib)
+
+    def _dart_commit(self, dart_ctx, trees, classes):
+        """Scale the new + dropped trees (xgboost normalize_type) and
+        restore margins. Scales are baked into leaf values so predict
+        and serialization need no per-tree weight array."""
+        dropped, contrib = dart_ctx
+        k = len(dropped)
+        if k == 0:
+            return
+        # xgboost dart.cc NormalizeTrees: leaves already carry eta, the
+        # weights below redistribute mass between new and dropped trees
+        lr = float(self.p.eta) / max(1, len(trees) // self.n_class)
+        if self.p.normalize_type == "forest":
+            new_scale = 1.0 / (1.0 + lr)
+            old_scale = 1.0 / (1.0 + lr)
+        else:  # "tree"
+            new_scale = 1.0 / (k + lr)
+            old_scale = k / (k + lr)
+        # scale the just-built trees' leaves (and their margin effect,
+        # which update() already added at full weight)
+        delta = torch.zeros_like(self.margin)
+        for t, cls in zip(trees, classes):
+            mv = delta if self.n_class == 1 else delta[:, cls]
+            self._add_tree_margin_raw(t, mv)
+            t.value *= new_scale
+        self.margin += (new_scale - 1.0) * delta
+        self.booster._flat_cache = None
+        # scale the dropped trees and re-add their shrunk contribution
+        for ti in dropped:
+            self.booster.trees[ti].value *= old_scale
+        self.margin += old_scale * contrib
 
     def _quantize(self, gpair: torch.Tensor) -> Tuple[torch.Tensor, float, float]:
         mx = torch.stack(
