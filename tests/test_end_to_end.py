@@ -382,3 +382,38 @@ def test_dart_distributed_equals_single():
     for ta, tb in zip(b1.trees, b2.trees):
         assert np.array_equal(ta.thr, tb.thr)
         assert np.array_equal(ta.value, tb.value)
+
+
+def _feval_neg_err(margin, dmat):
+    import numpy as np
+
+    label = np.asarray(dmat.get_label())
+    pred = (1.0 / (1.0 + np.exp(-np.asarray(margin)))) > 0.5
+    return "negerr", -float((pred != (label > 0.5)).mean())
+
+
+def test_custom_feval_maximize_early_stopping():
+    """Custom feval + explicit maximize drive early stopping
+    (xgboost.train(maximize=...) parity)."""
+    import numpy as np
+
+    from xgboost_ray_amd import RayDMatrix, RayParams, train
+
+    rng = np.random.RandomState(0)
+    X = rng.rand(3000, 5).astype(np.float32)
+    y = (X[:, 0] + X[:, 1] > 1).astype(np.float32)
+    res = {}
+    bst = train(
+        {"objective": "binary:logistic", "max_depth": 3,
+         "disable_default_eval_metric": True},
+        RayDMatrix(X, y),
+        num_boost_round=30,
+        ray_params=RayParams(num_actors=2),
+        evals=[(RayDMatrix(X, y), "train")],
+        feval=_feval_neg_err,
+        maximize=True,
+        early_stopping_rounds=5,
+        evals_result=res,
+    )
+    assert "negerr" in res["train"]
+    assert bst.best_iteration is not None

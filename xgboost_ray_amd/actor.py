@@ -245,6 +245,7 @@ class _ActorWorker:
                 verbose_eval=kwargs.get("verbose_eval", False),
                 obj=kwargs.get("obj"),
                 feval=kwargs.get("feval"),
+                maximize=kwargs.get("maximize"),
                 evals_result=evals_result,
             )
             if stop_cb.stopped:

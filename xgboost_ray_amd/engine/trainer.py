@@ -1330,6 +1330,7 @@ def run_training(
     verbose_eval: Union[bool, int] = False,
     obj: Optional[Callable] = None,
     feval: Optional[Callable] = None,
+    maximize: Optional[bool] = None,
     evals_result: Optional[Dict] = None,
 ) -> Booster:
     """The per-rank training loop (reference: xgb.train inside
@@ -1343,7 +1344,8 @@ def run_training(
     cb = CallbackList(callbacks)
     log: Dict[str, Dict[str, List[float]]] = {}
     best_score, best_iter, stall = None, None, 0
-    maximize = None
+    # explicit maximize (xgboost.train kwarg) wins; otherwise inferred
+    # from the last builtin metric, defaulting to False for custom feval
     start_iter = engine.iteration
 
     for rnd in range(num_boost_round):
@@ -1372,7 +1374,12 @@ def run_training(
                 last_metric = list(results[last_eval].keys())[-1]
                 score = results[last_eval][last_metric]
                 if maximize is None:
-                    maximize = get_metric(last_metric.split("@")[0]).higher_better
+                    try:
+                        maximize = get_metric(
+                            last_metric.split("@")[0]
+                        ).higher_better
+                    except ValueError:
+                        maximize = False  # custom feval: assume loss-like
                 improved = (
                     best_score is None
                     or (score > best_score if maximize else score < best_score)
