@@ -1046,7 +1046,19 @@ class BoostingEngine:
                     margin.cpu().numpy(),
                     _Shim(None if label is None else label.cpu().numpy()),
                 )
-                out.setdefault(ev.name, {})[name] = float(val)
+                val = float(val)
+                if self.coll.is_distributed:
+                    # a custom feval sees only the local shard; average
+                    # across ranks so every rank's early-stopping decision
+                    # is identical (rank-divergent stops break collectives)
+                    t = torch.tensor(
+                        [val], dtype=torch.float64,
+                        device=self.device
+                        if self.device.type == "cuda" else "cpu",
+                    )
+                    self.coll.allreduce_(t)
+                    val = float(t[0]) / self.coll.world_size
+                out.setdefault(ev.name, {})[name] = val
         return out
 
     def update_eval_margins(self, evals: Sequence[EvalPack], trees, classes):
