@@ -194,3 +194,27 @@ def test_gblinear_gpu():
     )
     w = bst.linear_weights[:, 0]
     assert np.abs(w[:6] - w_true).max() < 0.05
+
+
+@pytest.mark.gpu
+def test_dart_gpu_matches_cpu():
+    """DART on GPU grows the same trees as CPU (dropout replay path)."""
+    import torch
+    from xgboost_ray_amd.engine.quantile import BinnedMatrix
+    from xgboost_ray_amd.engine.trainer import run_training
+
+    X, y = create_data(100000, 8, 11, "reg")
+    params = {"objective": "reg:squarederror", "booster": "dart",
+              "rate_drop": 0.3, "max_depth": 4, "eta": 0.3, "seed": 2}
+    dm_g = BinnedMatrix.build(
+        torch.from_numpy(X).cuda(), label=torch.from_numpy(y).cuda(),
+        max_bin=64,
+    )
+    bst_g = run_training(dict(params), dm_g, 10)
+    dm_c = BinnedMatrix.build(
+        torch.from_numpy(X), label=torch.from_numpy(y), max_bin=64
+    )
+    bst_c = run_training(dict(params), dm_c, 10)
+    for tg, tc in zip(bst_g.trees, bst_c.trees):
+        assert np.array_equal(tg.feat, tc.feat)
+        assert np.array_equal(tg.thr, tc.thr)
