@@ -948,8 +948,7 @@ class BoostingEngine:
             prev_all_hist = all_hist
 
         # remaining frontier nodes (max depth reached) become leaves
-        for nd in frontier:
-            self._finalize_leaf(nd, val_l, cover_l, scale_h)
+        self._finalize_leaves_batch(frontier, val_l, cover_l, scale_h)
         # leaf margin update: walk all leaves via the segment structure -
         # every row's final node is its segment's node. Collect leaf segs.
         # (split nodes consumed their segments; leaves kept them)
@@ -959,6 +958,7 @@ class BoostingEngine:
         leaf_starts = [s for (s, c, v) in self._leaf_segs]
         leaf_counts = [c for (s, c, v) in self._leaf_segs]
         leaf_vals = [v for (s, c, v) in self._leaf_segs]
+        _tick("leaf_host")
         if leaf_starts:
             ops.update_margins(
                 self.margin if self.n_class == 1 else self.margin[:, cls],
@@ -987,6 +987,43 @@ class BoostingEngine:
         return tree
 
     _leaf_segs: List[Tuple[int, int, float]] = []
+
+
+    def _finalize_leaves_batch(self, nodes, val_l, cover_l, scale_h):
+        """Vectorized _finalize_leaf for a whole frontier (the per-node
+        Python loop was 5.6 ms/round on 4096-leaf depth-12 trees).
+        Same op order as _calc_weight, so values are bit-identical."""
+        if not nodes:
+            return
+        G = np.fromiter(
+            (nd.sum_g for nd in nodes), np.float64, len(nodes)
+        ) / self._scale_g_cur
+        H = np.fromiter(
+            (nd.sum_h for nd in nodes), np.float64, len(nodes)
+        ) / scale_h
+        lam = self.p.reg_lambda
+        alpha = self.p.reg_alpha
+        denom = H + lam
+        Gs = G
+        if alpha > 0:
+            Gs = np.copysign(np.maximum(np.abs(G) - alpha, 0.0), G)
+        w = np.where(denom > 0, -Gs / np.where(denom > 0, denom, 1.0), 0.0)
+        if self.mono is not None:
+            w = np.clip(
+                w,
+                np.fromiter((nd.w_lower for nd in nodes), np.float64,
+                            len(nodes)),
+                np.fromiter((nd.w_upper for nd in nodes), np.float64,
+                            len(nodes)),
+            )
+        if self.p.max_delta_step > 0:
+            w = np.clip(w, -self.p.max_delta_step, self.p.max_delta_step)
+        v = self.p.eta * w
+        segs = self._leaf_segs
+        for nd, vi, Hi in zip(nodes, v.tolist(), H.tolist()):
+            val_l[nd.nid] = vi
+            cover_l[nd.nid] = Hi
+            segs.append((nd.start, nd.count, vi))
 
     def _finalize_leaf(self, nd: _Node, val_l, cover_l, scale_h):
         G = nd.sum_g / self._scale_g_cur
