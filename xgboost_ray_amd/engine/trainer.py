@@ -118,7 +118,7 @@ class EvalPack:
     margin: Optional[torch.Tensor] = None
 
 
-@dataclass
+@dataclass(slots=True)
 class _Node:
     nid: int
     depth: int
@@ -833,9 +833,10 @@ class BoostingEngine:
                 thr_all[okf] = cuts_flat_cpu[
                     cut_ptr_cpu[bfeat[okf]] + bbin[okf]
                 ]
+            depth_leaves = []
             for k, nd in enumerate(order_nodes):
                 if not splits_ok[k]:
-                    self._finalize_leaf(nd, val_l, cover_l, scale_h)
+                    depth_leaves.append(nd)
                     continue
                 f = int(bfeat[k])
                 b = int(bbin[k])
@@ -867,6 +868,9 @@ class BoostingEngine:
                     (lid, int(blg[k]), int(blh[k]))
                 )
 
+            self._finalize_leaves_batch(
+                depth_leaves, val_l, cover_l, scale_h
+            )
             if not split_nodes:
                 frontier = []
                 break
@@ -892,9 +896,11 @@ class BoostingEngine:
             _tick("partition")
 
             new_frontier: List[_Node] = []
+            lc_list = left_counts.tolist()  # ONE conversion, not K tensor
+            # __getitem__ + int() pairs (those cost ~1 us each)
             for k, nd in enumerate(split_nodes):
                 lid, lg, lh = children_meta[k]
-                lcount = int(left_counts[k])
+                lcount = lc_list[k]
                 child_path = ()
                 if self.interaction_sets is not None:
                     child_path = tuple(
