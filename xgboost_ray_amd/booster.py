@@ -430,7 +430,8 @@ class Booster:
 
 
     def predict_contribs(self, X, iteration_range=None) -> np.ndarray:
-        """Exact TreeSHAP feature attributions (xgboost ``pred_contribs``).
+        """Exact TreeSHAP feature attributions (xgboost ``pred_contribs``;
+        the reference forwards predict kwargs to xgboost, main.py:795-810).
 
         Returns [n, F+1] (multi-class: [n, num_class, F+1]); the last
         column is the bias (expected value). Rows satisfy additivity:

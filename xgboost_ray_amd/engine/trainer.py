@@ -359,6 +359,10 @@ class BoostingEngine:
     def _update_linear(self, gpair):
         """One gblinear round: preconditioned full-batch coordinate step.
 
+        The reference reaches gblinear through xgboost's booster param
+        passthrough (reference main.py:745 xgb.train(params...); exercised
+        by reference tests/test_sklearn.py test_sklearn_api_gblinear).
+
         Deterministic replacement for xgboost's gblinear updaters
         (shotgun is non-deterministic; this is a damped Jacobi update
         with the same per-coordinate Newton preconditioner, L2 on
@@ -443,6 +447,10 @@ class BoostingEngine:
 
     def _dart_dropout(self, it):
         """Pick the dropped-tree set and remove it from the margins.
+
+        The reference reaches DART through xgboost's booster="dart"
+        passthrough (reference main.py:745); semantics follow xgboost
+        dart.cc (rate_drop/skip_drop/one_drop/normalize_type).
 
         Deterministic: seeded by (seed, iteration), identical on every
         rank. Returns (dropped_ids, dropped_contrib) or a no-drop marker.
