@@ -417,3 +417,66 @@ def test_custom_feval_maximize_early_stopping():
     )
     assert "negerr" in res["train"]
     assert bst.best_iteration is not None
+
+
+@pytest.mark.parametrize(
+    "extra",
+    [
+        {"colsample_bytree": 0.7, "colsample_bylevel": 0.8, "seed": 11},
+        {"grow_policy": "lossguide", "max_leaves": 24, "max_depth": 0},
+        {"booster": "dart", "rate_drop": 0.25, "seed": 4},
+        {"num_parallel_tree": 3, "eta": 0.2},
+        {"monotone_constraints": "(1,0,-1,0,0)", "seed": 2},
+        {"interaction_constraints": [[0, 1], [2, 3, 4]], "max_depth": 5},
+    ],
+)
+def test_distributed_equals_single_matrix(extra):
+    """Randomized differential: 2-actor training grows bitwise-identical
+    trees to 1-actor training across parameter variants (the invariant
+    every rank-divergence bug breaks). Row subsampling is excluded: each
+    rank samples its own shard (reference/xgboost behavior), so the
+    realized row sets - and therefore the trees - legitimately differ
+    between world sizes while remaining per-configuration deterministic
+    (see test_subsample_deterministic_per_world_size)."""
+    import numpy as np
+
+    from xgboost_ray_amd import RayDMatrix, RayParams, train
+
+    rng = np.random.RandomState(0)
+    X = rng.rand(4000, 5).astype(np.float32)
+    y = (X[:, 0] + X[:, 2] + 0.3 * rng.randn(4000) > 1).astype(np.float32)
+    params = {"objective": "binary:logistic", "max_depth": 4, "eta": 0.3}
+    params.update(extra)
+    b1 = train(dict(params), RayDMatrix(X, y), num_boost_round=6,
+               ray_params=RayParams(num_actors=1))
+    b2 = train(dict(params), RayDMatrix(X, y), num_boost_round=6,
+               ray_params=RayParams(num_actors=2))
+    assert len(b1.trees) == len(b2.trees)
+    for ta, tb in zip(b1.trees, b2.trees):
+        assert np.array_equal(ta.feat, tb.feat)
+        assert np.array_equal(ta.thr, tb.thr)
+        assert np.array_equal(ta.value, tb.value)
+
+
+def test_subsample_deterministic_per_world_size():
+    """Subsampled training is reproducible run-to-run at a fixed world
+    size (the checkpoint-resume contract), even though different world
+    sizes realize different row samples."""
+    import numpy as np
+
+    from xgboost_ray_amd import RayDMatrix, RayParams, train
+
+    rng = np.random.RandomState(0)
+    X = rng.rand(4000, 5).astype(np.float32)
+    y = (X[:, 0] + X[:, 2] + 0.3 * rng.randn(4000) > 1).astype(np.float32)
+    params = {"objective": "binary:logistic", "max_depth": 4, "eta": 0.3,
+              "subsample": 0.8, "seed": 11}
+    runs = [
+        train(dict(params), RayDMatrix(X, y), num_boost_round=5,
+              ray_params=RayParams(num_actors=2))
+        for _ in range(2)
+    ]
+    for ta, tb in zip(runs[0].trees, runs[1].trees):
+        assert np.array_equal(ta.feat, tb.feat)
+        assert np.array_equal(ta.thr, tb.thr)
+        assert np.array_equal(ta.value, tb.value)

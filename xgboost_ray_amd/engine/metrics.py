@@ -12,6 +12,8 @@ from typing import Optional
 
 import torch
 
+from xgboost_ray_amd.engine.objectives import sigmoid_sizeinv
+
 _AUC_BINS = 16384
 
 
@@ -62,7 +64,7 @@ class LogLoss(Metric):
     name = "logloss"
 
     def local_stats(self, margin, label, weight, qid, obj):
-        p = torch.sigmoid(margin.double()).clamp(1e-16, 1 - 1e-16)
+        p = sigmoid_sizeinv(margin.double()).clamp(1e-16, 1 - 1e-16)
         y = label.double()
         w = _w(label, weight)
         ll = -(y * torch.log(p) + (1 - y) * torch.log(1 - p))
@@ -81,7 +83,7 @@ class BinaryError(Metric):
             self.name = f"error@{threshold}"
 
     def local_stats(self, margin, label, weight, qid, obj):
-        p = torch.sigmoid(margin.double())
+        p = sigmoid_sizeinv(margin.double())
         w = _w(label, weight)
         wrong = (p > self.threshold).double() != label.double()
         return torch.stack([(w * wrong).sum(), w.sum()])
@@ -97,7 +99,7 @@ class AUC(Metric):
     higher_better = True
 
     def local_stats(self, margin, label, weight, qid, obj):
-        p = torch.sigmoid(margin.double())
+        p = sigmoid_sizeinv(margin.double())
         b = torch.clamp((p * _AUC_BINS).long(), max=_AUC_BINS - 1)
         if weight is None:
             # unweighted: integer scatter (native u64 atomics on GPU -

@@ -16,6 +16,25 @@ import torch
 _EPS = 1e-16
 
 
+def sigmoid_sizeinv(t: torch.Tensor) -> torch.Tensor:
+    """``torch.sigmoid`` whose CPU results do not depend on array length.
+
+    torch's CPU kernel runs a sleef vector body plus a scalar tail whose
+    results can differ by 1 ulp for the same input value; a row that sits
+    in the tail at one shard size and in the body at another broke the
+    bitwise distributed==single invariant (7 quantized-gradient flips on
+    a 4000-row matrix split 2 ways). Padding to a vector-width multiple
+    keeps every real element in the body. GPU elementwise kernels are
+    position-independent, so no padding there.
+    """
+    flat = t.reshape(-1)
+    if t.device.type != "cpu" or flat.numel() % 64 == 0:
+        return torch.sigmoid(t)
+    pad = (-flat.numel()) % 64
+    tp = torch.cat([flat, torch.zeros(pad, dtype=t.dtype)])
+    return torch.sigmoid(tp)[: flat.numel()].reshape(t.shape)
+
+
 class Objective:
     name = "base"
     n_class = 0
@@ -81,7 +100,7 @@ class Logistic(Objective):
         return -math.log(1.0 / base_score - 1.0)
 
     def gradients(self, margin, label, weight=None, qid=None):
-        p = torch.sigmoid(margin)
+        p = sigmoid_sizeinv(margin)
         g = p - label
         h = torch.clamp(p * (1.0 - p), min=_EPS)
         if self.scale_pos_weight != 1.0:
