@@ -428,6 +428,10 @@ def test_custom_feval_maximize_early_stopping():
         {"num_parallel_tree": 3, "eta": 0.2},
         {"monotone_constraints": "(1,0,-1,0,0)", "seed": 2},
         {"interaction_constraints": [[0, 1], [2, 3, 4]], "max_depth": 5},
+        {"booster": "dart", "rate_drop": 0.3, "colsample_bytree": 0.7,
+         "seed": 9},
+        {"_actors": 3, "colsample_bylevel": 0.6, "max_depth": 5,
+         "seed": 13},
     ],
 )
 def test_distributed_equals_single_matrix(extra):
@@ -447,10 +451,11 @@ def test_distributed_equals_single_matrix(extra):
     y = (X[:, 0] + X[:, 2] + 0.3 * rng.randn(4000) > 1).astype(np.float32)
     params = {"objective": "binary:logistic", "max_depth": 4, "eta": 0.3}
     params.update(extra)
+    n_actors = params.pop("_actors", 2)
     b1 = train(dict(params), RayDMatrix(X, y), num_boost_round=6,
                ray_params=RayParams(num_actors=1))
     b2 = train(dict(params), RayDMatrix(X, y), num_boost_round=6,
-               ray_params=RayParams(num_actors=2))
+               ray_params=RayParams(num_actors=n_actors))
     assert len(b1.trees) == len(b2.trees)
     for ta, tb in zip(b1.trees, b2.trees):
         assert np.array_equal(ta.feat, tb.feat)
