@@ -485,3 +485,35 @@ def test_subsample_deterministic_per_world_size():
         assert np.array_equal(ta.feat, tb.feat)
         assert np.array_equal(ta.thr, tb.thr)
         assert np.array_equal(ta.value, tb.value)
+
+
+def test_ranking_distributed_equals_single():
+    """Group-aware qid sharding: whole query groups per actor makes
+    distributed ranking bitwise-equal to single-actor training (the
+    reference fragments groups across row shards), and qid-matrix
+    predictions recombine through the recorded group indices."""
+    import numpy as np
+
+    from xgboost_ray_amd import RayDMatrix, RayParams, predict, train
+
+    rng = np.random.RandomState(0)
+    n = 4000
+    qid = np.repeat(np.arange(n // 20), 20).astype(np.int64)
+    X = rng.rand(n, 5).astype(np.float32)
+    rel = np.clip(
+        (X[:, 0] * 3 + 0.3 * rng.randn(n)).round(), 0, 3
+    ).astype(np.float32)
+    params = {"objective": "rank:ndcg", "max_depth": 4, "eta": 0.3}
+    b1 = train(dict(params), RayDMatrix(X, rel, qid=qid),
+               num_boost_round=4, ray_params=RayParams(num_actors=1))
+    b2 = train(dict(params), RayDMatrix(X, rel, qid=qid),
+               num_boost_round=4, ray_params=RayParams(num_actors=2))
+    for a, b in zip(b1.trees, b2.trees):
+        assert np.array_equal(a.feat, b.feat)
+        assert np.array_equal(a.thr, b.thr)
+        assert np.array_equal(a.value, b.value)
+    # prediction on a qid matrix reassembles in original row order
+    p2 = predict(b2, RayDMatrix(X, qid=qid),
+                 ray_params=RayParams(num_actors=2))
+    p1 = predict(b1, RayDMatrix(X), ray_params=RayParams(num_actors=1))
+    assert np.array_equal(p1, p2)

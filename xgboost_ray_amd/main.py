@@ -661,6 +661,17 @@ def _predict(
             a.shutdown()
     if data.sharding == RayShardingMode.FIXED:
         return np.concatenate(results, axis=0)
+    actor_idx = getattr(getattr(data, "loader", None), "actor_indices", None)
+    if actor_idx and data.sharding == RayShardingMode.INTERLEAVED:
+        # qid matrices shard whole query groups (not raw interleaved
+        # rows) - reassemble through the recorded per-actor indices
+        first = np.asarray(results[0])
+        n = sum(len(r) for r in results)
+        shape = (n,) + first.shape[1:]
+        out = np.empty(shape, dtype=first.dtype)
+        for rank, part in enumerate(results):
+            out[actor_idx[rank]] = np.asarray(part)
+        return out
     return combine_data(data.sharding, results)
 
 

@@ -60,9 +60,24 @@ class RayShardingMode(Enum):
 
 
 def _get_sharding_indices(
-    sharding: RayShardingMode, rank: int, num_actors: int, n: int
+    sharding: RayShardingMode, rank: int, num_actors: int, n: int,
+    qid: Optional[np.ndarray] = None,
 ) -> np.ndarray:
-    """Row indices of actor `rank` (reference matrix.py:1088-1110)."""
+    """Row indices of actor `rank` (reference matrix.py:1088-1110).
+
+    With ``qid`` given (ranking data, sorted by qid), whole query groups
+    are assigned round-robin instead of raw rows: interleaved ROW
+    sharding would fragment every group across actors, silently
+    degrading pairwise gradients (the reference inherits that flaw from
+    its row sharding; group-aware sharding also makes distributed
+    ranking bitwise-equal to single-actor training).
+    """
+    if qid is not None and sharding == RayShardingMode.INTERLEAVED:
+        qid = np.asarray(qid).reshape(-1)
+        change = np.ones(n, dtype=bool)
+        change[1:] = qid[1:] != qid[:-1]
+        group_id = np.cumsum(change) - 1
+        return np.nonzero(group_id % num_actors == rank)[0]
     if sharding == RayShardingMode.BATCH:
         # np.array_split semantics
         splits = np.array_split(np.arange(n), num_actors)
@@ -281,8 +296,13 @@ class _CentralRayDMatrixLoader(_RayDMatrixLoader):
         shard = self._to_shard(local_df)
         n = shard["data"].shape[0]
         refs: Dict[int, Dict[str, Any]] = {}
+        qid_arr = shard.get("qid")
+        self.actor_indices = {}
         for actor_rank in range(num_actors):
-            idx = _get_sharding_indices(sharding, actor_rank, num_actors, n)
+            idx = _get_sharding_indices(
+                sharding, actor_rank, num_actors, n, qid=qid_arr
+            )
+            self.actor_indices[actor_rank] = idx
             actor_refs = {}
             for key, val in shard.items():
                 if key == "feature_names":
