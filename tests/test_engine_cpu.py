@@ -777,3 +777,59 @@ class TestDart:
         b2 = Booster()
         b2.load_model(f)
         assert np.allclose(b2.predict(X[:200]), b.predict(X[:200]))
+
+
+class TestCountObjectives:
+    def test_poisson_learns_rate(self):
+        rng = np.random.RandomState(0)
+        X = rng.rand(20000, 4).astype(np.float32)
+        lam = np.exp(1.0 + X[:, 0] * 2)
+        y = rng.poisson(lam).astype(np.float32)
+        dm = BinnedMatrix.build(
+            torch.from_numpy(X), label=torch.from_numpy(y), max_bin=64
+        )
+        res = {}
+        bst = run_training(
+            {"objective": "count:poisson", "max_depth": 4, "eta": 0.3},
+            dm, 20, evals=[EvalPack(name="t", X=None)], evals_result=res,
+        )
+        pred = bst.predict(X[:5000])
+        assert np.corrcoef(pred, lam[:5000])[0, 1] > 0.95
+        nll = res["t"]["poisson-nloglik"]
+        assert nll[-1] < nll[0]
+
+    def test_gamma_tweedie_sle(self):
+        rng = np.random.RandomState(1)
+        X = rng.rand(10000, 4).astype(np.float32)
+        lam = np.exp(1.0 + X[:, 0] * 2)
+        yg = (rng.gamma(2.0, np.exp(X[:, 0]))).astype(np.float32) + 1e-3
+        dmg = BinnedMatrix.build(
+            torch.from_numpy(X), label=torch.from_numpy(yg), max_bin=64
+        )
+        bg = run_training(
+            {"objective": "reg:gamma", "max_depth": 3, "eta": 0.2}, dmg, 20
+        )
+        assert np.corrcoef(
+            bg.predict(X), 2.0 * np.exp(X[:, 0])
+        )[0, 1] > 0.9
+
+        yp = rng.poisson(lam).astype(np.float32)
+        dmt = BinnedMatrix.build(
+            torch.from_numpy(X), label=torch.from_numpy(yp), max_bin=64
+        )
+        bt = run_training(
+            {"objective": "reg:tweedie", "max_depth": 3, "eta": 0.2,
+             "tweedie_variance_power": 1.3}, dmt, 15,
+        )
+        assert np.corrcoef(bt.predict(X), lam)[0, 1] > 0.95
+
+        ysl = (X[:, 0] * 5).astype(np.float32)
+        dms = BinnedMatrix.build(
+            torch.from_numpy(X), label=torch.from_numpy(ysl), max_bin=64
+        )
+        bs = run_training(
+            {"objective": "reg:squaredlogerror", "max_depth": 3,
+             "eta": 0.3, "base_score": 1.0}, dms, 25,
+        )
+        rmse = float(np.sqrt(np.mean((bs.predict(X) - ysl) ** 2)))
+        assert rmse < 0.2

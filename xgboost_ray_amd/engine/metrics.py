@@ -248,11 +248,82 @@ class MAP(_PerGroupMetric):
         return float(ap)
 
 
+
+class PoissonNLogLik(Metric):
+    name = "poisson-nloglik"
+    higher_better = False
+
+    def local_stats(self, margin, label, weight, qid, obj):
+        mu = torch.exp(margin.double()).clamp(min=1e-16)
+        y = label.double()
+        nll = mu - y * torch.log(mu) + torch.lgamma(y + 1.0)
+        w = _w(label, weight)
+        return torch.stack([(w * nll).sum(), w.sum()])
+
+    def finalize(self, s):
+        return float(s[0] / s[1])
+
+
+class GammaNLogLik(Metric):
+    name = "gamma-nloglik"
+    higher_better = False
+
+    def local_stats(self, margin, label, weight, qid, obj):
+        mu = torch.exp(margin.double()).clamp(min=1e-16)
+        y = label.double()
+        nll = y / mu + torch.log(mu)
+        w = _w(label, weight)
+        return torch.stack([(w * nll).sum(), w.sum()])
+
+    def finalize(self, s):
+        return float(s[0] / s[1])
+
+
+class TweedieNLogLik(Metric):
+    higher_better = False
+
+    def __init__(self, rho: float = 1.5):
+        self.rho = float(rho)
+        self.name = f"tweedie-nloglik@{self.rho}"
+
+    def local_stats(self, margin, label, weight, qid, obj):
+        m = margin.double()
+        y = label.double()
+        r = self.rho
+        nll = -y * torch.exp((1.0 - r) * m) / (1.0 - r) + torch.exp(
+            (2.0 - r) * m
+        ) / (2.0 - r)
+        w = _w(label, weight)
+        return torch.stack([(w * nll).sum(), w.sum()])
+
+    def finalize(self, s):
+        return float(s[0] / s[1])
+
+
+class RMSLE(Metric):
+    name = "rmsle"
+    higher_better = False
+
+    def local_stats(self, margin, label, weight, qid, obj):
+        p = torch.clamp(margin.double(), min=-1.0 + 1e-6)
+        d = torch.log1p(p) - torch.log1p(label.double())
+        w = _w(label, weight)
+        return torch.stack([(w * d * d).sum(), w.sum()])
+
+    def finalize(self, s):
+        import math
+
+        return float(math.sqrt(s[0] / s[1]))
+
+
 def get_metric(name: str) -> Metric:
     if name.startswith("error@"):
         return BinaryError(float(name.split("@")[1]))
     if name.startswith("ndcg@"):
         return NDCG(int(name.split("@")[1]))
+    if name.startswith("tweedie-nloglik"):
+        parts = name.split("@")
+        return TweedieNLogLik(float(parts[1]) if len(parts) > 1 else 1.5)
     table = {
         "rmse": RMSE,
         "mae": MAE,
@@ -264,6 +335,9 @@ def get_metric(name: str) -> Metric:
         "merror": MError,
         "ndcg": NDCG,
         "map": MAP,
+        "poisson-nloglik": PoissonNLogLik,
+        "gamma-nloglik": GammaNLogLik,
+        "rmsle": RMSLE,
     }
     if name not in table:
         raise ValueError(f"Unsupported eval_metric: {name}")

@@ -27,12 +27,19 @@ def sigmoid_sizeinv(t: torch.Tensor) -> torch.Tensor:
     keeps every real element in the body. GPU elementwise kernels are
     position-independent, so no padding there.
     """
+    return elemwise_sizeinv(torch.sigmoid, t)
+
+
+def elemwise_sizeinv(fn, t: torch.Tensor) -> torch.Tensor:
+    """Apply an elementwise transcendental with length-invariant CPU
+    results (see sigmoid_sizeinv) - used for sigmoid/exp/log1p in
+    gradient paths that must be bitwise world-size invariant."""
     flat = t.reshape(-1)
     if t.device.type != "cpu" or flat.numel() % 64 == 0:
-        return torch.sigmoid(t)
+        return fn(t)
     pad = (-flat.numel()) % 64
     tp = torch.cat([flat, torch.zeros(pad, dtype=t.dtype)])
-    return torch.sigmoid(tp)[: flat.numel()].reshape(t.shape)
+    return fn(tp)[: flat.numel()].reshape(t.shape)
 
 
 class Objective:
@@ -344,6 +351,103 @@ class CustomObjective(Objective):
         return torch.stack([g.reshape(margin.shape), h.reshape(margin.shape)], dim=-1)
 
 
+
+class Poisson(Objective):
+    """count:poisson - margin is log(mean); xgboost semantics incl. the
+    max_delta_step=0.7 default applied by the trainer."""
+
+    name = "count:poisson"
+    default_metric = "poisson-nloglik"
+
+    def prob_to_margin(self, base_score):
+        return math.log(max(base_score, _EPS))
+
+    def gradients(self, margin, label, weight=None, qid=None):
+        mu = elemwise_sizeinv(torch.exp, margin)
+        g = mu - label
+        h = mu
+        return self._apply_weight(g, h, weight)
+
+    def transform_prediction(self, margin):
+        return torch.exp(margin)
+
+    def validate_label(self, label):
+        if bool((label < 0).any()):
+            raise ValueError("count:poisson requires non-negative labels")
+
+
+class Gamma(Objective):
+    """reg:gamma - log-link gamma deviance."""
+
+    name = "reg:gamma"
+    default_metric = "gamma-nloglik"
+
+    def prob_to_margin(self, base_score):
+        return math.log(max(base_score, _EPS))
+
+    def gradients(self, margin, label, weight=None, qid=None):
+        ratio = label * elemwise_sizeinv(torch.exp, -margin)
+        g = 1.0 - ratio
+        h = ratio
+        return self._apply_weight(g, h, weight)
+
+    def transform_prediction(self, margin):
+        return torch.exp(margin)
+
+    def validate_label(self, label):
+        if bool((label <= 0).any()):
+            raise ValueError("reg:gamma requires positive labels")
+
+
+class Tweedie(Objective):
+    """reg:tweedie - log-link, variance power rho (default 1.5)."""
+
+    name = "reg:tweedie"
+    default_metric = "tweedie-nloglik"
+
+    def __init__(self, rho: float = 1.5):
+        self.rho = float(rho)
+        self.default_metric = f"tweedie-nloglik@{self.rho}"
+
+    def prob_to_margin(self, base_score):
+        return math.log(max(base_score, _EPS))
+
+    def gradients(self, margin, label, weight=None, qid=None):
+        r = self.rho
+        a = elemwise_sizeinv(torch.exp, (1.0 - r) * margin)
+        b = elemwise_sizeinv(torch.exp, (2.0 - r) * margin)
+        g = -label * a + b
+        h = -label * (1.0 - r) * a + (2.0 - r) * b
+        return self._apply_weight(g, h, weight)
+
+    def transform_prediction(self, margin):
+        return torch.exp(margin)
+
+    def validate_label(self, label):
+        if bool((label < 0).any()):
+            raise ValueError("reg:tweedie requires non-negative labels")
+
+
+class SquaredLogError(Objective):
+    """reg:squaredlogerror - 0.5*(log1p(pred)-log1p(y))^2 on raw margin."""
+
+    name = "reg:squaredlogerror"
+    default_metric = "rmsle"
+
+    def gradients(self, margin, label, weight=None, qid=None):
+        p1 = torch.clamp(margin + 1.0, min=1e-6)
+        d = elemwise_sizeinv(torch.log, p1) - elemwise_sizeinv(
+            torch.log, torch.clamp(label + 1.0, min=1e-6)
+        )
+        g = d / p1
+        h = torch.clamp((1.0 - d) / (p1 * p1), min=1e-6)
+        return self._apply_weight(g, h, weight)
+
+    def validate_label(self, label):
+        if bool((label < -1).any()):
+            raise ValueError("reg:squaredlogerror requires label > -1")
+
+
 _REGISTRY = {
     "reg:squarederror": SquaredError,
     "reg:linear": SquaredError,  # legacy alias
@@ -353,15 +457,22 @@ _REGISTRY = {
     "binary:logitraw": LogisticRaw,
     "rank:pairwise": RankPairwise,
     "rank:ndcg": RankNDCG,
+    "count:poisson": Poisson,
+    "reg:gamma": Gamma,
+    "reg:tweedie": Tweedie,
+    "reg:squaredlogerror": SquaredLogError,
 }
 
 
 def get_objective(
-    name_or_fn, num_class: int = 0, scale_pos_weight: float = 1.0
+    name_or_fn, num_class: int = 0, scale_pos_weight: float = 1.0,
+    tweedie_variance_power: float = 1.5,
 ) -> Objective:
     if callable(name_or_fn):
         return CustomObjective(name_or_fn, num_class)
     name = name_or_fn or "reg:squarederror"
+    if name == "reg:tweedie":
+        return Tweedie(tweedie_variance_power)
     if name in ("multi:softmax",):
         return SoftmaxClass(num_class)
     if name in ("multi:softprob",):

@@ -56,6 +56,7 @@ class TrainParams:
     max_leaves: int = 0
     monotone_constraints: object = None
     interaction_constraints: object = None
+    tweedie_variance_power: float = 1.5
     tree_method: str = "hist"
     booster: str = "gbtree"
     # DART (booster="dart") dropout parameters (xgboost semantics)
@@ -177,8 +178,13 @@ class BoostingEngine:
         self._pin_bufs = {}
         obj_spec = custom_objective or self.p.objective
         self.obj: Objective = get_objective(
-            obj_spec, self.p.num_class, float(self.p.scale_pos_weight)
+            obj_spec, self.p.num_class, float(self.p.scale_pos_weight),
+            float(self.p.tweedie_variance_power),
         )
+        # xgboost defaults max_delta_step to 0.7 for Poisson regression
+        if (self.p.objective == "count:poisson"
+                and not self.p.max_delta_step):
+            self.p.max_delta_step = 0.7
         self.n_class = max(1, self.p.num_class)
         if self.p.base_score is None:
             self.p.base_score = 0.5
