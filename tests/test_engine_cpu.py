@@ -833,3 +833,52 @@ class TestCountObjectives:
         )
         rmse = float(np.sqrt(np.mean((bs.predict(X) - ysl) ** 2)))
         assert rmse < 0.2
+
+
+class TestAFT:
+    def test_aft_censored_all_distributions(self):
+        rng = np.random.RandomState(0)
+        n = 15000
+        X = rng.rand(n, 4).astype(np.float32)
+        t = np.exp(1.0 + 2 * X[:, 0] + 0.3 * rng.randn(n)).astype(
+            np.float32
+        )
+        cens = np.exp(1.0 + 2 * X[:, 0] + rng.rand(n)).astype(np.float32)
+        right = rng.rand(n) < 0.3
+        yl = np.where(right, np.minimum(t, cens), t).astype(np.float32)
+        yu = np.where(right, np.inf, t).astype(np.float32)
+        dm = BinnedMatrix.build(torch.from_numpy(X), max_bin=64)
+        dm.label_lower_bound = torch.from_numpy(yl)
+        dm.label_upper_bound = torch.from_numpy(yu)
+        for dist in ("normal", "logistic", "extreme"):
+            res = {}
+            bst = run_training(
+                {"objective": "survival:aft", "max_depth": 4, "eta": 0.3,
+                 "aft_loss_distribution": dist, "base_score": 1.0},
+                dm, 15, evals=[EvalPack(name="t", X=None)],
+                evals_result=res,
+            )
+            pred = bst.predict(X[:4000], output_margin=True)
+            assert np.corrcoef(pred, 1 + 2 * X[:4000, 0])[0, 1] > 0.95
+            nll = res["t"]["aft-nloglik"]
+            assert nll[-1] < nll[0]
+
+    def test_aft_through_raydmatrix(self):
+        from xgboost_ray_amd import RayDMatrix, RayParams, train
+
+        rng = np.random.RandomState(1)
+        n = 6000
+        X = rng.rand(n, 4).astype(np.float32)
+        t = np.exp(1.0 + 2 * X[:, 0]).astype(np.float32)
+        right = rng.rand(n) < 0.2
+        yl = t.copy()
+        yu = np.where(right, np.inf, t).astype(np.float32)
+        bst = train(
+            {"objective": "survival:aft", "max_depth": 3, "eta": 0.3,
+             "base_score": 1.0},
+            RayDMatrix(X, label_lower_bound=yl, label_upper_bound=yu),
+            num_boost_round=10,
+            ray_params=RayParams(num_actors=2),
+        )
+        pred = bst.predict(X[:2000], output_margin=True)
+        assert np.corrcoef(pred, 1 + 2 * X[:2000, 0])[0, 1] > 0.9

@@ -187,6 +187,11 @@ class _ActorWorker:
                     dm.feature_weights = _to_dev(
                         shard["feature_weights"], self.device
                     )
+                for bkey in ("label_lower_bound", "label_upper_bound"):
+                    if shard.get(bkey) is not None:
+                        setattr(
+                            dm, bkey, _to_dev(shard[bkey], self.device)
+                        )
 
             eval_packs = []
             for uid, name in evals:

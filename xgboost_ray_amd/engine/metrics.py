@@ -316,6 +316,27 @@ class RMSLE(Metric):
         return float(math.sqrt(s[0] / s[1]))
 
 
+
+class AFTNLogLik(Metric):
+    """survival:aft negative log likelihood; label is the [n, 2]
+    (lower, upper) bound stack the engine builds for AFT training."""
+
+    name = "aft-nloglik"
+    higher_better = False
+
+    def local_stats(self, margin, label, weight, qid, obj):
+        from xgboost_ray_amd.engine.objectives import AFT
+
+        o = obj if isinstance(obj, AFT) else AFT()
+        nll = o.nll(margin.double(), label[:, 0].double(),
+                    label[:, 1].double())
+        w = _w(label[:, 0], weight)
+        return torch.stack([(w * nll).sum(), w.sum()])
+
+    def finalize(self, s):
+        return float(s[0] / s[1])
+
+
 def get_metric(name: str) -> Metric:
     if name.startswith("error@"):
         return BinaryError(float(name.split("@")[1]))
@@ -338,6 +359,7 @@ def get_metric(name: str) -> Metric:
         "poisson-nloglik": PoissonNLogLik,
         "gamma-nloglik": GammaNLogLik,
         "rmsle": RMSLE,
+        "aft-nloglik": AFTNLogLik,
     }
     if name not in table:
         raise ValueError(f"Unsupported eval_metric: {name}")

@@ -57,6 +57,8 @@ class TrainParams:
     monotone_constraints: object = None
     interaction_constraints: object = None
     tweedie_variance_power: float = 1.5
+    aft_loss_distribution: str = "normal"
+    aft_loss_distribution_scale: float = 1.0
     tree_method: str = "hist"
     booster: str = "gbtree"
     # DART (booster="dart") dropout parameters (xgboost semantics)
@@ -185,6 +187,11 @@ class BoostingEngine:
         if (self.p.objective == "count:poisson"
                 and not self.p.max_delta_step):
             self.p.max_delta_step = 0.7
+        from xgboost_ray_amd.engine.objectives import AFT
+
+        if isinstance(self.obj, AFT):
+            self.obj.dist = str(self.p.aft_loss_distribution)
+            self.obj.sigma = float(self.p.aft_loss_distribution_scale)
         self.n_class = max(1, self.p.num_class)
         if self.p.base_score is None:
             self.p.base_score = 0.5
@@ -200,6 +207,21 @@ class BoostingEngine:
             self.margin = self.margin + bm
         if dtrain.label is not None:
             self.obj.validate_label(dtrain.label)
+        self._train_label = dtrain.label
+        if isinstance(self.obj, AFT):
+            yl = getattr(dtrain, "label_lower_bound", None)
+            yu = getattr(dtrain, "label_upper_bound", None)
+            if yl is None or yu is None:
+                if dtrain.label is None:
+                    raise ValueError(
+                        "survival:aft needs label_lower_bound/"
+                        "label_upper_bound (or a plain label for the "
+                        "uncensored case)"
+                    )
+                yl = yu = dtrain.label
+            self._train_label = torch.stack(
+                [yl.float(), yu.float()], dim=1
+            )
         self.feat_bins = dtrain.cuts.feat_bins().to(self.device)
         self.n_bins = dtrain.cuts.max_bins
         self.mono = None
@@ -332,7 +354,7 @@ class BoostingEngine:
     # -- one boosting round --------------------------------------------------
     def update(self):
         it = self.iteration
-        label = self.dtrain.label
+        label = self._train_label
         dart_ctx = None
         if self.p.booster == "dart":
             dart_ctx = self._dart_dropout(it)
@@ -1067,7 +1089,7 @@ class BoostingEngine:
         specs = []
         for ev in evals:
             margin = self.margin if ev.X is None else ev.margin
-            label = self.dtrain.label if ev.X is None else ev.label
+            label = self._train_label if ev.X is None else ev.label
             weight = self.dtrain.weight if ev.X is None else ev.weight
             qid = self.dtrain.qid if ev.X is None else ev.qid
             for mname in metric_names:
