@@ -882,3 +882,29 @@ class TestAFT:
         )
         pred = bst.predict(X[:2000], output_margin=True)
         assert np.corrcoef(pred, 1 + 2 * X[:2000, 0])[0, 1] > 0.9
+
+    def test_cox_recovers_hazard(self):
+        rng = np.random.RandomState(0)
+        n = 15000
+        X = rng.rand(n, 4).astype(np.float32)
+        true_risk = 2.0 * X[:, 0] - 1.0 * X[:, 1]
+        t = rng.exponential(1.0 / np.exp(true_risk)).astype(np.float32)
+        cens = rng.exponential(2.0, n).astype(np.float32)
+        event = t <= cens
+        label = np.where(event, np.minimum(t, cens), -cens).astype(
+            np.float32
+        )
+        label[label == 0] = 1e-6
+        dm = BinnedMatrix.build(
+            torch.from_numpy(X), label=torch.from_numpy(label), max_bin=64
+        )
+        res = {}
+        bst = run_training(
+            {"objective": "survival:cox", "max_depth": 4, "eta": 0.3,
+             "base_score": 1.0},
+            dm, 15, evals=[EvalPack(name="t", X=None)], evals_result=res,
+        )
+        pred = bst.predict(X[:4000], output_margin=True)
+        assert np.corrcoef(pred, true_risk[:4000])[0, 1] > 0.9
+        nll = res["t"]["cox-nloglik"]
+        assert nll[-1] < nll[0]

@@ -337,6 +337,34 @@ class AFTNLogLik(Metric):
         return float(s[0] / s[1])
 
 
+
+class CoxNLogLik(Metric):
+    """Breslow negative partial log likelihood per event (local shard)."""
+
+    name = "cox-nloglik"
+    higher_better = False
+
+    def local_stats(self, margin, label, weight, qid, obj):
+        t = label.abs().double()
+        event = label > 0
+        m = margin.double()
+        shift = m.max()
+        eta = torch.exp(m - shift)
+        order = torch.argsort(-t, stable=True)
+        eta_s = eta[order]
+        ev_s = event[order]
+        m_s = m[order]
+        cum = torch.cumsum(eta_s, 0)
+        ll = (m_s[ev_s] - shift - torch.log(
+            torch.clamp(cum[ev_s], min=1e-300)
+        )).sum()
+        n_ev = ev_s.sum()
+        return torch.stack([-ll, n_ev.double()])
+
+    def finalize(self, s):
+        return float(s[0] / max(float(s[1]), 1.0))
+
+
 def get_metric(name: str) -> Metric:
     if name.startswith("error@"):
         return BinaryError(float(name.split("@")[1]))
@@ -360,6 +388,7 @@ def get_metric(name: str) -> Metric:
         "gamma-nloglik": GammaNLogLik,
         "rmsle": RMSLE,
         "aft-nloglik": AFTNLogLik,
+        "cox-nloglik": CoxNLogLik,
     }
     if name not in table:
         raise ValueError(f"Unsupported eval_metric: {name}")

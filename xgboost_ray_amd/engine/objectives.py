@@ -635,6 +635,88 @@ class AFT(Objective):
         pass  # bounds validated at stack time
 
 
+
+class Cox(Objective):
+    """survival:cox - Breslow partial likelihood on right-censored data.
+
+    xgboost label encoding: label > 0 is an event at time t=label;
+    label < 0 is right-censored at t=|label|. Risk sets are computed over
+    the local shard (xgboost's distributed behavior); margins are
+    log-hazard ratios, predictions exp(margin). All ops are sorted
+    cumsums - deterministic on CPU and GPU, no custom kernels.
+    """
+
+    name = "survival:cox"
+    default_metric = "cox-nloglik"
+
+    def gradients(self, margin, label, weight=None, qid=None):
+        t = label.abs().double()
+        event = (label > 0).double()
+        m = margin.double()
+        eta = elemwise_sizeinv(torch.exp, m - m.max())
+        # sort by time DESCENDING: risk set of time t = all rows with
+        # t_j >= t = prefix of the sorted order
+        order = torch.argsort(-t, stable=True)
+        t_s = t[order]
+        eta_s = eta[order]
+        ev_s = event[order]
+        cum_eta = torch.cumsum(eta_s, 0)
+        # rows tied on time share one risk set: take the cumsum at the
+        # LAST row of each tied block
+        change = torch.ones_like(t_s, dtype=torch.bool)
+        change[:-1] = t_s[:-1] != t_s[1:]
+        block_last = torch.cumsum(change.long(), 0)  # 1-based block id
+        # last index of each block via scatter of positions
+        n = t_s.numel()
+        pos = torch.arange(n, device=t.device)
+        nblocks = int(block_last[-1])
+        last_idx = torch.zeros(
+            nblocks + 1, dtype=torch.long, device=t.device
+        )
+        last_idx.scatter_reduce_(
+            0, block_last, pos, reduce="amax", include_self=False
+        )
+        R_s = cum_eta[last_idx[block_last]]
+        R_s = torch.clamp(R_s, min=1e-300)
+        # event terms: per sorted row, a_k = ev/R, b_k = ev/R^2;
+        # each row i accumulates over events with t_k <= t_i, i.e. the
+        # SUFFIX of the descending order
+        a = ev_s / R_s
+        b = ev_s / (R_s * R_s)
+        # suffix sums respecting ties: all rows of a tied block get the
+        # same suffix (events at equal time include each other's risk)
+        suf_a = torch.flip(torch.cumsum(torch.flip(a, [0]), 0), [0])
+        suf_b = torch.flip(torch.cumsum(torch.flip(b, [0]), 0), [0])
+        # align tied blocks to their FIRST row's suffix value
+        first_idx = torch.zeros(
+            nblocks + 1, dtype=torch.long, device=t.device
+        )
+        first_idx.scatter_reduce_(
+            0, block_last, pos, reduce="amin", include_self=False
+        )
+        suf_a = suf_a[first_idx[block_last]]
+        suf_b = suf_b[first_idx[block_last]]
+        g_s = -ev_s + eta_s * suf_a
+        h_s = torch.clamp(
+            eta_s * suf_a - eta_s * eta_s * suf_b, min=1e-6
+        )
+        g = torch.empty_like(g_s)
+        h = torch.empty_like(h_s)
+        g[order] = g_s
+        h[order] = h_s
+        return self._apply_weight(g.float(), h.float(), weight)
+
+    def transform_prediction(self, margin):
+        return torch.exp(margin)
+
+    def validate_label(self, label):
+        if bool((label == 0).any()):
+            raise ValueError(
+                "survival:cox labels must be nonzero (+t event, -t "
+                "censored)"
+            )
+
+
 _REGISTRY = {
     "reg:squarederror": SquaredError,
     "reg:linear": SquaredError,  # legacy alias
@@ -649,6 +731,7 @@ _REGISTRY = {
     "reg:tweedie": Tweedie,
     "reg:squaredlogerror": SquaredLogError,
     "survival:aft": AFT,
+    "survival:cox": Cox,
 }
 
 
