@@ -908,3 +908,22 @@ class TestAFT:
         assert np.corrcoef(pred, true_risk[:4000])[0, 1] > 0.9
         nll = res["t"]["cox-nloglik"]
         assert nll[-1] < nll[0]
+
+    def test_rank_map_objective(self):
+        rng = np.random.RandomState(0)
+        n = 8000
+        qid = np.repeat(np.arange(n // 20), 20).astype(np.int64)
+        X = rng.rand(n, 5).astype(np.float32)
+        rel = (X[:, 0] + 0.3 * rng.randn(n) > 0.7).astype(np.float32)
+        dm = BinnedMatrix.build(
+            torch.from_numpy(X), label=torch.from_numpy(rel),
+            qid=torch.from_numpy(qid), max_bin=64,
+        )
+        res = {}
+        run_training(
+            {"objective": "rank:map", "max_depth": 4, "eta": 0.3},
+            dm, 15, evals=[EvalPack(name="t", X=None)], evals_result=res,
+        )
+        mp = res["t"]["map"]
+        assert mp[-1] > mp[0]          # optimizes its own metric
+        assert mp[-1] > 0.76           # reaches the data's MAP ceiling

@@ -198,9 +198,12 @@ class LambdaRankBase(Objective):
     def gradients(self, margin, label, weight=None, qid=None):
         if qid is None:
             raise ValueError("rank objectives require qid/group information")
-        device = margin.device
-        if device.type == "cuda":
+        if margin.device.type == "cuda":
             return self._gradients_gpu(margin, label, weight, qid)
+        return self._gradients_generic(margin, label, weight, qid)
+
+    def _gradients_generic(self, margin, label, weight=None, qid=None):
+        device = margin.device
         g = torch.zeros_like(margin)
         h = torch.zeros_like(margin)
         qid = qid.to(device)
@@ -222,7 +225,41 @@ class LambdaRankBase(Objective):
             rho = torch.sigmoid(-sij)  # d/ds_i of log-loss for pair (i,j)
             lam = rho
             hess = torch.clamp(rho * (1.0 - rho), min=_EPS)
-            if self.ndcg_weighting:
+            if getattr(self, "map_weighting", False):
+                # |delta AP| for swapping i, j at the current ranking
+                # (binary relevance; closed form via 1/k prefix sums)
+                rel = (y > 0).to(m.dtype)
+                R = rel.sum()
+                if float(R) <= 0:
+                    continue
+                order = torch.argsort(m, descending=True)
+                rank = torch.empty_like(order)
+                rank[order] = torch.arange(e - s, device=device)
+                rel_sorted = rel[order]
+                prefix = torch.cumsum(rel_sorted, 0)  # rel count in top k+1
+                invk = 1.0 / torch.arange(
+                    1, e - s + 1, device=device, dtype=m.dtype
+                )
+                S = torch.cumsum(rel_sorted * invk, 0)  # sum 1/k over rel
+                # per-doc terms at its own rank (0-based rank r -> k=r+1)
+                term = prefix[rank] * invk[rank]       # prefix[k]/k
+                Sr = S[rank]                           # S up to own rank
+                # pair (i above j): delta = |term_j' - term_i - (S_{j-1}-S_i)|
+                ra = rank.unsqueeze(1)
+                rb = rank.unsqueeze(0)
+                lo_r = torch.minimum(ra, rb)
+                hi_r = torch.maximum(ra, rb)
+                S_lo = S[lo_r]
+                S_hi_m1 = torch.where(
+                    hi_r > 0, S[torch.clamp(hi_r - 1, min=0)],
+                    torch.zeros_like(S_lo),
+                )
+                t_lo = (prefix[lo_r]) * invk[lo_r]
+                t_hi = (prefix[hi_r]) * invk[hi_r]
+                delta = torch.abs(t_hi - t_lo - (S_hi_m1 - S_lo)) / R
+                lam = lam * delta
+                hess = hess * delta
+            elif self.ndcg_weighting:
                 # |delta NDCG| for swapping i, j at current ranking
                 order = torch.argsort(m, descending=True)
                 rank = torch.empty_like(order)
@@ -299,6 +336,21 @@ class LambdaRankBase(Objective):
         if weight is not None:
             gpair = gpair * weight.unsqueeze(1)
         return gpair
+
+
+class RankMAP(LambdaRankBase):
+    """rank:map - pairwise lambdas weighted by |delta-AP| (binary
+    relevance). Uses the generic per-group path on every device (no
+    dedicated HIP kernel; MAP weighting is a rarely-hot objective)."""
+
+    name = "rank:map"
+    default_metric = "map"
+    map_weighting = True
+
+    def _gradients_gpu(self, margin, label, weight, qid):
+        # no dedicated HIP kernel: run the generic group loop on the
+        # device tensors (correct, just not the fast path)
+        return self._gradients_generic(margin, label, weight, qid)
 
 
 class RankPairwise(LambdaRankBase):
@@ -726,6 +778,7 @@ _REGISTRY = {
     "binary:logitraw": LogisticRaw,
     "rank:pairwise": RankPairwise,
     "rank:ndcg": RankNDCG,
+    "rank:map": RankMAP,
     "count:poisson": Poisson,
     "reg:gamma": Gamma,
     "reg:tweedie": Tweedie,
