@@ -927,3 +927,27 @@ class TestAFT:
         mp = res["t"]["map"]
         assert mp[-1] > mp[0]          # optimizes its own metric
         assert mp[-1] > 0.76           # reaches the data's MAP ceiling
+
+    def test_pseudohuber_and_hinge(self):
+        rng = np.random.RandomState(0)
+        X = rng.rand(8000, 4).astype(np.float32)
+        y = (X[:, 0] * 3 + 0.1 * rng.randn(8000)).astype(np.float32)
+        y[::50] += 20  # outliers shrugged off by the robust loss
+        dm = BinnedMatrix.build(
+            torch.from_numpy(X), label=torch.from_numpy(y), max_bin=64
+        )
+        b = run_training(
+            {"objective": "reg:pseudohubererror", "max_depth": 4,
+             "eta": 0.3, "base_score": 0.0}, dm, 30,
+        )
+        med = np.median(np.abs(b.predict(X) - X[:, 0] * 3))
+        assert med < 0.25
+        yb = (X[:, 0] > 0.5).astype(np.float32)
+        dmb = BinnedMatrix.build(
+            torch.from_numpy(X), label=torch.from_numpy(yb), max_bin=64
+        )
+        bh = run_training(
+            {"objective": "binary:hinge", "max_depth": 3, "eta": 0.3,
+             "base_score": 0.0}, dmb, 15,
+        )
+        assert float((bh.predict(X) == yb).mean()) > 0.9

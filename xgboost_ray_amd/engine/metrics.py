@@ -365,6 +365,23 @@ class CoxNLogLik(Metric):
         return float(s[0] / max(float(s[1]), 1.0))
 
 
+
+class MPHE(Metric):
+    """Mean pseudo-Huber error (delta = 1)."""
+
+    name = "mphe"
+    higher_better = False
+
+    def local_stats(self, margin, label, weight, qid, obj):
+        r = (margin - label).double()
+        loss = torch.sqrt(1.0 + r * r) - 1.0
+        w = _w(label, weight)
+        return torch.stack([(w * loss).sum(), w.sum()])
+
+    def finalize(self, s):
+        return float(s[0] / s[1])
+
+
 def get_metric(name: str) -> Metric:
     if name.startswith("error@"):
         return BinaryError(float(name.split("@")[1]))
@@ -389,6 +406,7 @@ def get_metric(name: str) -> Metric:
         "rmsle": RMSLE,
         "aft-nloglik": AFTNLogLik,
         "cox-nloglik": CoxNLogLik,
+        "mphe": MPHE,
     }
     if name not in table:
         raise ValueError(f"Unsupported eval_metric: {name}")

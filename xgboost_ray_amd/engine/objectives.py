@@ -769,6 +769,49 @@ class Cox(Objective):
             )
 
 
+
+class PseudoHuber(Objective):
+    """reg:pseudohubererror - smooth Huber, slope delta (huber_slope)."""
+
+    name = "reg:pseudohubererror"
+    default_metric = "mphe"
+
+    def __init__(self, delta: float = 1.0):
+        self.delta = float(delta)
+
+    def gradients(self, margin, label, weight=None, qid=None):
+        r = margin - label
+        z = 1.0 + (r / self.delta) ** 2
+        sq = torch.sqrt(z)
+        g = r / sq
+        h = torch.clamp(1.0 / (z * sq), min=_EPS)
+        return self._apply_weight(g, h, weight)
+
+
+class Hinge(Objective):
+    """binary:hinge - predicts 0/1 directly."""
+
+    name = "binary:hinge"
+    default_metric = "error"
+
+    def gradients(self, margin, label, weight=None, qid=None):
+        y = 2.0 * label - 1.0  # {0,1} -> {-1,+1}
+        active = margin * y < 1.0
+        g = torch.where(active, -y, torch.zeros_like(margin))
+        h = torch.where(
+            active, torch.ones_like(margin),
+            torch.full_like(margin, _EPS),
+        )
+        return self._apply_weight(g, h, weight)
+
+    def transform_prediction(self, margin):
+        return (margin > 0).to(torch.float32)
+
+    def validate_label(self, label):
+        if bool((label < 0).any()) or bool((label > 1).any()):
+            raise ValueError("binary:hinge labels must be 0/1")
+
+
 _REGISTRY = {
     "reg:squarederror": SquaredError,
     "reg:linear": SquaredError,  # legacy alias
@@ -785,6 +828,8 @@ _REGISTRY = {
     "reg:squaredlogerror": SquaredLogError,
     "survival:aft": AFT,
     "survival:cox": Cox,
+    "reg:pseudohubererror": PseudoHuber,
+    "binary:hinge": Hinge,
 }
 
 
