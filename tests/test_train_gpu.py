@@ -218,3 +218,60 @@ def test_dart_gpu_matches_cpu():
     for tg, tc in zip(bst_g.trees, bst_c.trees):
         assert np.array_equal(tg.feat, tc.feat)
         assert np.array_equal(tg.thr, tc.thr)
+
+
+@pytest.mark.gpu
+def test_extended_objectives_gpu():
+    """Poisson / tweedie / AFT / cox / hinge run on device tensors."""
+    import torch
+    from xgboost_ray_amd.engine.quantile import BinnedMatrix
+    from xgboost_ray_amd.engine.trainer import run_training
+
+    rng = np.random.RandomState(0)
+    n = 100000
+    X = rng.rand(n, 4).astype(np.float32)
+    lam = np.exp(1.0 + 2 * X[:, 0])
+    Xg = torch.from_numpy(X).cuda()
+
+    yp = rng.poisson(lam).astype(np.float32)
+    dmp = BinnedMatrix.build(
+        Xg, label=torch.from_numpy(yp).cuda(), max_bin=64
+    )
+    bp = run_training(
+        {"objective": "count:poisson", "max_depth": 4, "eta": 0.3}, dmp, 10
+    )
+    assert np.corrcoef(bp.predict(X[:5000]), lam[:5000])[0, 1] > 0.9
+
+    bt = run_training(
+        {"objective": "reg:tweedie", "max_depth": 4, "eta": 0.3}, dmp, 10
+    )
+    assert np.corrcoef(bt.predict(X[:5000]), lam[:5000])[0, 1] > 0.9
+
+    t = np.exp(1.0 + 2 * X[:, 0] + 0.3 * rng.randn(n)).astype(np.float32)
+    right = rng.rand(n) < 0.3
+    yl = t.copy()
+    yu = np.where(right, np.inf, t).astype(np.float32)
+    dma = BinnedMatrix.build(Xg, max_bin=64)
+    dma.label_lower_bound = torch.from_numpy(yl).cuda()
+    dma.label_upper_bound = torch.from_numpy(yu).cuda()
+    ba = run_training(
+        {"objective": "survival:aft", "max_depth": 4, "eta": 0.3,
+         "base_score": 1.0}, dma, 10,
+    )
+    pr = ba.predict(X[:5000], output_margin=True)
+    assert np.corrcoef(pr, 1 + 2 * X[:5000, 0])[0, 1] > 0.9
+
+    risk = 2.0 * X[:, 0]
+    tc = rng.exponential(1.0 / np.exp(risk)).astype(np.float32)
+    ev = rng.rand(n) < 0.7
+    lab = np.where(ev, tc, -tc).astype(np.float32)
+    lab[lab == 0] = 1e-6
+    dmc = BinnedMatrix.build(
+        Xg, label=torch.from_numpy(lab).cuda(), max_bin=64
+    )
+    bc = run_training(
+        {"objective": "survival:cox", "max_depth": 4, "eta": 0.3,
+         "base_score": 1.0}, dmc, 10,
+    )
+    pc = bc.predict(X[:5000], output_margin=True)
+    assert np.corrcoef(pc, risk[:5000])[0, 1] > 0.85
