@@ -951,3 +951,26 @@ class TestAFT:
              "base_score": 0.0}, dmb, 15,
         )
         assert float((bh.predict(X) == yb).mean()) > 0.9
+
+    def test_iteration_range_with_parallel_trees(self):
+        rng = np.random.RandomState(0)
+        X = rng.rand(2000, 4).astype(np.float32)
+        y = (X[:, 0] > 0.5).astype(np.float32)
+        dm = BinnedMatrix.build(
+            torch.from_numpy(X), label=torch.from_numpy(y), max_bin=32
+        )
+        b = run_training(
+            {"objective": "binary:logistic", "max_depth": 3,
+             "num_parallel_tree": 2}, dm, 4,
+        )
+        assert len(b.trees) == 8
+        from xgboost_ray_amd.engine.objectives import get_objective
+
+        bm = float(get_objective(b.objective, 0).prob_to_margin(
+            b.base_score
+        ))
+        m_all = b.predict(X[:100], output_margin=True)
+        m_a = b.predict(X[:100], output_margin=True, iteration_range=(0, 2))
+        m_b = b.predict(X[:100], output_margin=True, iteration_range=(2, 4))
+        # range margins are additive (each includes the base once)
+        assert np.abs((m_a - bm) + (m_b - bm) + bm - m_all).max() < 1e-5
