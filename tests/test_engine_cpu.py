@@ -974,3 +974,61 @@ class TestAFT:
         m_b = b.predict(X[:100], output_margin=True, iteration_range=(2, 4))
         # range margins are additive (each includes the base once)
         assert np.abs((m_a - bm) + (m_b - bm) + bm - m_all).max() < 1e-5
+
+
+class TestEvalSemantics:
+    """xgboost EvalTransform parity: metrics on transformed predictions,
+    and dart eval margins tracking the real (rescaled) model."""
+
+    def test_rmse_on_transformed_predictions(self):
+        """reg:logistic's default rmse is computed on probabilities, not
+        margins (XGBoost applies obj EvalTransform before metrics)."""
+        dm, X, y = _binned()
+        res = {}
+        bst = run_training(
+            {"objective": "reg:logistic", "max_depth": 4, "eta": 0.3,
+             "eval_metric": ["rmse", "mae"]},
+            dm, 8, evals=[EvalPack(name="train", X=None)], evals_result=res,
+        )
+        p = bst.predict(X)  # transformed (sigmoid) predictions
+        rmse = float(np.sqrt(np.mean((p - y) ** 2)))
+        mae = float(np.mean(np.abs(p - y)))
+        assert res["train"]["rmse"][-1] == pytest.approx(rmse, rel=1e-5)
+        assert res["train"]["mae"][-1] == pytest.approx(mae, rel=1e-5)
+        # sanity: probabilities live in [0,1] so rmse must be < 1
+        assert res["train"]["rmse"][-1] < 1.0
+
+    def test_rmse_identity_for_squarederror(self):
+        dm, X, y = _binned(kind="reg")
+        res = {}
+        bst = run_training(
+            {"objective": "reg:squarederror", "max_depth": 4, "eta": 0.3,
+             "eval_metric": ["rmse"]},
+            dm, 8, evals=[EvalPack(name="train", X=None)], evals_result=res,
+        )
+        p = bst.predict(X)
+        rmse = float(np.sqrt(np.mean((p - y) ** 2)))
+        assert res["train"]["rmse"][-1] == pytest.approx(rmse, rel=1e-5)
+
+    def test_dart_eval_margin_tracks_rescaled_model(self):
+        """_dart_commit rescales committed trees in place each round; the
+        eval-set margin must track the REAL model, not drift (advisor
+        finding: reported 0.181 vs true 0.237 logloss in 20 rounds)."""
+        dm, X, y = _binned()
+        Xe, ye = create_data(500, 6, seed=7, kind="binary")
+        res = {}
+        bst = run_training(
+            {"objective": "binary:logistic", "booster": "dart",
+             "max_depth": 4, "eta": 0.3, "rate_drop": 0.5,
+             "eval_metric": ["logloss"]},
+            dm, 20,
+            evals=[EvalPack(
+                name="eval", X=torch.from_numpy(Xe),
+                label=torch.from_numpy(ye))],
+            evals_result=res,
+        )
+        p = np.clip(bst.predict(Xe), 1e-16, 1 - 1e-16)
+        true_ll = float(np.mean(
+            -(ye * np.log(p) + (1 - ye) * np.log(1 - p))
+        ))
+        assert res["eval"]["logloss"][-1] == pytest.approx(true_ll, rel=1e-4)

@@ -192,3 +192,25 @@ def test_uid_identity(xy):
     assert a != b
     assert a == a
     assert len({a, b}) == 2
+
+
+def test_reshard_on_num_actors_change(xy):
+    """A loaded matrix reused with a different world size must re-shard
+    (reference matrix.py raises on the mismatch; we re-shard, which is
+    strictly safer than silently reusing stale shards)."""
+    X, y = xy
+    dm = RayDMatrix(X, y, num_actors=2)
+    assert dm.loaded and len(dm.refs) == 2
+    # shrink the world: all 100 rows must still be covered by 3 shards
+    dm.load_data(num_actors=3)
+    assert len(dm.refs) == 3
+    parts_x, parts_y = _gather_all(dm, 3)
+    assert sum(p.shape[0] for p in parts_x) == 100
+    np.testing.assert_array_equal(
+        np.sort(np.concatenate(parts_y)), np.sort(y)
+    )
+    # grow back: no KeyError, 2 shards again
+    dm.load_data(num_actors=2)
+    assert len(dm.refs) == 2
+    parts_x, _ = _gather_all(dm, 2)
+    assert sum(p.shape[0] for p in parts_x) == 100

@@ -82,7 +82,9 @@ class DMatrix:
 def _as_float32_matrix(data, missing=None):
     if data is None:
         return None
-    if hasattr(data, "values"):  # pandas
+    if isinstance(data, torch.Tensor):
+        data = data.detach().cpu().numpy()
+    elif hasattr(data, "values") and not callable(data.values):  # pandas
         data = data.values
     arr = np.ascontiguousarray(np.asarray(data), dtype=np.float32)
     if arr.ndim == 1:
@@ -96,7 +98,9 @@ def _as_float32_matrix(data, missing=None):
 def _as_float32_vec(v):
     if v is None:
         return None
-    if hasattr(v, "values"):
+    if isinstance(v, torch.Tensor):
+        v = v.detach().cpu().numpy()
+    elif hasattr(v, "values") and not callable(v.values):
         v = v.values
     return np.ascontiguousarray(np.asarray(v), dtype=np.float32).reshape(-1)
 

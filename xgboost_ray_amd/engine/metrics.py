@@ -34,11 +34,28 @@ def _w(label, weight):
     return weight.double()
 
 
+def _eval_pred(margin, obj):
+    """xgboost's EvalTransform semantics: metrics defined on predictions
+    (rmse/mae/mphe) receive the objective's transformed output — sigmoid for
+    logistic, exp for log-link objectives — before the metric is computed
+    (XGBoost learner.cc EvalOneIter -> obj_->EvalTransform). Falls back to
+    the raw margin when no objective is known or the transform changes the
+    shape (multiclass prob matrices, custom objectives)."""
+    p = margin.double()
+    if obj is None:
+        return p
+    try:
+        t = obj.transform_prediction(p)
+    except Exception:
+        return p
+    return t.double() if t.shape == p.shape else p
+
+
 class RMSE(Metric):
     name = "rmse"
 
     def local_stats(self, margin, label, weight, qid, obj):
-        pred = margin.double()
+        pred = _eval_pred(margin, obj)
         w = _w(label, weight)
         se = (w * (pred - label.double()) ** 2).sum()
         return torch.stack([se, w.sum()])
@@ -52,8 +69,9 @@ class MAE(Metric):
 
     def local_stats(self, margin, label, weight, qid, obj):
         w = _w(label, weight)
+        pred = _eval_pred(margin, obj)
         return torch.stack(
-            [(w * (margin.double() - label.double()).abs()).sum(), w.sum()]
+            [(w * (pred - label.double()).abs()).sum(), w.sum()]
         )
 
     def finalize(self, s):
@@ -305,7 +323,7 @@ class RMSLE(Metric):
     higher_better = False
 
     def local_stats(self, margin, label, weight, qid, obj):
-        p = torch.clamp(margin.double(), min=-1.0 + 1e-6)
+        p = torch.clamp(_eval_pred(margin, obj), min=-1.0 + 1e-6)
         d = torch.log1p(p) - torch.log1p(label.double())
         w = _w(label, weight)
         return torch.stack([(w * d * d).sum(), w.sum()])

@@ -118,7 +118,7 @@ class Logistic(Objective):
         return self._apply_weight(g, h, weight)
 
     def transform_prediction(self, margin):
-        return torch.sigmoid(margin)
+        return sigmoid_sizeinv(margin)
 
     def validate_label(self, label):
         if bool((label < 0).any()) or bool((label > 1).any()):
@@ -421,7 +421,7 @@ class Poisson(Objective):
         return self._apply_weight(g, h, weight)
 
     def transform_prediction(self, margin):
-        return torch.exp(margin)
+        return elemwise_sizeinv(torch.exp, margin)
 
     def validate_label(self, label):
         if bool((label < 0).any()):
@@ -444,7 +444,7 @@ class Gamma(Objective):
         return self._apply_weight(g, h, weight)
 
     def transform_prediction(self, margin):
-        return torch.exp(margin)
+        return elemwise_sizeinv(torch.exp, margin)
 
     def validate_label(self, label):
         if bool((label <= 0).any()):
@@ -473,7 +473,7 @@ class Tweedie(Objective):
         return self._apply_weight(g, h, weight)
 
     def transform_prediction(self, margin):
-        return torch.exp(margin)
+        return elemwise_sizeinv(torch.exp, margin)
 
     def validate_label(self, label):
         if bool((label < 0).any()):
@@ -681,7 +681,7 @@ class AFT(Objective):
         return self._apply_weight(g, h, weight)
 
     def transform_prediction(self, margin):
-        return torch.exp(margin)
+        return elemwise_sizeinv(torch.exp, margin)
 
     def validate_label(self, label):
         pass  # bounds validated at stack time
@@ -759,7 +759,7 @@ class Cox(Objective):
         return self._apply_weight(g.float(), h.float(), weight)
 
     def transform_prediction(self, margin):
-        return torch.exp(margin)
+        return elemwise_sizeinv(torch.exp, margin)
 
     def validate_label(self, label):
         if bool((label == 0).any()):

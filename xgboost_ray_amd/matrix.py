@@ -492,13 +492,15 @@ class RayDMatrix:
 
     def load_data(self, num_actors: Optional[int] = None, rank: Optional[int] = None):
         """Centrally materialize + shard the data (no-op if distributed)."""
-        if self.loaded:
-            return
         if num_actors is not None:
             if self.num_actors is not None and num_actors != self.num_actors:
-                # re-shard for a different world size
+                # re-shard for a different world size (reference matrix.py
+                # raises here; re-sharding is strictly more useful and keeps
+                # old-shard reuse impossible: unload frees the shm refs)
                 self.unload_data()
             self.num_actors = num_actors
+        if self.loaded:
+            return
         if self.num_actors is None:
             raise ValueError("num_actors must be set before load_data()")
         if not self.distributed:
