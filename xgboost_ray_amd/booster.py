@@ -15,6 +15,8 @@ from typing import Dict, List, Optional, Sequence, Union
 import numpy as np
 import torch
 
+from xgboost_ray_amd import ubjson
+
 
 @dataclass
 class Tree:
@@ -579,25 +581,45 @@ class Booster:
         self.__dict__.update(state)
 
     def save_model(self, fname: str):
-        doc = self._to_json_dict()
-        if str(fname).endswith(".ubj") or str(fname).endswith(".xgb"):
-            # best-effort: we always write the JSON document; XGBoost >= 1.6
-            # accepts JSON content regardless of extension when loading via
-            # load_model with schema sniffing.
-            pass
-        with open(fname, "w") as f:
-            json.dump(doc, f)
+        """Persist in the XGBoost model format the extension selects:
+        ``.json`` -> JSON, ``.ubj``/``.xgb`` -> UBJSON (the binary format
+        stock XGBoost >= 1.6 dispatches ``.ubj`` to with no content
+        sniffing; the reference's user contract saves ``model.xgb``,
+        reference README.md:78 and xgboost_ray/tune.py:130-156)."""
+        name = str(fname)
+        if name.endswith(".ubj") or name.endswith(".xgb"):
+            with open(fname, "wb") as f:
+                f.write(ubjson.dumps(self._to_json_dict()))
+        else:
+            with open(fname, "w") as f:
+                json.dump(self._to_json_dict(), f)
 
     def save_raw(self, raw_format: str = "json") -> bytes:
-        return json.dumps(self._to_json_dict()).encode()
+        if raw_format == "ubj":
+            return ubjson.dumps(self._to_json_dict())
+        if raw_format == "json":
+            return json.dumps(self._to_json_dict()).encode()
+        raise ValueError(
+            f"raw_format must be 'json' or 'ubj', got {raw_format!r}"
+        )
+
+    @staticmethod
+    def _parse_model_bytes(raw: bytes):
+        """Content-sniffed parse: both formats open with ``{`` (0x7b), but
+        in UBJSON the next byte is an integer-length marker for the first
+        key (or a ``$``/``#`` container header) while JSON follows with
+        whitespace, ``"`` or ``}``."""
+        if raw[:1] == b"{" and raw[1:2] in b"iUIlL$#N":
+            return ubjson.loads(raw)
+        return json.loads(raw.decode())
 
     def load_model(self, fname):
         if isinstance(fname, (bytes, bytearray)):
-            doc = json.loads(bytes(fname).decode())
+            raw = bytes(fname)
         else:
-            with open(fname) as f:
-                doc = json.load(f)
-        self._from_json_dict(doc)
+            with open(fname, "rb") as f:
+                raw = f.read()
+        self._from_json_dict(self._parse_model_bytes(raw))
         return self
 
     def _to_json_dict(self) -> Dict:
@@ -713,6 +735,11 @@ class Booster:
             self.tree_info = []
             self._flat_cache = None
             return
+        gbm_param = model.get("gbtree_model_param", {})
+        if "num_parallel_tree" in gbm_param:
+            self.params["num_parallel_tree"] = int(
+                gbm_param["num_parallel_tree"]
+            )
         self.tree_info = [int(x) for x in model.get("tree_info", [])]
         self.trees = []
         for tj in model["trees"]:
