@@ -156,6 +156,25 @@ class CallbackList:
                 stop |= bool(cb(booster, iteration, evals_log))
         return stop
 
+    def after_training(self, booster, iteration, evals_log):
+        """End-of-training hook. Accepts both this engine's 3-arg form
+        (booster, final_iteration, log) and xgboost's 1-arg
+        TrainingCallback.after_training(model) signature."""
+        import inspect
+
+        for cb in self.callbacks:
+            fn = getattr(cb, "after_training", None)
+            if fn is None:
+                continue
+            try:
+                n_params = len(inspect.signature(fn).parameters)
+            except (TypeError, ValueError):
+                n_params = 3
+            if n_params >= 3:
+                fn(booster, iteration, evals_log)
+            else:
+                fn(booster)
+
 
 class BoostingEngine:
     """Single-rank engine; collectives make it data-parallel."""
@@ -1515,4 +1534,5 @@ def run_training(
         )
     if evals_result is not None:
         evals_result.update(log)
+    cb.after_training(engine.booster, engine.iteration - 1, log)
     return engine.booster
