@@ -24,6 +24,7 @@ if "--cpu-only" not in sys.argv:
             os.path.join(src_dir, f)
             for f in sorted(os.listdir(src_dir))
             if f.endswith((".cpp", ".hip"))
+            and not f.endswith("_hip.hip")  # hipify build artifacts
         ]
         if sources:
             ext_modules = [

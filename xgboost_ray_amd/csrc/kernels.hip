@@ -19,7 +19,7 @@
 
 #include <hip/hip_runtime.h>
 #include <torch/extension.h>
-#include <ATen/cuda/CUDAContext.h>
+#include <c10/hip/HIPStream.h>
 
 #include <cstdint>
 #include <cstdlib>
@@ -34,6 +34,10 @@
   } while (0)
 
 static inline int64_t ceil_div(int64_t a, int64_t b) { return (a + b - 1) / b; }
+
+// torch spells the ROCm GPU device type with its legacy vendor name in the
+// public tensor API; isolate that single spelling here.
+static inline bool on_gpu(const torch::Tensor& t) { return t.is_cuda(); }
 
 // ---------------------------------------------------------------------------
 // quantize_gpair: [n,2] f32 -> [n,2] i64 (round to nearest)
@@ -1131,11 +1135,11 @@ static torch::Tensor cat_start_count(const torch::Tensor& starts_cpu,
 }
 
 torch::Tensor quantize_gpair(torch::Tensor gpair, double scale_g, double scale_h) {
-  TORCH_CHECK(gpair.is_cuda() && gpair.dtype() == torch::kFloat32);
+  TORCH_CHECK(on_gpu(gpair) && gpair.dtype() == torch::kFloat32);
   auto out = torch::empty_like(gpair, gpair.options().dtype(torch::kInt32));
   int64_t n = gpair.size(0);
   if (n == 0) return out;
-  auto stream = at::cuda::getCurrentCUDAStream();
+  auto stream = c10::hip::getCurrentHIPStream();
   int64_t blocks = std::min<int64_t>(ceil_div(n, 256), 8192);
   hipLaunchKernelGGL(quantize_gpair_kernel, dim3(blocks), dim3(256), 0,
                      stream.stream(), (const float2*)gpair.data_ptr<float>(),
@@ -1145,7 +1149,7 @@ torch::Tensor quantize_gpair(torch::Tensor gpair, double scale_g, double scale_h
 
 torch::Tensor bin_matrix(torch::Tensor values, torch::Tensor cuts_flat,
                          torch::Tensor cut_ptr) {
-  TORCH_CHECK(values.is_cuda() && values.dim() == 2);
+  TORCH_CHECK(on_gpu(values) && values.dim() == 2);
   int64_t n = values.size(0);
   int F = (int)values.size(1);
   // padded allocation: 16B-aligned row stride, pad bytes = 255 (missing)
@@ -1156,7 +1160,7 @@ torch::Tensor bin_matrix(torch::Tensor values, torch::Tensor cuts_flat,
   auto out = full.narrow(1, 0, F);
   if (n == 0) return out;
   int total_cuts = (int)cuts_flat.size(0);
-  auto stream = at::cuda::getCurrentCUDAStream();
+  auto stream = c10::hip::getCurrentHIPStream();
   int64_t blocks = std::min<int64_t>(ceil_div(n * F, 256), 16384);
   size_t lds = (size_t)total_cuts * sizeof(float);
   if (lds <= 64 * 1024) {
@@ -1213,7 +1217,7 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
   // builds features [f_lo, f_hi) into the caller-provided [K, F, n_bins, 2]
   // histogram (zeroed by the caller); ranges let the driver overlap the
   // RCCL AllReduce of one feature block with the build of the next.
-  TORCH_CHECK(bins.is_cuda() && bins.dtype() == torch::kUInt8);
+  TORCH_CHECK(on_gpu(bins) && bins.dtype() == torch::kUInt8);
   const int K = (int)starts.size(0);
   const int F = (int)bins.size(1);
   auto dev = bins.device();
@@ -1283,7 +1287,7 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
     max_end = std::max(max_end, starts_acc[k] + counts_acc[k]);
   }
   if (min_start == INT64_MAX) { min_start = 0; max_end = 0; }
-  auto stream = at::cuda::getCurrentCUDAStream();
+  auto stream = c10::hip::getCurrentHIPStream();
   int64_t span = max_end - min_start;
   TORCH_CHECK(gpair_q.dtype() == torch::kInt32,
               "gpair_q must be int32 packed pairs");
@@ -1439,7 +1443,7 @@ std::vector<torch::Tensor> find_splits(torch::Tensor hist, torch::Tensor parent_
   auto kf_dl = torch::empty({(int64_t)K * F}, optsb);
   auto kf_lg = torch::empty({(int64_t)K * F}, optsl);
   auto kf_lh = torch::empty({(int64_t)K * F}, optsl);
-  auto stream = at::cuda::getCurrentCUDAStream();
+  auto stream = c10::hip::getCurrentHIPStream();
   int64_t total = (int64_t)K * F;
   // 8 segments of 16 lanes per 128-thread block: each hardware wave
   // runs four independent (k, f) scans
@@ -1505,7 +1509,7 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
   if (total_chunks == 0) return {ridx_out, left_counts, gseg_out};
   // ONE H2D copy: [starts(K) | counts(K) | chunk_off(K+1) | feat(K) |
   // bin(K) | default_left(K)] - staged through a pinned buffer
-  auto stream0 = at::cuda::getCurrentCUDAStream();
+  auto stream0 = c10::hip::getCurrentHIPStream();
   static thread_local PinnedStager part_meta_stager;
   auto meta_cpu = part_meta_stager.get(5 * K + 1);
   torch::cat_out(meta_cpu, {starts_cpu, counts_cpu, chunk_off_cpu,
@@ -1521,7 +1525,7 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
   int64_t* sb_p = mp + 4 * K + 1;
   int64_t* dl_p = mp + 5 * K + 1;
 
-  auto stream = at::cuda::getCurrentCUDAStream();
+  auto stream = c10::hip::getCurrentHIPStream();
   auto block_counts = torch::empty({total_chunks},
       torch::TensorOptions().dtype(torch::kInt32).device(dev));
   auto flags = torch::empty({ridx.size(0)},
@@ -1576,7 +1580,7 @@ void predict_trees(torch::Tensor X, torch::Tensor feat, torch::Tensor thr,
   int F = (int)X.size(1);
   int T = (int)tree_ptr.size(0) - 1;
   if (n == 0 || T <= 0) return;
-  auto stream = at::cuda::getCurrentCUDAStream();
+  auto stream = c10::hip::getCurrentHIPStream();
   int64_t n_nodes = feat.size(0);
   auto packed = torch::empty({n_nodes, 4},
       torch::TensorOptions().dtype(torch::kInt32).device(X.device()));
@@ -1614,7 +1618,7 @@ void update_margins(torch::Tensor margin, torch::Tensor ridx,
     total_chunks = acc[K];
   }
   if (total_chunks == 0) return;
-  auto stream0 = at::cuda::getCurrentCUDAStream();
+  auto stream0 = c10::hip::getCurrentHIPStream();
   static thread_local PinnedStager margin_meta_stager;
   auto meta_cpu = margin_meta_stager.get(3 * K + 1);
   torch::cat_out(meta_cpu, {starts_cpu, counts_cpu, chunk_off_cpu});
@@ -1622,7 +1626,7 @@ void update_margins(torch::Tensor margin, torch::Tensor ridx,
   margin_meta_stager.mark(stream0.stream());
   int64_t* mp = meta.data_ptr<int64_t>();
   auto lv = leaf_vals.to(dev).to(torch::kFloat32);
-  auto stream = at::cuda::getCurrentCUDAStream();
+  auto stream = c10::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(update_margins_kernel, dim3((uint32_t)total_chunks),
                      dim3(PART_THREADS), 0, stream.stream(),
                      margin.data_ptr<float>(), ridx.data_ptr<int32_t>(),
@@ -1638,7 +1642,7 @@ torch::Tensor lambdarank_grad(torch::Tensor margin, torch::Tensor label,
   auto out = torch::zeros({n, 2},
       torch::TensorOptions().dtype(torch::kFloat32).device(margin.device()));
   if (n == 0 || G <= 0) return out;
-  auto stream = at::cuda::getCurrentCUDAStream();
+  auto stream = c10::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(lambdarank_kernel, dim3(G), dim3(256), 0,
                      stream.stream(), margin.data_ptr<float>(),
                      label.data_ptr<float>(), group_ptr.data_ptr<int64_t>(),
