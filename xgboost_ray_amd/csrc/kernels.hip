@@ -1191,7 +1191,7 @@ struct PinnedStager {
   bool pending = false;
   torch::Tensor get(int64_t n, torch::Dtype dt = torch::kInt64) {
     if (pending) {
-      hipEventSynchronize(ev);
+      (void)hipEventSynchronize(ev);
       pending = false;
     }
     if (!buf.defined() || buf.numel() < n || buf.scalar_type() != dt) {
@@ -1208,6 +1208,20 @@ struct PinnedStager {
     pending = true;
   }
 };
+
+// cat into a pinned stager view. The view's length MUST equal the summed
+// input length: torch::cat_out resizes a mismatched output, and the
+// resized allocation is pageable — silently defeating the pinned staging
+// (and flooding stderr with Resize.cpp warnings). Checked, not assumed.
+static void cat_into_pinned(torch::Tensor out,
+                            std::vector<torch::Tensor> parts) {
+  int64_t total = 0;
+  for (auto& p : parts) total += p.numel();
+  TORCH_CHECK(out.numel() == total, "pinned stager size mismatch: have ",
+              out.numel(), " need ", total);
+  TORCH_CHECK(out.is_pinned(), "stager buffer lost pinning");
+  torch::cat_out(out, parts);
+}
 
 torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                               torch::Tensor ridx, torch::Tensor starts,
@@ -1314,8 +1328,8 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
   // the training loop's dominant host cost.
   auto starts_adj = starts_cpu - min_start;
   static thread_local PinnedStager hist_meta_stager;
-  auto meta_cpu = hist_meta_stager.get(2 * K + K + 1);
-  torch::cat_out(meta_cpu, {starts_adj, counts_cpu, chunk_off_cpu});
+  auto meta_cpu = hist_meta_stager.get(3 * K + 1);
+  cat_into_pinned(meta_cpu, {starts_adj, counts_cpu, chunk_off_cpu});
   auto meta = meta_cpu.to(dev, /*non_blocking=*/true);
   hist_meta_stager.mark(stream.stream());
   int64_t* mp = meta.data_ptr<int64_t>();
@@ -1511,11 +1525,11 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
   // bin(K) | default_left(K)] - staged through a pinned buffer
   auto stream0 = c10::hip::getCurrentHIPStream();
   static thread_local PinnedStager part_meta_stager;
-  auto meta_cpu = part_meta_stager.get(5 * K + 1);
-  torch::cat_out(meta_cpu, {starts_cpu, counts_cpu, chunk_off_cpu,
-                            split_feat.to(torch::kCPU).to(torch::kInt64),
-                            split_bin.to(torch::kCPU).to(torch::kInt64),
-                            default_left.to(torch::kCPU).to(torch::kInt64)});
+  auto meta_cpu = part_meta_stager.get(6 * K + 1);
+  cat_into_pinned(meta_cpu, {starts_cpu, counts_cpu, chunk_off_cpu,
+                             split_feat.to(torch::kCPU).to(torch::kInt64),
+                             split_bin.to(torch::kCPU).to(torch::kInt64),
+                             default_left.to(torch::kCPU).to(torch::kInt64)});
   auto meta = meta_cpu.to(dev, /*non_blocking=*/true);
   part_meta_stager.mark(stream0.stream());
   int64_t* mp = meta.data_ptr<int64_t>();
@@ -1621,7 +1635,7 @@ void update_margins(torch::Tensor margin, torch::Tensor ridx,
   auto stream0 = c10::hip::getCurrentHIPStream();
   static thread_local PinnedStager margin_meta_stager;
   auto meta_cpu = margin_meta_stager.get(3 * K + 1);
-  torch::cat_out(meta_cpu, {starts_cpu, counts_cpu, chunk_off_cpu});
+  cat_into_pinned(meta_cpu, {starts_cpu, counts_cpu, chunk_off_cpu});
   auto meta = meta_cpu.to(dev, /*non_blocking=*/true);
   margin_meta_stager.mark(stream0.stream());
   int64_t* mp = meta.data_ptr<int64_t>();
