@@ -517,3 +517,17 @@ def test_ranking_distributed_equals_single():
                  ray_params=RayParams(num_actors=2))
     p1 = predict(b1, RayDMatrix(X), ray_params=RayParams(num_actors=1))
     assert np.array_equal(p1, p2)
+
+
+def test_staged_work_writes_back():
+    """_StagedWork.wait() must complete the host reduce then write the
+    result into the original tensor (gloo transport w/ GPU compute)."""
+    import torch
+
+    from xgboost_ray_amd.engine.collective import _StagedWork
+
+    dest = torch.zeros(4)
+    host = torch.arange(4, dtype=torch.float32)
+    w = _StagedWork(None, host, dest)
+    assert w.wait() is True
+    assert torch.equal(dest, host)

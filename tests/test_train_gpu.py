@@ -277,3 +277,64 @@ def test_extended_objectives_gpu():
     )
     pc = bc.predict(X[:5000], output_margin=True)
     assert np.corrcoef(pc, risk[:5000])[0, 1] > 0.85
+
+
+def _train_higgs_like(num_actors, rounds=8, env=None):
+    import os
+
+    X, y = create_data(120_000, 10)
+    old = {}
+    for k, v in (env or {}).items():
+        old[k] = os.environ.get(k)
+        os.environ[k] = v
+    try:
+        bst = train(
+            {"objective": "binary:logistic", "tree_method": "gpu_hist",
+             "max_depth": 6, "eta": 0.3},
+            RayDMatrix(X, label=y),
+            rounds,
+            ray_params=RayParams(num_actors=num_actors, gpus_per_actor=1,
+                                 max_actor_restarts=0),
+        )
+    finally:
+        for k, v in old.items():
+            if v is None:
+                os.environ.pop(k, None)
+            else:
+                os.environ[k] = v
+    return bst, X
+
+
+def test_multirank_gloo_staged_one_gpu():
+    """World-size-2 training with BOTH ranks on this MI355X: GPU compute,
+    host-staged gloo transport (RXGB_COLL_BACKEND=gloo) through the SAME
+    chunked/overlapped per-depth histogram allreduce path
+    (trainer.py _build_depth_histograms). The model must be bitwise
+    identical to single-actor training (distributed==single invariant;
+    reference's core collective semantics, reference README.md:345-349)."""
+    bst1, X = _train_higgs_like(1)
+    bst2, _ = _train_higgs_like(2, env={"RXGB_COLL_BACKEND": "gloo"})
+    np.testing.assert_array_equal(
+        bst1.predict(X, output_margin=True),
+        bst2.predict(X, output_margin=True),
+    )
+
+
+def test_multirank_rccl_one_gpu():
+    """Two RCCL ranks sharing one MI355X. RCCL may refuse duplicate
+    devices in a communicator; skip (with the refusal recorded) where it
+    does — the gloo-staged test above covers the trainer path either
+    way, and the torchrun bench path covers 1-GPU-per-rank RCCL."""
+    bst1, X = _train_higgs_like(1)
+    try:
+        bst2, _ = _train_higgs_like(2)
+    except Exception as e:  # noqa: BLE001
+        msg = repr(e)
+        if any(s in msg.lower() for s in
+               ("duplicate", "invalid", "nccl", "rccl", "actor")):
+            pytest.skip(f"RCCL refused 2 ranks on one GPU: {msg[:200]}")
+        raise
+    np.testing.assert_array_equal(
+        bst1.predict(X, output_margin=True),
+        bst2.predict(X, output_margin=True),
+    )

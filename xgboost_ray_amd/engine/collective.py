@@ -26,6 +26,23 @@ import torch
 import torch.distributed as dist
 
 
+class _StagedWork:
+    """Async-allreduce handle for host-staged gloo transport: wait()
+    completes the reduce on the host copy, then writes back to the
+    original GPU tensor (same .wait() surface as a dist.Work)."""
+
+    def __init__(self, work, host: torch.Tensor, dest: torch.Tensor):
+        self._work = work
+        self._host = host
+        self._dest = dest
+
+    def wait(self):
+        if self._work is not None:
+            self._work.wait()
+        self._dest.copy_(self._host)
+        return True
+
+
 class Collective:
     def __init__(
         self,
@@ -42,10 +59,18 @@ class Collective:
         self.device = device or torch.device("cpu")
         self._group = None
         self._initialized_here = False
+        self.backend = None
         if world_size <= 1:
             return
         if backend is None:
-            backend = "nccl" if self.device.type == "cuda" else "gloo"
+            # RXGB_COLL_BACKEND=gloo forces host-staged gloo collectives
+            # with GPU compute: used when several ranks share one MI355X
+            # (RCCL refuses duplicate devices in one communicator) and as
+            # a diagnostic mode; the trainer's chunked/overlapped
+            # allreduce path is identical, only the transport changes.
+            backend = os.environ.get("RXGB_COLL_BACKEND") or (
+                "nccl" if self.device.type == "cuda" else "gloo"
+            )
         self.backend = backend
         if dist.is_initialized():
             # reuse the outer process group (e.g. launched via torchrun)
@@ -72,11 +97,21 @@ class Collective:
         self._group = None
         self._initialized_here = False
 
+    def _stage_host(self, tensor: torch.Tensor) -> bool:
+        """True when the transport is gloo but the tensor lives on GPU:
+        collectives then run on a host copy, result copied back."""
+        return self.backend == "gloo" and tensor.is_cuda
+
     # -- collectives -------------------------------------------------------
     def allreduce_(self, tensor: torch.Tensor, op: str = "sum") -> torch.Tensor:
         if not self.is_distributed:
             return tensor
         ops = {"sum": dist.ReduceOp.SUM, "max": dist.ReduceOp.MAX, "min": dist.ReduceOp.MIN}
+        if self._stage_host(tensor):
+            host = tensor.cpu()
+            dist.all_reduce(host, op=ops[op], group=self._group)
+            tensor.copy_(host)
+            return tensor
         dist.all_reduce(tensor, op=ops[op], group=self._group)
         return tensor
 
@@ -85,10 +120,21 @@ class Collective:
         if not self.is_distributed:
             return None
         ops = {"sum": dist.ReduceOp.SUM, "max": dist.ReduceOp.MAX}
+        if self._stage_host(tensor):
+            host = tensor.cpu()
+            work = dist.all_reduce(
+                host, op=ops[op], group=self._group, async_op=True
+            )
+            return _StagedWork(work, host, tensor)
         return dist.all_reduce(tensor, op=ops[op], group=self._group, async_op=True)
 
     def broadcast_(self, tensor: torch.Tensor, src: int = 0) -> torch.Tensor:
         if not self.is_distributed:
+            return tensor
+        if self._stage_host(tensor):
+            host = tensor.cpu()
+            dist.broadcast(host, src=src, group=self._group)
+            tensor.copy_(host)
             return tensor
         dist.broadcast(tensor, src=src, group=self._group)
         return tensor
