@@ -136,6 +136,50 @@ class _Node:
     path_feats: tuple = ()  # split features on the root path (sorted)
 
 
+class _TreeArrays:
+    """Growable SoA node storage for depthwise growth (replaces python
+    lists of per-node appends; fields are written by vectorized numpy
+    fancy-indexing per depth)."""
+
+    __slots__ = ("feat", "thr", "left", "dl", "val", "gain", "cover",
+                 "parent", "n", "cap")
+    _FIELDS = (
+        ("feat", -1, np.int64), ("thr", 0.0, np.float64),
+        ("left", -1, np.int64), ("dl", 0, np.int64),
+        ("val", 0.0, np.float64), ("gain", 0.0, np.float64),
+        ("cover", 0.0, np.float64), ("parent", -1, np.int64),
+    )
+
+    def __init__(self, cap: int = 64):
+        self.cap = int(cap)
+        self.n = 1  # root preallocated (all defaults)
+        for name, fill, dt in self._FIELDS:
+            setattr(self, name, np.full(self.cap, fill, dt))
+
+    def ensure(self, need: int):
+        if need <= self.cap:
+            return
+        cap = self.cap
+        while cap < need:
+            cap *= 2
+        for name, fill, dt in self._FIELDS:
+            old = getattr(self, name)
+            new = np.full(cap, fill, dt)
+            new[: self.n] = old[: self.n]
+            setattr(self, name, new)
+        self.cap = cap
+
+
+def _calc_weight_vec(G, H, lam, alpha):
+    """Vectorized _calc_weight with the identical operation order (the
+    scalar path's results are part of the bitwise determinism contract)."""
+    denom = H + lam
+    if alpha > 0:
+        G = np.copysign(np.maximum(np.abs(G) - alpha, 0.0), G)
+    bad = denom <= 0
+    return np.where(bad, 0.0, -G / np.where(bad, 1.0, denom))
+
+
 class CallbackList:
     def __init__(self, callbacks):
         self.callbacks = list(callbacks or [])
@@ -693,68 +737,76 @@ class BoostingEngine:
             root_sum = gq[ridx.long()].sum(dim=0, dtype=torch.int64)
         if self.coll.is_distributed:
             self.coll.allreduce_(root_sum)
-        root = _Node(
-            nid=0,
-            depth=0,
-            start=0,
-            count=n_local,
-            sum_g=int(root_sum[0]),
-            sum_h=int(root_sum[1]),
-        )
 
-        # growing tree arrays (python lists; tree sizes are tiny)
-        feat_l, thr_l, sbin_l = [-1], [0.0], [-1]
-        left_l, dl_l, val_l, gain_l, cover_l = [-1], [0], [0.0], [0.0], [0.0]
-        parent_l = [-1]
-
+        # ---- SoA tree storage (grow-doubling numpy arrays; the previous
+        # python-list-per-field representation plus _Node objects cost
+        # ~8 ms/round of host bookkeeping at depth 12 / 8191 nodes)
+        ta = _TreeArrays(max(64, 2 ** min(self.p.max_depth + 2, 16)))
         cuts_flat_cpu = self.dtrain.cuts.cuts_flat.cpu().numpy()
         cut_ptr_cpu = self.dtrain.cuts.cut_ptr.cpu().numpy()
+        track_paths = self.interaction_sets is not None
+        mono_np = (
+            np.asarray(self.mono, dtype=np.int64)
+            if self.mono is not None else None
+        )
 
-        frontier: List[_Node] = [root]
-        prev_all_hist: Optional[torch.Tensor] = None  # [prev_frontier, F, B, 2]
+        # ---- SoA frontier: one numpy array per field. Children are
+        # always appended as (left, right) pairs, so siblings sit at
+        # positions 2i / 2i+1 and the build/derive pairing vectorizes.
+        fr_nid = np.zeros(1, np.int64)
+        fr_start = np.zeros(1, np.int64)
+        fr_count = np.array([n_local], np.int64)
+        fr_sumg = np.array([int(root_sum[0])], np.int64)
+        fr_sumh = np.array([int(root_sum[1])], np.int64)
+        fr_pslot = np.zeros(1, np.int64)
+        fr_wlo = np.full(1, -np.inf)
+        fr_whi = np.full(1, np.inf)
+        fr_paths = [()] if track_paths else None
+
+        import os as _os2
+
+        lam = float(self.p.reg_lambda)
+        alpha = float(self.p.reg_alpha)
+        prev_all_hist: Optional[torch.Tensor] = None
         for depth in range(self.p.max_depth):
-            if not frontier:
+            nF = fr_nid.size
+            if nF == 0:
                 break
-            # ---- build order: built (smaller) children first, derived
-            # siblings after, so the built block is a CONTIGUOUS leading
-            # slice (one AllReduce, batched subtraction for the rest)
-            build_nodes: List[_Node] = []
-            derive_big: List[_Node] = []
-            derive_parent_slot: List[int] = []
-            derive_sib_pos: List[int] = []
+            # ---- build order: per sibling pair build the globally
+            # smaller child (quantized hessian sum as the size proxy -
+            # identical on every rank; local counts are not), derive the
+            # sibling by subtraction. Built block = contiguous leading
+            # slice (one AllReduce, batched subtraction for the rest).
             if depth == 0:
-                build_nodes = [root]
+                order_idx = np.zeros(1, np.int64)
+                K = 1
+                derive_pslot = np.zeros(0, np.int64)
+                derive_sib_pos = np.zeros(0, np.int64)
             else:
-                i = 0
-                while i < len(frontier):
-                    a = frontier[i]
-                    b = frontier[i + 1] if i + 1 < len(frontier) else None
-                    if b is not None and a.parent_slot == b.parent_slot:
-                        # siblings: build the globally smaller one. Use the
-                        # quantized hessian sum as the size proxy - it is
-                        # identical on every rank (local counts are not).
-                        small, big = (a, b) if a.sum_h <= b.sum_h else (b, a)
-                        derive_parent_slot.append(big.parent_slot)
-                        derive_sib_pos.append(len(build_nodes))
-                        build_nodes.append(small)
-                        derive_big.append(big)
-                        i += 2
-                    else:
-                        build_nodes.append(a)
-                        i += 1
-            order_nodes: List[_Node] = build_nodes + derive_big
-            for pos, nd in enumerate(order_nodes):
-                nd.slot = pos
+                ia = np.arange(0, nF, 2)
+                ib = ia + 1
+                small_is_a = fr_sumh[ia] <= fr_sumh[ib]
+                idx_small = np.where(small_is_a, ia, ib)
+                idx_big = np.where(small_is_a, ib, ia)
+                order_idx = np.concatenate([idx_small, idx_big])
+                K = idx_small.size
+                derive_pslot = fr_pslot[idx_big]
+                derive_sib_pos = np.arange(K, dtype=np.int64)
 
-            K = len(build_nodes)
-            nF = len(frontier)
-            starts = torch.tensor(
-                [nd.start for nd in build_nodes], dtype=torch.int64
+            # scan-slot order: built block first, derived siblings after
+            nid_ord = fr_nid[order_idx]
+            start_ord = fr_start[order_idx]
+            count_ord = fr_count[order_idx]
+            sumg_ord = fr_sumg[order_idx]
+            sumh_ord = fr_sumh[order_idx]
+            wlo_ord = fr_wlo[order_idx]
+            whi_ord = fr_whi[order_idx]
+            paths_ord = (
+                [fr_paths[i] for i in order_idx] if track_paths else None
             )
-            counts = torch.tensor(
-                [nd.count for nd in build_nodes], dtype=torch.int64
-            )
-            import os as _os2
+
+            starts = torch.from_numpy(np.ascontiguousarray(start_ord[:K]))
+            counts = torch.from_numpy(np.ascontiguousarray(count_ord[:K]))
 
             F = self.dtrain.n_features
             all_hist = torch.empty(
@@ -800,18 +852,14 @@ class BoostingEngine:
                 if self.coll.is_distributed:
                     self.coll.allreduce_(hist)
             _tick("allreduce")
-            if derive_big:
+            if derive_sib_pos.size:
                 # sibling = parent - built, batched over all pairs
                 if self.device.type == "cuda":
-                    pslots = self._stage_i64(
-                        np.asarray(derive_parent_slot), "pslots"
-                    )
-                    spos = self._stage_i64(np.asarray(derive_sib_pos), "spos")
+                    pslots = self._stage_i64(derive_pslot, "pslots")
+                    spos = self._stage_i64(derive_sib_pos, "spos")
                 else:
-                    pslots = torch.tensor(
-                        derive_parent_slot, dtype=torch.int64
-                    )
-                    spos = torch.tensor(derive_sib_pos, dtype=torch.int64)
+                    pslots = torch.from_numpy(derive_pslot)
+                    spos = torch.from_numpy(derive_sib_pos)
                 torch.sub(
                     prev_all_hist.index_select(0, pslots),
                     all_hist.index_select(0, spos),
@@ -820,11 +868,7 @@ class BoostingEngine:
 
             # ---- split scan over the whole frontier (scan-slot order)
             # ONE H2D for both parent-sum vectors
-            psums_np = np.array(
-                [[nd.sum_g for nd in order_nodes],
-                 [nd.sum_h for nd in order_nodes]],
-                dtype=np.int64,
-            )
+            psums_np = np.ascontiguousarray(np.stack([sumg_ord, sumh_ord]))
             if self.device.type == "cuda":
                 psums = self._stage_i64(psums_np, "psums")
                 KK = psums_np.shape[1]
@@ -844,9 +888,10 @@ class BoostingEngine:
             _tick("stack")
             mono_bounds = None
             if self.mono is not None:
-                mono_bounds = torch.tensor(
-                    [[nd.w_lower, nd.w_upper] for nd in order_nodes],
-                    dtype=torch.float64,
+                mono_bounds = torch.from_numpy(
+                    np.ascontiguousarray(
+                        np.stack([wlo_ord, whi_ord], axis=1)
+                    )
                 )
             best = ops.find_splits(
                 all_hist,
@@ -861,9 +906,8 @@ class BoostingEngine:
                 self.p.min_child_weight,
                 monotone=self.mono,
                 bounds=mono_bounds,
-                allowed=self._allowed_mask(
-                    [nd.path_feats for nd in order_nodes]
-                ),
+                allowed=self._allowed_mask(paths_ord)
+                if track_paths else None,
             )
             _tick("scan")
             gain = best["gain"]
@@ -874,148 +918,141 @@ class BoostingEngine:
             blg = best["left_g"]
             blh = best["left_h"]
 
-            split_nodes: List[_Node] = []
-            sf, sb, sdl = [], [], []
-            children_meta = []
-            # vectorized split/leaf decision + threshold lookup for the
-            # whole frontier (the per-node Python loop was ~0.5 ms/round)
-            splits_ok = (
-                (gain > 0) & (bfeat >= 0) & np.isfinite(gain)
-            )
-            thr_all = np.zeros(len(order_nodes), dtype=np.float64)
-            okf = np.nonzero(splits_ok)[0]
-            if okf.size:
-                thr_all[okf] = cuts_flat_cpu[
-                    cut_ptr_cpu[bfeat[okf]] + bbin[okf]
-                ]
-            depth_leaves = []
-            for k, nd in enumerate(order_nodes):
-                if not splits_ok[k]:
-                    depth_leaves.append(nd)
-                    continue
-                f = int(bfeat[k])
-                b = int(bbin[k])
-                # split after bin b: left iff bin <= b; bin b covers
-                # [cut[b-1], cut[b]) so "bin <= b" <=> v < cut[b] = thr.
-                thr = float(thr_all[k])
-                lid = len(feat_l)
-                feat_l[nd.nid] = f
-                thr_l[nd.nid] = thr
-                sbin_l[nd.nid] = b
-                left_l[nd.nid] = lid
-                dl_l[nd.nid] = int(bdl[k])
-                gain_l[nd.nid] = float(gain[k])
-                cover_l[nd.nid] = float(nd.sum_h) / scale_h
-                feat_l.extend((-1, -1))
-                thr_l.extend((0.0, 0.0))
-                sbin_l.extend((-1, -1))
-                left_l.extend((-1, -1))
-                dl_l.extend((0, 0))
-                val_l.extend((0.0, 0.0))
-                gain_l.extend((0.0, 0.0))
-                cover_l.extend((0.0, 0.0))
-                parent_l.extend((nd.nid, nd.nid))
-                split_nodes.append(nd)
-                sf.append(f)
-                sb.append(b)
-                sdl.append(int(bdl[k]))
-                children_meta.append(
-                    (lid, int(blg[k]), int(blh[k]))
-                )
+            # ---- vectorized split/leaf decision for the whole frontier
+            splits_ok = (gain > 0) & (bfeat >= 0) & np.isfinite(gain)
+            okf = np.nonzero(splits_ok)[0]  # ascending == scan-slot order
+            leaf_idx = np.nonzero(~splits_ok)[0]
+            n_split = int(okf.size)
 
-            self._finalize_leaves_batch(
-                depth_leaves, val_l, cover_l, scale_h
-            )
-            if not split_nodes:
-                frontier = []
+            if leaf_idx.size:
+                self._finalize_leaves_soa(
+                    ta, nid_ord[leaf_idx], sumg_ord[leaf_idx],
+                    sumh_ord[leaf_idx], start_ord[leaf_idx],
+                    count_ord[leaf_idx], wlo_ord[leaf_idx],
+                    whi_ord[leaf_idx], scale_h,
+                )
+            if n_split == 0:
+                fr_nid = np.zeros(0, np.int64)
                 break
 
-            sstarts = torch.tensor(
-                [nd.start for nd in split_nodes], dtype=torch.int64
+            # child ids in scan-slot order (matches the sequential
+            # numbering the list-based builder produced)
+            base = ta.n
+            lids = base + 2 * np.arange(n_split, dtype=np.int64)
+            ta.ensure(base + 2 * n_split)
+            nids_ok = nid_ord[okf]
+            f_ok = bfeat[okf].astype(np.int64)
+            b_ok = bbin[okf].astype(np.int64)
+            # split after bin b: left iff bin <= b; bin b covers
+            # [cut[b-1], cut[b]) so "bin <= b" <=> v < cut[b] = thr.
+            thr_ok = cuts_flat_cpu[cut_ptr_cpu[f_ok] + b_ok].astype(
+                np.float64
             )
-            scounts = torch.tensor(
-                [nd.count for nd in split_nodes], dtype=torch.int64
-            )
+            ta.feat[nids_ok] = f_ok
+            ta.thr[nids_ok] = thr_ok
+            ta.left[nids_ok] = lids
+            ta.dl[nids_ok] = bdl[okf]
+            ta.gain[nids_ok] = gain[okf].astype(np.float64)
+            ta.cover[nids_ok] = sumh_ord[okf].astype(np.float64) / scale_h
+            ta.parent[lids] = nids_ok
+            ta.parent[lids + 1] = nids_ok
+            ta.n = base + 2 * n_split
+
+            sstarts = torch.from_numpy(np.ascontiguousarray(start_ord[okf]))
+            scounts = torch.from_numpy(np.ascontiguousarray(count_ord[okf]))
+            sf32 = bfeat[okf].astype(np.int32)
+            sb32 = bbin[okf].astype(np.int32)
+            sdl8 = bdl[okf].astype(np.uint8)
             _tick("tree_host")
             ridx, left_counts, gseg = ops.partition_rows(
                 self.dtrain.bins,
                 ridx,
                 sstarts,
                 scounts,
-                torch.tensor(sf, dtype=torch.int32),
-                torch.tensor(sb, dtype=torch.int32),
-                torch.tensor(sdl, dtype=torch.uint8),
+                torch.from_numpy(sf32),
+                torch.from_numpy(sb32),
+                torch.from_numpy(sdl8),
                 gpair_seg=gseg,
                 bins_t=getattr(self.dtrain, "bins_t", None),
             )
             _tick("partition")
 
-            new_frontier: List[_Node] = []
-            lc_list = left_counts.tolist()  # ONE conversion, not K tensor
-            # __getitem__ + int() pairs (those cost ~1 us each)
-            for k, nd in enumerate(split_nodes):
-                lid, lg, lh = children_meta[k]
-                lcount = lc_list[k]
-                child_path = ()
-                if self.interaction_sets is not None:
-                    child_path = tuple(
-                        sorted(set(nd.path_feats) | {int(sf[k])})
-                    )
-                lnode = _Node(
-                    nid=lid,
-                    depth=depth + 1,
-                    start=nd.start,
-                    count=lcount,
-                    sum_g=lg,
-                    sum_h=lh,
-                    parent_slot=nd.slot,
-                    w_lower=nd.w_lower,
-                    w_upper=nd.w_upper,
-                    path_feats=child_path,
-                )
-                rnode = _Node(
-                    nid=lid + 1,
-                    depth=depth + 1,
-                    start=nd.start + lcount,
-                    count=nd.count - lcount,
-                    sum_g=nd.sum_g - lg,
-                    sum_h=nd.sum_h - lh,
-                    parent_slot=nd.slot,
-                    w_lower=nd.w_lower,
-                    w_upper=nd.w_upper,
-                    path_feats=child_path,
-                )
-                if self.mono is not None:
-                    c = int(self.mono[sf[k]])
-                    if c != 0:
-                        wl = _clamp(
-                            _calc_weight(lg / scale_g, lh / scale_h,
-                                         self.p.reg_lambda, self.p.reg_alpha),
-                            nd.w_lower, nd.w_upper)
-                        wr = _clamp(
-                            _calc_weight((nd.sum_g - lg) / scale_g,
-                                         (nd.sum_h - lh) / scale_h,
-                                         self.p.reg_lambda, self.p.reg_alpha),
-                            nd.w_lower, nd.w_upper)
-                        mid = 0.5 * (wl + wr)
-                        if c > 0:
-                            lnode.w_upper = min(lnode.w_upper, mid)
-                            rnode.w_lower = max(rnode.w_lower, mid)
-                        else:
-                            lnode.w_lower = max(lnode.w_lower, mid)
-                            rnode.w_upper = min(rnode.w_upper, mid)
-                new_frontier.extend([lnode, rnode])
-            frontier = new_frontier
+            lc = left_counts.cpu().numpy().astype(np.int64)
+            l_start = start_ord[okf]
+            l_sumg = blg[okf]
+            l_sumh = blh[okf]
+            r_start = l_start + lc
+            r_count = count_ord[okf] - lc
+            r_sumg = sumg_ord[okf] - l_sumg
+            r_sumh = sumh_ord[okf] - l_sumh
+
+            # monotone child bounds (exact op order of the scalar path)
+            wlo_p = wlo_ord[okf]
+            whi_p = whi_ord[okf]
+            l_wlo, l_whi = wlo_p.copy(), whi_p.copy()
+            r_wlo, r_whi = wlo_p.copy(), whi_p.copy()
+            if mono_np is not None:
+                cvec = mono_np[sf32]
+                act = cvec != 0
+                if act.any():
+                    wl = np.maximum(wlo_p, np.minimum(
+                        whi_p, _calc_weight_vec(
+                            l_sumg / scale_g, l_sumh / scale_h, lam, alpha
+                        )))
+                    wr = np.maximum(wlo_p, np.minimum(
+                        whi_p, _calc_weight_vec(
+                            r_sumg / scale_g, r_sumh / scale_h, lam, alpha
+                        )))
+                    mid = 0.5 * (wl + wr)
+                    pos = act & (cvec > 0)
+                    neg = act & (cvec < 0)
+                    l_whi = np.where(pos, np.minimum(l_whi, mid), l_whi)
+                    r_wlo = np.where(pos, np.maximum(r_wlo, mid), r_wlo)
+                    l_wlo = np.where(neg, np.maximum(l_wlo, mid), l_wlo)
+                    r_whi = np.where(neg, np.minimum(r_whi, mid), r_whi)
+
+            n2 = 2 * n_split
+            fr_nid = np.empty(n2, np.int64)
+            fr_nid[0::2] = lids
+            fr_nid[1::2] = lids + 1
+            fr_start = np.empty(n2, np.int64)
+            fr_start[0::2] = l_start
+            fr_start[1::2] = r_start
+            fr_count = np.empty(n2, np.int64)
+            fr_count[0::2] = lc
+            fr_count[1::2] = r_count
+            fr_sumg = np.empty(n2, np.int64)
+            fr_sumg[0::2] = l_sumg
+            fr_sumg[1::2] = r_sumg
+            fr_sumh = np.empty(n2, np.int64)
+            fr_sumh[0::2] = l_sumh
+            fr_sumh[1::2] = r_sumh
+            fr_pslot = np.empty(n2, np.int64)
+            fr_pslot[0::2] = okf
+            fr_pslot[1::2] = okf
+            fr_wlo = np.empty(n2, np.float64)
+            fr_wlo[0::2] = l_wlo
+            fr_wlo[1::2] = r_wlo
+            fr_whi = np.empty(n2, np.float64)
+            fr_whi[0::2] = l_whi
+            fr_whi[1::2] = r_whi
+            if track_paths:
+                new_paths = []
+                for k_i, f_i in zip(okf.tolist(), sf32.tolist()):
+                    cp = tuple(sorted(set(paths_ord[k_i]) | {int(f_i)}))
+                    new_paths.append(cp)
+                    new_paths.append(cp)
+                fr_paths = new_paths
             prev_all_hist = all_hist
 
         # remaining frontier nodes (max depth reached) become leaves
-        self._finalize_leaves_batch(frontier, val_l, cover_l, scale_h)
-        # leaf margin update: walk all leaves via the segment structure -
-        # every row's final node is its segment's node. Collect leaf segs.
-        # (split nodes consumed their segments; leaves kept them)
-        # We track them via a second pass: recompute from the tree by
-        # replaying partitions is wasteful; instead we collected leaf info
-        # in _finalize_leaf.
+        if fr_nid.size:
+            self._finalize_leaves_soa(
+                ta, fr_nid, fr_sumg, fr_sumh, fr_start, fr_count,
+                fr_wlo, fr_whi, scale_h,
+            )
+        # leaf margin update: every row's final node is its segment's
+        # node; _finalize_leaves_soa collected (start, count, value) segs
         leaf_starts = [s for (s, c, v) in self._leaf_segs]
         leaf_counts = [c for (s, c, v) in self._leaf_segs]
         leaf_vals = [v for (s, c, v) in self._leaf_segs]
@@ -1033,22 +1070,50 @@ class BoostingEngine:
         if _prof:
             print("PROF", {k: round(v*1000, 2) for k, v in _t.items()}, flush=True)
 
+        n = ta.n
         tree = Tree(
-            feat=np.asarray(feat_l, np.int32),
-            thr=np.asarray(thr_l, np.float32),
-            left=np.asarray(left_l, np.int32),
-            default_left=np.asarray(dl_l, np.uint8),
-            value=np.asarray(val_l, np.float32),
-            gain=np.asarray(gain_l, np.float32),
-            cover=np.asarray(cover_l, np.float32),
-            parent=np.asarray(
-                [2147483647 if p < 0 else p for p in parent_l], np.int32
-            ),
+            feat=ta.feat[:n].astype(np.int32),
+            thr=ta.thr[:n].astype(np.float32),
+            left=ta.left[:n].astype(np.int32),
+            default_left=ta.dl[:n].astype(np.uint8),
+            value=ta.val[:n].astype(np.float32),
+            gain=ta.gain[:n].astype(np.float32),
+            cover=ta.cover[:n].astype(np.float32),
+            parent=np.where(
+                ta.parent[:n] < 0, 2147483647, ta.parent[:n]
+            ).astype(np.int32),
         )
         return tree
 
     _leaf_segs: List[Tuple[int, int, float]] = []
 
+
+    def _finalize_leaves_soa(self, ta, nids, sumg, sumh, starts, counts,
+                             wlo, whi, scale_h):
+        """SoA leaf finalization: same math/op order as
+        _finalize_leaves_batch, writing into _TreeArrays and collecting
+        (start, count, value) margin segments."""
+        if nids.size == 0:
+            return
+        G = sumg.astype(np.float64) / self._scale_g_cur
+        H = sumh.astype(np.float64) / scale_h
+        lam = self.p.reg_lambda
+        alpha = self.p.reg_alpha
+        denom = H + lam
+        Gs = G
+        if alpha > 0:
+            Gs = np.copysign(np.maximum(np.abs(G) - alpha, 0.0), G)
+        w = np.where(denom > 0, -Gs / np.where(denom > 0, denom, 1.0), 0.0)
+        if self.mono is not None:
+            w = np.clip(w, wlo, whi)
+        if self.p.max_delta_step > 0:
+            w = np.clip(w, -self.p.max_delta_step, self.p.max_delta_step)
+        v = self.p.eta * w
+        ta.val[nids] = v
+        ta.cover[nids] = H
+        self._leaf_segs.extend(
+            zip(starts.tolist(), counts.tolist(), v.tolist())
+        )
 
     def _finalize_leaves_batch(self, nodes, val_l, cover_l, scale_h):
         """Vectorized _finalize_leaf for a whole frontier (the per-node
