@@ -347,3 +347,63 @@ def test_monotone_training_gpu():
     grid[:, 2] = 0.5
     pred = bst.predict(grid, output_margin=True)
     assert (np.diff(pred) >= -1e-6).all()
+
+
+def test_build_histogram_direct_small_nodes(binned, monkeypatch):
+    """Tiny-node direct-to-global path must produce the EXACT histogram
+    the LDS route produces (int64 atomics, any order). Many small
+    segments force every node under the direct threshold."""
+    gpu = _gpu_ops()
+    Xt, yt, cuts, bins_cpu = binned
+    n = Xt.shape[0]
+    gp = torch.stack([yt - 0.3, torch.rand(n) + 0.1], dim=1).float()
+    gq = cpu_ops.quantize_gpair(gp, 2.0**28, 2.0**27)
+    rng = np.random.RandomState(3)
+    ridx = torch.from_numpy(rng.permutation(n).astype(np.int32))
+    # 512 random segments (~390 rows avg): most take the direct path,
+    # the larger tail stays on the LDS route - both compose into one hist
+    starts, counts = _segments(n, 512)
+    ref = cpu_ops.build_histogram(
+        bins_cpu, gq, ridx, starts, counts, cuts.max_bins
+    )
+    monkeypatch.setenv("RXGB_HIST_DIRECT_ROWS", "512")
+    out = gpu.build_histogram(
+        bins_cpu.cuda(), gq.cuda(), ridx.cuda(), starts, counts,
+        cuts.max_bins,
+    ).cpu()
+    torch.testing.assert_close(ref, out, rtol=0, atol=0)
+    # and disabled (LDS route) agrees too
+    monkeypatch.setenv("RXGB_HIST_DIRECT_ROWS", "0")
+    out2 = gpu.build_histogram(
+        bins_cpu.cuda(), gq.cuda(), ridx.cuda(), starts, counts,
+        cuts.max_bins,
+    ).cpu()
+    torch.testing.assert_close(ref, out2, rtol=0, atol=0)
+
+
+def test_deep_tree_direct_ab_bitwise():
+    """Depth-12 GPU training with the small-node direct path on vs off:
+    identical models (the path only changes atomics' destination)."""
+    import os
+
+    from xgboost_ray_amd.engine.quantile import BinnedMatrix
+    from xgboost_ray_amd.engine.trainer import run_training
+
+    X, y = create_data(300_000, 12, seed=5)
+    dm = BinnedMatrix.build(
+        torch.from_numpy(X).cuda(), label=torch.from_numpy(y).cuda(),
+        max_bin=256,
+    )
+    preds = {}
+    for thresh in ("0", "512"):
+        os.environ["RXGB_HIST_DIRECT_ROWS"] = thresh
+        try:
+            bst = run_training(
+                {"objective": "binary:logistic", "max_depth": 12,
+                 "eta": 0.3, "tree_method": "gpu_hist"},
+                dm, 4,
+            )
+            preds[thresh] = bst.predict(X, output_margin=True)
+        finally:
+            os.environ.pop("RXGB_HIST_DIRECT_ROWS", None)
+    np.testing.assert_array_equal(preds["0"], preds["512"])

@@ -207,6 +207,37 @@ __device__ __forceinline__ void hist_accum_row16(
   hist_accum_packed16(lds_hist, packed, n_bins, lane, gp);
 }
 
+// Tiny-node fast path: accumulate a node's rows straight into the
+// GLOBAL histogram. For a node with c rows the LDS route pays a
+// 16*n_bins*2-cell tile zero + scan per workgroup regardless of c; for
+// c*2*16 atomics < that cost (c below ~512 at 256 bins) direct global
+// int64 atomics are cheaper - and identically deterministic (integer
+// adds in any order). The caller zeroes hist, so partial adds compose.
+__device__ inline void hist_direct_rows(
+    const uint8_t* __restrict__ bins, const int2* __restrict__ gpair_seg,
+    const int32_t* __restrict__ ridx, long long* __restrict__ ghist_node,
+    int64_t seg_start, int64_t row_lo, int64_t row_hi, int64_t row_stride,
+    int f0, int fcount, int n_bins, int tid, int nthreads) {
+  for (int64_t i = row_lo + tid; i < row_hi; i += nthreads) {
+    const int64_t seg_i = seg_start + i;
+    const int2 gpi = gpair_seg[seg_i];
+    const uint64_t r = (uint32_t)ridx[seg_i];
+    const uint8_t* rowb = bins + r * row_stride + f0;
+    #pragma unroll 4
+    for (int f = 0; f < fcount; ++f) {
+      const int b = rowb[f];
+      if (b != 255) {
+        long long* cell =
+            &ghist_node[(((size_t)(f0 + f)) * n_bins + b) * 2];
+        atomicAdd((unsigned long long*)cell,
+                  (unsigned long long)(long long)gpi.x);
+        atomicAdd((unsigned long long*)cell + 1,
+                  (unsigned long long)(long long)gpi.y);
+      }
+    }
+  }
+}
+
 template <int VFB>  // 16: uint4 row loads; 8: uint2; 0: byte fallback
 __global__ __launch_bounds__(HIST_THREADS) void build_histogram_kernel(
     const uint8_t* __restrict__ bins,        // [n_rows_total, row_stride]
@@ -216,7 +247,7 @@ __global__ __launch_bounds__(HIST_THREADS) void build_histogram_kernel(
     const int64_t* __restrict__ chunk_off,   // [K+1] cumulative chunks
     long long* __restrict__ hist,            // [K, F, n_bins, 2]
     int K, int F, int n_bins, int fb_size, int64_t row_stride, int f_base,
-    int rows_per_wg) {
+    int rows_per_wg, int direct_rows) {
   // locate (node, chunk) from blockIdx.x via binary search on chunk_off
   int wg = blockIdx.x;
   int lo = 0, hi = K;
@@ -233,6 +264,19 @@ __global__ __launch_bounds__(HIST_THREADS) void build_histogram_kernel(
   const int fb = blockIdx.y;
   const int f0 = f_base + fb * fb_size;
   const int fcount = fb_size < (F - f0) ? fb_size : (F - f0);
+
+  {
+    const int64_t node_count_e = node_start[K + node];
+    if (node_count_e < (int64_t)direct_rows) {
+      int64_t row_hi_e = row_lo + rows_per_wg;
+      if (row_hi_e > node_count_e) row_hi_e = node_count_e;
+      hist_direct_rows(bins, gpair_seg, ridx,
+                       hist + ((size_t)node * F) * (size_t)n_bins * 2,
+                       seg_start, row_lo, row_hi_e, row_stride, f0, fcount,
+                       n_bins, threadIdx.x, blockDim.x);
+      return;
+    }
+  }
 
   extern __shared__ unsigned long long lds_hist[];  // [fb_size][n_bins][2]
   // VFB==16 uses the SoA g/h-plane layout (see hist_accum_row16): zero
@@ -437,7 +481,7 @@ __global__ __launch_bounds__(THREADS) void build_histogram_xcd_kernel(
     const int64_t* __restrict__ chunk_off,   // [K+1] cumulative chunks
     long long* __restrict__ hist,            // [K, F, n_bins, 2]
     int K, int F, int n_bins, int n_fb, int64_t row_stride, int f_base,
-    int rows_per_wg, int total_chunks) {
+    int rows_per_wg, int total_chunks, int direct_rows) {
   const int flat = blockIdx.x;
   const int octet = flat / (8 * n_fb);
   const int rem = flat % (8 * n_fb);
@@ -456,6 +500,19 @@ __global__ __launch_bounds__(THREADS) void build_histogram_xcd_kernel(
   const int64_t row_lo = chunk_in_node * (int64_t)rows_per_wg;
   const int f0 = f_base + fb * 16;
   const int fcount = 16 < (F - f0) ? 16 : (F - f0);
+
+  {
+    const int64_t node_count_e = node_start[K + node];
+    if (node_count_e < (int64_t)direct_rows) {
+      int64_t row_hi_e = row_lo + rows_per_wg;
+      if (row_hi_e > node_count_e) row_hi_e = node_count_e;
+      hist_direct_rows(bins, gpair_seg, ridx,
+                       hist + ((size_t)node * F) * (size_t)n_bins * 2,
+                       seg_start, row_lo, row_hi_e, row_stride, f0, fcount,
+                       n_bins, threadIdx.x, blockDim.x);
+      return;
+    }
+  }
 
   extern __shared__ unsigned long long lds_hist[];
   const int hplane = 16 * n_bins;
@@ -1276,6 +1333,12 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
   }
   // R=8 (4096 rows/WG) measured best at 100M x 200: finer chunks load-
   // balance deep depths better than R=16/32, and occupancy is LDS-bound.
+  // tiny-node threshold for the direct-to-global path (0 disables)
+  int direct_rows = 512;
+  if (const char* e = getenv("RXGB_HIST_DIRECT_ROWS")) {
+    int v = atoi(e);
+    if (v >= 0 && v <= 65536) direct_rows = v;
+  }
   int mfb_r = 8;
   if (const char* e = getenv("RXGB_HIST_MULTIFB_R")) {
     int v = atoi(e);
@@ -1375,7 +1438,7 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                          chunk_off_p,
                          reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
                          K, F, (int)n_bins, n_fb, row_stride, (int)f_lo,
-                         rows_per_wg, (int)total_chunks);
+                         rows_per_wg, (int)total_chunks, direct_rows);
     };
     if (xt == 512)
       launch_xcd(std::integral_constant<int, 512>{});
@@ -1410,7 +1473,7 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                        sc_adj_p, chunk_off_p,
                        reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
                        K, F, (int)n_bins, fb_size, row_stride, (int)f_lo,
-                       rows_per_wg);
+                       rows_per_wg, direct_rows);
   } else if (vec16) {
     hipLaunchKernelGGL((build_histogram_kernel<16>),
                        dim3((uint32_t)total_chunks, n_fb),
@@ -1421,7 +1484,7 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                        sc_adj_p, chunk_off_p,
                        reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
                        K, F, (int)n_bins, fb_size, row_stride, (int)f_lo,
-                       rows_per_wg);
+                       rows_per_wg, direct_rows);
   } else {
     hipLaunchKernelGGL((build_histogram_kernel<0>),
                        dim3((uint32_t)total_chunks, n_fb),
@@ -1432,7 +1495,7 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
                        sc_adj_p, chunk_off_p,
                        reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
                        K, F, (int)n_bins, fb_size, row_stride, (int)f_lo,
-                       rows_per_wg);
+                       rows_per_wg, direct_rows);
   }
   return hist;
 }
