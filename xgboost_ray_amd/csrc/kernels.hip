@@ -1333,8 +1333,13 @@ torch::Tensor build_histogram(torch::Tensor bins, torch::Tensor gpair_q,
   }
   // R=8 (4096 rows/WG) measured best at 100M x 200: finer chunks load-
   // balance deep depths better than R=16/32, and occupancy is LDS-bound.
-  // tiny-node threshold for the direct-to-global path (0 disables)
-  int direct_rows = 512;
+  // tiny-node threshold for the direct-to-global path. Default OFF:
+  // measured on HIGGS-11Mx28 depth 12, direct=512/1024/2048 gave
+  // 11.84/12.14/12.88 ms/round vs 11.65 with the LDS route - the
+  // skip-if-zero merge already makes empty tiles cheap, and the direct
+  // path's global-atomic latency is not hidden at 256-thread occupancy.
+  // Kept as an env knob for wider matrices / other shapes.
+  int direct_rows = 0;
   if (const char* e = getenv("RXGB_HIST_DIRECT_ROWS")) {
     int v = atoi(e);
     if (v >= 0 && v <= 65536) direct_rows = v;

@@ -840,6 +840,14 @@ def get_objective(
     if callable(name_or_fn):
         return CustomObjective(name_or_fn, num_class)
     name = name_or_fn or "reg:squarederror"
+    if name == "custom":
+        # a model TRAINED with a custom python objective: predictions are
+        # raw margins (xgboost semantics - no PredTransform is known), and
+        # base_score passes through untransformed.
+        o = Objective()
+        o.name = "custom"
+        o.n_class = num_class
+        return o
     if name == "reg:tweedie":
         return Tweedie(tweedie_variance_power)
     if name == "survival:aft":

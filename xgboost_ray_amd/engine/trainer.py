@@ -746,7 +746,7 @@ class BoostingEngine:
         cut_ptr_cpu = self.dtrain.cuts.cut_ptr.cpu().numpy()
         track_paths = self.interaction_sets is not None
         mono_np = (
-            np.asarray(self.mono, dtype=np.int64)
+            self.mono.cpu().numpy().astype(np.int64)
             if self.mono is not None else None
         )
 

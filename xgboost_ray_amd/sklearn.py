@@ -117,6 +117,28 @@ def _check_if_params_are_ray_dmatrix(X, sample_weight, base_margin, eval_set,
 class _RayXGBModel(RayXGBMixin):
     _estimator_type = "regressor"
 
+    def __sklearn_tags__(self):
+        """sklearn >= 1.6 tags protocol (BaseEstimator is not inherited
+        because its get_params cannot see through **kwargs estimators;
+        stock XGBoost implements this the same way)."""
+        from sklearn.utils._tags import (
+            ClassifierTags,
+            RegressorTags,
+            TargetTags,
+            default_tags,
+        )
+
+        tags = default_tags(self)
+        tags.estimator_type = self._estimator_type
+        tags.target_tags = TargetTags(required=True)
+        tags.input_tags.allow_nan = True
+        if self._estimator_type == "classifier":
+            tags.classifier_tags = ClassifierTags()
+        elif self._estimator_type == "regressor":
+            tags.regressor_tags = RegressorTags()
+        tags.non_deterministic = False
+        return tags
+
     def __init__(
         self,
         max_depth: Optional[int] = None,
