@@ -64,3 +64,82 @@ class TestAssignPartitions:
         total = sum(len(v) for v in out.values())
         assert total == 2
         assert max(len(v) for v in out.values()) == 1
+
+
+class _FakeFuture:
+    def __init__(self, key):
+        self.key = key
+
+
+class _FakeDaskClient:
+    """Mocked dask.distributed client with a multi-node who_has map."""
+
+    def __init__(self, key_to_workers):
+        self._who = key_to_workers
+
+    def persist(self, data):
+        return data
+
+    def who_has(self, futures):
+        return {f.key: self._who.get(f.key, ()) for f in futures}
+
+
+class _FakeDaskDF:
+    def __init__(self, n):
+        self.npartitions = n
+
+    def get_partition(self, i):
+        return ("part", i)
+
+
+def test_dask_ip_probe_multi_node(monkeypatch):
+    """get_ip_to_parts must map partitions to their dask-worker hosts
+    (reference dask.py:136-167 probes object locations the same way) and
+    the greedy assigner must then keep local partitions local."""
+    import sys
+    import types
+
+    from xgboost_ray_amd.data_sources import dask as dask_src
+
+    keymap = {
+        ("p", 0): ("tcp://10.0.0.1:4000",),
+        ("p", 1): ("tcp://10.0.0.2:4000",),
+        ("p", 2): ("tcp://10.0.0.1:4001",),
+        ("p", 3): ("tcp://10.0.0.2:4001",),
+    }
+    client = _FakeDaskClient(keymap)
+    fake_mod = types.ModuleType("dask.distributed")
+    fake_mod.default_client = lambda: client
+    fake_mod.futures_of = lambda persisted: [
+        _FakeFuture(("p", i)) for i in range(persisted.npartitions)
+    ]
+    fake_pkg = types.ModuleType("dask")
+    fake_pkg.distributed = fake_mod
+    monkeypatch.setitem(sys.modules, "dask", fake_pkg)
+    monkeypatch.setitem(sys.modules, "dask.distributed", fake_mod)
+
+    data = _FakeDaskDF(4)
+    actor_ips = {0: "10.0.0.1", 1: "10.0.0.2"}
+    ip_to_parts = dask_src.get_ip_to_parts(data, actor_ips)
+    assert set(ip_to_parts) == {"10.0.0.1", "10.0.0.2"}
+    assert ip_to_parts["10.0.0.1"] == [("part", 0), ("part", 2)]
+    assert ip_to_parts["10.0.0.2"] == [("part", 1), ("part", 3)]
+
+    from xgboost_ray_amd.data_sources._distributed import (
+        assign_partitions_to_actors,
+    )
+
+    assigned = assign_partitions_to_actors(ip_to_parts, actor_ips)
+    assert sorted(assigned[0]) == [("part", 0), ("part", 2)]
+    assert sorted(assigned[1]) == [("part", 1), ("part", 3)]
+
+
+def test_dask_ip_probe_local_fallback():
+    """No dask.distributed client: every partition maps to the first
+    actor's IP (single-process scheduler == process-local data)."""
+    from xgboost_ray_amd.data_sources import dask as dask_src
+
+    data = _FakeDaskDF(3)
+    out = dask_src.get_ip_to_parts(data, {0: "127.0.0.1", 1: "127.0.0.1"})
+    assert list(out) == ["127.0.0.1"]
+    assert len(out["127.0.0.1"]) == 3

@@ -60,11 +60,50 @@ class Dask(DataSource):
         data: Any, actors: Sequence
     ) -> Tuple[Any, Optional[Dict[int, Any]]]:
         actor_rank_ips = get_actor_rank_ips(actors)
-        # single-node deployment: all partitions are local to every actor
-        parts = [data.get_partition(i) for i in range(data.npartitions)]
-        ip = next(iter(actor_rank_ips.values()), "127.0.0.1")
-        return data, assign_partitions_to_actors({ip: parts}, actor_rank_ips)
+        ip_to_parts = get_ip_to_parts(data, actor_rank_ips)
+        return data, assign_partitions_to_actors(
+            ip_to_parts, actor_rank_ips
+        )
 
     @staticmethod
     def get_n(data: Any) -> int:
         return data.npartitions
+
+
+def get_ip_to_parts(data: Any, actor_rank_ips: Dict[int, str]) -> Dict:
+    """Partition -> host-IP map (reference dask.py:136-167 probes Ray's
+    object locations; here the probe asks dask.distributed directly).
+
+    - With a dask.distributed Client: ``client.who_has`` on the persisted
+      partition futures yields each partition's worker address, so a
+      multi-node dask cluster gets true locality-aware assignment.
+    - Without one (threaded/synchronous scheduler): every partition is
+      process-local, so they are all mapped to the first actor's IP (the
+      greedy assigner then splits them evenly).
+    """
+    parts = [data.get_partition(i) for i in range(data.npartitions)]
+    local_ip = next(iter(actor_rank_ips.values()), "127.0.0.1")
+    try:
+        from dask.distributed import default_client, futures_of
+
+        client = default_client()
+    except Exception:
+        return {local_ip: parts}
+    try:
+        persisted = client.persist(data)
+        futures = futures_of(persisted)
+        who = client.who_has(futures)
+        out: Dict[str, list] = {}
+        for i, fut in enumerate(futures):
+            workers = who.get(fut.key) or ()
+            # worker address "tcp://10.0.0.3:43211" -> "10.0.0.3"
+            ip = (
+                str(next(iter(workers))).rsplit(":", 1)[0].split("//")[-1]
+                if workers else local_ip
+            )
+            out.setdefault(ip, []).append(
+                persisted.get_partition(i)
+            )
+        return out
+    except Exception:
+        return {local_ip: parts}
