@@ -379,6 +379,13 @@ def _train(
 
         # collect results; re-raises actor exceptions
         results = {r: f.result() for r, f in train_futures.items()}
+        # final drain: queue items (last checkpoint, after_training
+        # reports, callback returns) ride the mp.Queue feeder thread, a
+        # DIFFERENT channel than the RPC pipe the futures completed on -
+        # items put before the actor returned can surface here AFTER the
+        # future resolves, and queue.empty() can read stale-True. Drain
+        # until the queue stays quiet.
+        _drain_queue_settled(state, callback_returns)
     except (ActorError, TrainingError, TrainingStoppedError,
             RayXGBoostActorAvailable) as err:
         # failure path (reference main.py:1302-1316)
@@ -393,14 +400,9 @@ def _train(
                 pass
             state.actors[rank] = None
             state.failed_actor_ranks.add(rank)
-        # drain remaining queue items (late checkpoints)
-        while not state.queue.empty():
-            try:
-                _handle_queue_item(
-                    state.queue.get_nowait(), state, callback_returns
-                )
-            except Exception:
-                break
+        # drain remaining queue items (late checkpoints) - settled drain:
+        # the dying rank-0's last checkpoint may still be in the feeder
+        _drain_queue_settled(state, callback_returns)
         if isinstance(err, RayXGBoostActorAvailable):
             raise
         raise ActorError(str(err)) from err
@@ -413,6 +415,37 @@ def _train(
     total_n = sum(res["train_n"] for res in results.values())
     state.additional_results["total_n"] = total_n
     return bst, evals_result, state.additional_results
+
+
+def _drain_queue_settled(
+    state: "_TrainingState",
+    callback_returns: Dict,
+    settle_s: float = 0.25,
+    max_s: float = 3.0,
+):
+    """Drain the actor queue until it stays empty for ``settle_s``.
+
+    mp.Queue delivery is asynchronous (feeder thread); a plain
+    empty()/get_nowait() loop can finish while items are still in
+    flight from an actor whose RPC already returned."""
+    import queue as _q
+
+    deadline = time.monotonic() + max_s
+    last_item = time.monotonic()
+    while time.monotonic() < deadline:
+        try:
+            item = state.queue.get(timeout=0.05)
+        except _q.Empty:
+            if time.monotonic() - last_item > settle_s:
+                return
+            continue
+        except Exception:
+            return
+        try:
+            _handle_queue_item(item, state, callback_returns)
+        except Exception:
+            pass
+        last_item = time.monotonic()
 
 
 def _wait_all_settled(train_futures: Dict[int, Future], timeout: float):
