@@ -1,6 +1,9 @@
 """Locality assignment unit tests with mocked partition->node maps
 (reference test_data_source.py:16-166 technique: no cluster needed)."""
 
+import numpy as np
+import pandas as pd
+
 from xgboost_ray_amd.data_sources._distributed import (
     assign_partitions_to_actors,
 )
@@ -143,3 +146,90 @@ def test_dask_ip_probe_local_fallback():
     out = dask_src.get_ip_to_parts(data, {0: "127.0.0.1", 1: "127.0.0.1"})
     assert list(out) == ["127.0.0.1"]
     assert len(out["127.0.0.1"]) == 3
+
+
+class _FakePartitioned:
+    """Object implementing the __partitioned__ dict protocol
+    (reference data_sources/partitioned.py:18-99)."""
+
+    def __init__(self, frames, locations):
+        self._frames = frames
+        self.__partitioned__ = {
+            "get": lambda objs: objs,
+            "shape": (sum(len(f) for f in frames), frames[0].shape[1]),
+            "partition_tiling": (len(frames), 1),
+            "partitions": {
+                (i, 0): {
+                    "start": (sum(len(f) for f in frames[:i]), 0),
+                    "shape": f.shape,
+                    "data": f,
+                    "location": [locations[i]],
+                }
+                for i, f in enumerate(frames)
+            },
+        }
+
+
+def _make_partitioned(n_parts=4, rows=50, locations=None):
+    rng = np.random.RandomState(0)
+    frames = []
+    for i in range(n_parts):
+        df = pd.DataFrame(
+            rng.randn(rows, 3).astype(np.float32), columns=["a", "b", "c"]
+        )
+        df["label"] = (df["a"] > 0).astype(np.float32)
+        frames.append(df)
+    locations = locations or ["127.0.0.1"] * n_parts
+    return _FakePartitioned(frames, locations), frames
+
+
+def test_partitioned_protocol_load():
+    from xgboost_ray_amd.data_sources.partitioned import Partitioned
+
+    data, frames = _make_partitioned()
+    assert Partitioned.is_data_type(data)
+    assert Partitioned.get_n(data) == 4
+    df = Partitioned.load_data(data)
+    assert len(df) == 200
+    pd.testing.assert_frame_equal(
+        df.iloc[:50].reset_index(drop=True), frames[0]
+    )
+
+
+def test_partitioned_locality_assignment():
+    from xgboost_ray_amd.data_sources.partitioned import Partitioned
+
+    data, frames = _make_partitioned(
+        locations=["10.0.0.1", "10.0.0.2", "10.0.0.1", "10.0.0.2"]
+    )
+
+    class _FakeActor:
+        def __init__(self, ip):
+            self._ip = ip
+
+        def ip(self):
+            return self._ip
+
+    actors = [_FakeActor("10.0.0.1"), _FakeActor("10.0.0.2")]
+    _, assigned = Partitioned.get_actor_shards(data, actors)
+    # each actor gets exactly its two co-located partitions
+    for rank, ip in ((0, "10.0.0.1"), (1, "10.0.0.2")):
+        got = assigned[rank]
+        assert len(got) == 2
+        want_idx = [i for i, loc in enumerate(
+            ["10.0.0.1", "10.0.0.2", "10.0.0.1", "10.0.0.2"]) if loc == ip]
+        for g, wi in zip(got, want_idx):
+            pd.testing.assert_frame_equal(g, frames[wi])
+
+
+def test_partitioned_end_to_end_train():
+    """A __partitioned__ object trains through RayDMatrix + train()."""
+    from xgboost_ray_amd import RayDMatrix, RayParams, train
+
+    data, _ = _make_partitioned(n_parts=4, rows=200)
+    dm = RayDMatrix(data, label="label")
+    bst = train(
+        {"objective": "binary:logistic", "max_depth": 3},
+        dm, 3, ray_params=RayParams(num_actors=2),
+    )
+    assert bst.num_boosted_rounds() == 3
