@@ -599,7 +599,6 @@ __device__ inline double calc_weight_d(double G, double H, double lam,
   return denom > 0.0 ? -G / denom : 0.0;
 }
 
-#define SCAN_SEG 8
 __global__ __launch_bounds__(256) void find_splits_kf_kernel(
     const long long* __restrict__ hist,  // [K, F, B, 2]
     const long long* __restrict__ parent_g, const long long* __restrict__ parent_h,
@@ -614,20 +613,18 @@ __global__ __launch_bounds__(256) void find_splits_kf_kernel(
     long long* __restrict__ out_lg,    // [K, F]
     long long* __restrict__ out_lh,    // [K, F]
     int K, int F, int B) {
-  // One SCAN_SEG-lane segment per (node, feature): each hardware wave
-  // carries 64/SCAN_SEG independent (k, f) scans, multiplying the
-  // interleaved f64 dependency chains per wave (the scan is
-  // chain-latency bound) with the FP order unchanged. The 32 lanes stage the histogram
-  // into LDS AND precompute the dequantized doubles in parallel (each
-  // element rounds independently, so parallel precompute is
-  // bitwise-identical to the CPU oracle's elementwise multiply). Then
-  // the two missing-value directions scan on lanes 0 and 1 of the half
-  // concurrently - the sequential dependent f64 chain was 52%
-  // issue-stall as a single-lane loop.
-  extern __shared__ long long lds_h[];  // per segment: [B*2 i64][B*2 f64]
-  const int wave = threadIdx.x / SCAN_SEG;  // scan-segment id in the block
-  const int lane = threadIdx.x % SCAN_SEG;
-  const int waves_per_block = blockDim.x / SCAN_SEG;
+  // ONE 64-lane wave per (node, feature). The per-bin left sums come
+  // from an INT64 inclusive wave-scan (lane-local serial prefix over a
+  // contiguous bin chunk + shfl_up scan of lane totals) - integer adds
+  // are exact under any order, so the scan reassociation cannot change a
+  // single bit. Each bin's gain is then an independent f64 expression of
+  // its exact prefix (mirrored 1:1 by the CPU oracle, which dequantizes
+  // the same int cumsum), so 64 lanes evaluate 4 bins each in parallel
+  // where the previous design ran a 255-step dependent f64 chain on 2
+  // lanes (52% issue-stall, and nearly histogram-sized total time).
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int waves_per_block = blockDim.x / WAVE;
   int64_t kf = (int64_t)blockIdx.x * waves_per_block + wave;
   if (kf >= (int64_t)K * F) return;
   const int k = (int)(kf / F);
@@ -647,93 +644,129 @@ __global__ __launch_bounds__(256) void find_splits_kf_kernel(
   const double inv_g = 1.0 / scale_g;
   const double inv_h = 1.0 / scale_h;
   const long long* gh = hist + ((size_t)k * F + f) * B * 2;
-  long long* h = lds_h + (size_t)wave * B * 4;
-  double* hd = reinterpret_cast<double*>(h + B * 2);
-  for (int i = lane; i < B * 2; i += SCAN_SEG) {
-    const long long v = gh[i];
-    h[i] = v;
-    hd[i] = (double)v * ((i & 1) ? inv_h : inv_g);
-  }
-  // wave-local LDS visibility: all lanes' stores above precede this wait
-  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-  if (lane >= 2) return;
   const int nb = feat_bins[f];
+
+  // lane-local inclusive prefixes over this lane's contiguous bin chunk
+  const int bpl = (B + WAVE - 1) / WAVE;  // <= 4 for B <= 256
+  long long lpg[4], lph[4];
+  long long gsum = 0, hsum = 0;
+  #pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    if (j < bpl) {
+      const int b = lane * bpl + j;
+      if (b < B) {
+        gsum += gh[b * 2];
+        hsum += gh[b * 2 + 1];
+      }
+      lpg[j] = gsum;
+      lph[j] = hsum;
+    }
+  }
+  // inclusive wave-scan of lane totals -> exclusive offset per lane
+  long long gscan = gsum, hscan = hsum;
+  #pragma unroll
+  for (int d = 1; d < WAVE; d <<= 1) {
+    const long long ug = __shfl_up(gscan, d, WAVE);
+    const long long uh = __shfl_up(hscan, d, WAVE);
+    if (lane >= d) {
+      gscan += ug;
+      hscan += uh;
+    }
+  }
+  const long long goff = gscan - gsum;   // exclusive prefix offset
+  const long long hoff = hscan - hsum;
+  const long long Gtot_q = __shfl(gscan, WAVE - 1, WAVE);
+  const long long Htot_q = __shfl(hscan, WAVE - 1, WAVE);
 
   const double Gp = (double)parent_g[k] * inv_g;
   const double Hp = (double)parent_h[k] * inv_h;
   const double parent_score = calc_score(Gp, Hp, lam, alpha);
-
-  // feature totals (same order as cumsum's last element); computed
-  // redundantly by both lanes in lockstep - cheaper than a broadcast
-  double Gtot = 0.0, Htot = 0.0;
-  long long Gtot_q = 0, Htot_q = 0;
-  for (int b = 0; b < B; ++b) {
-    Gtot += hd[b * 2];
-    Htot += hd[b * 2 + 1];
-    Gtot_q += h[b * 2];
-    Htot_q += h[b * 2 + 1];
-  }
-  const double Gmiss = Gp - Gtot;
-  const double Hmiss = Hp - Htot;
+  const double Gmiss = Gp - (double)Gtot_q * inv_g;
+  const double Hmiss = Hp - (double)Htot_q * inv_h;
   const long long Gmiss_q = parent_g[k] - Gtot_q;
   const long long Hmiss_q = parent_h[k] - Htot_q;
 
-  // lane 0 scans default_left=1, lane 1 scans default_left=0
-  const int dl = 1 - lane;
-  SplitCand best = {-1.0, 0, dl, 0, 0};
-  {
-    double GL = 0.0, HL = 0.0;
-    long long GLq = 0, HLq = 0;
-    for (int b = 0; b < nb - 1 && b < B; ++b) {
-      GL += hd[b * 2];
-      HL += hd[b * 2 + 1];
-      GLq += h[b * 2];
-      HLq += h[b * 2 + 1];
-      double gl = dl ? GL + Gmiss : GL;
-      double hl = dl ? HL + Hmiss : HL;
-      double gr = Gp - gl, hr = Hp - hl;
+  const int c = (mono != nullptr) ? mono[f] : 0;
+  double blo = 0.0, bup = 0.0;
+  if (c != 0) {
+    blo = bounds[(size_t)k * 2];
+    bup = bounds[(size_t)k * 2 + 1];
+  }
+
+  // candidate ordering = CPU oracle order: gain desc, then dl=1 before
+  // dl=0 (strictly-greater replacement across the two passes), then
+  // lowest bin (ascending scan, strictly-greater replacement)
+  SplitCand best = {-1.0, 0x7fffffff, -1, 0, 0};
+  #pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    if (j >= bpl) break;
+    const int b = lane * bpl + j;
+    if (b >= nb - 1 || b >= B) continue;
+    const long long GLq = goff + lpg[j];
+    const long long HLq = hoff + lph[j];
+    const double GLd = (double)GLq * inv_g;
+    const double HLd = (double)HLq * inv_h;
+    #pragma unroll
+    for (int dl = 1; dl >= 0; --dl) {
+      const double gl = dl ? GLd + Gmiss : GLd;
+      const double hl = dl ? HLd + Hmiss : HLd;
+      const double gr = Gp - gl, hr = Hp - hl;
       if (hl < mcw || hr < mcw) continue;
-      if (mono != nullptr) {
-        const int c = mono[f];
-        if (c != 0) {
-          const double blo = bounds[(size_t)k * 2];
-          const double bup = bounds[(size_t)k * 2 + 1];
-          double wl = calc_weight_d(gl, hl, lam, alpha);
-          double wr = calc_weight_d(gr, hr, lam, alpha);
-          wl = fmin(fmax(wl, blo), bup);
-          wr = fmin(fmax(wr, blo), bup);
-          if (c > 0 ? (wl > wr) : (wl < wr)) continue;
-        }
+      if (c != 0) {
+        double wl = calc_weight_d(gl, hl, lam, alpha);
+        double wr = calc_weight_d(gr, hr, lam, alpha);
+        wl = fmin(fmax(wl, blo), bup);
+        wr = fmin(fmax(wr, blo), bup);
+        if (c > 0 ? (wl > wr) : (wl < wr)) continue;
       }
-      double gain = 0.5 * (calc_score(gl, hl, lam, alpha) +
-                           calc_score(gr, hr, lam, alpha) - parent_score) -
-                    gamma;
-      if (gain > best.gain) {
+      const double gain = 0.5 * (calc_score(gl, hl, lam, alpha) +
+                                 calc_score(gr, hr, lam, alpha) -
+                                 parent_score) -
+                          gamma;
+      const bool take =
+          (gain > best.gain) ||
+          (gain == best.gain &&
+           (dl > best.default_left ||
+            (dl == best.default_left && b < best.bin)));
+      if (take) {
         best.gain = gain;
         best.bin = b;
+        best.default_left = dl;
         best.left_g = dl ? GLq + Gmiss_q : GLq;
         best.left_h = dl ? HLq + Hmiss_q : HLq;
       }
     }
   }
-  // merge with the CPU oracle's preference: dl=0 wins only on a STRICTLY
-  // greater gain (dl=1 was evaluated first there)
-  const double g1 = __shfl(best.gain, 1, SCAN_SEG);
-  const int b1 = __shfl(best.bin, 1, SCAN_SEG);
-  const long long lg1 = __shfl(best.left_g, 1, SCAN_SEG);
-  const long long lh1 = __shfl(best.left_h, 1, SCAN_SEG);
-  if (lane != 0) return;
-  if (g1 > best.gain) {
-    best.gain = g1;
-    best.bin = b1;
-    best.default_left = 0;
-    best.left_g = lg1;
-    best.left_h = lh1;
-  } else {
-    best.default_left = 1;
+
+  // wave argmax reduction with the same comparator
+  #pragma unroll
+  for (int off = WAVE / 2; off > 0; off >>= 1) {
+    const double og = __shfl_down(best.gain, off, WAVE);
+    const int ob = __shfl_down(best.bin, off, WAVE);
+    const int odl = __shfl_down(best.default_left, off, WAVE);
+    const long long olg = __shfl_down(best.left_g, off, WAVE);
+    const long long olh = __shfl_down(best.left_h, off, WAVE);
+    const bool take =
+        (og > best.gain) ||
+        (og == best.gain &&
+         (odl > best.default_left ||
+          (odl == best.default_left && ob < best.bin)));
+    if (take) {
+      best.gain = og;
+      best.bin = ob;
+      best.default_left = odl;
+      best.left_g = olg;
+      best.left_h = olh;
+    }
   }
-  // no-split sentinel keeps the CPU contract (gain -1, dl 0)
-  if (best.gain <= -1.0) best.default_left = 0;
+  if (lane != 0) return;
+  if (best.gain <= -1.0) {
+    // no-split sentinel keeps the CPU contract (gain -1, dl 0, bin 0)
+    best.bin = 0;
+    best.default_left = 0;
+    best.left_g = 0;
+    best.left_h = 0;
+  }
   out_gain[kf] = best.gain;
   out_bin[kf] = best.bin;
   out_dl[kf] = (uint8_t)best.default_left;
@@ -1527,13 +1560,11 @@ std::vector<torch::Tensor> find_splits(torch::Tensor hist, torch::Tensor parent_
   auto kf_lh = torch::empty({(int64_t)K * F}, optsl);
   auto stream = c10::hip::getCurrentHIPStream();
   int64_t total = (int64_t)K * F;
-  // 8 segments of 16 lanes per 128-thread block: each hardware wave
-  // runs four independent (k, f) scans
-  const int waves_per_block = 8;
-  const size_t scan_lds = (size_t)waves_per_block * B * 4 * sizeof(long long);
+  // one 64-lane wave per (k, f), four waves per 256-thread block; no LDS
+  const int waves_per_block = 4;
   hipLaunchKernelGGL(find_splits_kf_kernel,
                      dim3((uint32_t)ceil_div(total, waves_per_block)),
-                     dim3(waves_per_block * SCAN_SEG), scan_lds,
+                     dim3(waves_per_block * WAVE), 0,
                      stream.stream(),
                      reinterpret_cast<const long long*>(hist.data_ptr<int64_t>()),
                      reinterpret_cast<const long long*>(parent_g.data_ptr<int64_t>()),

@@ -124,15 +124,18 @@ def find_splits(
     # mirrors this exact op order.
     inv_g = 1.0 / scale_g
     inv_h = 1.0 / scale_h
-    g = hist[..., 0].double() * inv_g  # [K, F, B]
-    h = hist[..., 1].double() * inv_h
     gq = hist[..., 0]
     hq = hist[..., 1]
 
-    GL = torch.cumsum(g, dim=2)
-    HL = torch.cumsum(h, dim=2)
+    # Left-sum prefixes are computed in INT64 (exact under any summation
+    # order) and dequantized once per bin: every bin's gain is then
+    # independent of its neighbours, which is what lets the GPU scan run
+    # one 64-lane wave per (node, feature) with a parallel int wave-scan
+    # and still be bitwise-identical to this oracle.
     GLq = torch.cumsum(gq, dim=2)
     HLq = torch.cumsum(hq, dim=2)
+    GL = GLq.double() * inv_g
+    HL = HLq.double() * inv_h
     Gp = (parent_g.double() * inv_g).view(K, 1, 1)
     Hp = (parent_h.double() * inv_h).view(K, 1, 1)
     # Missing mass per (node, feature) = parent - feature total.
