@@ -1738,7 +1738,13 @@ void update_margins(torch::Tensor margin, torch::Tensor ridx,
   auto meta = meta_cpu.to(dev, /*non_blocking=*/true);
   margin_meta_stager.mark(stream0.stream());
   int64_t* mp = meta.data_ptr<int64_t>();
-  auto lv = leaf_vals.to(dev).to(torch::kFloat32);
+  // leaf values ride their own pinned stager (fp32; the int64 meta
+  // stager cannot carry them without a bitcast)
+  static thread_local PinnedStager margin_lv_stager;
+  auto lv_cpu = margin_lv_stager.get(K, torch::kFloat32);
+  lv_cpu.copy_(leaf_vals.to(torch::kFloat32));
+  auto lv = lv_cpu.to(dev, /*non_blocking=*/true);
+  margin_lv_stager.mark(stream0.stream());
   auto stream = c10::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(update_margins_kernel, dim3((uint32_t)total_chunks),
                      dim3(PART_THREADS), 0, stream.stream(),

@@ -163,10 +163,12 @@ def lambdarank_grad(margin, label, group_ptr, rank, idcg, use_ndcg):
 
 
 def update_margins(margin, ridx, starts, counts, leaf_values):
-    dev = margin.device
-    lv = torch.as_tensor(leaf_values, dtype=torch.float32, device=dev)
+    # starts/counts/leaf_values stay on the HOST: the extension stages
+    # all control data through ONE pinned H2D (a .to(dev) here forced a
+    # pointless H2D + synchronizing D2H round-trip per round)
+    lv = torch.as_tensor(leaf_values, dtype=torch.float32)
     target = margin if margin.is_contiguous() else margin.contiguous()
-    _load().update_margins(target, ridx, starts.to(dev), counts.to(dev), lv)
+    _load().update_margins(target, ridx, starts, counts, lv)
     if target is not margin:
         margin.copy_(target)
     return margin
