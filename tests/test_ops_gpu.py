@@ -407,3 +407,40 @@ def test_deep_tree_direct_ab_bitwise():
         finally:
             os.environ.pop("RXGB_HIST_DIRECT_ROWS", None)
     np.testing.assert_array_equal(preds["0"], preds["512"])
+
+
+@pytest.mark.parametrize("mode,spw,weighted", [
+    (1, 1.0, False), (1, 3.0, True), (0, 1.0, False), (0, 1.0, True),
+])
+def test_grad_fused_matches_torch_composition(mode, spw, weighted):
+    """The fused gradient kernel must be BITWISE equal to the torch GPU
+    composition it replaces (same f32 op order), and its absmax must
+    equal abs().max()."""
+    import os
+
+    from xgboost_ray_amd import ops
+    from xgboost_ray_amd.engine.objectives import Logistic, SquaredError
+
+    torch.manual_seed(0)
+    n = 1_000_003  # odd size: exercises the grid-stride tail
+    margin = (torch.randn(n, device="cuda") * 4).float()
+    label = (torch.rand(n, device="cuda") < 0.4).float()
+    weight = (
+        (torch.rand(n, device="cuda") * 2 + 0.1).float()
+        if weighted else None
+    )
+    obj = Logistic(spw) if mode == 1 else SquaredError()
+    fused = ops.grad_fused(margin, label, weight, spw, mode)
+    assert fused is not None
+    gp_f, mx = fused
+    os.environ["RXGB_FUSED_GRAD"] = "0"
+    try:
+        gp_t = obj.gradients(margin, label, weight)
+    finally:
+        os.environ.pop("RXGB_FUSED_GRAD", None)
+    torch.testing.assert_close(gp_f, gp_t, rtol=0, atol=0)
+    torch.testing.assert_close(
+        mx.cpu(),
+        torch.stack([gp_t[:, 0].abs().max(), gp_t[:, 1].abs().max()]).cpu(),
+        rtol=0, atol=0,
+    )

@@ -1771,8 +1771,105 @@ torch::Tensor lambdarank_grad(torch::Tensor margin, torch::Tensor label,
   return out;
 }
 
+
+// ---------------------------------------------------------------------------
+// grad_fused: objective gradient + hessian + |g|/|h| max in ONE pass.
+// Replaces the 5-kernel torch chain (sigmoid, sub, mul, clamp, stack)
+// plus the two abs().max() reduction passes for the hot objectives -
+// the margin/label are read once and the fp32 gpair written once.
+// The f32 op ORDER replicates torch's eager kernels exactly
+// (p = 1/(1+expf(-x)); h = fmaxf(p*(1-p), eps); per-factor multiplies in
+// the same sequence), so models stay bitwise-identical to the CPU
+// oracle. The max reduction is order-free (fmax is associative).
+// MODE 0: reg:squarederror (g = m - y, h = 1)
+// MODE 1: binary:logistic  (g = p - y, h = max(p(1-p), 1e-16))
+// ---------------------------------------------------------------------------
+template <int MODE>
+__global__ __launch_bounds__(256) void grad_fused_kernel(
+    const float* __restrict__ margin, const float* __restrict__ label,
+    const float* __restrict__ weight,  // nullable
+    float spw, float2* __restrict__ gpair,
+    unsigned int* __restrict__ absmax,  // [2] f32 bits (monotone for >=0)
+    int64_t n) {
+  float gmax = 0.0f, hmax = 0.0f;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    const float x = margin[i];
+    const float y = label[i];
+    float g, h;
+    if constexpr (MODE == 1) {
+      const float p = 1.0f / (1.0f + expf(-x));
+      g = p - y;
+      h = fmaxf(p * (1.0f - p), 1e-16f);
+      if (spw != 1.0f) {
+        const float w = 1.0f + (spw - 1.0f) * y;
+        g *= w;
+        h *= w;
+      }
+    } else {
+      g = x - y;
+      h = 1.0f;
+    }
+    if (weight != nullptr) {
+      const float w = weight[i];
+      g *= w;
+      h *= w;
+    }
+    gpair[i] = {g, h};
+    gmax = fmaxf(gmax, fabsf(g));
+    hmax = fmaxf(hmax, fabsf(h));
+  }
+  // wave reduce, then one atomic per wave (non-negative f32 bits are
+  // monotone as uints)
+  #pragma unroll
+  for (int off = WAVE / 2; off > 0; off >>= 1) {
+    gmax = fmaxf(gmax, __shfl_down(gmax, off, WAVE));
+    hmax = fmaxf(hmax, __shfl_down(hmax, off, WAVE));
+  }
+  if ((threadIdx.x & (WAVE - 1)) == 0) {
+    atomicMax(&absmax[0], __float_as_uint(gmax));
+    atomicMax(&absmax[1], __float_as_uint(hmax));
+  }
+}
+
+std::vector<torch::Tensor> grad_fused(torch::Tensor margin,
+                                      torch::Tensor label,
+                                      torch::Tensor weight, double spw,
+                                      int64_t mode) {
+  TORCH_CHECK(on_gpu(margin) && margin.dtype() == torch::kFloat32);
+  const int64_t n = margin.numel();
+  auto dev = margin.device();
+  auto gpair = torch::empty(
+      {n, 2}, torch::TensorOptions().dtype(torch::kFloat32).device(dev));
+  auto absmax = torch::zeros(
+      {2}, torch::TensorOptions().dtype(torch::kFloat32).device(dev));
+  if (n == 0) return {gpair, absmax};
+  auto stream = c10::hip::getCurrentHIPStream();
+  const int64_t blocks = std::min<int64_t>(ceil_div(n, 256 * 8), 8192);
+  const float* wp =
+      weight.numel() ? weight.data_ptr<float>() : nullptr;
+  if (mode == 1) {
+    hipLaunchKernelGGL((grad_fused_kernel<1>), dim3((uint32_t)blocks),
+                       dim3(256), 0, stream.stream(),
+                       margin.data_ptr<float>(), label.data_ptr<float>(),
+                       wp, (float)spw,
+                       (float2*)gpair.data_ptr<float>(),
+                       (unsigned int*)absmax.data_ptr<float>(), n);
+  } else {
+    hipLaunchKernelGGL((grad_fused_kernel<0>), dim3((uint32_t)blocks),
+                       dim3(256), 0, stream.stream(),
+                       margin.data_ptr<float>(), label.data_ptr<float>(),
+                       wp, (float)spw,
+                       (float2*)gpair.data_ptr<float>(),
+                       (unsigned int*)absmax.data_ptr<float>(), n);
+  }
+  return {gpair, absmax};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("quantize_gpair", &quantize_gpair, "quantize gradient pairs");
+  m.def("grad_fused", &grad_fused, "fused objective gradient + absmax");
   m.def("bin_matrix", &bin_matrix, "bin feature matrix");
   m.def("build_histogram", &build_histogram, "build gradient histograms");
   m.def("find_splits", &find_splits, "best-split scan");

@@ -78,6 +78,14 @@ class SquaredError(Objective):
     default_metric = "rmse"
 
     def gradients(self, margin, label, weight=None, qid=None):
+        if margin.is_cuda and margin.dim() == 1:
+            from xgboost_ray_amd import ops
+
+            out = ops.grad_fused(margin, label, weight, 1.0, mode=0)
+            if out is not None:
+                gp, mx = out
+                gp._rxgb_absmax = mx  # consumed by _quantize
+                return gp
         g = margin - label
         h = torch.ones_like(margin)
         return self._apply_weight(g, h, weight)
@@ -107,6 +115,16 @@ class Logistic(Objective):
         return -math.log(1.0 / base_score - 1.0)
 
     def gradients(self, margin, label, weight=None, qid=None):
+        if margin.is_cuda and margin.dim() == 1:
+            from xgboost_ray_amd import ops
+
+            out = ops.grad_fused(
+                margin, label, weight, self.scale_pos_weight, mode=1
+            )
+            if out is not None:
+                gp, mx = out
+                gp._rxgb_absmax = mx  # consumed by _quantize
+                return gp
         p = sigmoid_sizeinv(margin)
         g = p - label
         h = torch.clamp(p * (1.0 - p), min=_EPS)

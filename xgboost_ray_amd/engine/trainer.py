@@ -594,9 +594,15 @@ class BoostingEngine:
         self.margin += old_scale * contrib
 
     def _quantize(self, gpair: torch.Tensor) -> Tuple[torch.Tensor, float, float]:
-        mx = torch.stack(
-            [gpair[:, 0].abs().max(), gpair[:, 1].abs().max()]
-        ).double()
+        fused_mx = getattr(gpair, "_rxgb_absmax", None)
+        if fused_mx is not None:
+            # grad_fused already reduced |g|/|h| maxes in its single pass
+            # (f32 max is order-free, so the value equals abs().max())
+            mx = fused_mx.double()
+        else:
+            mx = torch.stack(
+                [gpair[:, 0].abs().max(), gpair[:, 1].abs().max()]
+            ).double()
         if self.coll.is_distributed:
             mx_d = mx.to(self.device) if self.device.type == "cuda" else mx
             self.coll.allreduce_(mx_d, op="max")

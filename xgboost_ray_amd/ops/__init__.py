@@ -158,3 +158,17 @@ def predict_trees(
 def update_margins(margin, ridx, starts, counts, leaf_values):
     """margin[ridx[seg_k]] += leaf_values[k] for each final-leaf segment."""
     return _impl(margin).update_margins(margin, ridx, starts, counts, leaf_values)
+
+
+def grad_fused(margin, label, weight, scale_pos_weight, mode):
+    """Fused objective gradient + |g|/|h| max for the hot objectives
+    (mode 0 = reg:squarederror, 1 = binary:logistic). GPU only: returns
+    (fp32 gpair [n,2], f32 absmax [2]) or None on CPU (the torch
+    composition is the CPU path and the numerics oracle)."""
+    import os
+
+    if not margin.is_cuda or os.environ.get("RXGB_FUSED_GRAD") == "0":
+        return None
+    from xgboost_ray_amd.ops import gpu
+
+    return gpu.grad_fused(margin, label, weight, scale_pos_weight, mode)

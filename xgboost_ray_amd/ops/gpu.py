@@ -172,3 +172,14 @@ def update_margins(margin, ridx, starts, counts, leaf_values):
     if target is not margin:
         margin.copy_(target)
     return margin
+
+
+def grad_fused(margin, label, weight, scale_pos_weight, mode):
+    w = weight if weight is not None else torch.zeros(
+        0, dtype=torch.float32, device=margin.device)
+    gpair, absmax = _load().grad_fused(
+        margin.contiguous(), label.contiguous().float(),
+        w.contiguous().float() if w.numel() else w,
+        float(scale_pos_weight), int(mode),
+    )
+    return gpair, absmax
