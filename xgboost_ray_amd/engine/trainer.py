@@ -858,30 +858,35 @@ class BoostingEngine:
                 if self.coll.is_distributed:
                     self.coll.allreduce_(hist)
             _tick("allreduce")
-            if derive_sib_pos.size:
+            # ---- ONE H2D for ALL of this depth's control data:
+            # [parent_g (KK) | parent_h (KK) | derive parent slots (nd) |
+            #  derive sibling positions (nd)] - separate stages cost one
+            # pinned copy + stream op each
+            KK = sumg_ord.size
+            nd = derive_sib_pos.size
+            if self.device.type == "cuda":
+                depth_meta = self._stage_i64(
+                    np.concatenate(
+                        [sumg_ord, sumh_ord, derive_pslot,
+                         derive_sib_pos]
+                    ),
+                    "depth_meta",
+                )
+                pg, ph = depth_meta[:KK], depth_meta[KK : 2 * KK]
+                pslots = depth_meta[2 * KK : 2 * KK + nd]
+                spos = depth_meta[2 * KK + nd :]
+            else:
+                pg = torch.from_numpy(sumg_ord)
+                ph = torch.from_numpy(sumh_ord)
+                pslots = torch.from_numpy(derive_pslot)
+                spos = torch.from_numpy(derive_sib_pos)
+            if nd:
                 # sibling = parent - built, batched over all pairs
-                if self.device.type == "cuda":
-                    pslots = self._stage_i64(derive_pslot, "pslots")
-                    spos = self._stage_i64(derive_sib_pos, "spos")
-                else:
-                    pslots = torch.from_numpy(derive_pslot)
-                    spos = torch.from_numpy(derive_sib_pos)
                 torch.sub(
                     prev_all_hist.index_select(0, pslots),
                     all_hist.index_select(0, spos),
                     out=all_hist[K:],
                 )
-
-            # ---- split scan over the whole frontier (scan-slot order)
-            # ONE H2D for both parent-sum vectors
-            psums_np = np.ascontiguousarray(np.stack([sumg_ord, sumh_ord]))
-            if self.device.type == "cuda":
-                psums = self._stage_i64(psums_np, "psums")
-                KK = psums_np.shape[1]
-                pg, ph = psums[:KK], psums[KK : 2 * KK]
-            else:
-                psums = torch.from_numpy(psums_np)
-                pg, ph = psums[0], psums[1]
             fb = self.feat_bins
             mask = feat_mask
             if self.p.colsample_bylevel < 1.0:
