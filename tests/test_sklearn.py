@@ -155,7 +155,9 @@ class TestSklearnInterop:
     def test_ray_dmatrix_passthrough(self):
         X, y = create_data(800, 4)
         dm = RayDMatrix(X, label=y)
-        clf = RayXGBClassifier(n_estimators=4, max_depth=3)
+        # num_class is required for RayDMatrix classifier fits (labels
+        # cannot be inspected; reference test_sklearn_matrix.py:48)
+        clf = RayXGBClassifier(n_estimators=4, max_depth=3, num_class=2)
         clf.fit(dm, None, ray_params=RP)
         pred = clf.predict(RayDMatrix(X), ray_params=RP)
         assert len(pred) == 800

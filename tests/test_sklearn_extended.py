@@ -508,3 +508,48 @@ class TestEarlyStopping:
         assert getattr(clf, "best_iteration", None) is not None
         it = len(clf.evals_result_["validation_0"]["logloss"])
         assert clf.best_iteration < it
+
+
+class TestRayDMatrixValidation:
+    """reference test_sklearn_matrix.py:34-78: classifier + RayDMatrix
+    validation and the mixed eval_set errors."""
+
+    def test_classifier_requires_num_class(self):
+        X, y = create_data(300, 4)
+        dm = RayDMatrix(X, label=y)
+        with pytest.raises(Exception, match="num_class"):
+            RayXGBClassifier(n_estimators=2).fit(dm, None)
+
+    def test_mixed_eval_set_errors(self):
+        X, y = create_data(300, 4)
+        Xe, ye = create_data(100, 4, seed=5)
+        dm = RayDMatrix(X, label=y)
+        dme = RayDMatrix(Xe, label=ye)
+        with pytest.raises(Exception, match=r"RayDMatrix, str"):
+            RayXGBClassifier(n_estimators=2, num_class=2).fit(
+                dm, None, eval_set=[(Xe, ye)])
+        with pytest.raises(Exception, match=r"array_like, array_like"):
+            RayXGBClassifier(n_estimators=2).fit(
+                X, y, eval_set=[(dme, "eval")])
+
+    def test_classifier_with_num_class_works(self):
+        X, y = create_data(600, 4)
+        Xe, ye = create_data(200, 4, seed=3)
+        dm = RayDMatrix(X, label=y)
+        dme = RayDMatrix(Xe, label=ye)
+        clf = RayXGBClassifier(n_estimators=3, num_class=2,
+                               eval_metric=["logloss"])
+        clf.fit(dm, None, eval_set=[(dme, "eval")], ray_params=RP1,
+                verbose=False)
+        assert "eval" in clf.evals_result_
+        pred = clf.predict(RayDMatrix(Xe), ray_params=RP1)
+        proba = clf.predict_proba(RayDMatrix(Xe), ray_params=RP1)
+        assert len(pred) == 200 and proba.shape == (200, 2)
+
+    def test_multiclass_raydmatrix_num_class(self):
+        X, y = create_data(600, 4, kind="multi")
+        dm = RayDMatrix(X, label=y)
+        clf = RayXGBClassifier(n_estimators=3, num_class=4)
+        clf.fit(dm, None, ray_params=RP1)
+        proba = clf.predict_proba(RayDMatrix(X), ray_params=RP1)
+        assert proba.shape == (600, 4)

@@ -108,7 +108,7 @@ def _check_if_params_are_ray_dmatrix(X, sample_weight, base_margin, eval_set,
             if any(not isinstance(e[0], RayDMatrix) for e in eval_set):
                 raise ValueError(
                     "When X is a RayDMatrix, eval_set entries must be "
-                    "(RayDMatrix, name) tuples."
+                    "(RayDMatrix, str) tuples."
                 )
             evals = tuple(eval_set)
     return train_dmatrix, evals
@@ -264,6 +264,13 @@ class _RayXGBModel(RayXGBMixin):
                 X, label=y, weight=sample_weight, base_margin=base_margin,
                 missing=self.missing, qid=qid, **dm_params,
             )
+            if any(
+                isinstance(e[0], RayDMatrix) for e in (eval_set or [])
+            ):
+                raise ValueError(
+                    "When X is array-like, eval_set entries must be "
+                    "(array_like, array_like) tuples, not RayDMatrix."
+                )
             evals = []
             for i, (ex, ey) in enumerate(eval_set or []):
                 ew = (sample_weight_eval_set or [None] * len(eval_set))[i]
@@ -416,13 +423,29 @@ class RayXGBClassifier(_RayXGBModel):
     ):
         extra_params = {}
         if isinstance(X, RayDMatrix):
-            # cannot inspect labels; trust user-provided objective/num_class
+            # labels cannot be inspected inside a RayDMatrix, so the
+            # class count must be explicit (reference
+            # test_sklearn_matrix.py:48 expects an error naming
+            # `num_class`)
+            nc = (self.kwargs or {}).get("num_class")
+            if not nc:
+                raise ValueError(
+                    "Fitting a classifier on a RayDMatrix requires the "
+                    "`num_class` parameter (labels cannot be inspected "
+                    "for encoding): RayXGBClassifier(num_class=...)"
+                )
             if self.objective and self.objective.startswith("multi"):
-                nc = (self.kwargs or {}).get("num_class")
-                if nc:
-                    extra_params["num_class"] = nc
+                extra_params["num_class"] = nc
+            elif nc > 2:
+                extra_params["objective"] = "multi:softprob"
+                extra_params["num_class"] = nc
+            else:
+                # binary: the engine must NOT see num_class=2 (that means
+                # a 2-column multiclass margin); kwargs carry it, so
+                # override it back to the binary sentinel
+                extra_params["num_class"] = 0
             self.classes_ = None
-            self.n_classes_ = (self.kwargs or {}).get("num_class", 2)
+            self.n_classes_ = nc
         else:
             y = np.asarray(y)
             self.classes_ = np.unique(y)
