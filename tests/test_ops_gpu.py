@@ -444,3 +444,44 @@ def test_grad_fused_matches_torch_composition(mode, spw, weighted):
         torch.stack([gp_t[:, 0].abs().max(), gp_t[:, 1].abs().max()]).cpu(),
         rtol=0, atol=0,
     )
+
+
+def test_predict_lds_matches_plain():
+    """Tree-tiled LDS predictor vs the plain walk: identical outputs
+    (same per-row accumulation order), including NaN rows and
+    deep (over-tile fallback) trees."""
+    import os
+
+    from xgboost_ray_amd.engine.quantile import BinnedMatrix
+    from xgboost_ray_amd.engine.trainer import run_training
+    from xgboost_ray_amd.ops import gpu as gops
+
+    X, y = create_data(60_000, 10, seed=2)
+    X = X.copy()
+    X[::13, 3] = np.nan
+    dm = BinnedMatrix.build(
+        torch.from_numpy(X).cuda(), label=torch.from_numpy(y).cuda(),
+        max_bin=64,
+    )
+    for depth in (8, 12):  # depth 12 can exceed the 4096-node tile
+        bst = run_training(
+            {"objective": "binary:logistic", "max_depth": depth,
+             "eta": 0.3, "tree_method": "gpu_hist"},
+            dm, 6,
+        )
+        Xp = torch.from_numpy(X[:20_000]).cuda()
+        flat = bst._flat_trees(Xp.device)
+        outs = {}
+        for m in ("0", "1"):
+            os.environ["RXGB_PREDICT_LDS"] = m
+            try:
+                out = torch.zeros(20_000, device="cuda")
+                gops.predict_trees(
+                    Xp, flat["feat"], flat["thr"], flat["left"],
+                    flat["default_left"], flat["value"],
+                    flat["tree_ptr"], out,
+                )
+                outs[m] = out.cpu()
+            finally:
+                os.environ.pop("RXGB_PREDICT_LDS", None)
+        torch.testing.assert_close(outs["0"], outs["1"], rtol=0, atol=0)
