@@ -829,19 +829,32 @@ __global__ void find_splits_reduce_kernel(
 // gather from the binned matrix), cache it as one flag byte per row, and
 // count per-256-row-block lefts. Pass 2 then re-reads 1 B/row instead of
 // repeating the gather - halves the partition's memory cost.
+// meta layout (both modes): [starts Ks | counts Ks | chunk_off Ks+1 |
+// feat Ks | bin Ks | dl Ks]. Legacy: Ks = K (host-staged meta),
+// limits == nullptr. Device mode: limits = {n_split, total_chunks}
+// written by plan_partition_kernel; grids are BOUNDS and excess
+// workgroups exit here.
 __global__ void partition_count_kernel(
     const uint8_t* __restrict__ bins, const int32_t* __restrict__ ridx,
-    const int64_t* __restrict__ node_start,  // [K] then counts at [K..2K)
-    const int64_t* __restrict__ chunk_off,   // [K+1]
-    const int64_t* __restrict__ split_feat, const int64_t* __restrict__ split_bin,
-    const int64_t* __restrict__ default_left,
+    const int64_t* __restrict__ meta,
+    const int64_t* __restrict__ limits,  // nullptr or [2]
     int32_t* __restrict__ block_counts,  // [total_chunks]
     uint8_t* __restrict__ flags,         // [n] go-left per segment position
     int K, int64_t row_stride,
     const uint8_t* __restrict__ bins_T,  // [F, n] column-major copy or null
     int64_t n_rows_total) {
   int wg = blockIdx.x;
-  int lo = 0, hi = K;
+  int64_t Ks = K;
+  if (limits != nullptr) {
+    if (wg >= limits[1]) return;
+    Ks = limits[0];
+  }
+  const int64_t* node_start = meta;            // starts, counts at [Ks..2Ks)
+  const int64_t* chunk_off = meta + 2 * Ks;
+  const int64_t* split_feat = meta + 3 * Ks + 1;
+  const int64_t* split_bin = meta + 4 * Ks + 1;
+  const int64_t* default_left = meta + 5 * Ks + 1;
+  int lo = 0, hi = (int)Ks;
   while (lo + 1 < hi) {
     int m = (lo + hi) >> 1;
     if (chunk_off[m] <= wg) lo = m; else hi = m;
@@ -849,7 +862,7 @@ __global__ void partition_count_kernel(
   const int node = lo;
   const int64_t chunk_in_node = wg - chunk_off[node];
   const int64_t row_lo = chunk_in_node * PART_CHUNK;
-  const int64_t count = node_start[K + node];
+  const int64_t count = node_start[Ks + node];
   const int64_t seg_start = node_start[node];
   const int feat = (int)split_feat[node], sbin = (int)split_bin[node];
   const int dl = (int)default_left[node];
@@ -906,12 +919,16 @@ __global__ void partition_count_kernel(
 // K-element total pull.
 __global__ void partition_prefix_kernel(
     const int32_t* __restrict__ block_counts,  // [total_chunks]
-    const int64_t* __restrict__ chunk_off,     // [K+1]
+    const int64_t* __restrict__ meta,          // layout as count kernel
+    const int64_t* __restrict__ limits,        // nullptr or [2]
     int64_t* __restrict__ left_before,         // [total_chunks]
     int64_t* __restrict__ node_left_total,     // [K]
     int K) {
   const int k = blockIdx.x;
-  if (k >= K) return;
+  int64_t Ks = K;
+  if (limits != nullptr) Ks = limits[0];
+  if (k >= Ks) return;
+  const int64_t* chunk_off = meta + 2 * Ks;
   const int64_t c0 = chunk_off[k], c1 = chunk_off[k + 1];
   __shared__ int64_t sc[256];
   __shared__ int64_t sc2[256];
@@ -945,13 +962,20 @@ __global__ void partition_scatter_kernel(
     int32_t* __restrict__ ridx_out,
     const int2* __restrict__ gseg,      // segment-ordered gradient pairs
     int2* __restrict__ gseg_out,        // permuted alongside ridx
-    const int64_t* __restrict__ node_start,  // [K] + counts at [K..2K)
-    const int64_t* __restrict__ chunk_off,   // [K+1]
+    const int64_t* __restrict__ meta,   // layout as count kernel
+    const int64_t* __restrict__ limits,  // nullptr or [2]
     const int64_t* __restrict__ left_before,   // [total_chunks] excl. prefix within node
     const int64_t* __restrict__ node_left_total,  // [K]
     int K) {
   int wg = blockIdx.x;
-  int lo = 0, hi = K;
+  int64_t Ks = K;
+  if (limits != nullptr) {
+    if (wg >= limits[1]) return;
+    Ks = limits[0];
+  }
+  const int64_t* node_start = meta;
+  const int64_t* chunk_off = meta + 2 * Ks;
+  int lo = 0, hi = (int)Ks;
   while (lo + 1 < hi) {
     int m = (lo + hi) >> 1;
     if (chunk_off[m] <= wg) lo = m; else hi = m;
@@ -959,7 +983,7 @@ __global__ void partition_scatter_kernel(
   const int node = lo;
   const int64_t chunk_in_node = wg - chunk_off[node];
   const int64_t row_lo = chunk_in_node * PART_CHUNK;
-  const int64_t count = node_start[K + node];
+  const int64_t count = node_start[Ks + node];
   const int64_t seg_start = node_start[node];
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave_id = threadIdx.x / WAVE;
@@ -1544,7 +1568,7 @@ std::vector<torch::Tensor> find_splits(torch::Tensor hist, torch::Tensor parent_
                                        double scale_h, double lam, double alpha,
                                        double gamma, double mcw,
                                        torch::Tensor mono, torch::Tensor bounds,
-                                       torch::Tensor allowed) {
+                                       torch::Tensor allowed, bool pull) {
   const int K = (int)hist.size(0);
   const int F = (int)hist.size(1);
   const int B = (int)hist.size(2);
@@ -1586,12 +1610,169 @@ std::vector<torch::Tensor> find_splits(torch::Tensor hist, torch::Tensor parent_
                      reinterpret_cast<const long long*>(kf_lh.data_ptr<int64_t>()),
                      reinterpret_cast<long long*>(out_packed.data_ptr<int64_t>()),
                      K, F);
+  if (!pull) return {out_packed};  // device [K,6]: consumed by the
+                                   // fused partition (single-sync path)
   // pinned D2H: the caller consumes (copies out of) the result
   // immediately, so returning a view of the cached buffer is safe
   static thread_local PinnedStager split_out_stager;
   auto out_cpu = split_out_stager.get((int64_t)K * 6).view({K, 6});
   out_cpu.copy_(out_packed);
   return {out_cpu};
+}
+
+
+// ---------------------------------------------------------------------------
+// plan_partition: ONE device kernel turns the depth's packed split
+// results ([K,6]: f32-gain bits, feat, bin, dl, lg, lh) into the
+// compacted partition metadata the partition kernels consume
+// (starts|counts|chunk_off|feat|bin|dl of the SPLIT nodes only), so the
+// partition can launch without the host ever seeing the splits. The
+// split predicate (gain > 0, feat >= 0, finite) is bit-identical to the
+// host replay's numpy predicate - both read the same f32 bits.
+// Single workgroup: K <= 8192 and the work per node is trivial.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(1024) void plan_partition_kernel(
+    const long long* __restrict__ packed,      // [K, 6]
+    const int64_t* __restrict__ starts_ord,    // [K] scan-slot order
+    const int64_t* __restrict__ counts_ord,    // [K]
+    int64_t* __restrict__ meta,     // [6K+1] compacted partition meta
+    int64_t* __restrict__ scalars,  // [2]: n_split, total_chunks
+    int K) {
+  __shared__ int64_t s_ok[1024];
+  __shared__ int64_t s_ch[1024];
+  const int per = (K + blockDim.x - 1) / blockDim.x;
+  const int lo = threadIdx.x * per;
+  const int hi = min(K, lo + per);
+  // thread-local pass: ok flags + chunk counts
+  int64_t my_ok = 0, my_ch = 0;
+  for (int k = lo; k < hi; ++k) {
+    const float gain = __int_as_float((int)packed[(size_t)k * 6]);
+    const long long f = packed[(size_t)k * 6 + 1];
+    const bool ok = (gain > 0.0f) && (f >= 0) && isfinite(gain);
+    if (ok) {
+      ++my_ok;
+      my_ch += (counts_ord[k] + PART_CHUNK - 1) / PART_CHUNK;
+    }
+  }
+  s_ok[threadIdx.x] = my_ok;
+  s_ch[threadIdx.x] = my_ch;
+  __syncthreads();
+  // inclusive block scan (Hillis-Steele over 1024)
+  for (int off = 1; off < 1024; off <<= 1) {
+    int64_t a = 0, b = 0;
+    if ((int)threadIdx.x >= off) {
+      a = s_ok[threadIdx.x - off];
+      b = s_ch[threadIdx.x - off];
+    }
+    __syncthreads();
+    s_ok[threadIdx.x] += a;
+    s_ch[threadIdx.x] += b;
+    __syncthreads();
+  }
+  const int64_t n_split = s_ok[1023];
+  const int64_t total_chunks = s_ch[1023];
+  int64_t pos = s_ok[threadIdx.x] - my_ok;   // exclusive prefixes
+  int64_t coff = s_ch[threadIdx.x] - my_ch;
+  // meta layout (Ks = n_split): [starts Ks | counts Ks | chunk_off Ks+1
+  //  | feat Ks | bin Ks | dl Ks] packed back-to-back at n_split-based
+  // offsets; chunk_off tail padded with total_chunks so the kernels'
+  // binary search over a BOUND-sized node range never maps a chunk to a
+  // phantom node.
+  int64_t* m_starts = meta;
+  int64_t* m_counts = meta + n_split;
+  int64_t* m_coff = meta + 2 * n_split;
+  int64_t* m_feat = meta + 3 * n_split + 1;
+  int64_t* m_bin = meta + 4 * n_split + 1;
+  int64_t* m_dl = meta + 5 * n_split + 1;
+  for (int k = lo; k < hi; ++k) {
+    const float gain = __int_as_float((int)packed[(size_t)k * 6]);
+    const long long f = packed[(size_t)k * 6 + 1];
+    const bool ok = (gain > 0.0f) && (f >= 0) && isfinite(gain);
+    if (ok) {
+      m_starts[pos] = starts_ord[k];
+      m_counts[pos] = counts_ord[k];
+      m_coff[pos] = coff;
+      m_feat[pos] = f;
+      m_bin[pos] = packed[(size_t)k * 6 + 2];
+      m_dl[pos] = packed[(size_t)k * 6 + 3];
+      coff += (counts_ord[k] + PART_CHUNK - 1) / PART_CHUNK;
+      ++pos;
+    }
+  }
+  if (threadIdx.x == 1023) m_coff[n_split] = total_chunks;
+  if (threadIdx.x == 0) {
+    scalars[0] = n_split;
+    scalars[1] = total_chunks;
+  }
+}
+
+// Fused scan-consume partition: find_splits' packed output feeds the
+// partition directly on device; the host gets {packed | left_counts}
+// in ONE pinned D2H it reads after a single stream sync per depth
+// (replacing the 2 syncs/depth of the host-planned path).
+std::vector<torch::Tensor> partition_rows_from_packed(
+    torch::Tensor bins, torch::Tensor ridx, torch::Tensor starts_ord,
+    torch::Tensor counts_ord, torch::Tensor packed, torch::Tensor gseg,
+    torch::Tensor bins_t, int64_t chunk_bound) {
+  const int K = (int)packed.size(0);     // frontier bound (scan-slot count)
+  auto dev = bins.device();
+  auto ridx_out = ridx.clone();
+  auto gseg_out = gseg.clone();
+  auto optsl = torch::TensorOptions().dtype(torch::kInt64).device(dev);
+  auto meta = torch::empty({6 * (int64_t)K + 2}, optsl);
+  auto scalars = torch::zeros({2}, optsl);
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(plan_partition_kernel, dim3(1), dim3(1024), 0,
+                     stream.stream(),
+                     reinterpret_cast<const long long*>(packed.data_ptr<int64_t>()),
+                     starts_ord.data_ptr<int64_t>(),
+                     counts_ord.data_ptr<int64_t>(),
+                     meta.data_ptr<int64_t>(), scalars.data_ptr<int64_t>(),
+                     K);
+  int64_t* mp = meta.data_ptr<int64_t>();
+  const int64_t* sc2 = scalars.data_ptr<int64_t>();
+  auto block_counts = torch::empty({std::max<int64_t>(chunk_bound, 1)},
+      torch::TensorOptions().dtype(torch::kInt32).device(dev));
+  auto flags = torch::empty({ridx.size(0)},
+      torch::TensorOptions().dtype(torch::kUInt8).device(dev));
+  // device-limit form: node/chunk layout pointers are n_split-based
+  // inside `meta`, discovered by each kernel from scalars[0]
+  hipLaunchKernelGGL(partition_count_kernel,
+                     dim3((uint32_t)std::max<int64_t>(chunk_bound, 1)),
+                     dim3(PART_THREADS), 0, stream.stream(),
+                     bins.data_ptr<uint8_t>(), ridx.data_ptr<int32_t>(),
+                     mp, sc2, block_counts.data_ptr<int32_t>(),
+                     flags.data_ptr<uint8_t>(), K, bins.stride(0),
+                     bins_t.numel() ? bins_t.data_ptr<uint8_t>() : nullptr,
+                     bins.size(0));
+  auto left_before = torch::empty({std::max<int64_t>(chunk_bound, 1)}, optsl);
+  auto node_left_total = torch::zeros({K}, optsl);
+  hipLaunchKernelGGL(partition_prefix_kernel,
+                     dim3((uint32_t)std::max(K, 1)), dim3(256), 0,
+                     stream.stream(),
+                     block_counts.data_ptr<int32_t>(), mp, sc2,
+                     left_before.data_ptr<int64_t>(),
+                     node_left_total.data_ptr<int64_t>(), K);
+  hipLaunchKernelGGL(partition_scatter_kernel,
+                     dim3((uint32_t)std::max<int64_t>(chunk_bound, 1)),
+                     dim3(PART_THREADS), 0, stream.stream(),
+                     flags.data_ptr<uint8_t>(), ridx.data_ptr<int32_t>(),
+                     ridx_out.data_ptr<int32_t>(),
+                     gseg.numel() ? (const int2*)gseg.data_ptr<int32_t>()
+                                  : nullptr,
+                     gseg_out.numel() ? (int2*)gseg_out.data_ptr<int32_t>()
+                                      : nullptr,
+                     mp, sc2, left_before.data_ptr<int64_t>(),
+                     node_left_total.data_ptr<int64_t>(), K);
+  // ONE async pinned D2H of everything the host replay needs
+  static thread_local PinnedStager pull_stager;
+  auto pull = pull_stager.get(7 * (int64_t)K);
+  pull.narrow(0, 0, 6 * (int64_t)K)
+      .copy_(packed.view({-1}), /*non_blocking=*/true);
+  pull.narrow(0, 6 * (int64_t)K, K)
+      .copy_(node_left_total, /*non_blocking=*/true);
+  pull_stager.mark(stream.stream());
+  return {ridx_out, gseg_out, pull};
 }
 
 std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx,
@@ -1632,11 +1813,7 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
   auto meta = meta_cpu.to(dev, /*non_blocking=*/true);
   part_meta_stager.mark(stream0.stream());
   int64_t* mp = meta.data_ptr<int64_t>();
-  int64_t* sc_p = mp;
-  int64_t* chunk_off_p = mp + 2 * K;
-  int64_t* sf_p = mp + 3 * K + 1;
-  int64_t* sb_p = mp + 4 * K + 1;
-  int64_t* dl_p = mp + 5 * K + 1;
+  // layout inside `meta` (Ks = K): see partition_count_kernel
 
   auto stream = c10::hip::getCurrentHIPStream();
   auto block_counts = torch::empty({total_chunks},
@@ -1646,7 +1823,7 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
   hipLaunchKernelGGL(partition_count_kernel, dim3((uint32_t)total_chunks),
                      dim3(PART_THREADS), 0, stream.stream(),
                      bins.data_ptr<uint8_t>(), ridx.data_ptr<int32_t>(),
-                     sc_p, chunk_off_p, sf_p, sb_p, dl_p,
+                     mp, (const int64_t*)nullptr,
                      block_counts.data_ptr<int32_t>(),
                      flags.data_ptr<uint8_t>(), K, bins.stride(0),
                      bins_t.numel() ? bins_t.data_ptr<uint8_t>() : nullptr,
@@ -1664,7 +1841,8 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
       torch::TensorOptions().dtype(torch::kInt64).device(dev));
   hipLaunchKernelGGL(partition_prefix_kernel, dim3((uint32_t)K), dim3(256),
                      0, stream0.stream(),
-                     block_counts.data_ptr<int32_t>(), chunk_off_p,
+                     block_counts.data_ptr<int32_t>(), mp,
+                     (const int64_t*)nullptr,
                      left_before.data_ptr<int64_t>(),
                      node_left_total.data_ptr<int64_t>(), K);
   static thread_local PinnedStager nl_stager;
@@ -1679,7 +1857,7 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
                                   : nullptr,
                      gseg_out.numel() ? (int2*)gseg_out.data_ptr<int32_t>()
                                       : nullptr,
-                     sc_p, chunk_off_p,
+                     mp, (const int64_t*)nullptr,
                      left_before.data_ptr<int64_t>(),
                      node_left_total.data_ptr<int64_t>(), K);
   return {ridx_out, node_left_total_cpu, gseg_out};
@@ -1993,6 +2171,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("build_histogram", &build_histogram, "build gradient histograms");
   m.def("find_splits", &find_splits, "best-split scan");
   m.def("partition_rows", &partition_rows, "stable row partition");
+  m.def("partition_rows_from_packed", &partition_rows_from_packed,
+        "device-planned partition consuming find_splits packed output");
   m.def("predict_trees", &predict_trees, "tree-walk prediction");
   m.def("update_margins", &update_margins, "leaf margin update");
   m.def("lambdarank_grad", &lambdarank_grad, "pairwise lambdarank gradients");

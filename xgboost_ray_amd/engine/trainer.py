@@ -867,14 +867,16 @@ class BoostingEngine:
             if self.device.type == "cuda":
                 depth_meta = self._stage_i64(
                     np.concatenate(
-                        [sumg_ord, sumh_ord, derive_pslot,
-                         derive_sib_pos]
+                        [sumg_ord, sumh_ord, start_ord, count_ord,
+                         derive_pslot, derive_sib_pos]
                     ),
                     "depth_meta",
                 )
                 pg, ph = depth_meta[:KK], depth_meta[KK : 2 * KK]
-                pslots = depth_meta[2 * KK : 2 * KK + nd]
-                spos = depth_meta[2 * KK + nd :]
+                d_starts = depth_meta[2 * KK : 3 * KK]
+                d_counts = depth_meta[3 * KK : 4 * KK]
+                pslots = depth_meta[4 * KK : 4 * KK + nd]
+                spos = depth_meta[4 * KK + nd :]
             else:
                 pg = torch.from_numpy(sumg_ord)
                 ph = torch.from_numpy(sumh_ord)
@@ -904,30 +906,71 @@ class BoostingEngine:
                         np.stack([wlo_ord, whi_ord], axis=1)
                     )
                 )
-            best = ops.find_splits(
-                all_hist,
-                pg,
-                ph,
-                fb,
-                scale_g,
-                scale_h,
-                self.p.reg_lambda,
-                self.p.reg_alpha,
-                self.p.gamma,
-                self.p.min_child_weight,
-                monotone=self.mono,
-                bounds=mono_bounds,
-                allowed=self._allowed_mask(paths_ord)
-                if track_paths else None,
+            # single-sync fused path (GPU): the split scan's packed
+            # output feeds a device-planned partition directly; ONE
+            # pinned D2H brings {splits | left_counts} to the host, which
+            # then replays the identical split predicate for tree
+            # bookkeeping. Halves the per-depth host syncs (2 -> 1).
+            use_fused = (
+                self.device.type == "cuda"
+                and _os2.environ.get("RXGB_ONE_SYNC", "1") != "0"
             )
-            _tick("scan")
-            gain = best["gain"]
-            _tick("scan_pull")
-            bfeat = best["feature"]
-            bbin = best["bin"]
-            bdl = best["default_left"]
-            blg = best["left_g"]
-            blh = best["left_h"]
+            lc_full = None
+            allowed_m = (
+                self._allowed_mask(paths_ord) if track_paths else None
+            )
+            if use_fused:
+                packed_dev = ops.find_splits(
+                    all_hist, pg, ph, fb, scale_g, scale_h,
+                    self.p.reg_lambda, self.p.reg_alpha, self.p.gamma,
+                    self.p.min_child_weight, monotone=self.mono,
+                    bounds=mono_bounds, allowed=allowed_m, pull=False,
+                )
+                _tick("scan")
+                # grid BOUND: every split node adds at most one
+                # partial chunk (PART_CHUNK = 2048 rows/workgroup)
+                frontier_rows = int(count_ord.sum())
+                chunk_bound = (frontier_rows + 2047) // 2048 + KK
+                ridx, gseg, pull = ops.partition_rows_from_packed(
+                    self.dtrain.bins, ridx, d_starts, d_counts,
+                    packed_dev, gseg,
+                    getattr(self.dtrain, "bins_t", None), chunk_bound,
+                )
+                torch.cuda.current_stream().synchronize()
+                arr = pull[: 7 * KK].numpy()
+                pk = arr[: 6 * KK].reshape(KK, 6)
+                gain = pk[:, 0].astype(np.int32).view(np.float32).copy()
+                bfeat = pk[:, 1].astype(np.int32)
+                bbin = pk[:, 2].astype(np.int32)
+                bdl = pk[:, 3].astype(np.uint8)
+                blg = pk[:, 4].copy()
+                blh = pk[:, 5].copy()
+                lc_full = arr[6 * KK : 7 * KK].copy()
+                _tick("scan_pull")
+            else:
+                best = ops.find_splits(
+                    all_hist,
+                    pg,
+                    ph,
+                    fb,
+                    scale_g,
+                    scale_h,
+                    self.p.reg_lambda,
+                    self.p.reg_alpha,
+                    self.p.gamma,
+                    self.p.min_child_weight,
+                    monotone=self.mono,
+                    bounds=mono_bounds,
+                    allowed=allowed_m,
+                )
+                _tick("scan")
+                gain = best["gain"]
+                _tick("scan_pull")
+                bfeat = best["feature"]
+                bbin = best["bin"]
+                bdl = best["default_left"]
+                blg = best["left_g"]
+                blh = best["left_h"]
 
             # ---- vectorized split/leaf decision for the whole frontier
             splits_ok = (gain > 0) & (bfeat >= 0) & np.isfinite(gain)
@@ -969,26 +1012,35 @@ class BoostingEngine:
             ta.parent[lids + 1] = nids_ok
             ta.n = base + 2 * n_split
 
-            sstarts = torch.from_numpy(np.ascontiguousarray(start_ord[okf]))
-            scounts = torch.from_numpy(np.ascontiguousarray(count_ord[okf]))
             sf32 = bfeat[okf].astype(np.int32)
             sb32 = bbin[okf].astype(np.int32)
             sdl8 = bdl[okf].astype(np.uint8)
             _tick("tree_host")
-            ridx, left_counts, gseg = ops.partition_rows(
-                self.dtrain.bins,
-                ridx,
-                sstarts,
-                scounts,
-                torch.from_numpy(sf32),
-                torch.from_numpy(sb32),
-                torch.from_numpy(sdl8),
-                gpair_seg=gseg,
-                bins_t=getattr(self.dtrain, "bins_t", None),
-            )
+            if lc_full is not None:
+                # fused path: the partition already ran on device; its
+                # per-split-node left counts rode the single pull, in
+                # the same ascending-okf order the plan kernel compacted
+                lc = lc_full[:n_split].astype(np.int64)
+            else:
+                sstarts = torch.from_numpy(
+                    np.ascontiguousarray(start_ord[okf])
+                )
+                scounts = torch.from_numpy(
+                    np.ascontiguousarray(count_ord[okf])
+                )
+                ridx, left_counts, gseg = ops.partition_rows(
+                    self.dtrain.bins,
+                    ridx,
+                    sstarts,
+                    scounts,
+                    torch.from_numpy(sf32),
+                    torch.from_numpy(sb32),
+                    torch.from_numpy(sdl8),
+                    gpair_seg=gseg,
+                    bins_t=getattr(self.dtrain, "bins_t", None),
+                )
+                lc = left_counts.cpu().numpy().astype(np.int64)
             _tick("partition")
-
-            lc = left_counts.cpu().numpy().astype(np.int64)
             l_start = start_ord[okf]
             l_sumg = blg[okf]
             l_sumh = blh[okf]

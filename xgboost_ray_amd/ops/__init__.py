@@ -95,6 +95,7 @@ def find_splits(
     monotone=None,
     bounds=None,
     allowed=None,
+    pull=True,
 ):
     """Best-split scan over histograms.
 
@@ -105,6 +106,9 @@ def find_splits(
       left_g/left_h int64 (quantized left-child sums incl. missing if
       default_left).
     """
+    kw = {}
+    if hist.is_cuda:
+        kw["pull"] = pull  # CPU impl always returns host arrays
     return _impl(hist).find_splits(
         hist,
         parent_g,
@@ -119,6 +123,18 @@ def find_splits(
         monotone,
         bounds,
         allowed,
+        **kw,
+    )
+
+
+def partition_rows_from_packed(bins, ridx, starts_ord, counts_ord, packed,
+                               gseg, bins_t, chunk_bound):
+    """Single-sync fused partition (GPU only): see ops.gpu."""
+    from xgboost_ray_amd.ops import gpu
+
+    return gpu.partition_rows_from_packed(
+        bins, ridx, starts_ord, counts_ord, packed, gseg, bins_t,
+        chunk_bound,
     )
 
 

@@ -82,6 +82,7 @@ def find_splits(
     monotone=None,
     bounds=None,
     allowed=None,
+    pull=True,
 ):
     import numpy as np
 
@@ -107,7 +108,10 @@ def find_splits(
         monotone.to(dev).to(torch.int8),
         bounds.to(dev).to(torch.float64),
         allowed,
+        pull,
     )
+    if not pull:
+        return packed  # device [K,6] for the fused single-sync partition
     arr = packed.cpu().numpy()  # ONE D2H for the whole depth's splits
     return {
         "gain": arr[:, 0].astype(np.int32).view(np.float32),
@@ -183,3 +187,27 @@ def grad_fused(margin, label, weight, scale_pos_weight, mode):
         float(scale_pos_weight), int(mode),
     )
     return gpair, absmax
+
+
+# partition chunk geometry (PART_THREADS * PART_ROWS_PER_THREAD in
+# kernels.hip); the python side needs it only to BOUND the fused
+# partition grid
+PART_CHUNK = 2048
+
+
+def partition_rows_from_packed(bins, ridx, starts_ord, counts_ord, packed,
+                               gseg, bins_t, chunk_bound):
+    """Single-sync partition: consumes find_splits' device packed output
+    (plan + count + prefix + scatter all device-planned); returns
+    (ridx_out, gseg_out, pull) where `pull` is a pinned i64 [7K] buffer
+    [packed rows | left_counts] valid after a stream sync."""
+    dev = bins.device
+    if bins_t is None:
+        bins_t = torch.zeros(0, dtype=torch.uint8, device=dev)
+    if gseg is None:
+        gseg = torch.zeros((0, 2), dtype=torch.int32, device=dev)
+    r, g, pull = _load().partition_rows_from_packed(
+        bins, ridx, starts_ord, counts_ord, packed, gseg, bins_t,
+        int(chunk_bound),
+    )
+    return r, g, pull
