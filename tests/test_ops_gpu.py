@@ -485,3 +485,33 @@ def test_predict_lds_matches_plain():
             finally:
                 os.environ.pop("RXGB_PREDICT_LDS", None)
         torch.testing.assert_close(outs["0"], outs["1"], rtol=0, atol=0)
+
+
+def test_one_sync_path_bitwise():
+    """Opt-in single-sync depth loop (RXGB_ONE_SYNC=1) vs the default
+    2-sync path: identical models (the device plan kernel's split
+    predicate must replay exactly on the host)."""
+    import os
+
+    from xgboost_ray_amd.engine.quantile import BinnedMatrix
+    from xgboost_ray_amd.engine.trainer import run_training
+
+    X, y = create_data(150_000, 10, seed=4)
+    dm = BinnedMatrix.build(
+        torch.from_numpy(X).cuda(), label=torch.from_numpy(y).cuda(),
+        max_bin=256,
+    )
+    preds = {}
+    for m in ("0", "1"):
+        os.environ["RXGB_ONE_SYNC"] = m
+        try:
+            bst = run_training(
+                {"objective": "binary:logistic", "max_depth": 9,
+                 "eta": 0.3, "tree_method": "gpu_hist",
+                 "subsample": 0.8},
+                dm, 5,
+            )
+            preds[m] = bst.predict(X, output_margin=True)
+        finally:
+            os.environ.pop("RXGB_ONE_SYNC", None)
+    np.testing.assert_array_equal(preds["0"], preds["1"])
