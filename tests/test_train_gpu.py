@@ -344,3 +344,19 @@ def test_multirank_rccl_one_gpu():
         bst1.predict(X, output_margin=True),
         bst2.predict(X, output_margin=True),
     )
+
+
+def test_sklearn_estimator_gpu():
+    """RayXGBClassifier with tree_method=gpu_hist through the actor
+    layer (reference sklearn GPU path)."""
+    X, y = create_data(150_000, 8)
+    clf_kwargs = dict(n_estimators=8, max_depth=5,
+                      tree_method="gpu_hist")
+    from xgboost_ray_amd import RayXGBClassifier
+
+    clf = RayXGBClassifier(**clf_kwargs)
+    clf.fit(X, y, ray_params=RayParams(num_actors=1, gpus_per_actor=1))
+    acc = (clf.predict(
+        X, ray_params=RayParams(num_actors=1, gpus_per_actor=1)) == y
+    ).mean()
+    assert acc > 0.9
