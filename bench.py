@@ -206,8 +206,14 @@ def main():
 
     if world > 1:
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        # RXGB_COLL_BACKEND=gloo: host-staged transport with GPU compute
+        # (e.g. several ranks sharing one MI355X, where RCCL refuses
+        # duplicate devices in a communicator)
+        backend = os.environ.get("RXGB_COLL_BACKEND") or (
+            "nccl" if use_gpu else "gloo"
+        )
         dist.init_process_group(
-            backend="nccl" if use_gpu else "gloo",
+            backend=backend,
             init_method="env://",
             rank=rank,
             world_size=world,
