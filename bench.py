@@ -199,8 +199,11 @@ def main():
 
     use_gpu = torch.cuda.is_available() and not args.cpu
     if use_gpu:
-        torch.cuda.set_device(local_rank)
-        device = torch.device("cuda", local_rank)
+        # modulo: several ranks may share one GPU (gloo-staged transport
+        # experiments on a 1-GPU box)
+        dev_id = local_rank % max(1, torch.cuda.device_count())
+        torch.cuda.set_device(dev_id)
+        device = torch.device("cuda", dev_id)
     else:
         device = torch.device("cpu")
 
