@@ -1775,20 +1775,24 @@ std::vector<torch::Tensor> partition_rows_from_packed(
   return {ridx_out, gseg_out, pull};
 }
 
-std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx,
-                                          torch::Tensor starts, torch::Tensor counts,
-                                          torch::Tensor split_feat,
-                                          torch::Tensor split_bin,
-                                          torch::Tensor default_left,
-                                          torch::Tensor gseg,
-                                          torch::Tensor bins_t) {
+// Two-phase partition: `begin` launches count+prefix and returns
+// immediately so the caller's host-side tree bookkeeping overlaps those
+// kernels; `finish` pulls the per-node left totals (the one host sync)
+// and launches the scatter (which the caller's subsequent bookkeeping
+// then overlaps). partition_rows keeps the one-call form.
+std::vector<torch::Tensor> partition_rows_begin(
+    torch::Tensor bins, torch::Tensor ridx, torch::Tensor starts,
+    torch::Tensor counts, torch::Tensor split_feat,
+    torch::Tensor split_bin, torch::Tensor default_left,
+    torch::Tensor gseg, torch::Tensor bins_t) {
   const int K = (int)starts.size(0);
-  const int F = (int)bins.size(1);
   auto dev = bins.device();
   auto ridx_out = ridx.clone();
   auto gseg_out = gseg.clone();
-  auto left_counts = torch::zeros({K}, torch::kInt64);
-  if (K == 0) return {ridx_out, left_counts, gseg_out};
+  auto empty_i64 = torch::zeros({K > 0 ? K : 1}, torch::kInt64);
+  if (K == 0)
+    return {ridx_out, gseg_out, empty_i64, empty_i64, empty_i64,
+            torch::zeros({1}, torch::kInt64)};
   auto starts_cpu = starts.to(torch::kCPU).to(torch::kInt64);
   auto counts_cpu = counts.to(torch::kCPU).to(torch::kInt64);
   int64_t total_chunks = 0;
@@ -1800,9 +1804,9 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
       acc[k + 1] = acc[k] + (cacc[k] + PART_CHUNK - 1) / PART_CHUNK;
     total_chunks = acc[K];
   }
-  if (total_chunks == 0) return {ridx_out, left_counts, gseg_out};
-  // ONE H2D copy: [starts(K) | counts(K) | chunk_off(K+1) | feat(K) |
-  // bin(K) | default_left(K)] - staged through a pinned buffer
+  if (total_chunks == 0)
+    return {ridx_out, gseg_out, empty_i64, empty_i64, empty_i64,
+            torch::zeros({1}, torch::kInt64)};
   auto stream0 = c10::hip::getCurrentHIPStream();
   static thread_local PinnedStager part_meta_stager;
   auto meta_cpu = part_meta_stager.get(6 * K + 1);
@@ -1813,7 +1817,6 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
   auto meta = meta_cpu.to(dev, /*non_blocking=*/true);
   part_meta_stager.mark(stream0.stream());
   int64_t* mp = meta.data_ptr<int64_t>();
-  // layout inside `meta` (Ks = K): see partition_count_kernel
 
   auto stream = c10::hip::getCurrentHIPStream();
   auto block_counts = torch::empty({total_chunks},
@@ -1828,13 +1831,6 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
                      flags.data_ptr<uint8_t>(), K, bins.stride(0),
                      bins_t.numel() ? bins_t.data_ptr<uint8_t>() : nullptr,
                      bins.size(0));
-  // Per-node exclusive prefix of block counts. ONE D2H pull + a host
-  // loop + ONE H2D push: the torch-op chain this replaces (device
-  // cumsum + repeat_interleave + index_selects, ~8 tiny transfers and
-  // kernels per call) stalled the training loop ~15 ms per depth -
-  // measured 7x slower rounds end to end.
-  // device-side rebase: per-node exclusive prefix of block counts; the
-  // host only pulls the K node totals it needs for tree bookkeeping
   auto left_before = torch::empty({total_chunks},
       torch::TensorOptions().dtype(torch::kInt64).device(dev));
   auto node_left_total = torch::empty({K},
@@ -1845,10 +1841,26 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
                      (const int64_t*)nullptr,
                      left_before.data_ptr<int64_t>(),
                      node_left_total.data_ptr<int64_t>(), K);
+  auto tc = torch::full({1}, total_chunks, torch::kInt64);
+  return {ridx_out, gseg_out, meta, left_before, node_left_total, tc,
+          flags};
+}
+
+std::vector<torch::Tensor> partition_rows_finish(
+    torch::Tensor ridx, torch::Tensor ridx_out, torch::Tensor gseg,
+    torch::Tensor gseg_out, torch::Tensor meta, torch::Tensor left_before,
+    torch::Tensor node_left_total, torch::Tensor tc_t,
+    torch::Tensor flags) {
+  const int K = (int)node_left_total.size(0);
+  const int64_t total_chunks = tc_t.item<int64_t>();
+  if (total_chunks == 0)
+    return {ridx_out, torch::zeros({K}, torch::kInt64), gseg_out};
+  auto stream = c10::hip::getCurrentHIPStream();
   static thread_local PinnedStager nl_stager;
   auto node_left_total_cpu_pin = nl_stager.get(K);
   node_left_total_cpu_pin.copy_(node_left_total);  // sync: host needs it
   auto node_left_total_cpu = node_left_total_cpu_pin.clone();
+  int64_t* mp = meta.data_ptr<int64_t>();
   hipLaunchKernelGGL(partition_scatter_kernel, dim3((uint32_t)total_chunks),
                      dim3(PART_THREADS), 0, stream.stream(),
                      flags.data_ptr<uint8_t>(), ridx.data_ptr<int32_t>(),
@@ -1861,6 +1873,23 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
                      left_before.data_ptr<int64_t>(),
                      node_left_total.data_ptr<int64_t>(), K);
   return {ridx_out, node_left_total_cpu, gseg_out};
+}
+
+std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx,
+                                          torch::Tensor starts, torch::Tensor counts,
+                                          torch::Tensor split_feat,
+                                          torch::Tensor split_bin,
+                                          torch::Tensor default_left,
+                                          torch::Tensor gseg,
+                                          torch::Tensor bins_t) {
+  auto st = partition_rows_begin(bins, ridx, starts, counts, split_feat,
+                                 split_bin, default_left, gseg, bins_t);
+  if (st.size() == 6) {  // K==0 / no chunks: nothing to scatter
+    return {st[0], torch::zeros({(int64_t)starts.size(0)}, torch::kInt64),
+            st[1]};
+  }
+  return partition_rows_finish(ridx, st[0], gseg, st[1], st[2], st[3],
+                               st[4], st[5], st[6]);
 }
 
 
@@ -2171,6 +2200,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("build_histogram", &build_histogram, "build gradient histograms");
   m.def("find_splits", &find_splits, "best-split scan");
   m.def("partition_rows", &partition_rows, "stable row partition");
+  m.def("partition_rows_begin", &partition_rows_begin,
+        "count+prefix phase (returns before the host sync)");
+  m.def("partition_rows_finish", &partition_rows_finish,
+        "left-totals pull + scatter phase");
   m.def("partition_rows_from_packed", &partition_rows_from_packed,
         "device-planned partition consuming find_splits packed output");
   m.def("predict_trees", &predict_trees, "tree-walk prediction");

@@ -984,6 +984,38 @@ class BoostingEngine:
             leaf_idx = np.nonzero(~splits_ok)[0]
             n_split = int(okf.size)
 
+            if n_split == 0:
+                if leaf_idx.size:
+                    self._finalize_leaves_soa(
+                        ta, nid_ord[leaf_idx], sumg_ord[leaf_idx],
+                        sumh_ord[leaf_idx], start_ord[leaf_idx],
+                        count_ord[leaf_idx], wlo_ord[leaf_idx],
+                        whi_ord[leaf_idx], scale_h,
+                    )
+                fr_nid = np.zeros(0, np.int64)
+                break
+
+            # launch the partition's count+prefix FIRST (GPU): the leaf
+            # finalization and tree-array writes below then run on the
+            # host while those kernels execute, and the children
+            # bookkeeping after finish() overlaps the scatter
+            sf32 = bfeat[okf].astype(np.int32)
+            sb32 = bbin[okf].astype(np.int32)
+            sdl8 = bdl[okf].astype(np.uint8)
+            part_ctx = None
+            if lc_full is None and self.device.type == "cuda":
+                part_ctx = ops.partition_begin(
+                    self.dtrain.bins,
+                    ridx,
+                    torch.from_numpy(np.ascontiguousarray(start_ord[okf])),
+                    torch.from_numpy(np.ascontiguousarray(count_ord[okf])),
+                    torch.from_numpy(sf32),
+                    torch.from_numpy(sb32),
+                    torch.from_numpy(sdl8),
+                    gpair_seg=gseg,
+                    bins_t=getattr(self.dtrain, "bins_t", None),
+                )
+
             if leaf_idx.size:
                 self._finalize_leaves_soa(
                     ta, nid_ord[leaf_idx], sumg_ord[leaf_idx],
@@ -991,9 +1023,6 @@ class BoostingEngine:
                     count_ord[leaf_idx], wlo_ord[leaf_idx],
                     whi_ord[leaf_idx], scale_h,
                 )
-            if n_split == 0:
-                fr_nid = np.zeros(0, np.int64)
-                break
 
             # child ids in scan-slot order (matches the sequential
             # numbering the list-based builder produced)
@@ -1018,15 +1047,15 @@ class BoostingEngine:
             ta.parent[lids + 1] = nids_ok
             ta.n = base + 2 * n_split
 
-            sf32 = bfeat[okf].astype(np.int32)
-            sb32 = bbin[okf].astype(np.int32)
-            sdl8 = bdl[okf].astype(np.uint8)
             _tick("tree_host")
             if lc_full is not None:
                 # fused path: the partition already ran on device; its
                 # per-split-node left counts rode the single pull, in
                 # the same ascending-okf order the plan kernel compacted
                 lc = lc_full[:n_split].astype(np.int64)
+            elif part_ctx is not None:
+                ridx, left_counts, gseg = ops.partition_finish(part_ctx)
+                lc = left_counts.numpy().astype(np.int64)
             else:
                 sstarts = torch.from_numpy(
                     np.ascontiguousarray(start_ord[okf])

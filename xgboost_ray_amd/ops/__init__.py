@@ -127,6 +127,22 @@ def find_splits(
     )
 
 
+def partition_begin(bins, ridx, starts, counts, split_feat, split_bin,
+                    default_left, gpair_seg=None, bins_t=None):
+    """Two-phase partition, phase 1 (GPU only): see ops.gpu."""
+    from xgboost_ray_amd.ops import gpu
+
+    return gpu.partition_begin(bins, ridx, starts, counts, split_feat,
+                               split_bin, default_left, gpair_seg, bins_t)
+
+
+def partition_finish(ctx):
+    """Two-phase partition, phase 2 (GPU only): see ops.gpu."""
+    from xgboost_ray_amd.ops import gpu
+
+    return gpu.partition_finish(ctx)
+
+
 def partition_rows_from_packed(bins, ridx, starts_ord, counts_ord, packed,
                                gseg, bins_t, chunk_bound):
     """Single-sync fused partition (GPU only): see ops.gpu."""

@@ -211,3 +211,34 @@ def partition_rows_from_packed(bins, ridx, starts_ord, counts_ord, packed,
         int(chunk_bound),
     )
     return r, g, pull
+
+
+def partition_begin(bins, ridx, starts, counts, split_feat, split_bin,
+                    default_left, gpair_seg=None, bins_t=None):
+    """Launch the partition's count+prefix kernels and return a context
+    immediately: host bookkeeping between begin and finish overlaps
+    them (and the tree writes after finish overlap the scatter)."""
+    dev = bins.device
+    if bins_t is None:
+        bins_t = torch.zeros(0, dtype=torch.uint8, device=dev)
+    if gpair_seg is None:
+        gpair_seg = torch.zeros((0, 2), dtype=torch.int32, device=dev)
+    st = _load().partition_rows_begin(
+        bins, ridx, starts, counts, split_feat, split_bin, default_left,
+        gpair_seg, bins_t,
+    )
+    return (ridx, gpair_seg, st)
+
+
+def partition_finish(ctx):
+    ridx, gseg, st = ctx
+    if len(st) == 6:  # no chunks: nothing moved
+        import torch as _t
+
+        return st[0], _t.zeros(
+            st[2].numel(), dtype=_t.int64
+        ), st[1]
+    r, lc, g = _load().partition_rows_finish(
+        ridx, st[0], gseg, st[1], st[2], st[3], st[4], st[5], st[6]
+    )
+    return r, lc, g
