@@ -573,3 +573,20 @@ def test_queue_drain_no_lost_items():
         assert len(items) == rounds * per_round, (r, len(items))
         # per-rank FIFO order preserved
         assert items == sorted(items, key=lambda t: (t[2], t[3]))
+
+
+def test_invalid_train_kwarg_raises():
+    """Unknown train() kwargs must fail fast with a TypeError naming the
+    argument (reference testKwargsValidation, test_end_to_end.py:355) -
+    not be silently dropped inside the actors."""
+    from tests.utils import create_data
+    from xgboost_ray_amd import RayDMatrix, RayParams, train
+
+    X, y = create_data(200, 4)
+    with pytest.raises(TypeError, match="totally_invalid_kwarg"):
+        train(
+            {"objective": "binary:logistic"},
+            RayDMatrix(X, label=y), 2,
+            ray_params=RayParams(num_actors=1),
+            totally_invalid_kwarg="",
+        )

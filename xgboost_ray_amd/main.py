@@ -476,6 +476,21 @@ def train(
     start_time = time.time()
     ray_params = _validate_ray_params(ray_params)
 
+    # fail fast on unknown keyword arguments instead of silently ignoring
+    # them inside the actors (reference testKwargsValidation,
+    # test_end_to_end.py:355 expects a TypeError naming the kwarg)
+    _ALLOWED_TRAIN_KWARGS = {
+        "early_stopping_rounds", "verbose_eval", "obj", "feval",
+        "maximize", "xgb_model", "callbacks", "num_boost_round",
+        "checkpoint_frequency",
+    }
+    unknown = set(kwargs) - _ALLOWED_TRAIN_KWARGS
+    if unknown:
+        raise TypeError(
+            "train() got unexpected keyword argument(s): "
+            + ", ".join(sorted(unknown))
+        )
+
     if not isinstance(dtrain, RayDMatrix):
         raise ValueError(
             f"The `dtrain` argument must be a RayDMatrix, got {type(dtrain)}."
