@@ -9,8 +9,10 @@
 //     associativity (no float atomics anywhere)
 //   - find_splits:    sequential-per-(node,feature) scan, bitwise-identical
 //     to the CPU torch reference (same add/divide order)
-//   - partition_rows: stable two-pass partition (ballot prefix + scatter)
-//   - predict_trees:  SoA tree-walk
+//   - partition_rows: stable two-pass partition (ballot prefix + scatter;
+//     two-phase begin/finish entry, device-planned variant)
+//   - grad_fused:     fused objective gradient + |g|/|h| block maxes
+//   - predict_trees:  packed-node tree walk (LDS tree-tiled serving path)
 //
 // Design notes (per /opt/skills/guides/cdna_hip_programming.md):
 //   wave = 64 lanes; LDS = 160 KiB/CU; histogram tiles sized so 1-2

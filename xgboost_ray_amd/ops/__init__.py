@@ -49,7 +49,8 @@ def _impl(t: torch.Tensor):
 
 
 def quantize_gpair(gpair: torch.Tensor, scale_g: float, scale_h: float) -> torch.Tensor:
-    """fp32 [n,2] gradient pairs -> int64 [n,2] fixed point."""
+    """fp32 [n,2] gradient pairs -> int32 [n,2] fixed point (|q| <= 2^30
+    by scale construction; int64 accumulators hold any sum)."""
     return _impl(gpair).quantize_gpair(gpair, scale_g, scale_h)
 
 
