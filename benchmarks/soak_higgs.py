@@ -1,6 +1,11 @@
 """200-round HIGGS realism soak: per-round eval metrics + a lossguide
 number (profiles evidence; not the driver-facing bench)."""
 
+import os as _os
+import sys as _sys
+
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
+
 import time
 
 import torch
