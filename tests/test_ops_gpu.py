@@ -515,3 +515,36 @@ def test_one_sync_path_bitwise():
         finally:
             os.environ.pop("RXGB_ONE_SYNC", None)
     np.testing.assert_array_equal(preds["0"], preds["1"])
+
+
+def test_fused_eval_matches_torch():
+    """Fused logloss/AUC eval kernels vs the torch composition: logloss
+    to reduction-order tolerance, AUC histograms exactly."""
+    import os
+
+    from xgboost_ray_amd.engine.metrics import AUC, AUCPR, LogLoss
+
+    torch.manual_seed(1)
+    n = 2_000_003
+    margin = (torch.randn(n, device="cuda") * 3).float()
+    label = (torch.rand(n, device="cuda") < 0.35).float()
+    for metric in (LogLoss(), AUC(), AUCPR()):
+        vals = {}
+        for m in ("0", "1"):
+            os.environ["RXGB_FUSED_EVAL"] = "1" if m == "1" else "0"
+            try:
+                st = metric.local_stats(margin, label, None, None, None)
+                vals[m] = metric.finalize(st.cpu())
+            finally:
+                os.environ.pop("RXGB_FUSED_EVAL", None)
+        assert vals["1"] == pytest.approx(vals["0"], rel=1e-9), metric.name
+    # weighted logloss path too
+    w = (torch.rand(n, device="cuda") + 0.1).float()
+    m = LogLoss()
+    a = m.finalize(m.local_stats(margin, label, w, None, None).cpu())
+    os.environ["RXGB_FUSED_EVAL"] = "0"
+    try:
+        b = m.finalize(m.local_stats(margin, label, w, None, None).cpu())
+    finally:
+        os.environ.pop("RXGB_FUSED_EVAL", None)
+    assert a == pytest.approx(b, rel=1e-9)

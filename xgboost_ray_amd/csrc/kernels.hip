@@ -2195,9 +2195,112 @@ std::vector<torch::Tensor> grad_fused(torch::Tensor margin,
   return {gpair, absmax};
 }
 
+
+// ---------------------------------------------------------------------------
+// Fused eval metrics: ONE pass over [margin, label(, weight)] replaces
+// the 4-6 torch f64 elementwise/reduction kernels per metric per round.
+// logloss: per-block (sum w*ll, sum w) f64 partials -> torch .sum(0)
+// (fixed-order reduction, deterministic). AUC: unweighted int64
+// pos/neg score-bin histograms via global atomics (exact).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void eval_logloss_kernel(
+    const float* __restrict__ margin, const float* __restrict__ label,
+    const float* __restrict__ weight,  // nullable
+    double* __restrict__ block_out,    // [gridDim.x, 2]
+    int64_t n) {
+  double ll = 0.0, wsum = 0.0;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    // same math as metrics.LogLoss: sigmoid in f64, clamped p
+    double p = 1.0 / (1.0 + exp(-(double)margin[i]));
+    p = fmin(fmax(p, 1e-16), 1.0 - 1e-16);
+    const double y = (double)label[i];
+    const double w = weight ? (double)weight[i] : 1.0;
+    ll += w * -(y * log(p) + (1.0 - y) * log(1.0 - p));
+    wsum += w;
+  }
+  __shared__ double red[2 * 256 / WAVE];
+  #pragma unroll
+  for (int off = WAVE / 2; off > 0; off >>= 1) {
+    ll += __shfl_down(ll, off, WAVE);
+    wsum += __shfl_down(wsum, off, WAVE);
+  }
+  const int wid = threadIdx.x / WAVE;
+  if ((threadIdx.x & (WAVE - 1)) == 0) {
+    red[2 * wid] = ll;
+    red[2 * wid + 1] = wsum;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    const int nw = blockDim.x / WAVE;
+    double a = red[0], b = red[1];
+    for (int w = 1; w < nw; ++w) {
+      a += red[2 * w];
+      b += red[2 * w + 1];
+    }
+    block_out[2 * (size_t)blockIdx.x] = a;
+    block_out[2 * (size_t)blockIdx.x + 1] = b;
+  }
+}
+
+torch::Tensor eval_logloss(torch::Tensor margin, torch::Tensor label,
+                           torch::Tensor weight) {
+  const int64_t n = margin.numel();
+  auto dev = margin.device();
+  const int64_t blocks = std::min<int64_t>(ceil_div(n, 256 * 16), 2048);
+  auto block_out = torch::empty(
+      {std::max<int64_t>(blocks, 1), 2},
+      torch::TensorOptions().dtype(torch::kFloat64).device(dev));
+  if (n == 0) return torch::zeros({2}, block_out.options());
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(eval_logloss_kernel, dim3((uint32_t)blocks),
+                     dim3(256), 0, stream.stream(),
+                     margin.data_ptr<float>(), label.data_ptr<float>(),
+                     weight.numel() ? weight.data_ptr<float>() : nullptr,
+                     block_out.data_ptr<double>(), n);
+  return block_out.sum(0);  // fixed-order torch reduction
+}
+
+__global__ __launch_bounds__(256) void eval_auc_hist_kernel(
+    const float* __restrict__ margin, const float* __restrict__ label,
+    long long* __restrict__ hist,  // [2*B]: neg plane | pos plane
+    int B, int64_t n) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    const double p = 1.0 / (1.0 + exp(-(double)margin[i]));
+    long long b = (long long)(p * (double)B);  // trunc, matches .long()
+    if (b > B - 1) b = B - 1;
+    if (b < 0) b = 0;
+    const int pos = label[i] > 0.5f ? 1 : 0;
+    atomicAdd((unsigned long long*)&hist[(size_t)pos * B + b], 1ull);
+  }
+}
+
+torch::Tensor eval_auc_hist(torch::Tensor margin, torch::Tensor label,
+                            int64_t n_bins) {
+  const int64_t n = margin.numel();
+  auto dev = margin.device();
+  auto hist = torch::zeros(
+      {2 * n_bins},
+      torch::TensorOptions().dtype(torch::kInt64).device(dev));
+  if (n == 0) return hist;
+  auto stream = c10::hip::getCurrentHIPStream();
+  const int64_t blocks = std::min<int64_t>(ceil_div(n, 256 * 16), 4096);
+  hipLaunchKernelGGL(eval_auc_hist_kernel, dim3((uint32_t)blocks),
+                     dim3(256), 0, stream.stream(),
+                     margin.data_ptr<float>(), label.data_ptr<float>(),
+                     reinterpret_cast<long long*>(hist.data_ptr<int64_t>()),
+                     (int)n_bins, n);
+  return hist;  // [neg | pos] planes
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("quantize_gpair", &quantize_gpair, "quantize gradient pairs");
   m.def("grad_fused", &grad_fused, "fused objective gradient + absmax");
+  m.def("eval_logloss", &eval_logloss, "fused logloss eval");
+  m.def("eval_auc_hist", &eval_auc_hist, "fused AUC score histograms");
   m.def("bin_matrix", &bin_matrix, "bin feature matrix");
   m.def("build_histogram", &build_histogram, "build gradient histograms");
   m.def("find_splits", &find_splits, "best-split scan");

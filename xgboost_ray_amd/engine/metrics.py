@@ -17,6 +17,18 @@ from xgboost_ray_amd.engine.objectives import sigmoid_sizeinv
 _AUC_BINS = 16384
 
 
+def _fused_eval_ok(margin):
+    """GPU fused metric kernels: one pass instead of 4-6 f64 torch
+    kernels over the shard (RXGB_FUSED_EVAL=0 disables)."""
+    import os
+
+    return (
+        margin.is_cuda
+        and margin.dim() == 1
+        and os.environ.get("RXGB_FUSED_EVAL") != "0"
+    )
+
+
 class Metric:
     name = "base"
     higher_better = False
@@ -82,6 +94,10 @@ class LogLoss(Metric):
     name = "logloss"
 
     def local_stats(self, margin, label, weight, qid, obj):
+        if _fused_eval_ok(margin):
+            from xgboost_ray_amd.ops import gpu
+
+            return gpu.eval_logloss(margin, label, weight)
         p = sigmoid_sizeinv(margin.double()).clamp(1e-16, 1 - 1e-16)
         y = label.double()
         w = _w(label, weight)
@@ -117,6 +133,14 @@ class AUC(Metric):
     higher_better = True
 
     def local_stats(self, margin, label, weight, qid, obj):
+        if weight is None and _fused_eval_ok(margin):
+            from xgboost_ray_amd.ops import gpu
+
+            hist = gpu.eval_auc_hist(margin, label, _AUC_BINS)
+            # kernel layout [neg | pos] -> metric layout [pos | neg]
+            return torch.cat(
+                [hist[_AUC_BINS:], hist[:_AUC_BINS]]
+            ).double()
         p = sigmoid_sizeinv(margin.double())
         b = torch.clamp((p * _AUC_BINS).long(), max=_AUC_BINS - 1)
         if weight is None:

@@ -242,3 +242,18 @@ def partition_finish(ctx):
         ridx, st[0], gseg, st[1], st[2], st[3], st[4], st[5], st[6]
     )
     return r, lc, g
+
+
+def eval_logloss(margin, label, weight=None):
+    w = weight if weight is not None else torch.zeros(
+        0, dtype=torch.float32, device=margin.device)
+    return _load().eval_logloss(
+        margin.contiguous(), label.contiguous().float(),
+        w.contiguous().float() if w.numel() else w,
+    )
+
+
+def eval_auc_hist(margin, label, n_bins):
+    return _load().eval_auc_hist(
+        margin.contiguous(), label.contiguous().float(), int(n_bins)
+    )
