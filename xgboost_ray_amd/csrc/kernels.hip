@@ -1759,6 +1759,35 @@ std::vector<torch::Tensor> partition_rows_from_packed(
                      block_counts.data_ptr<int32_t>(), mp, sc2,
                      left_before.data_ptr<int64_t>(),
                      node_left_total.data_ptr<int64_t>(), K);
+  // ONE async pinned D2H of everything the host replay needs -
+  // enqueued BEFORE the scatter, so the host's single sync wakes as
+  // soon as scan+plan+count+prefix are done and ALL its bookkeeping
+  // overlaps the scatter (the first version synced past the scatter,
+  // which serialized the bookkeeping after it and measured slower
+  // than the 2-sync structure)
+  static thread_local PinnedStager pull_stager;
+  auto pull = pull_stager.get(7 * (int64_t)K);
+  pull.narrow(0, 0, 6 * (int64_t)K)
+      .copy_(packed.view({-1}), /*non_blocking=*/true);
+  pull.narrow(0, 6 * (int64_t)K, K)
+      .copy_(node_left_total, /*non_blocking=*/true);
+  pull_stager.mark(stream.stream());
+  auto cb = torch::full({1}, chunk_bound, torch::kInt64);
+  return {ridx_out, gseg_out, pull, meta, scalars, left_before,
+          node_left_total, flags, cb};
+}
+
+// Scatter phase of the device-planned partition: launched AFTER the
+// caller records its pull event, so the host's event-wait wakes before
+// the scatter and all bookkeeping overlaps it.
+void partition_scatter_from_packed(
+    torch::Tensor ridx, torch::Tensor ridx_out, torch::Tensor gseg,
+    torch::Tensor gseg_out, torch::Tensor meta, torch::Tensor scalars,
+    torch::Tensor left_before, torch::Tensor node_left_total,
+    torch::Tensor flags, torch::Tensor cb) {
+  const int K = (int)node_left_total.size(0);
+  const int64_t chunk_bound = cb.item<int64_t>();
+  auto stream = c10::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(partition_scatter_kernel,
                      dim3((uint32_t)std::max<int64_t>(chunk_bound, 1)),
                      dim3(PART_THREADS), 0, stream.stream(),
@@ -1768,17 +1797,9 @@ std::vector<torch::Tensor> partition_rows_from_packed(
                                   : nullptr,
                      gseg_out.numel() ? (int2*)gseg_out.data_ptr<int32_t>()
                                       : nullptr,
-                     mp, sc2, left_before.data_ptr<int64_t>(),
+                     meta.data_ptr<int64_t>(), scalars.data_ptr<int64_t>(),
+                     left_before.data_ptr<int64_t>(),
                      node_left_total.data_ptr<int64_t>(), K);
-  // ONE async pinned D2H of everything the host replay needs
-  static thread_local PinnedStager pull_stager;
-  auto pull = pull_stager.get(7 * (int64_t)K);
-  pull.narrow(0, 0, 6 * (int64_t)K)
-      .copy_(packed.view({-1}), /*non_blocking=*/true);
-  pull.narrow(0, 6 * (int64_t)K, K)
-      .copy_(node_left_total, /*non_blocking=*/true);
-  pull_stager.mark(stream.stream());
-  return {ridx_out, gseg_out, pull};
 }
 
 // Two-phase partition: `begin` launches count+prefix and returns
@@ -2338,6 +2359,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "left-totals pull + scatter phase");
   m.def("partition_rows_from_packed", &partition_rows_from_packed,
         "device-planned partition consuming find_splits packed output");
+  m.def("partition_scatter_from_packed", &partition_scatter_from_packed,
+        "scatter phase of the device-planned partition");
   m.def("predict_trees", &predict_trees, "tree-walk prediction");
   m.def("update_margins", &update_margins, "leaf margin update");
   m.def("lambdarank_grad", &lambdarank_grad, "pairwise lambdarank gradients");

@@ -937,12 +937,18 @@ class BoostingEngine:
                 # partial chunk (PART_CHUNK = 2048 rows/workgroup)
                 frontier_rows = int(count_ord.sum())
                 chunk_bound = (frontier_rows + 2047) // 2048 + KK
-                ridx, gseg, pull = ops.partition_rows_from_packed(
-                    self.dtrain.bins, ridx, d_starts, d_counts,
-                    packed_dev, gseg,
-                    getattr(self.dtrain, "bins_t", None), chunk_bound,
+                ridx, gseg, pull, pull_ev = (
+                    ops.partition_rows_from_packed(
+                        self.dtrain.bins, ridx, d_starts, d_counts,
+                        packed_dev, gseg,
+                        getattr(self.dtrain, "bins_t", None),
+                        chunk_bound,
+                    )
                 )
-                torch.cuda.current_stream().synchronize()
+                # event recorded between the pull D2H and the scatter:
+                # the host wakes before the scatter and every piece of
+                # bookkeeping below overlaps it
+                pull_ev.synchronize()
                 arr = pull[: 7 * KK].numpy()
                 pk = arr[: 6 * KK].reshape(KK, 6)
                 gain = pk[:, 0].astype(np.int32).view(np.float32).copy()

@@ -206,11 +206,19 @@ def partition_rows_from_packed(bins, ridx, starts_ord, counts_ord, packed,
         bins_t = torch.zeros(0, dtype=torch.uint8, device=dev)
     if gseg is None:
         gseg = torch.zeros((0, 2), dtype=torch.int32, device=dev)
-    r, g, pull = _load().partition_rows_from_packed(
+    out = _load().partition_rows_from_packed(
         bins, ridx, starts_ord, counts_ord, packed, gseg, bins_t,
         int(chunk_bound),
     )
-    return r, g, pull
+    # out: [ridx_out, gseg_out, pull, meta, scalars, left_before,
+    #       node_left_total, flags, chunk_bound]
+    ev = torch.cuda.Event()
+    ev.record()  # after plan+count+prefix+pull, BEFORE the scatter
+    _load().partition_scatter_from_packed(
+        ridx, out[0], gseg, out[1], out[3], out[4], out[5], out[6],
+        out[7], out[8],
+    )
+    return out[0], out[1], out[2], ev
 
 
 def partition_begin(bins, ridx, starts, counts, split_feat, split_bin,
