@@ -906,20 +906,21 @@ class BoostingEngine:
                         np.stack([wlo_ord, whi_ord], axis=1)
                     )
                 )
-            # single-sync fused path (GPU, opt-in): the split scan's
+            # single-sync fused path (GPU, default): the split scan's
             # packed output feeds a device-planned partition directly;
             # ONE pinned D2H brings {splits | left_counts} to the host,
             # which replays the identical split predicate for
-            # bookkeeping. Measured SLOWER than the 2-sync structure on
-            # HIGGS (d8 4.04-4.15 vs 3.80 ms/round; d12 7.57 vs 7.08):
-            # with two syncs the host's tree bookkeeping after the scan
-            # pull runs concurrently with the partition kernels, and
-            # that overlap outweighs the saved sync. Kept as
-            # RXGB_ONE_SYNC=1 for shapes where sync latency dominates
-            # (many tiny depths / MI300-class virtualized setups).
+            # bookkeeping. The pull is enqueued BEFORE the scatter and
+            # the host waits on an inter-launch EVENT, so it wakes
+            # right after count+prefix and every piece of bookkeeping
+            # overlaps the scatter: measured d8 3.56-3.61 vs 3.72-3.73
+            # ms/round for the 2-sync structure (the first version
+            # synced past the scatter and was SLOWER - see
+            # profiles/README.md pass 5/8). RXGB_ONE_SYNC=0 restores
+            # the 2-sync path.
             use_fused = (
                 self.device.type == "cuda"
-                and _os2.environ.get("RXGB_ONE_SYNC") == "1"
+                and _os2.environ.get("RXGB_ONE_SYNC", "1") != "0"
             )
             lc_full = None
             allowed_m = (
