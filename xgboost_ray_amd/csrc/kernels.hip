@@ -1723,7 +1723,11 @@ std::vector<torch::Tensor> partition_rows_from_packed(
   const int K = (int)packed.size(0);     // frontier bound (scan-slot count)
   auto dev = bins.device();
   auto ridx_out = ridx.clone();
-  auto gseg_out = gseg.clone();
+  // gseg needs no clone: the scatter fully rewrites every SPLIT
+  // segment, and a leaf segment's gradient pairs are never read again
+  // (ridx IS read for the end-of-round leaf margin update, so it keeps
+  // the clone). Saves an 88 MB device copy per depth at 11M rows.
+  auto gseg_out = torch::empty_like(gseg);
   auto optsl = torch::TensorOptions().dtype(torch::kInt64).device(dev);
   auto meta = torch::empty({6 * (int64_t)K + 2}, optsl);
   auto scalars = torch::zeros({2}, optsl);
@@ -1811,11 +1815,16 @@ std::vector<torch::Tensor> partition_rows_begin(
     torch::Tensor bins, torch::Tensor ridx, torch::Tensor starts,
     torch::Tensor counts, torch::Tensor split_feat,
     torch::Tensor split_bin, torch::Tensor default_left,
-    torch::Tensor gseg, torch::Tensor bins_t) {
+    torch::Tensor gseg, torch::Tensor bins_t, bool gseg_full_rewrite) {
   const int K = (int)starts.size(0);
   auto dev = bins.device();
   auto ridx_out = ridx.clone();
-  auto gseg_out = gseg.clone();
+  // Depthwise growth rewrites every still-live gseg segment each depth,
+  // so its output needs no clone (see partition_rows_from_packed).
+  // Lossguide partitions ONE node per expansion and keeps reading the
+  // other candidates''' segments -> it must clone.
+  auto gseg_out =
+      gseg_full_rewrite ? torch::empty_like(gseg) : gseg.clone();
   auto empty_i64 = torch::zeros({K > 0 ? K : 1}, torch::kInt64);
   if (K == 0)
     return {ridx_out, gseg_out, empty_i64, empty_i64, empty_i64,
@@ -1910,7 +1919,8 @@ std::vector<torch::Tensor> partition_rows(torch::Tensor bins, torch::Tensor ridx
                                           torch::Tensor gseg,
                                           torch::Tensor bins_t) {
   auto st = partition_rows_begin(bins, ridx, starts, counts, split_feat,
-                                 split_bin, default_left, gseg, bins_t);
+                                 split_bin, default_left, gseg, bins_t,
+                                 /*gseg_full_rewrite=*/false);
   if (st.size() == 6) {  // K==0 / no chunks: nothing to scatter
     return {st[0], torch::zeros({(int64_t)starts.size(0)}, torch::kInt64),
             st[1]};

@@ -222,7 +222,8 @@ def partition_rows_from_packed(bins, ridx, starts_ord, counts_ord, packed,
 
 
 def partition_begin(bins, ridx, starts, counts, split_feat, split_bin,
-                    default_left, gpair_seg=None, bins_t=None):
+                    default_left, gpair_seg=None, bins_t=None,
+                    gseg_full_rewrite=True):
     """Launch the partition's count+prefix kernels and return a context
     immediately: host bookkeeping between begin and finish overlaps
     them (and the tree writes after finish overlap the scatter)."""
@@ -233,7 +234,7 @@ def partition_begin(bins, ridx, starts, counts, split_feat, split_bin,
         gpair_seg = torch.zeros((0, 2), dtype=torch.int32, device=dev)
     st = _load().partition_rows_begin(
         bins, ridx, starts, counts, split_feat, split_bin, default_left,
-        gpair_seg, bins_t,
+        gpair_seg, bins_t, bool(gseg_full_rewrite),
     )
     return (ridx, gpair_seg, st)
 
