@@ -1135,8 +1135,10 @@ __global__ void predict_trees_kernel(
 #define MARGIN_CHUNK (PART_THREADS * MARGIN_ROWS_PER_THREAD)
 __global__ void update_margins_kernel(
     float* __restrict__ margin, const int32_t* __restrict__ ridx,
+    const int32_t* __restrict__ ridx_b,      // second ping-pong buffer
     const int64_t* __restrict__ node_start,  // [K] + counts at [K..2K)
     const int64_t* __restrict__ chunk_off,   // [K+1]
+    const int64_t* __restrict__ parity,      // [K] 0 -> ridx, 1 -> ridx_b
     const float* __restrict__ leaf_vals, int K) {
   int wg = blockIdx.x;
   int lo = 0, hi = K;
@@ -1149,6 +1151,7 @@ __global__ void update_margins_kernel(
   const int64_t row_lo = chunk_in_node * MARGIN_CHUNK;
   const int64_t count = node_start[K + node];
   const int64_t seg_start = node_start[node];
+  if (parity != nullptr && parity[node]) ridx = ridx_b;
   const float v = leaf_vals[node];
   int32_t rv[MARGIN_ROWS_PER_THREAD];
   bool valid[MARGIN_ROWS_PER_THREAD];
@@ -1719,10 +1722,14 @@ __global__ __launch_bounds__(1024) void plan_partition_kernel(
 std::vector<torch::Tensor> partition_rows_from_packed(
     torch::Tensor bins, torch::Tensor ridx, torch::Tensor starts_ord,
     torch::Tensor counts_ord, torch::Tensor packed, torch::Tensor gseg,
-    torch::Tensor bins_t, int64_t chunk_bound) {
+    torch::Tensor bins_t, int64_t chunk_bound, torch::Tensor ridx_dest) {
   const int K = (int)packed.size(0);     // frontier bound (scan-slot count)
   auto dev = bins.device();
-  auto ridx_out = ridx.clone();
+  // ping-pong destination: the caller tracks which buffer each leaf
+  // segment's rows last landed in (parity) for the margin update, so
+  // no full clone is needed (the 44 MB stream-ordered copy per depth
+  // serialized the loop like the gseg clone did)
+  auto ridx_out = ridx_dest.numel() ? ridx_dest : ridx.clone();
   // gseg needs no clone: the scatter fully rewrites every SPLIT
   // segment, and a leaf segment's gradient pairs are never read again
   // (ridx IS read for the end-of-round leaf margin update, so it keeps
@@ -2056,7 +2063,8 @@ void predict_trees(torch::Tensor X, torch::Tensor feat, torch::Tensor thr,
 
 void update_margins(torch::Tensor margin, torch::Tensor ridx,
                     torch::Tensor starts, torch::Tensor counts,
-                    torch::Tensor leaf_vals) {
+                    torch::Tensor leaf_vals, torch::Tensor ridx_b,
+                    torch::Tensor parity) {
   const int K = (int)starts.size(0);
   if (K == 0) return;
   auto dev = margin.device();
@@ -2074,8 +2082,14 @@ void update_margins(torch::Tensor margin, torch::Tensor ridx,
   if (total_chunks == 0) return;
   auto stream0 = c10::hip::getCurrentHIPStream();
   static thread_local PinnedStager margin_meta_stager;
-  auto meta_cpu = margin_meta_stager.get(3 * K + 1);
-  cat_into_pinned(meta_cpu, {starts_cpu, counts_cpu, chunk_off_cpu});
+  const bool have_parity = parity.numel() > 0;
+  auto meta_cpu = margin_meta_stager.get((have_parity ? 4 : 3) * K + 1);
+  if (have_parity) {
+    cat_into_pinned(meta_cpu, {starts_cpu, counts_cpu, chunk_off_cpu,
+                               parity.to(torch::kCPU).to(torch::kInt64)});
+  } else {
+    cat_into_pinned(meta_cpu, {starts_cpu, counts_cpu, chunk_off_cpu});
+  }
   auto meta = meta_cpu.to(dev, /*non_blocking=*/true);
   margin_meta_stager.mark(stream0.stream());
   int64_t* mp = meta.data_ptr<int64_t>();
@@ -2090,7 +2104,10 @@ void update_margins(torch::Tensor margin, torch::Tensor ridx,
   hipLaunchKernelGGL(update_margins_kernel, dim3((uint32_t)total_chunks),
                      dim3(PART_THREADS), 0, stream.stream(),
                      margin.data_ptr<float>(), ridx.data_ptr<int32_t>(),
+                     ridx_b.numel() ? ridx_b.data_ptr<int32_t>()
+                                    : ridx.data_ptr<int32_t>(),
                      mp, mp + 2 * K,
+                     have_parity ? mp + 3 * K + 1 : nullptr,
                      lv.data_ptr<float>(), K);
 }
 

@@ -771,6 +771,19 @@ class BoostingEngine:
 
         import os as _os2
 
+        # single-sync fused loop decision is per-tree (see the in-loop
+        # comment); the ping-pong ridx buffers exist ONLY for it - the
+        # 2-sync path reassigns ridx to fresh clones each depth, which
+        # would leave buffer 0 stale for the margin update
+        use_fused_loop = (
+            self.device.type == "cuda"
+            and _os2.environ.get("RXGB_ONE_SYNC", "1") != "0"
+        )
+        ridx_bufs = None
+        cur_buf = 0
+        if use_fused_loop:
+            ridx_bufs = [ridx, torch.empty_like(ridx)]
+
         lam = float(self.p.reg_lambda)
         alpha = float(self.p.reg_alpha)
         prev_all_hist: Optional[torch.Tensor] = None
@@ -918,10 +931,7 @@ class BoostingEngine:
             # synced past the scatter and was SLOWER - see
             # profiles/README.md pass 5/8). RXGB_ONE_SYNC=0 restores
             # the 2-sync path.
-            use_fused = (
-                self.device.type == "cuda"
-                and _os2.environ.get("RXGB_ONE_SYNC", "1") != "0"
-            )
+            use_fused = use_fused_loop
             lc_full = None
             allowed_m = (
                 self._allowed_mask(paths_ord) if track_paths else None
@@ -944,8 +954,11 @@ class BoostingEngine:
                         packed_dev, gseg,
                         getattr(self.dtrain, "bins_t", None),
                         chunk_bound,
+                        ridx_dest=ridx_bufs[1 - cur_buf],
                     )
                 )
+                depth_in_buf = cur_buf
+                cur_buf = 1 - cur_buf
                 # event recorded between the pull D2H and the scatter:
                 # the host wakes before the scatter and every piece of
                 # bookkeeping below overlaps it
@@ -998,6 +1011,7 @@ class BoostingEngine:
                         sumh_ord[leaf_idx], start_ord[leaf_idx],
                         count_ord[leaf_idx], wlo_ord[leaf_idx],
                         whi_ord[leaf_idx], scale_h,
+                        parity=(depth_in_buf if use_fused else 0),
                     )
                 fr_nid = np.zeros(0, np.int64)
                 break
@@ -1029,6 +1043,7 @@ class BoostingEngine:
                     sumh_ord[leaf_idx], start_ord[leaf_idx],
                     count_ord[leaf_idx], wlo_ord[leaf_idx],
                     whi_ord[leaf_idx], scale_h,
+                    parity=(depth_in_buf if use_fused else 0),
                 )
 
             # child ids in scan-slot order (matches the sequential
@@ -1154,21 +1169,24 @@ class BoostingEngine:
         if fr_nid.size:
             self._finalize_leaves_soa(
                 ta, fr_nid, fr_sumg, fr_sumh, fr_start, fr_count,
-                fr_wlo, fr_whi, scale_h,
+                fr_wlo, fr_whi, scale_h, parity=cur_buf,
             )
         # leaf margin update: every row's final node is its segment's
         # node; _finalize_leaves_soa collected (start, count, value) segs
-        leaf_starts = [s for (s, c, v) in self._leaf_segs]
-        leaf_counts = [c for (s, c, v) in self._leaf_segs]
-        leaf_vals = [v for (s, c, v) in self._leaf_segs]
+        leaf_starts = [s for (s, c, v, bp) in self._leaf_segs]
+        leaf_counts = [c for (s, c, v, bp) in self._leaf_segs]
+        leaf_vals = [v for (s, c, v, bp) in self._leaf_segs]
+        leaf_parity = [bp for (s, c, v, bp) in self._leaf_segs]
         _tick("leaf_host")
         if leaf_starts:
             ops.update_margins(
                 self.margin if self.n_class == 1 else self.margin[:, cls],
-                ridx,
+                ridx_bufs[0] if ridx_bufs is not None else ridx,
                 torch.tensor(leaf_starts, dtype=torch.int64),
                 torch.tensor(leaf_counts, dtype=torch.int64),
                 np.asarray(leaf_vals, dtype=np.float32),
+                ridx_b=(ridx_bufs[1] if ridx_bufs is not None else None),
+                parity=(leaf_parity if ridx_bufs is not None else None),
             )
         self._leaf_segs = []
         _tick("margins")
@@ -1194,7 +1212,7 @@ class BoostingEngine:
 
 
     def _finalize_leaves_soa(self, ta, nids, sumg, sumh, starts, counts,
-                             wlo, whi, scale_h):
+                             wlo, whi, scale_h, parity=0):
         """SoA leaf finalization: same math/op order as
         _finalize_leaves_batch, writing into _TreeArrays and collecting
         (start, count, value) margin segments."""
@@ -1217,7 +1235,8 @@ class BoostingEngine:
         ta.val[nids] = v
         ta.cover[nids] = H
         self._leaf_segs.extend(
-            zip(starts.tolist(), counts.tolist(), v.tolist())
+            zip(starts.tolist(), counts.tolist(), v.tolist(),
+                [parity] * int(nids.size))
         )
 
     def _finalize_leaves_batch(self, nodes, val_l, cover_l, scale_h):
@@ -1254,7 +1273,7 @@ class BoostingEngine:
         for nd, vi, Hi in zip(nodes, v.tolist(), H.tolist()):
             val_l[nd.nid] = vi
             cover_l[nd.nid] = Hi
-            segs.append((nd.start, nd.count, vi))
+            segs.append((nd.start, nd.count, vi, 0))
 
     def _finalize_leaf(self, nd: _Node, val_l, cover_l, scale_h):
         G = nd.sum_g / self._scale_g_cur
@@ -1266,7 +1285,7 @@ class BoostingEngine:
             w = max(-self.p.max_delta_step, min(self.p.max_delta_step, w))
         val_l[nd.nid] = self.p.eta * w
         cover_l[nd.nid] = H
-        self._leaf_segs.append((nd.start, nd.count, self.p.eta * w))
+        self._leaf_segs.append((nd.start, nd.count, self.p.eta * w, 0))
 
     # -- evaluation ----------------------------------------------------------
     def eval_sets(self, evals: Sequence[EvalPack], feval=None) -> Dict[str, Dict[str, float]]:
@@ -1562,9 +1581,9 @@ class BoostingEngine:
             _, _, rec = heapq.heappop(heap)
             self._finalize_leaf_rec(rec, val_l, cover_l, scale_h)
 
-        leaf_starts = [st for (st, c, v) in self._leaf_segs]
-        leaf_counts = [c for (st, c, v) in self._leaf_segs]
-        leaf_vals = [v for (st, c, v) in self._leaf_segs]
+        leaf_starts = [st for (st, c, v, bp) in self._leaf_segs]
+        leaf_counts = [c for (st, c, v, bp) in self._leaf_segs]
+        leaf_vals = [v for (st, c, v, bp) in self._leaf_segs]
         if leaf_starts:
             ops.update_margins(
                 self.margin if self.n_class == 1 else self.margin[:, cls],
@@ -1596,7 +1615,9 @@ class BoostingEngine:
             w = max(-self.p.max_delta_step, min(self.p.max_delta_step, w))
         val_l[nid] = self.p.eta * w
         cover_l[nid] = H
-        self._leaf_segs.append((rec["start"], rec["count"], self.p.eta * w))
+        self._leaf_segs.append(
+            (rec["start"], rec["count"], self.p.eta * w, 0)
+        )
 
     # quantization scale of the current tree (set in _grow_tree via _quantize)
     _scale_g_cur: float = 1.0

@@ -145,13 +145,13 @@ def partition_finish(ctx):
 
 
 def partition_rows_from_packed(bins, ridx, starts_ord, counts_ord, packed,
-                               gseg, bins_t, chunk_bound):
+                               gseg, bins_t, chunk_bound, ridx_dest=None):
     """Single-sync fused partition (GPU only): see ops.gpu."""
     from xgboost_ray_amd.ops import gpu
 
     return gpu.partition_rows_from_packed(
         bins, ridx, starts_ord, counts_ord, packed, gseg, bins_t,
-        chunk_bound,
+        chunk_bound, ridx_dest,
     )
 
 
@@ -188,9 +188,22 @@ def predict_trees(
     )
 
 
-def update_margins(margin, ridx, starts, counts, leaf_values):
-    """margin[ridx[seg_k]] += leaf_values[k] for each final-leaf segment."""
-    return _impl(margin).update_margins(margin, ridx, starts, counts, leaf_values)
+def update_margins(margin, ridx, starts, counts, leaf_values,
+                   ridx_b=None, parity=None):
+    """margin[ridx[seg_k]] += leaf_values[k] for each final-leaf segment.
+
+    ridx_b/parity (GPU): ping-pong leaf buffers - parity[k] selects the
+    buffer node k's rows last landed in."""
+    if margin.is_cuda:
+        from xgboost_ray_amd.ops import gpu
+
+        return gpu.update_margins(
+            margin, ridx, starts, counts, leaf_values,
+            ridx_b=ridx_b, parity=parity,
+        )
+    return _impl(margin).update_margins(
+        margin, ridx, starts, counts, leaf_values
+    )
 
 
 def grad_fused(margin, label, weight, scale_pos_weight, mode):

@@ -166,13 +166,21 @@ def lambdarank_grad(margin, label, group_ptr, rank, idcg, use_ndcg):
     )
 
 
-def update_margins(margin, ridx, starts, counts, leaf_values):
+def update_margins(margin, ridx, starts, counts, leaf_values,
+                   ridx_b=None, parity=None):
     # starts/counts/leaf_values stay on the HOST: the extension stages
     # all control data through ONE pinned H2D (a .to(dev) here forced a
-    # pointless H2D + synchronizing D2H round-trip per round)
+    # pointless H2D + synchronizing D2H round-trip per round).
+    # ridx_b/parity: ping-pong leaf buffers (parity[k] selects which
+    # buffer node k's rows last landed in).
     lv = torch.as_tensor(leaf_values, dtype=torch.float32)
+    rb = ridx_b if ridx_b is not None else torch.zeros(
+        0, dtype=torch.int32, device=margin.device)
+    par = (torch.as_tensor(parity, dtype=torch.int64)
+           if parity is not None
+           else torch.zeros(0, dtype=torch.int64))
     target = margin if margin.is_contiguous() else margin.contiguous()
-    _load().update_margins(target, ridx, starts, counts, lv)
+    _load().update_margins(target, ridx, starts, counts, lv, rb, par)
     if target is not margin:
         margin.copy_(target)
     return margin
@@ -196,7 +204,7 @@ PART_CHUNK = 2048
 
 
 def partition_rows_from_packed(bins, ridx, starts_ord, counts_ord, packed,
-                               gseg, bins_t, chunk_bound):
+                               gseg, bins_t, chunk_bound, ridx_dest=None):
     """Single-sync partition: consumes find_splits' device packed output
     (plan + count + prefix + scatter all device-planned); returns
     (ridx_out, gseg_out, pull) where `pull` is a pinned i64 [7K] buffer
@@ -206,9 +214,11 @@ def partition_rows_from_packed(bins, ridx, starts_ord, counts_ord, packed,
         bins_t = torch.zeros(0, dtype=torch.uint8, device=dev)
     if gseg is None:
         gseg = torch.zeros((0, 2), dtype=torch.int32, device=dev)
+    rd = ridx_dest if ridx_dest is not None else torch.zeros(
+        0, dtype=torch.int32, device=bins.device)
     out = _load().partition_rows_from_packed(
         bins, ridx, starts_ord, counts_ord, packed, gseg, bins_t,
-        int(chunk_bound),
+        int(chunk_bound), rd,
     )
     # out: [ridx_out, gseg_out, pull, meta, scalars, left_before,
     #       node_left_total, flags, chunk_bound]
