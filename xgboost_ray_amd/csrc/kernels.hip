@@ -2371,6 +2371,77 @@ torch::Tensor eval_auc_hist(torch::Tensor margin, torch::Tensor label,
   return hist;  // [neg | pos] planes
 }
 
+
+// ---------------------------------------------------------------------------
+// depth_step: one C++ call runs a whole depth of the fused single-sync
+// loop - stage control data, zero+build the histogram, derive siblings,
+// scan, plan+partition, enqueue the pull and record the pull event,
+// then launch the scatter. The python loop's ~30 torch/pybind dispatches
+// per depth (~100 us of host time each depth) collapse to one call +
+// one wait + numpy bookkeeping. Single-GPU only: the distributed path
+// keeps the python structure (it interleaves the RCCL allreduce).
+// ---------------------------------------------------------------------------
+static hipEvent_t g_pull_event = nullptr;
+
+void wait_pull_event() {
+  if (g_pull_event) (void)hipEventSynchronize(g_pull_event);
+}
+
+std::vector<torch::Tensor> depth_step(
+    torch::Tensor bins, torch::Tensor bins_t, torch::Tensor gseg,
+    torch::Tensor ridx, torch::Tensor ridx_dest, torch::Tensor prev_hist,
+    torch::Tensor hist, torch::Tensor depth_meta_cpu, int64_t KK,
+    int64_t nd, int64_t K_built, torch::Tensor starts_cpu,
+    torch::Tensor counts_cpu, torch::Tensor fb, double scale_g,
+    double scale_h, double lam, double alpha, double gamma, double mcw,
+    torch::Tensor mono, torch::Tensor bounds, torch::Tensor allowed,
+    int64_t n_bins, int64_t chunk_bound) {
+  auto dev = bins.device();
+  auto stream = c10::hip::getCurrentHIPStream();
+  const int F = (int)bins.size(1);
+
+  // ONE pinned H2D for all of this depth's control data
+  static thread_local PinnedStager depth_meta_stager;
+  auto meta_cpu_pin = depth_meta_stager.get(depth_meta_cpu.numel());
+  meta_cpu_pin.copy_(depth_meta_cpu);
+  auto dmeta = meta_cpu_pin.to(dev, /*non_blocking=*/true);
+  depth_meta_stager.mark(stream.stream());
+  // layout: [sumg KK | sumh KK | starts KK | counts KK | pslot nd | sib nd]
+  auto pg = dmeta.narrow(0, 0, KK);
+  auto ph = dmeta.narrow(0, KK, KK);
+  auto d_starts = dmeta.narrow(0, 2 * KK, KK);
+  auto d_counts = dmeta.narrow(0, 3 * KK, KK);
+
+  // built block: zero + histogram
+  hist.narrow(0, 0, K_built).zero_();
+  build_histogram(bins, gseg, ridx, starts_cpu, counts_cpu, n_bins, 0, F,
+                  hist, /*pregathered=*/true);
+  // sibling = parent - built
+  if (nd > 0) {
+    auto pslots = dmeta.narrow(0, 4 * KK, nd);
+    auto spos = dmeta.narrow(0, 4 * KK + nd, nd);
+    auto derived = hist.narrow(0, K_built, nd);
+    torch::sub_out(derived, prev_hist.index_select(0, pslots),
+                   hist.index_select(0, spos));
+  }
+  // scan
+  auto fs = find_splits(hist, pg, ph, fb, scale_g, scale_h, lam, alpha,
+                        gamma, mcw, mono, bounds, allowed,
+                        /*pull=*/false);
+  auto packed = fs[0];
+  // device-planned partition (writes into ridx_dest, no clones)
+  auto st = partition_rows_from_packed(bins, ridx, d_starts, d_counts,
+                                       packed, gseg, bins_t, chunk_bound,
+                                       ridx_dest);
+  // record the pull event BETWEEN the pull copies and the scatter
+  if (!g_pull_event)
+    (void)hipEventCreateWithFlags(&g_pull_event, hipEventDisableTiming);
+  (void)hipEventRecord(g_pull_event, stream.stream());
+  partition_scatter_from_packed(ridx, st[0], gseg, st[1], st[3], st[4],
+                                st[5], st[6], st[7], st[8]);
+  return {st[0], st[1], st[2]};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("quantize_gpair", &quantize_gpair, "quantize gradient pairs");
   m.def("grad_fused", &grad_fused, "fused objective gradient + absmax");
@@ -2388,6 +2459,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "device-planned partition consuming find_splits packed output");
   m.def("partition_scatter_from_packed", &partition_scatter_from_packed,
         "scatter phase of the device-planned partition");
+  m.def("depth_step", &depth_step,
+        "one fused single-sync depth (single-GPU fast path)");
+  m.def("wait_pull_event", &wait_pull_event,
+        "host wait for the depth_step pull event");
   m.def("predict_trees", &predict_trees, "tree-walk prediction");
   m.def("update_margins", &update_margins, "leaf margin update");
   m.def("lambdarank_grad", &lambdarank_grad, "pairwise lambdarank gradients");

@@ -826,82 +826,10 @@ class BoostingEngine:
 
             starts = torch.from_numpy(np.ascontiguousarray(start_ord[:K]))
             counts = torch.from_numpy(np.ascontiguousarray(count_ord[:K]))
-
-            F = self.dtrain.n_features
-            all_hist = torch.empty(
-                (nF, F, self.n_bins, 2), dtype=torch.int64, device=self.device
-            )
-            hist = all_hist[:K]  # built block (contiguous leading slice)
-            hist.zero_()  # derived rows are fully overwritten by the sub
-            overlap = self.coll.is_distributed and (
-                (
-                    self.device.type == "cuda"
-                    and self.dtrain.bins.stride(0) % 16 == 0
-                    and F > 16
-                )
-                or _os2.environ.get("RXGB_FORCE_OVERLAP_ALLREDUCE") == "1"
-            )
-            if overlap:
-                # overlap the RCCL AllReduce of each finished feature block
-                # with the build of the next (BASELINE north-star: side-
-                # stream collective over xGMI behind the histogram build)
-                n_chunks = min(4, (F + 31) // 32)
-                step = ((F + n_chunks - 1) // n_chunks + 15) // 16 * 16
-                pending = []
-                f0 = 0
-                while f0 < F:
-                    f1 = min(f0 + step, F)
-                    ops.build_histogram(
-                        self.dtrain.bins, gseg, ridx, starts, counts,
-                        self.n_bins, f_range=(f0, f1), out=hist,
-                        pregathered=True,
-                    )
-                    sl = hist[:, f0:f1].contiguous()
-                    pending.append((self.coll.allreduce_async(sl), sl, f0, f1))
-                    f0 = f1
-                for h, sl, c0, c1 in pending:
-                    h.wait()
-                    hist[:, c0:c1].copy_(sl)
-            else:
-                ops.build_histogram(
-                    self.dtrain.bins, gseg, ridx, starts, counts,
-                    self.n_bins, out=hist, pregathered=True,
-                )
-                _tick("hist")
-                if self.coll.is_distributed:
-                    self.coll.allreduce_(hist)
-            _tick("allreduce")
-            # ---- ONE H2D for ALL of this depth's control data:
-            # [parent_g (KK) | parent_h (KK) | derive parent slots (nd) |
-            #  derive sibling positions (nd)] - separate stages cost one
-            # pinned copy + stream op each
             KK = sumg_ord.size
             nd = derive_sib_pos.size
-            if self.device.type == "cuda":
-                depth_meta = self._stage_i64(
-                    np.concatenate(
-                        [sumg_ord, sumh_ord, start_ord, count_ord,
-                         derive_pslot, derive_sib_pos]
-                    ),
-                    "depth_meta",
-                )
-                pg, ph = depth_meta[:KK], depth_meta[KK : 2 * KK]
-                d_starts = depth_meta[2 * KK : 3 * KK]
-                d_counts = depth_meta[3 * KK : 4 * KK]
-                pslots = depth_meta[4 * KK : 4 * KK + nd]
-                spos = depth_meta[4 * KK + nd :]
-            else:
-                pg = torch.from_numpy(sumg_ord)
-                ph = torch.from_numpy(sumh_ord)
-                pslots = torch.from_numpy(derive_pslot)
-                spos = torch.from_numpy(derive_sib_pos)
-            if nd:
-                # sibling = parent - built, batched over all pairs
-                torch.sub(
-                    prev_all_hist.index_select(0, pslots),
-                    all_hist.index_select(0, spos),
-                    out=all_hist[K:],
-                )
+
+            # feature gate needed by both the python path and depth_step
             fb = self.feat_bins
             mask = feat_mask
             if self.p.colsample_bylevel < 1.0:
@@ -911,58 +839,80 @@ class BoostingEngine:
                 fb = torch.where(
                     mask, self.feat_bins, torch.zeros_like(self.feat_bins)
                 )
-            _tick("stack")
-            mono_bounds = None
-            if self.mono is not None:
-                mono_bounds = torch.from_numpy(
-                    np.ascontiguousarray(
-                        np.stack([wlo_ord, whi_ord], axis=1)
-                    )
-                )
-            # single-sync fused path (GPU, default): the split scan's
-            # packed output feeds a device-planned partition directly;
-            # ONE pinned D2H brings {splits | left_counts} to the host,
-            # which replays the identical split predicate for
-            # bookkeeping. The pull is enqueued BEFORE the scatter and
-            # the host waits on an inter-launch EVENT, so it wakes
-            # right after count+prefix and every piece of bookkeeping
-            # overlaps the scatter: measured d8 3.56-3.61 vs 3.72-3.73
-            # ms/round for the 2-sync structure (the first version
-            # synced past the scatter and was SLOWER - see
-            # profiles/README.md pass 5/8). RXGB_ONE_SYNC=0 restores
-            # the 2-sync path.
-            use_fused = use_fused_loop
-            lc_full = None
             allowed_m = (
                 self._allowed_mask(paths_ord) if track_paths else None
             )
-            if use_fused:
-                packed_dev = ops.find_splits(
-                    all_hist, pg, ph, fb, scale_g, scale_h,
-                    self.p.reg_lambda, self.p.reg_alpha, self.p.gamma,
-                    self.p.min_child_weight, monotone=self.mono,
-                    bounds=mono_bounds, allowed=allowed_m, pull=False,
+
+            F = self.dtrain.n_features
+            all_hist = torch.empty(
+                (nF, F, self.n_bins, 2), dtype=torch.int64, device=self.device
+            )
+
+            use_depth_step = (
+                use_fused_loop and not self.coll.is_distributed
+            )
+            if use_depth_step:
+                # one C++ call per depth: stage + zero + hist + derive +
+                # scan + device-planned partition + pull + scatter. The
+                # ~30 per-depth python/torch dispatches were ~0.8 ms of
+                # wall per round at depth 8.
+                dm_t = torch.from_numpy(
+                    np.concatenate(
+                        [sumg_ord, sumh_ord, start_ord, count_ord,
+                         derive_pslot, derive_sib_pos]
+                    )
                 )
-                _tick("scan")
-                # grid BOUND: every split node adds at most one
-                # partial chunk (PART_CHUNK = 2048 rows/workgroup)
+                # C++ find_splits takes raw pointers: everything must
+                # already be the right dtype ON DEVICE
+                mono_t = (
+                    self.mono.to(self.device).to(torch.int8)
+                    if self.mono is not None
+                    else torch.zeros(0, dtype=torch.int8,
+                                     device=self.device)
+                )
+                bounds_t = (
+                    torch.from_numpy(np.ascontiguousarray(
+                        np.stack([wlo_ord, whi_ord], axis=1)
+                    )).to(self.device)
+                    if self.mono is not None
+                    else torch.zeros((0, 2), dtype=torch.float64,
+                                     device=self.device)
+                )
+                allowed_t = (
+                    allowed_m.to(self.device).to(torch.uint8).contiguous()
+                    if allowed_m is not None
+                    else torch.zeros(0, dtype=torch.uint8,
+                                     device=self.device)
+                )
+                bins_t_t = getattr(self.dtrain, "bins_t", None)
+                if bins_t_t is None:
+                    bins_t_t = torch.zeros(0, dtype=torch.uint8,
+                                           device=self.device)
+                gseg_in = (
+                    gseg if gseg is not None
+                    else torch.zeros((0, 2), dtype=torch.int32,
+                                     device=self.device)
+                )
                 frontier_rows = int(count_ord.sum())
                 chunk_bound = (frontier_rows + 2047) // 2048 + KK
-                ridx, gseg, pull, pull_ev = (
-                    ops.partition_rows_from_packed(
-                        self.dtrain.bins, ridx, d_starts, d_counts,
-                        packed_dev, gseg,
-                        getattr(self.dtrain, "bins_t", None),
-                        chunk_bound,
-                        ridx_dest=ridx_bufs[1 - cur_buf],
-                    )
+                from xgboost_ray_amd.ops import gpu as _gpu
+
+                ridx, gseg, pull = _gpu._load().depth_step(
+                    self.dtrain.bins, bins_t_t, gseg_in, ridx,
+                    ridx_bufs[1 - cur_buf],
+                    prev_all_hist if prev_all_hist is not None
+                    else all_hist,
+                    all_hist, dm_t, KK, int(nd), K, starts, counts,
+                    fb.to(self.device).to(torch.int32), scale_g,
+                    scale_h, self.p.reg_lambda,
+                    self.p.reg_alpha, self.p.gamma,
+                    self.p.min_child_weight, mono_t, bounds_t,
+                    allowed_t, self.n_bins, chunk_bound,
                 )
                 depth_in_buf = cur_buf
                 cur_buf = 1 - cur_buf
-                # event recorded between the pull D2H and the scatter:
-                # the host wakes before the scatter and every piece of
-                # bookkeeping below overlaps it
-                pull_ev.synchronize()
+                _tick("scan")
+                _gpu._load().wait_pull_event()
                 arr = pull[: 7 * KK].numpy()
                 pk = arr[: 6 * KK].reshape(KK, 6)
                 gain = pk[:, 0].astype(np.int32).view(np.float32).copy()
@@ -972,31 +922,163 @@ class BoostingEngine:
                 blg = pk[:, 4].copy()
                 blh = pk[:, 5].copy()
                 lc_full = arr[6 * KK : 7 * KK].copy()
+                use_fused = True
                 _tick("scan_pull")
+                return_to_decode = True
             else:
-                best = ops.find_splits(
-                    all_hist,
-                    pg,
-                    ph,
-                    fb,
-                    scale_g,
-                    scale_h,
-                    self.p.reg_lambda,
-                    self.p.reg_alpha,
-                    self.p.gamma,
-                    self.p.min_child_weight,
-                    monotone=self.mono,
-                    bounds=mono_bounds,
-                    allowed=allowed_m,
+                return_to_decode = False
+
+            if not return_to_decode:
+                hist = all_hist[:K]  # built block (contiguous slice)
+                hist.zero_()  # derived rows overwritten by the sub
+                overlap = self.coll.is_distributed and (
+                    (
+                        self.device.type == "cuda"
+                        and self.dtrain.bins.stride(0) % 16 == 0
+                        and F > 16
+                    )
+                    or _os2.environ.get("RXGB_FORCE_OVERLAP_ALLREDUCE") == "1"
                 )
-                _tick("scan")
-                gain = best["gain"]
-                _tick("scan_pull")
-                bfeat = best["feature"]
-                bbin = best["bin"]
-                bdl = best["default_left"]
-                blg = best["left_g"]
-                blh = best["left_h"]
+                if overlap:
+                    # overlap the RCCL AllReduce of each finished feature block
+                    # with the build of the next (BASELINE north-star: side-
+                    # stream collective over xGMI behind the histogram build)
+                    n_chunks = min(4, (F + 31) // 32)
+                    step = ((F + n_chunks - 1) // n_chunks + 15) // 16 * 16
+                    pending = []
+                    f0 = 0
+                    while f0 < F:
+                        f1 = min(f0 + step, F)
+                        ops.build_histogram(
+                            self.dtrain.bins, gseg, ridx, starts, counts,
+                            self.n_bins, f_range=(f0, f1), out=hist,
+                            pregathered=True,
+                        )
+                        sl = hist[:, f0:f1].contiguous()
+                        pending.append((self.coll.allreduce_async(sl), sl, f0, f1))
+                        f0 = f1
+                    for h, sl, c0, c1 in pending:
+                        h.wait()
+                        hist[:, c0:c1].copy_(sl)
+                else:
+                    ops.build_histogram(
+                        self.dtrain.bins, gseg, ridx, starts, counts,
+                        self.n_bins, out=hist, pregathered=True,
+                    )
+                    _tick("hist")
+                    if self.coll.is_distributed:
+                        self.coll.allreduce_(hist)
+                _tick("allreduce")
+                # ONE H2D for this depth's control data (layout in comment
+                # above depth_step)
+                if self.device.type == "cuda":
+                    depth_meta = self._stage_i64(
+                        np.concatenate(
+                            [sumg_ord, sumh_ord, start_ord, count_ord,
+                             derive_pslot, derive_sib_pos]
+                        ),
+                        "depth_meta",
+                    )
+                    pg, ph = depth_meta[:KK], depth_meta[KK : 2 * KK]
+                    d_starts = depth_meta[2 * KK : 3 * KK]
+                    d_counts = depth_meta[3 * KK : 4 * KK]
+                    pslots = depth_meta[4 * KK : 4 * KK + nd]
+                    spos = depth_meta[4 * KK + nd :]
+                else:
+                    pg = torch.from_numpy(sumg_ord)
+                    ph = torch.from_numpy(sumh_ord)
+                    pslots = torch.from_numpy(derive_pslot)
+                    spos = torch.from_numpy(derive_sib_pos)
+                if nd:
+                    # sibling = parent - built, batched over all pairs
+                    torch.sub(
+                        prev_all_hist.index_select(0, pslots),
+                        all_hist.index_select(0, spos),
+                        out=all_hist[K:],
+                    )
+                _tick("stack")
+                mono_bounds = None
+                if self.mono is not None:
+                    mono_bounds = torch.from_numpy(
+                        np.ascontiguousarray(
+                            np.stack([wlo_ord, whi_ord], axis=1)
+                        )
+                    )
+                # single-sync fused path (GPU, default): the split scan's
+                # packed output feeds a device-planned partition directly;
+                # ONE pinned D2H brings {splits | left_counts} to the host,
+                # which replays the identical split predicate for
+                # bookkeeping. The pull is enqueued BEFORE the scatter and
+                # the host waits on an inter-launch EVENT, so it wakes
+                # right after count+prefix and every piece of bookkeeping
+                # overlaps the scatter: measured d8 3.56-3.61 vs 3.72-3.73
+                # ms/round for the 2-sync structure (the first version
+                # synced past the scatter and was SLOWER - see
+                # profiles/README.md pass 5/8). RXGB_ONE_SYNC=0 restores
+                # the 2-sync path.
+                use_fused = use_fused_loop
+                lc_full = None
+                if use_fused:
+                    packed_dev = ops.find_splits(
+                        all_hist, pg, ph, fb, scale_g, scale_h,
+                        self.p.reg_lambda, self.p.reg_alpha, self.p.gamma,
+                        self.p.min_child_weight, monotone=self.mono,
+                        bounds=mono_bounds, allowed=allowed_m, pull=False,
+                    )
+                    _tick("scan")
+                    # grid BOUND: every split node adds at most one
+                    # partial chunk (PART_CHUNK = 2048 rows/workgroup)
+                    frontier_rows = int(count_ord.sum())
+                    chunk_bound = (frontier_rows + 2047) // 2048 + KK
+                    ridx, gseg, pull, pull_ev = (
+                        ops.partition_rows_from_packed(
+                            self.dtrain.bins, ridx, d_starts, d_counts,
+                            packed_dev, gseg,
+                            getattr(self.dtrain, "bins_t", None),
+                            chunk_bound,
+                            ridx_dest=ridx_bufs[1 - cur_buf],
+                        )
+                    )
+                    depth_in_buf = cur_buf
+                    cur_buf = 1 - cur_buf
+                    # event recorded between the pull D2H and the scatter:
+                    # the host wakes before the scatter and every piece of
+                    # bookkeeping below overlaps it
+                    pull_ev.synchronize()
+                    arr = pull[: 7 * KK].numpy()
+                    pk = arr[: 6 * KK].reshape(KK, 6)
+                    gain = pk[:, 0].astype(np.int32).view(np.float32).copy()
+                    bfeat = pk[:, 1].astype(np.int32)
+                    bbin = pk[:, 2].astype(np.int32)
+                    bdl = pk[:, 3].astype(np.uint8)
+                    blg = pk[:, 4].copy()
+                    blh = pk[:, 5].copy()
+                    lc_full = arr[6 * KK : 7 * KK].copy()
+                    _tick("scan_pull")
+                else:
+                    best = ops.find_splits(
+                        all_hist,
+                        pg,
+                        ph,
+                        fb,
+                        scale_g,
+                        scale_h,
+                        self.p.reg_lambda,
+                        self.p.reg_alpha,
+                        self.p.gamma,
+                        self.p.min_child_weight,
+                        monotone=self.mono,
+                        bounds=mono_bounds,
+                        allowed=allowed_m,
+                    )
+                    _tick("scan")
+                    gain = best["gain"]
+                    _tick("scan_pull")
+                    bfeat = best["feature"]
+                    bbin = best["bin"]
+                    bdl = best["default_left"]
+                    blg = best["left_g"]
+                    blh = best["left_h"]
 
             # ---- vectorized split/leaf decision for the whole frontier
             splits_ok = (gain > 0) & (bfeat >= 0) & np.isfinite(gain)
