@@ -850,6 +850,7 @@ class BoostingEngine:
 
             use_depth_step = (
                 use_fused_loop and not self.coll.is_distributed
+                and _os2.environ.get("RXGB_DEPTH_STEP", "1") != "0"
             )
             if use_depth_step:
                 # one C++ call per depth: stage + zero + hist + derive +
