@@ -548,3 +548,35 @@ def test_fused_eval_matches_torch():
     finally:
         os.environ.pop("RXGB_FUSED_EVAL", None)
     assert a == pytest.approx(b, rel=1e-9)
+
+
+def test_depth_step_path_bitwise():
+    """Opt-in C++ depth_step vs the python-orchestrated fused loop:
+    identical models, including monotone constraints (exercises the
+    device mono/bounds pointer plumbing)."""
+    import os
+
+    from xgboost_ray_amd.engine.quantile import BinnedMatrix
+    from xgboost_ray_amd.engine.trainer import run_training
+
+    X, y = create_data(150_000, 8, seed=6, kind="reg")
+    dm = BinnedMatrix.build(
+        torch.from_numpy(X).cuda(), label=torch.from_numpy(y).cuda(),
+        max_bin=256,
+    )
+    for params in (
+        {"objective": "reg:squarederror", "max_depth": 8, "eta": 0.3},
+        {"objective": "reg:squarederror", "max_depth": 6, "eta": 0.3,
+         "monotone_constraints": [1, -1, 0, 0, 0, 0, 0, 0]},
+    ):
+        preds = {}
+        for m in ("0", "1"):
+            os.environ["RXGB_DEPTH_STEP"] = m
+            try:
+                bst = run_training(
+                    dict(params, tree_method="gpu_hist"), dm, 5,
+                )
+                preds[m] = bst.predict(X, output_margin=True)
+            finally:
+                os.environ.pop("RXGB_DEPTH_STEP", None)
+        np.testing.assert_array_equal(preds["0"], preds["1"])
