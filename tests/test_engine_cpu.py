@@ -1115,3 +1115,24 @@ class TestBinnedAUCTolerance:
             margin, torch.from_numpy(y), w)
         assert abs(ea - ba) < 5e-3
         assert abs(ep - bp) < 5e-2
+
+
+class TestDartSampleType:
+    def test_weighted_differs_and_deterministic(self):
+        """sample_type=weighted (drop prob proportional to tree weight,
+        xgboost dart.cc) differs from uniform and stays deterministic."""
+        dm, X, y = _binned()
+        outs = {}
+        for st in ("uniform", "weighted"):
+            runs = []
+            for _ in range(2):
+                bst = run_training(
+                    {"objective": "binary:logistic", "booster": "dart",
+                     "rate_drop": 0.4, "sample_type": st,
+                     "max_depth": 4, "eta": 0.3, "seed": 5},
+                    dm, 15,
+                )
+                runs.append(bst.predict(X, output_margin=True))
+            np.testing.assert_array_equal(runs[0], runs[1])
+            outs[st] = runs[0]
+        assert not np.array_equal(outs["uniform"], outs["weighted"])

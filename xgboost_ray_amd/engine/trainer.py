@@ -66,6 +66,7 @@ class TrainParams:
     skip_drop: float = 0.0
     one_drop: bool = False
     normalize_type: str = "tree"
+    sample_type: str = "uniform"  # or "weighted" (xgboost dart.cc)
     verbosity: int = 1
     nthread: int = 0
     disable_default_eval_metric: bool = False
@@ -552,7 +553,23 @@ class BoostingEngine:
         )
         if n_trees == 0 or rng.rand() < float(self.p.skip_drop):
             return ([], None)
-        drop = np.nonzero(rng.rand(n_trees) < float(self.p.rate_drop))[0]
+        if self.p.sample_type == "weighted":
+            # xgboost dart.cc weighted selection: drop probability
+            # proportional to the tree's current weight,
+            # p_i = w_i * rate_drop * n / sum(w), clamped to 1
+            w = np.asarray(self._dart_weights(n_trees), dtype=np.float64)
+            sw = float(w.sum())
+            if sw <= 0:
+                p_i = np.full(n_trees, float(self.p.rate_drop))
+            else:
+                p_i = np.clip(
+                    w * float(self.p.rate_drop) * n_trees / sw, 0.0, 1.0
+                )
+            drop = np.nonzero(rng.rand(n_trees) < p_i)[0]
+        else:
+            drop = np.nonzero(
+                rng.rand(n_trees) < float(self.p.rate_drop)
+            )[0]
         if drop.size == 0 and self.p.one_drop:
             drop = np.array([int(rng.randint(n_trees))])
         if drop.size == 0:
@@ -561,6 +578,20 @@ class BoostingEngine:
         contrib = self._dart_predict_trees(dropped)
         self.margin -= contrib
         return (dropped, contrib)
+
+    def _dart_weights(self, n_trees):
+        """Per-tree weights for weighted dropout. Scales are baked into
+        leaf values at commit time, so the weight ledger is maintained
+        alongside (new trees enter at their new_scale; dropped trees are
+        rescaled by old_scale; untouched trees keep weight 1 from their
+        no-drop round)."""
+        w = getattr(self, "_dart_w", None)
+        if w is None:
+            w = []
+            self._dart_w = w
+        while len(w) < n_trees:
+            w.append(1.0)
+        return w  # the LIVE ledger (callers mutate it in place)
 
     def _dart_commit(self, dart_ctx, trees, classes):
         """Scale the new + dropped trees (xgboost normalize_type) and
@@ -592,6 +623,12 @@ class BoostingEngine:
         for ti in dropped:
             self.booster.trees[ti].value *= old_scale
         self.margin += old_scale * contrib
+        # weight ledger for sample_type="weighted"
+        wl = self._dart_weights(len(self.booster.trees))
+        for ti in dropped:
+            wl[ti] *= old_scale
+        for i in range(len(trees)):
+            wl[len(self.booster.trees) - len(trees) + i] = new_scale
 
     def _quantize(self, gpair: torch.Tensor) -> Tuple[torch.Tensor, float, float]:
         fused_mx = getattr(gpair, "_rxgb_absmax", None)
