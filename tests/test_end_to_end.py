@@ -432,6 +432,9 @@ def test_custom_feval_maximize_early_stopping():
          "seed": 9},
         {"_actors": 3, "colsample_bylevel": 0.6, "max_depth": 5,
          "seed": 13},
+        {"colsample_bynode": 0.6, "max_depth": 5, "seed": 17},
+        {"booster": "dart", "rate_drop": 0.3, "sample_type": "weighted",
+         "seed": 21},
     ],
 )
 def test_distributed_equals_single_matrix(extra):

@@ -1136,3 +1136,41 @@ class TestDartSampleType:
             np.testing.assert_array_equal(runs[0], runs[1])
             outs[st] = runs[0]
         assert not np.array_equal(outs["uniform"], outs["weighted"])
+
+
+class TestColsampleByNode:
+    def test_bynode_changes_model_and_deterministic(self):
+        dm, X, y = _binned()
+        outs = {}
+        for rate in (1.0, 0.5):
+            runs = []
+            for _ in range(2):
+                bst = run_training(
+                    {"objective": "binary:logistic", "max_depth": 5,
+                     "eta": 0.3, "colsample_bynode": rate, "seed": 3},
+                    dm, 6,
+                )
+                runs.append(bst.predict(X, output_margin=True))
+            np.testing.assert_array_equal(runs[0], runs[1])
+            outs[rate] = runs[0]
+        assert not np.array_equal(outs[1.0], outs[0.5])
+
+    def test_bynode_composes_with_interaction(self):
+        dm, X, y = _binned()
+        bst = run_training(
+            {"objective": "binary:logistic", "max_depth": 5, "eta": 0.3,
+             "colsample_bynode": 0.7,
+             "interaction_constraints": [[0, 1, 2], [3, 4, 5]]},
+            dm, 6,
+        )
+        sets = [frozenset({0, 1, 2}), frozenset({3, 4, 5})]
+        for t in bst.trees:
+            def walk(nid, feats):
+                f = int(t.feat[nid])
+                if f < 0:
+                    if feats:
+                        assert any(feats <= s for s in sets), feats
+                    return
+                walk(int(t.left[nid]), feats | {f})
+                walk(int(t.left[nid]) + 1, feats | {f})
+            walk(0, frozenset())

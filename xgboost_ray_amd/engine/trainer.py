@@ -47,6 +47,7 @@ class TrainParams:
     num_parallel_tree: int = 1
     colsample_bytree: float = 1.0
     colsample_bylevel: float = 1.0
+    colsample_bynode: float = 1.0
     scale_pos_weight: float = 1.0
     base_score: Optional[float] = None
     num_class: int = 0
@@ -709,6 +710,21 @@ class BoostingEngine:
         return view.to(self.device, non_blocking=True)
 
 
+    def _bynode_mask(self, KK, it, cls, depth):
+        """[KK, F] uint8 per-node feature gate for colsample_bynode.
+        Multiplies with colsample_bytree/bylevel (those zero feat_bins)
+        and interaction constraints (ANDed by the caller)."""
+        F = self.dtrain.n_features
+        k = max(1, int(round(float(self.p.colsample_bynode) * F)))
+        rng = np.random.RandomState(
+            (self.p.seed * 48271 + it * 2246822519
+             + cls * 3266489917 + depth * 668265263) % (2**31)
+        )
+        m = np.zeros((KK, F), np.uint8)
+        for row in range(KK):
+            m[row, rng.choice(F, size=k, replace=False)] = 1
+        return torch.from_numpy(m)
+
     def _allowed_mask(self, paths):
         """[K, F] uint8 gate from interaction constraints (None if off).
 
@@ -879,6 +895,15 @@ class BoostingEngine:
             allowed_m = (
                 self._allowed_mask(paths_ord) if track_paths else None
             )
+            if self.p.colsample_bynode < 1.0:
+                # per-node feature sampling (xgboost colsample_bynode):
+                # deterministic in (seed, iteration, class, depth) and
+                # in scan-slot order, so every rank draws the same masks
+                node_m = self._bynode_mask(KK, it, cls, depth)
+                allowed_m = (
+                    node_m if allowed_m is None
+                    else (allowed_m.to(torch.uint8) & node_m)
+                )
 
             F = self.dtrain.n_features
             all_hist = torch.empty(

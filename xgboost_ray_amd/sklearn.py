@@ -387,12 +387,12 @@ class RayXGBRFRegressor(RayXGBRegressor):
     """
 
     def __init__(
-        self, learning_rate=1.0, subsample=0.8, colsample_bytree=0.8,
+        self, learning_rate=1.0, subsample=0.8, colsample_bynode=0.8,
         reg_lambda=1e-5, **kwargs,
     ):
         super().__init__(
             learning_rate=learning_rate, subsample=subsample,
-            colsample_bytree=colsample_bytree, reg_lambda=reg_lambda, **kwargs,
+            colsample_bynode=colsample_bynode, reg_lambda=reg_lambda, **kwargs,
         )
 
     def get_xgb_params(self):
@@ -509,12 +509,12 @@ class RayXGBRFClassifier(RayXGBClassifier):
     of ``num_parallel_tree = n_estimators`` trees."""
 
     def __init__(
-        self, learning_rate=1.0, subsample=0.8, colsample_bytree=0.8,
+        self, learning_rate=1.0, subsample=0.8, colsample_bynode=0.8,
         reg_lambda=1e-5, **kwargs,
     ):
         super().__init__(
             learning_rate=learning_rate, subsample=subsample,
-            colsample_bytree=colsample_bytree, reg_lambda=reg_lambda, **kwargs,
+            colsample_bynode=colsample_bynode, reg_lambda=reg_lambda, **kwargs,
         )
 
     def get_xgb_params(self):
