@@ -408,6 +408,23 @@ class CoxNLogLik(Metric):
 
 
 
+class MAPE(Metric):
+    """Mean absolute percentage error (xgboost `mape`)."""
+
+    name = "mape"
+    higher_better = False
+
+    def local_stats(self, margin, label, weight, qid, obj):
+        pred = _eval_pred(margin, obj)
+        y = label.double()
+        w = _w(label, weight)
+        ape = (pred - y).abs() / y.abs().clamp(min=1e-10)
+        return torch.stack([(w * ape).sum(), w.sum()])
+
+    def finalize(self, s):
+        return float(s[0] / s[1])
+
+
 class MPHE(Metric):
     """Mean pseudo-Huber error (delta = 1)."""
 
@@ -449,6 +466,7 @@ def get_metric(name: str) -> Metric:
         "aft-nloglik": AFTNLogLik,
         "cox-nloglik": CoxNLogLik,
         "mphe": MPHE,
+        "mape": MAPE,
     }
     if name not in table:
         raise ValueError(f"Unsupported eval_metric: {name}")
