@@ -162,3 +162,36 @@ def test_tiny_dataset_more_actors_than_groups():
     )
     p = bst.predict(X)
     assert np.isfinite(p).all()
+
+
+def test_save_load_config_roundtrip():
+    """Booster.save_config/load_config (xgboost surface): params and
+    learner fields survive the JSON round trip."""
+    bst, X = _train_booster(max_depth=3, eta=0.123)
+    cfg = bst.save_config()
+    doc = json.loads(cfg)
+    assert doc["learner"]["learner_train_param"]["objective"] == \
+        "binary:logistic"
+    b2 = Booster()
+    b2.load_config(cfg)
+    assert b2.objective == "binary:logistic"
+    assert float(b2.params["eta"]) == 0.123
+
+
+def test_get_score_importance_types():
+    bst, X = _train_booster()
+    for t in ("weight", "gain", "total_gain", "cover", "total_cover"):
+        sc = bst.get_score(importance_type=t)
+        assert sc and all(v > 0 for v in sc.values()), t
+    w = bst.get_score(importance_type="weight")
+    tg = bst.get_score(importance_type="total_gain")
+    g = bst.get_score(importance_type="gain")
+    for k in g:
+        assert g[k] == pytest.approx(tg[k] / w[k])
+
+
+def test_best_ntree_limit():
+    bst, X = _train_booster()
+    assert bst.best_ntree_limit == 5  # no early stop: all rounds
+    bst.best_iteration = 2
+    assert bst.best_ntree_limit == 3

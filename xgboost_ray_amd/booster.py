@@ -594,6 +594,45 @@ class Booster:
             with open(fname, "w") as f:
                 json.dump(self._to_json_dict(), f)
 
+    def save_config(self) -> str:
+        """Learner configuration as a JSON string (xgboost
+        Booster.save_config surface: enough for config snapshots and
+        load_config round-trips; the model itself uses save_model)."""
+        return json.dumps({
+            "learner": {
+                "generic_param": {},
+                "gradient_booster": {
+                    "name": self.params.get("booster", "gbtree"),
+                },
+                "learner_model_param": {
+                    "base_score": repr(self.base_score),
+                    "num_class": str(self.num_class),
+                    "num_feature": str(self.num_features),
+                },
+                "learner_train_param": {
+                    "objective": self.objective,
+                },
+                "params": {
+                    k: v for k, v in self.params.items()
+                    if isinstance(v, (str, int, float, bool))
+                },
+            }
+        })
+
+    def load_config(self, config: str):
+        doc = json.loads(config)
+        learner = doc.get("learner", {})
+        lp = learner.get("learner_model_param", {})
+        for key in ("base_score", "num_class", "num_feature"):
+            if key in lp:
+                self.params[key] = lp[key]
+        if "learner_train_param" in learner:
+            obj = learner["learner_train_param"].get("objective")
+            if obj:
+                self.params["objective"] = obj
+        self.params.update(learner.get("params", {}))
+        return self
+
     def save_raw(self, raw_format: str = "json") -> bytes:
         if raw_format == "ubj":
             return ubjson.dumps(self._to_json_dict())
