@@ -424,6 +424,16 @@ class Booster:
         if pred_interactions:
             return self.predict_interactions(X, iteration_range)
         Xt = torch.from_numpy(np.ascontiguousarray(X, dtype=np.float32))
+        # MI355X-first routing: when a GPU is present, bulk prediction
+        # runs the LDS tree-walk kernel there (the CPU level-walk is for
+        # GPU-less test environments; at 5M rows x 50 trees it is
+        # minutes vs 30 ms on device)
+        if (
+            not Xt.is_cuda
+            and Xt.shape[0] >= 16384
+            and torch.cuda.is_available()
+        ):
+            Xt = Xt.cuda()
         margin = self.predict_margin_tensor(Xt, iteration_range)
         if bm is not None:
             margin = margin + torch.from_numpy(bm).to(margin.dtype).reshape(

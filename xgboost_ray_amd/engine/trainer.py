@@ -339,6 +339,10 @@ class BoostingEngine:
                 "max_depth": self.p.max_depth,
                 "eta": self.p.eta,
                 "num_parallel_tree": self.p.num_parallel_tree,
+                # retained so predict() can route to GPU actors
+                # (_is_gpu_params reads the model's tree_method)
+                "tree_method": self.p.tree_method,
+                "booster": self.p.booster,
             }
         )
 

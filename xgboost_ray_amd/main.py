@@ -736,7 +736,14 @@ def predict(
         raise ValueError(
             f"The `data` argument must be a RayDMatrix, got {type(data)}."
         )
-    use_gpu = _is_gpu_params(getattr(model, "params", {})) and torch.cuda.is_available()
+    # GPU routing: the model's tree_method OR an explicit GPU request in
+    # ray_params (a user passing gpus_per_actor=1 asked for GPU actors;
+    # round-2 finding: a model whose params lost tree_method predicted
+    # through the CPU walker - 174 s for 5M rows vs 0.03 s on device)
+    use_gpu = (
+        _is_gpu_params(getattr(model, "params", {}))
+        or ray_params.gpus_per_actor > 0
+    ) and torch.cuda.is_available()
     max_actor_restarts = (
         ray_params.max_actor_restarts
         if ray_params.max_actor_restarts >= 0

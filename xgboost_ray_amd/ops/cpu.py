@@ -274,16 +274,16 @@ def predict_trees(X, feat, thr, left, default_left, value, tree_ptr, out, tree_w
             is_leaf = f < 0
             if bool(is_leaf.all()):
                 break
-            active = ~is_leaf
-            ca = cur[active]
-            fa = f[active]
-            x = X[rows[active], fa]
-            miss = torch.isnan(x)
-            go_left = x < thr[ca]
-            go_left = torch.where(miss, default_left[ca].bool(), go_left)
-            nxt = torch.where(go_left, left[ca].long(), left[ca].long() + 1) + base
-            cur = cur.clone()
-            cur[active] = nxt
+            # full-width level step with self-looping leaves: the
+            # boolean-mask variant paid a nonzero + gather + clone per
+            # level, several times this version's cost
+            x = X[rows, f.clamp(min=0)]
+            go_left = torch.where(
+                torch.isnan(x), default_left[cur].bool(), x < thr[cur]
+            )
+            la = left[cur].long()
+            nxt = torch.where(go_left, la, la + 1) + base
+            cur = torch.where(is_leaf, cur, nxt)
         out += value[cur] * tree_weight
     return out
 
