@@ -1174,3 +1174,33 @@ class TestColsampleByNode:
                 walk(int(t.left[nid]), feats | {f})
                 walk(int(t.left[nid]) + 1, feats | {f})
             walk(0, frozenset())
+
+
+class TestApproxContribs:
+    def test_additivity_and_speed_path(self):
+        """Saabas approx contribs: additivity holds exactly (each path
+        telescopes from the tree's expected value to the leaf)."""
+        dm, X, y = _binned()
+        bst = run_training(
+            {"objective": "binary:logistic", "max_depth": 5, "eta": 0.3},
+            dm, 6,
+        )
+        C = bst.predict(X[:500], pred_contribs=True, approx_contribs=True)
+        margin = bst.predict(X[:500], output_margin=True)
+        np.testing.assert_allclose(C.sum(axis=1), margin, rtol=1e-5,
+                                   atol=1e-5)
+        # exact TreeSHAP and approx share the bias column
+        E = bst.predict(X[:20], pred_contribs=True)
+        np.testing.assert_allclose(C[:20, -1], E[:20, -1], rtol=1e-5)
+
+    def test_multiclass_additivity(self):
+        dm, X, y = _binned(kind="multi")
+        bst = run_training(
+            {"objective": "multi:softprob", "num_class": 4,
+             "max_depth": 4, "eta": 0.3},
+            dm, 4,
+        )
+        C = bst.predict(X[:200], pred_contribs=True, approx_contribs=True)
+        margin = bst.predict(X[:200], output_margin=True)
+        np.testing.assert_allclose(C.sum(axis=2), margin, rtol=1e-5,
+                                   atol=1e-5)

@@ -420,6 +420,8 @@ class Booster:
         if pred_leaf:
             return self.predict_leaf(X, iteration_range)
         if pred_contribs:
+            if kwargs.get("approx_contribs"):
+                return self.predict_contribs_approx(X, iteration_range)
             return self.predict_contribs(X, iteration_range)
         if pred_interactions:
             return self.predict_interactions(X, iteration_range)
@@ -444,6 +446,62 @@ class Booster:
         obj = get_objective(self.objective, self.num_class)
         return obj.transform_prediction(margin).cpu().numpy()
 
+
+    def predict_contribs_approx(self, X, iteration_range=None) -> np.ndarray:
+        """Saabas-style approximate contributions (xgboost
+        ``pred_contribs`` with ``approx_contribs=True``): each split on
+        the walked root-to-leaf path credits its feature with the change
+        in the cover-weighted expected value. Fully vectorized level
+        walk - orders of magnitude faster than exact TreeSHAP, satisfies
+        the same additivity identity contribs.sum(-1) == margin."""
+        from xgboost_ray_amd.engine.objectives import get_objective
+
+        X = _as_float32_matrix(X)
+        n, F = X.shape
+        lo, hi = 0, self.num_boosted_rounds()
+        if iteration_range is not None:
+            lo, hi = iteration_range
+            hi = hi or self.num_boosted_rounds()
+        obj = get_objective(self.objective, self.num_class)
+        base = float(obj.prob_to_margin(self.base_score))
+        k_cls = max(1, self.num_class)
+        k = k_cls * self.num_parallel_tree
+        if self.num_class > 1:
+            out = np.zeros((n, self.num_class, F + 1), np.float64)
+        else:
+            out = np.zeros((n, F + 1), np.float64)
+        out[..., F] += base  # global intercept joins the bias column
+        rows = np.arange(n)
+        for ti in range(lo * k, min(hi * k, len(self.trees))):
+            t = self.trees[ti]
+            cls = self.tree_info[ti] if self.num_class > 1 else 0
+            dest = out[:, cls, :] if self.num_class > 1 else out
+            mean_val = np.zeros(t.num_nodes, np.float64)
+            cover = t.cover.astype(np.float64)
+            _fill_node_means(t, mean_val, cover)
+            dest[:, F] += mean_val[0]  # bias: tree expected value
+            cur = np.zeros(n, np.int64)
+            while True:
+                f = t.feat[cur]
+                inner = f >= 0
+                if not inner.any():
+                    break
+                r = rows[inner]
+                fr = f[inner]
+                cr = cur[inner]
+                x = X[r, fr]
+                left = t.left[cr]
+                go_left = x < t.thr[cr]
+                go_left = np.where(
+                    np.isnan(x), t.default_left[cr].astype(bool), go_left
+                )
+                child = np.where(go_left, left, left + 1)
+                # credit the split feature with the expectation change
+                np.add.at(dest, (r, fr), mean_val[child] - mean_val[cr])
+                nxt = cur.copy()
+                nxt[inner] = child
+                cur = nxt
+        return out
 
     def predict_contribs(self, X, iteration_range=None) -> np.ndarray:
         """Exact TreeSHAP feature attributions (xgboost ``pred_contribs``;
