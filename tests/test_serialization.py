@@ -211,3 +211,29 @@ def test_booster_params_retain_gpu_routing_fields():
     import pickle as _p
 
     assert _is_gpu_params(_p.loads(_p.dumps(bst)).params)
+
+
+def test_predict_validates_feature_names():
+    """Named prediction inputs must match the training names in order
+    (xgboost validate_features semantics)."""
+    import pandas as pd
+
+    rng = np.random.RandomState(0)
+    df = pd.DataFrame(rng.randn(500, 4).astype(np.float32),
+                      columns=["a", "b", "c", "d"])
+    y = (df["a"] > 0).astype(np.float32)
+    from xgboost_ray_amd.engine.quantile import BinnedMatrix
+
+    dm = BinnedMatrix.build(
+        torch.from_numpy(df.values), label=torch.from_numpy(y.values),
+        max_bin=32,
+    )
+    bst = run_training({"objective": "binary:logistic", "max_depth": 3},
+                       dm, 3)
+    bst.feature_names = ["a", "b", "c", "d"]
+    bst.predict(df)  # matching names OK
+    bad = df.rename(columns={"d": "x"})
+    with pytest.raises(ValueError, match="feature_names mismatch"):
+        bst.predict(bad)
+    # opt-out preserved
+    bst.predict(bad, validate_features=False)

@@ -414,9 +414,22 @@ class Booster:
         if isinstance(data, DMatrix):
             X = data.data
             bm = data.base_margin
+            in_names = data.feature_names
         else:
+            in_names = (
+                [str(c) for c in data.columns]
+                if hasattr(data, "columns") else None
+            )
             X = _as_float32_matrix(data)
             bm = None
+        if validate_features and self.feature_names and in_names:
+            if list(in_names) != list(self.feature_names):
+                # xgboost semantics: named inputs must match the names
+                # the model was trained with, in order
+                raise ValueError(
+                    "feature_names mismatch: model expects "
+                    f"{self.feature_names}, got {in_names}"
+                )
         if pred_leaf:
             return self.predict_leaf(X, iteration_range)
         if pred_contribs:
