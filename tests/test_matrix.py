@@ -214,3 +214,44 @@ def test_reshard_on_num_actors_change(xy):
     assert len(dm.refs) == 2
     parts_x, _ = _gather_all(dm, 2)
     assert sum(p.shape[0] for p in parts_x) == 100
+
+
+class TestShmStoreEdges:
+    """shm object store robustness (the Ray object-store analogue)."""
+
+    def test_put_get_free_cycle(self):
+        import numpy as np
+
+        from xgboost_ray_amd import shm_store
+
+        store = shm_store.get_store()
+        arr = np.arange(100_000, dtype=np.float64)
+        ref = store.put(arr)
+        out = shm_store.get(ref)
+        np.testing.assert_array_equal(out, arr)
+        store.free(ref)
+        # double-free must be harmless (failure-path drains may race)
+        store.free(ref)
+
+    def test_many_small_objects(self):
+        import numpy as np
+
+        from xgboost_ray_amd import shm_store
+
+        store = shm_store.get_store()
+        refs = [store.put(np.full(10, i, np.int32)) for i in range(50)]
+        for i, r in enumerate(refs):
+            assert shm_store.get(r)[0] == i
+        for r in refs:
+            store.free(r)
+
+
+def test_env_overrides(monkeypatch):
+    """RXGB_* env tier: read-time typed overrides (reference _XGBoostEnv)."""
+    from xgboost_ray_amd.env import ENV
+
+    base = ENV.STATUS_FREQUENCY_S
+    monkeypatch.setenv("RXGB_STATUS_FREQUENCY_S", "123")
+    assert ENV.STATUS_FREQUENCY_S == 123  # int-typed field, coerced
+    monkeypatch.delenv("RXGB_STATUS_FREQUENCY_S")
+    assert ENV.STATUS_FREQUENCY_S == base
