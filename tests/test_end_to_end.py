@@ -593,3 +593,22 @@ def test_invalid_train_kwarg_raises():
             ray_params=RayParams(num_actors=1),
             totally_invalid_kwarg="",
         )
+
+
+def test_invalid_predict_kwarg_raises():
+    from tests.utils import create_data
+    from xgboost_ray_amd import RayDMatrix, RayParams, predict, train
+
+    X, y = create_data(300, 4)
+    bst = train({"objective": "binary:logistic"}, RayDMatrix(X, label=y),
+                2, ray_params=RayParams(num_actors=1))
+    with pytest.raises(TypeError, match="bogus_kwarg"):
+        predict(bst, RayDMatrix(X), ray_params=RayParams(num_actors=1),
+                bogus_kwarg=1)
+
+
+def test_tune_resources_requires_ray():
+    from xgboost_ray_amd.main import RayParams
+
+    with pytest.raises(RuntimeError, match="ray"):
+        RayParams(num_actors=2).get_tune_resources()

@@ -732,6 +732,16 @@ def predict(
 ) -> Optional[np.ndarray]:
     """Distributed prediction (reference main.py:1809-1896)."""
     ray_params = _validate_ray_params(ray_params)
+    _ALLOWED_PREDICT_KWARGS = {
+        "output_margin", "iteration_range", "pred_leaf", "pred_contribs",
+        "pred_interactions", "approx_contribs", "validate_features",
+    }
+    unknown = set(kwargs) - _ALLOWED_PREDICT_KWARGS
+    if unknown:
+        raise TypeError(
+            "predict() got unexpected keyword argument(s): "
+            + ", ".join(sorted(unknown))
+        )
     if not isinstance(data, RayDMatrix):
         raise ValueError(
             f"The `data` argument must be a RayDMatrix, got {type(data)}."
