@@ -1224,3 +1224,30 @@ class TestDartResumeTolerance:
         a = full.predict(X, output_margin=True)
         b = resumed.predict(X, output_margin=True)
         np.testing.assert_allclose(a, b, atol=2e-6, rtol=1e-6)
+
+
+class TestSubsampleMarginCache:
+    def test_all_rows_advance_under_subsample(self):
+        """xgboost UpdatePredictionCache semantics: with row sampling,
+        EVERY row's training margin advances each round (fuzz-found:
+        out-of-sample rows were left on stale margins)."""
+        dm, X, y = _binned()
+        engine_params = {"objective": "binary:logistic", "max_depth": 3,
+                         "eta": 0.1, "subsample": 0.6, "seed": 20}
+        eng = BoostingEngine(engine_params, dm)
+        for _ in range(4):
+            eng.update()
+        pm = eng.booster.predict(X, output_margin=True)
+        np.testing.assert_allclose(pm, eng.margin.numpy(), atol=1e-6)
+
+    def test_subsample_resume_bitwise(self):
+        dm, X, y = _binned()
+        cfg = {"objective": "binary:logistic", "max_depth": 5,
+               "eta": 0.1, "subsample": 0.7, "seed": 20}
+        full = run_training(dict(cfg), dm, 8)
+        half = run_training(dict(cfg), dm, 4)
+        resumed = run_training(dict(cfg), dm, 4, xgb_model=half)
+        np.testing.assert_array_equal(
+            full.predict(X, output_margin=True),
+            resumed.predict(X, output_margin=True),
+        )

@@ -1333,20 +1333,6 @@ class BoostingEngine:
         leaf_vals = [v for (s, c, v, bp) in self._leaf_segs]
         leaf_parity = [bp for (s, c, v, bp) in self._leaf_segs]
         _tick("leaf_host")
-        if leaf_starts:
-            ops.update_margins(
-                self.margin if self.n_class == 1 else self.margin[:, cls],
-                ridx_bufs[0] if ridx_bufs is not None else ridx,
-                torch.tensor(leaf_starts, dtype=torch.int64),
-                torch.tensor(leaf_counts, dtype=torch.int64),
-                np.asarray(leaf_vals, dtype=np.float32),
-                ridx_b=(ridx_bufs[1] if ridx_bufs is not None else None),
-                parity=(leaf_parity if ridx_bufs is not None else None),
-            )
-        self._leaf_segs = []
-        _tick("margins")
-        if _prof:
-            print("PROF", {k: round(v*1000, 2) for k, v in _t.items()}, flush=True)
 
         n = ta.n
         tree = Tree(
@@ -1361,6 +1347,31 @@ class BoostingEngine:
                 ta.parent[:n] < 0, 2147483647, ta.parent[:n]
             ).astype(np.int32),
         )
+
+        if n_local < self.dtrain.n_rows:
+            # row sampling: EVERY row's margin must advance (xgboost
+            # UpdatePredictionCache semantics) - the leaf-segment
+            # scatter only covers the sampled rows, which left
+            # out-of-sample gradients on stale margins (fuzz-found in
+            # round 2; also broke resume==continuous under subsample).
+            # The binned tree walk lands each row in the identical leaf
+            # the partition chose, so sampled rows get the same bits.
+            mv = self.margin if self.n_class == 1 else self.margin[:, cls]
+            self._add_tree_margin_raw(tree, mv)
+        elif leaf_starts:
+            ops.update_margins(
+                self.margin if self.n_class == 1 else self.margin[:, cls],
+                ridx_bufs[0] if ridx_bufs is not None else ridx,
+                torch.tensor(leaf_starts, dtype=torch.int64),
+                torch.tensor(leaf_counts, dtype=torch.int64),
+                np.asarray(leaf_vals, dtype=np.float32),
+                ridx_b=(ridx_bufs[1] if ridx_bufs is not None else None),
+                parity=(leaf_parity if ridx_bufs is not None else None),
+            )
+        self._leaf_segs = []
+        _tick("margins")
+        if _prof:
+            print("PROF", {k: round(v*1000, 2) for k, v in _t.items()}, flush=True)
         return tree
 
     _leaf_segs: List[Tuple[int, int, float]] = []
@@ -1739,16 +1750,7 @@ class BoostingEngine:
         leaf_starts = [st for (st, c, v, bp) in self._leaf_segs]
         leaf_counts = [c for (st, c, v, bp) in self._leaf_segs]
         leaf_vals = [v for (st, c, v, bp) in self._leaf_segs]
-        if leaf_starts:
-            ops.update_margins(
-                self.margin if self.n_class == 1 else self.margin[:, cls],
-                ridx,
-                torch.tensor(leaf_starts, dtype=torch.int64),
-                torch.tensor(leaf_counts, dtype=torch.int64),
-                np.asarray(leaf_vals, dtype=np.float32),
-            )
-        self._leaf_segs = []
-        return Tree(
+        tree = Tree(
             feat=np.asarray(feat_l, np.int32),
             thr=np.asarray(thr_l, np.float32),
             left=np.asarray(left_l, np.int32),
@@ -1760,6 +1762,20 @@ class BoostingEngine:
                 [2147483647 if pp < 0 else pp for pp in parent_l], np.int32
             ),
         )
+        if int(ridx.numel()) < self.dtrain.n_rows:
+            # row sampling: full-row margin advance (see depthwise note)
+            mv = self.margin if self.n_class == 1 else self.margin[:, cls]
+            self._add_tree_margin_raw(tree, mv)
+        elif leaf_starts:
+            ops.update_margins(
+                self.margin if self.n_class == 1 else self.margin[:, cls],
+                ridx,
+                torch.tensor(leaf_starts, dtype=torch.int64),
+                torch.tensor(leaf_counts, dtype=torch.int64),
+                np.asarray(leaf_vals, dtype=np.float32),
+            )
+        self._leaf_segs = []
+        return tree
 
     def _finalize_leaf_rec(self, rec, val_l, cover_l, scale_h, nid=None):
         nid = rec["nid"] if nid is None else nid
