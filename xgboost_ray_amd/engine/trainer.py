@@ -351,6 +351,10 @@ class BoostingEngine:
         """Continue training from an existing model (xgb_model kwarg)."""
         self.booster = model
         self.iteration = model.num_boosted_rounds()
+        # restore the dart weight ledger (weighted dropout resume)
+        dw = model.attr("dart_weights")
+        if dw:
+            self._dart_w = [float(x) for x in dw.split(",") if x]
         # recompute margins through the binned matrix (thresholds are exact
         # cut values for models we trained, so binned replay is exact)
         n = self.dtrain.n_rows
@@ -628,12 +632,17 @@ class BoostingEngine:
         for ti in dropped:
             self.booster.trees[ti].value *= old_scale
         self.margin += old_scale * contrib
-        # weight ledger for sample_type="weighted"
+        # weight ledger for sample_type="weighted"; persisted as a
+        # booster attribute so checkpoint/resume reproduces the same
+        # weighted-dropout probabilities (fuzz-found)
         wl = self._dart_weights(len(self.booster.trees))
         for ti in dropped:
             wl[ti] *= old_scale
         for i in range(len(trees)):
             wl[len(self.booster.trees) - len(trees) + i] = new_scale
+        self.booster.set_attr(
+            dart_weights=",".join(repr(w) for w in wl)
+        )
 
     def _quantize(self, gpair: torch.Tensor) -> Tuple[torch.Tensor, float, float]:
         fused_mx = getattr(gpair, "_rxgb_absmax", None)
