@@ -17,9 +17,15 @@ from xgboost_ray_amd.booster import Booster
 
 
 def main():
-    random.seed(7)
+    import argparse
+
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--seed", type=int, default=7)
+    ap.add_argument("--trials", type=int, default=24)
+    args = ap.parse_args()
+    random.seed(args.seed)
     fails = []
-    for trial in range(24):
+    for trial in range(args.trials):
         cfg = {"objective": "binary:logistic",
                "max_depth": random.choice([3, 5, 7]),
                "eta": random.choice([0.1, 0.3]), "seed": trial}
@@ -78,7 +84,8 @@ def main():
                 fails.append((trial, cfg, "resume"))
         except Exception as e:
             fails.append((trial, cfg, repr(e)[:120]))
-    print("FUZZ v2 RESULT:", "ALL 24 PASS" if not fails else fails)
+    print("FUZZ RESULT (seed %d, %d trials):" % (args.seed, args.trials),
+          "ALL PASS" if not fails else fails)
 
 
 if __name__ == "__main__":
