@@ -123,3 +123,38 @@ class TestElastic:
                     max_actor_restarts=1,
                 ),
             )
+
+
+class TestRestartDeterminismVariants:
+    """fail+checkpoint-resume == no-failure for the round-2 feature
+    combinations whose state must survive the checkpoint (dart weight
+    ledger, per-node column sampling, row sampling margin cache)."""
+
+    import pytest as _pytest
+
+    @_pytest.mark.parametrize("extra", [
+        {"booster": "dart", "rate_drop": 0.3, "sample_type": "weighted"},
+        {"colsample_bynode": 0.6},
+        {"subsample": 0.7},
+    ])
+    def test_variant(self, tmp_path, extra):
+        X, y = _data()
+        params = dict(PARAMS, **extra, seed=9)
+        kw = dict(checkpoint_frequency=5)
+        bst_noerr = train(
+            params, RayDMatrix(X, label=y), 16,
+            ray_params=RayParams(num_actors=2, max_actor_restarts=0, **kw),
+        )
+        bst_err = train(
+            params, RayDMatrix(X, label=y), 16,
+            ray_params=RayParams(num_actors=2, max_actor_restarts=1, **kw),
+            callbacks=[DieCallback(die_round=9, die_rank=1,
+                                   lock_dir=str(tmp_path))],
+        )
+        assert bst_err.num_boosted_rounds() == 16
+        a = bst_noerr.predict(X, output_margin=True)
+        b = bst_err.predict(X, output_margin=True)
+        if extra.get("booster") == "dart":
+            np.testing.assert_allclose(a, b, atol=2e-6, rtol=1e-6)
+        else:
+            np.testing.assert_array_equal(a, b)
